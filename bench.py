@@ -1,0 +1,208 @@
+#!/usr/bin/env python3
+"""bench.py — KNN queries/sec on the BASELINE workload.
+
+Workload (BASELINE.json configs[1]): 10M rows x 768-dim f32, brute-force
+cosine K=10, single query at a time, per GPU. A "step" is ONE query over the
+whole (sharded) corpus. Scaling is WEAK: each rank holds its own 10M-row
+shard, so the corpus grows with N while per-GPU work per query is fixed
+(BASELINE configs[4]-style sharding; rows_total = 10M * N).
+
+N>1 protocol (one process per GPU, launched by torch.distributed.run):
+every rank scans its shard (sdbv_knn_bruteforce), all ranks all-gather the
+per-shard top-K (K*(f64,i64) per rank — latency-bound over xGMI via RCCL),
+rank 0 merges with the reference tie-break (dist total_cmp asc, id asc).
+
+Output: ONE JSON line from rank 0 (see the driver contract).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=40)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--rows", type=int, default=10_000_000,
+                    help="rows per GPU shard (weak scaling)")
+    ap.add_argument("--dim", type=int, default=768)
+    ap.add_argument("--k", type=int, default=10)
+    ap.add_argument("--metric", default="cosine")
+    ap.add_argument("--seed", type=lambda x: int(x, 0), default=0x5DB1)
+    ap.add_argument("--cpu-sample-rows", type=int, default=2_000_000,
+                    help="bounded sample for the cpu_baseline leg")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, args.gpus)
+
+    import torch
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+
+    import surrealdb_amd
+    from surrealdb_amd.shard import merge_topk
+    from surrealdb_amd.synth import gen_f32
+
+    ctx = surrealdb_amd.Context(device=local_rank)
+
+    # --- stage this rank's shard (one-time, outside the timed region) ---
+    rows = args.rows
+    row_offset = rank * rows
+    t0 = time.perf_counter()
+    ctx.stage_synthetic(1, rows, args.dim, metric=args.metric, seed=args.seed,
+                        row_offset=row_offset, id_base=row_offset)
+    stage_s = time.perf_counter() - t0
+
+    queries = gen_f32(0xBEEF, 0, args.steps + args.warmup, args.dim)
+
+    device = torch.device(f"cuda:{local_rank}")
+
+    def one_query(qi):
+        ids, dists = ctx.knn_bruteforce(1, queries[qi], args.k)
+        if world > 1:
+            pad = args.k - len(ids)
+            if pad:
+                ids = np.concatenate(
+                    [ids, np.full(pad, np.iinfo(np.uint64).max, np.uint64)])
+                dists = np.concatenate([dists, np.full(pad, np.inf)])
+            local = torch.empty(args.k, 2, dtype=torch.float64, device=device)
+            local[:, 0] = torch.from_numpy(dists.copy()).to(device)
+            local[:, 1] = torch.from_numpy(
+                ids.view(np.float64).copy()).to(device)
+            gathered = [torch.empty_like(local) for _ in range(world)]
+            dist.all_gather(gathered, local)
+            if rank == 0:
+                g = [t.cpu().numpy() for t in gathered]
+                return merge_topk([x[:, 1].copy().view(np.uint64) for x in g],
+                                  [x[:, 0] for x in g], args.k)
+            return None
+        return ids, dists
+
+    # --- warmup ---
+    for i in range(args.warmup):
+        one_query(i)
+
+    # --- timed region: EXACTLY args.steps steps ---
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize(device) if torch.cuda.is_available() else None
+    step_times = []
+    scan_ms_acc = 0.0
+    t_begin = time.perf_counter()
+    for i in range(args.steps):
+        ts = time.perf_counter()
+        one_query(args.warmup + i)
+        step_times.append(time.perf_counter() - ts)
+        scan_ms_acc += ctx.stats()["last_scan_kernel_ms"]
+    torch.cuda.synchronize(device) if torch.cuda.is_available() else None
+    if dist:
+        dist.barrier()
+    t_total = time.perf_counter() - t_begin
+
+    # max over ranks
+    if dist:
+        tt = torch.tensor([t_total], dtype=torch.float64, device=device)
+        dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+        t_total = float(tt.item())
+
+    if rank != 0:
+        return
+
+    qps = args.steps / t_total
+    p50_ms = float(np.percentile(np.array(step_times) * 1e3, 50))
+    p95_ms = float(np.percentile(np.array(step_times) * 1e3, 95))
+
+    # --- roofline: dominant kernel = the distance scan ---
+    # algorithmic bytes per launch: rows * d * 4 B (corpus read)
+    scan_ms_avg = scan_ms_acc / args.steps
+    alg_bytes = rows * args.dim * 4
+    achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
+    traffic = os.environ.get("SDBV_TRAFFIC_BYTES_PER_LAUNCH")
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved_gbs, 1),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+        "traffic": float(traffic) if traffic else None,
+    }
+
+    # --- cpu_baseline: the oracle (kind "port") on host cores, rank0/N=1 ---
+    cpu_baseline = None
+    if world == 1 and not args.no_cpu_baseline:
+        import oracle
+        srows = min(args.cpu_sample_rows, rows)
+        sample = oracle.gen_f32(args.seed, 0, srows, args.dim)
+        q = queries[0]
+        # warmup + 3 timed queries over the bounded sample
+        oracle.topk_f32_mt(args.metric, sample, q, args.k)
+        tcs = time.perf_counter()
+        reps = 3
+        used = 0
+        for r in range(reps):
+            _, _, used = oracle.topk_f32_mt(args.metric, sample,
+                                            queries[r + 1], args.k)
+        t_cpu_sample = (time.perf_counter() - tcs) / reps
+        t_cpu_full = t_cpu_sample * (rows / srows)
+        cpu_baseline = {
+            "value": round(1.0 / t_cpu_full, 3),
+            "unit": "queries/s",
+            "cores": used,
+            "kind": "port",
+            "sample": f"{reps} queries x {srows} of {rows} rows "
+                      f"({t_cpu_sample*1e3:.0f} ms/query on the sample; "
+                      f"oracle orc_topk_f32_mt, OpenMP)",
+        }
+
+    out = {
+        "metric": "knn_qps",
+        "value": round(qps, 3),
+        "unit": "queries/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(t_total / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # no published reference number
+        "dtype": "f32",
+        "data": "synthetic",
+        "config": {
+            "workload": "brute-force cosine KNN, 10M rows/GPU x 768-dim f32, "
+                        "K=10, single query (BASELINE configs[1])",
+            "rows_total": rows * world,
+            "rows_per_gpu": rows,
+            "dim": args.dim,
+            "k": args.k,
+            "distance": args.metric,
+            "parallelism": f"rowshard{world}" if world > 1 else "single",
+            "stage_s": round(stage_s, 2),
+        },
+        "p50_ms": round(p50_ms, 3),
+        "p95_ms": round(p95_ms, 3),
+        "scan_kernel_ms_avg": round(scan_ms_avg, 3),
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,144 @@
+/* sdbv.h — C-ABI of the MI355X-native SurrealDB vector-KNN hot path.
+ *
+ * This is the drop-in boundary: the Rust host (surrealdb) would bind these
+ * entry points via hip-sys/bindgen (see INTEGRATION.md for the binding stub).
+ * Each function cites the reference interface it replaces:
+ *
+ *  - sdbv_knn_bruteforce / sdbv_knn_batch replace the distance+top-K core of
+ *    the brute-force KnnTopK operator
+ *    (reference: surrealdb/core/src/exec/operators/knn_topk.rs:166-267 and the
+ *    legacy KnnPriorityList path surrealdb/core/src/idx/planner/knn.rs:11-106).
+ *    The host keeps: record streaming, Value->vector extraction
+ *    (knn_topk.rs:274-288), WHERE filtering, and record materialisation.
+ *  - sdbv_stage_corpus replaces the per-search scattered vector access of
+ *    the reference (VectorCache LRU + He KV keys,
+ *    surrealdb/core/src/idx/trees/hnsw/cache.rs,
+ *    surrealdb/core/src/idx/trees/hnsw/elements.rs:94-140) with a one-time
+ *    HBM-resident staging of a table's vectors.
+ *  - sdbv_hnsw_* replace the layer-0 best-first search loop
+ *    (surrealdb/core/src/idx/trees/hnsw/layer.rs:184-223) behind
+ *    HnswIndex::knn_search (surrealdb/core/src/idx/trees/hnsw/index.rs:270-335).
+ *    The host keeps: pendings merge (index.rs:372-420), doc-id expansion
+ *    (Ids64, idx/trees/knn.rs:170-326), cond filtering (hnsw/filter.rs).
+ *
+ * Plain pointers and sizes only; no torch types. All calls are synchronous
+ * w.r.t. results and thread-safe (internal per-context mutex). The caller
+ * owns all host buffers; the context owns all device memory.
+ */
+#ifndef SDBV_H
+#define SDBV_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Error codes (negative) — 0 == OK. */
+enum {
+	SDBV_OK = 0,
+	SDBV_ERR_HIP = -1,          /* HIP runtime error; see sdbv_last_error */
+	SDBV_ERR_NO_TABLE = -2,     /* table id not staged */
+	SDBV_ERR_BAD_ARG = -3,      /* dimension mismatch, k out of range, ... */
+	SDBV_ERR_OOM = -4,          /* device allocation failed */
+	SDBV_ERR_UNSUPPORTED = -5,  /* metric not GPU-supported (host fallback is the
+	                               caller's oracle-free CPU path, not ours) */
+	SDBV_ERR_IDS_UNSORTED = -6, /* ids must be strictly increasing (tie-break contract) */
+};
+
+/* Distance metric — mirrors catalog::Distance
+ * (surrealdb/core/src/catalog/schema/index.rs:250-284).
+ * GPU kernels implement COSINE and EUCLIDEAN (the vector-search metrics);
+ * the remaining metrics stay on the reference's own CPU path. */
+enum {
+	SDBV_METRIC_COSINE = 0,
+	SDBV_METRIC_EUCLIDEAN = 1,
+};
+
+typedef struct sdbv_ctx sdbv_ctx; /* owns HIP stream, device pools, staged tables */
+
+typedef struct sdbv_stats {
+	double last_scan_kernel_ms;  /* HIP-event time of the last distance-scan kernel */
+	double last_merge_kernel_ms; /* HIP-event time of the last top-K merge kernel */
+	double last_total_ms;        /* host wall of the last search call (incl. D2H of K results) */
+	uint64_t bytes_staged;       /* device bytes held by staged tables */
+	uint64_t last_rows_scanned;  /* rows the last scan covered */
+} sdbv_stats;
+
+/* Create a context on HIP device `device` (-1 = current device). */
+int sdbv_init(int device, sdbv_ctx **out);
+void sdbv_shutdown(sdbv_ctx *);
+
+const char *sdbv_last_error(sdbv_ctx *);
+int sdbv_get_stats(sdbv_ctx *, sdbv_stats *out);
+
+/* Stage a table's vectors into HBM as a feature-major f32 store
+ * (column-major [d][n_pad]) plus per-row f64 norms precomputed with the
+ * reference's norm semantics (vector.rs:244-249).
+ * `rows` is row-major n x d host memory. `ids` maps row -> record/doc id and
+ * must be strictly increasing (NULL => 0..n-1); result tie-break is
+ * (distance asc, id asc), which equals the reference's orderings for
+ * monotone ids (knn.rs:363 BTreeSet<(FloatKey, VectorId)>, knn_topk.rs:61-73
+ * insertion-order seq). Restages (replaces) if `table` already staged. */
+int sdbv_stage_corpus(sdbv_ctx *, uint64_t table, const float *rows,
+                      const uint64_t *ids, uint64_t n, uint32_t d,
+                      uint8_t metric);
+
+/* Stage a deterministic synthetic corpus directly on-device (no PCIe):
+ * element (row_offset+i, j) = splitmix64-derived U[-20,20) f32 — the committed
+ * restatement of the BASELINE synthetic-data contract (matches
+ * RandomItemGenerator::Float(-20,20), knn.rs:633-643, by distribution;
+ * exact bit-stream defined by oracle/orc_gen_f32). ids = id_base + i. */
+int sdbv_stage_synthetic(sdbv_ctx *, uint64_t table, uint64_t n, uint32_t d,
+                         uint8_t metric, uint64_t seed, uint64_t row_offset,
+                         uint64_t id_base);
+
+uint64_t sdbv_table_rows(sdbv_ctx *, uint64_t table);
+int sdbv_drop_table(sdbv_ctx *, uint64_t table);
+
+/* Brute-force exact K-nearest-neighbour scan of a staged table.
+ * Results sorted ascending by (distance f64-total_cmp, id).
+ * Distances follow Distance::calculate F32 semantics (vector.rs:244-249 /
+ * 282-283): f32 accumulation per the restated ndarray contract, f64 finish.
+ * k <= 64 in this revision. *out_n = min(k, n). */
+int sdbv_knn_bruteforce(sdbv_ctx *, uint64_t table, const float *q, uint32_t d,
+                        uint32_t k, uint64_t *out_ids, double *out_dists,
+                        uint32_t *out_n);
+
+/* Batched-query brute-force: Q is b x d row-major; out_ids/out_dists are
+ * b x k row-major. The dense Q x corpus^T product runs on MFMA. */
+int sdbv_knn_batch(sdbv_ctx *, uint64_t table, const float *Q, uint32_t b,
+                   uint32_t d, uint32_t k, uint64_t *out_ids,
+                   double *out_dists);
+
+/* ---- HNSW (graph search; graph topology stays host-side) ---- */
+
+/* Batched gather+distance for HNSW layer-0 neighbour expansion
+ * (replaces the per-visit get_vector + Distance::calculate of
+ * layer.rs:184-223): for frontier rows[i] (row indices into the staged
+ * table), out_dists[i] = distance(q, row). */
+int sdbv_gather_distance(sdbv_ctx *, uint64_t table, const uint32_t *rows,
+                         uint32_t nrows, const float *q, uint32_t d,
+                         double *out_dists);
+
+/* Fully-on-GPU HNSW ef-search over a device-resident graph (CSR layer 0 +
+ * host-provided upper layers), one query per workgroup. Staged separately
+ * from the brute-force store. See DESIGN.md. */
+typedef struct sdbv_hnsw sdbv_hnsw;
+int sdbv_hnsw_upload(sdbv_ctx *, uint64_t table,
+                     const uint32_t *l0_offsets /* n+1 */,
+                     const uint32_t *l0_edges /* offsets[n] */, uint64_t n,
+                     uint32_t entry_point, sdbv_hnsw **out);
+int sdbv_hnsw_search(sdbv_hnsw *, const float *Q /* b x d */, uint32_t b,
+                     uint32_t d, uint32_t k, uint32_t ef,
+                     const uint32_t *entry_rows /* b entry rows (from host
+                     upper-layer descent), NULL => uploaded entry_point */,
+                     uint64_t *out_ids, double *out_dists, uint32_t *out_n);
+void sdbv_hnsw_free(sdbv_hnsw *);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* SDBV_H */

@@ -1,0 +1,238 @@
+# oracle — TEST INFRASTRUCTURE ONLY.
+# ctypes wrapper around liborcl.so (the CPU restatement of the reference's
+# vector-KNN semantics). Only tests/, __graft_entry__.smoke() and bench.py's
+# cpu_baseline leg may import this package. The product path (surrealdb_amd)
+# never touches it.
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "liborcl.so")
+
+METRICS = {
+    "cosine": 0,
+    "euclidean": 1,
+    "manhattan": 2,
+    "chebyshev": 3,
+    "hamming": 4,
+    "jaccard": 5,
+    "minkowski": 6,
+    "pearson": 7,
+}
+
+
+def build():
+    subprocess.run(["make", "-C", _DIR, "-s"], check=True)
+
+
+def _load():
+    if not os.path.exists(_SO):
+        build()
+    lib = ctypes.CDLL(_SO)
+    u64, u32, u8, f32p, f64p = (
+        ctypes.c_uint64,
+        ctypes.c_uint32,
+        ctypes.c_uint8,
+        ctypes.POINTER(ctypes.c_float),
+        ctypes.POINTER(ctypes.c_double),
+    )
+    u64p = ctypes.POINTER(ctypes.c_uint64)
+    u32p = ctypes.POINTER(ctypes.c_uint32)
+    lib.orc_gen_elem.restype = ctypes.c_float
+    lib.orc_gen_elem.argtypes = [u64, u64]
+    lib.orc_gen_f32.argtypes = [u64, u64, u64, u32, f32p]
+    lib.orc_dot_f32.restype = ctypes.c_float
+    lib.orc_dot_f32.argtypes = [f32p, f32p, u64]
+    lib.orc_dot_f64.restype = ctypes.c_double
+    lib.orc_dot_f64.argtypes = [f64p, f64p, u64]
+    lib.orc_sumsq_f32.restype = ctypes.c_float
+    lib.orc_sumsq_f32.argtypes = [f32p, u64]
+    lib.orc_sumsq_f64.restype = ctypes.c_double
+    lib.orc_sumsq_f64.argtypes = [f64p, u64]
+    for name in ("orc_dist_f32",):
+        fn = getattr(lib, name)
+        fn.restype = ctypes.c_double
+        fn.argtypes = [u8, ctypes.c_double, f32p, f32p, u64]
+    for name in ("orc_dist_f64", "orc_dist_number"):
+        fn = getattr(lib, name)
+        fn.restype = ctypes.c_double
+        fn.argtypes = [u8, ctypes.c_double, f64p, f64p, u64]
+    lib.orc_total_key.restype = u64
+    lib.orc_total_key.argtypes = [ctypes.c_double]
+    lib.orc_topk_f32.argtypes = [u8, ctypes.c_double, f32p, u64, u32, f32p, u32, u64p, f64p, u32p]
+    lib.orc_topk_number.argtypes = [u8, ctypes.c_double, f64p, u64, u32, f64p, u32, u64p, f64p, u32p]
+    lib.orc_topk_f32_mt.restype = ctypes.c_int
+    lib.orc_topk_f32_mt.argtypes = [
+        u8, ctypes.c_double, f32p, u64, u32, f32p, u32, u64p, f64p, u32p, ctypes.c_int,
+    ]
+    lib.orc_hnsw_new.restype = ctypes.c_void_p
+    lib.orc_hnsw_new.argtypes = [u32, u8, ctypes.c_double, u32, u32, u32,
+                                 ctypes.c_int, ctypes.c_int, u64, ctypes.c_double]
+    lib.orc_hnsw_free.argtypes = [ctypes.c_void_p]
+    lib.orc_hnsw_insert.argtypes = [ctypes.c_void_p, f32p]
+    lib.orc_hnsw_insert_level.argtypes = [ctypes.c_void_p, f32p, u32]
+    lib.orc_hnsw_search.restype = u32
+    lib.orc_hnsw_search.argtypes = [ctypes.c_void_p, f32p, u32, u32, u64p, f64p]
+    lib.orc_hnsw_check_props.restype = ctypes.c_int
+    lib.orc_hnsw_check_props.argtypes = [ctypes.c_void_p]
+    lib.orc_hnsw_num_layers.restype = u32
+    lib.orc_hnsw_num_layers.argtypes = [ctypes.c_void_p]
+    lib.orc_hnsw_num_elements.restype = u64
+    lib.orc_hnsw_num_elements.argtypes = [ctypes.c_void_p]
+    lib.orc_hnsw_l0_edge_count.restype = u64
+    lib.orc_hnsw_l0_edge_count.argtypes = [ctypes.c_void_p]
+    lib.orc_hnsw_l0_export.argtypes = [ctypes.c_void_p, u32p, u32p]
+    lib.orc_hnsw_search_ep.argtypes = [ctypes.c_void_p, f32p, u64p, f64p]
+    lib.orc_hnsw_entry_point.restype = ctypes.c_int64
+    lib.orc_hnsw_entry_point.argtypes = [ctypes.c_void_p]
+    return lib
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        _lib = _load()
+    return _lib
+
+
+def _f32p(a):
+    assert a.dtype == np.float32 and a.flags.c_contiguous
+    return a.ctypes.data_as(ctypes.POINTER(ctypes.c_float))
+
+
+def _f64p(a):
+    assert a.dtype == np.float64 and a.flags.c_contiguous
+    return a.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+
+
+def gen_f32(seed, row0, nrows, d):
+    """The committed synthetic-data contract (see sdbv_oracle.cpp)."""
+    out = np.empty((nrows, d), dtype=np.float32)
+    lib().orc_gen_f32(seed, row0, nrows, d, _f32p(out))
+    return out
+
+
+def dist_f32(metric, a, b, order=0.0):
+    return lib().orc_dist_f32(METRICS[metric], order, _f32p(a), _f32p(b), a.size)
+
+
+def dist_f64(metric, a, b, order=0.0):
+    return lib().orc_dist_f64(METRICS[metric], order, _f64p(a), _f64p(b), a.size)
+
+
+def dist_number(metric, a, b, order=0.0):
+    return lib().orc_dist_number(METRICS[metric], order, _f64p(a), _f64p(b), a.size)
+
+
+def topk_f32(metric, corpus, q, k, order=0.0):
+    n = corpus.shape[0]
+    ids = np.empty(k, dtype=np.uint64)
+    dists = np.empty(k, dtype=np.float64)
+    out_n = ctypes.c_uint32(0)
+    lib().orc_topk_f32(
+        METRICS[metric], order, _f32p(corpus), n, corpus.shape[1], _f32p(q), k,
+        ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.byref(out_n))
+    m = out_n.value
+    return ids[:m], dists[:m]
+
+
+def topk_number(metric, corpus, q, k, order=0.0):
+    n = corpus.shape[0]
+    ids = np.empty(k, dtype=np.uint64)
+    dists = np.empty(k, dtype=np.float64)
+    out_n = ctypes.c_uint32(0)
+    lib().orc_topk_number(
+        METRICS[metric], order, _f64p(corpus), n, corpus.shape[1], _f64p(q), k,
+        ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.byref(out_n))
+    m = out_n.value
+    return ids[:m], dists[:m]
+
+
+def topk_f32_mt(metric, corpus, q, k, order=0.0, nthreads=0):
+    n = corpus.shape[0]
+    ids = np.empty(k, dtype=np.uint64)
+    dists = np.empty(k, dtype=np.float64)
+    out_n = ctypes.c_uint32(0)
+    used = lib().orc_topk_f32_mt(
+        METRICS[metric], order, _f32p(corpus), n, corpus.shape[1], _f32p(q), k,
+        ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.byref(out_n), nthreads)
+    m = out_n.value
+    return ids[:m], dists[:m], used
+
+
+class Hnsw:
+    """Oracle HNSW (restates hnsw/mod.rs + layer.rs + heuristic.rs)."""
+
+    def __init__(self, d, metric="euclidean", order=0.0, m=12, m0=None,
+                 efc=150, extend=False, keep=False, seed=0x5DB1, ml=None):
+        import math
+        if m0 is None:
+            m0 = 2 * m
+        if ml is None:
+            ml = 1.0 / math.log(m)
+        self._h = lib().orc_hnsw_new(d, METRICS[metric], order, m, m0, efc,
+                                     int(extend), int(keep), seed, ml)
+        self.d = d
+
+    def insert(self, pt, level=None):
+        pt = np.ascontiguousarray(pt, dtype=np.float32)
+        if level is None:
+            lib().orc_hnsw_insert(self._h, _f32p(pt))
+        else:
+            lib().orc_hnsw_insert_level(self._h, _f32p(pt), level)
+
+    def search(self, q, k, ef):
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        cap = max(k, ef)
+        ids = np.empty(cap, dtype=np.uint64)
+        dists = np.empty(cap, dtype=np.float64)
+        n = lib().orc_hnsw_search(
+            self._h, _f32p(q), k, ef,
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        return ids[:n], dists[:n]
+
+    def search_ep(self, q):
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        ep = ctypes.c_uint64(0)
+        dist = ctypes.c_double(0)
+        lib().orc_hnsw_search_ep(self._h, _f32p(q), ctypes.byref(ep), ctypes.byref(dist))
+        return ep.value, dist.value
+
+    def check_props(self):
+        return lib().orc_hnsw_check_props(self._h)
+
+    def num_layers(self):
+        return lib().orc_hnsw_num_layers(self._h)
+
+    def l0_csr(self):
+        n = lib().orc_hnsw_num_elements(self._h)
+        ec = lib().orc_hnsw_l0_edge_count(self._h)
+        offsets = np.empty(n + 1, dtype=np.uint32)
+        edges = np.empty(max(ec, 1), dtype=np.uint32)
+        lib().orc_hnsw_l0_export(
+            self._h,
+            offsets.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+            edges.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)))
+        return offsets, edges[:ec]
+
+    def entry_point(self):
+        return lib().orc_hnsw_entry_point(self._h)
+
+    def __del__(self):
+        try:
+            lib().orc_hnsw_free(self._h)
+        except Exception:
+            pass

@@ -1,0 +1,948 @@
+// sdbv_oracle.cpp — TEST INFRASTRUCTURE ONLY.
+//
+// CPU restatement of the surrealdb/surrealdb vector-KNN hot path, used as the
+// parity oracle and as bench.py's cpu_baseline leg (kind "port"). It is NEVER
+// the product path: only tests/, __graft_entry__.smoke() and bench.py's
+// cpu_baseline may call into this library. The product path is the HIP
+// extension (surrealdb_amd/csrc) and must fail loudly when that is missing.
+//
+// Every function cites the reference file:line it restates (reference:
+// /root/reference/surrealdb @ 2026-05-29, Rust — not compilable in this
+// environment, no cargo/rustc; see DESIGN.md "Oracle pinning").
+//
+// Third-party arithmetic restated here (sources NOT vendored in the
+// reference; pinned by Cargo.lock):
+//  - ndarray 0.17.2 `unrolled_dot` / `unrolled_fold` (numeric_util.rs):
+//    eightfold-unrolled accumulation, restated in orc_dot_* / orc_sumsq_*.
+//    Call sites: vector.rs:237 (a.dot(b)), vector.rs:238-239 ((a*a).sum()).
+//  - ndarray-stats 0.7.0 DeviationExt l2_dist/l1_dist/linf_dist: sequential
+//    elementwise folds, final .to_f64().sqrt() for l2. Call sites:
+//    vector.rs:282-285 (euclidean), :379-382 (manhattan), :220-229 (chebyshev).
+// The reference's own exact-constant tests (vector.rs:723-772) pin results
+// across this boundary; residual last-ulp ambiguity (ndarray's pairwise-sum
+// threshold for `.sum()`) is documented in DESIGN.md and bounded far below
+// the 1e-5 relative score tolerance of the parity contract.
+//
+// Build: see oracle/Makefile. Compiled WITHOUT -ffast-math: every fp op here
+// is IEEE and order-preserving by construction.
+
+#include <cstdint>
+#include <cstring>
+#include <cfloat>
+#include <cmath>
+#include <cstdlib>
+#include <vector>
+#include <deque>
+#include <map>
+#include <set>
+#include <unordered_set>
+#include <algorithm>
+
+#if defined(_OPENMP)
+#include <omp.h>
+#endif
+
+extern "C" {
+
+// ---------------------------------------------------------------------------
+// Deterministic synthetic generator (committed data contract).
+// element(seed, gidx) for global element index gidx = row * d + j:
+//   splitmix64(seed + gidx) -> u64 -> f64 in [0,1) -> [-20, 20) -> f32.
+// Matches the reference's RandomItemGenerator::Float(-20.0, 20.0) by
+// distribution (knn.rs:633-643); the exact bit-stream is defined HERE and
+// replicated bit-identically by the device generator in
+// surrealdb_amd/csrc/scan_kernels.hip.
+// ---------------------------------------------------------------------------
+static inline uint64_t splitmix64(uint64_t z) {
+	z += 0x9E3779B97F4A7C15ULL;
+	z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+	z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+	return z ^ (z >> 31);
+}
+
+float orc_gen_elem(uint64_t seed, uint64_t gidx) {
+	uint64_t x = splitmix64(seed + gidx);
+	double u = (double)(x >> 11) * 0x1.0p-53; // [0,1)
+	return (float)(-20.0 + 40.0 * u);
+}
+
+void orc_gen_f32(uint64_t seed, uint64_t row0, uint64_t nrows, uint32_t d,
+                 float *out) {
+	for (uint64_t i = 0; i < nrows; i++)
+		for (uint32_t j = 0; j < d; j++)
+			out[i * d + j] = orc_gen_elem(seed, (row0 + i) * (uint64_t)d + j);
+}
+
+// ---------------------------------------------------------------------------
+// ndarray 0.17.2 numeric_util::unrolled_dot restatement (f32 / f64).
+// Eight independent partials p0..p7, each p += x*y (mul then add, NO fma),
+// combined as sum=0; sum+=(p0+p4); sum+=(p1+p5); sum+=(p2+p6); sum+=(p3+p7);
+// then sequential tail. Used by Array1::dot on contiguous arrays (no BLAS
+// feature in the reference's Cargo.lock).
+// ---------------------------------------------------------------------------
+#define ORC_UNROLLED_DOT(T)                                                  \
+	T p0 = 0, p1 = 0, p2 = 0, p3 = 0, p4 = 0, p5 = 0, p6 = 0, p7 = 0;       \
+	uint64_t i = 0;                                                          \
+	for (; i + 8 <= d; i += 8) {                                             \
+		p0 += a[i + 0] * b[i + 0];                                           \
+		p1 += a[i + 1] * b[i + 1];                                           \
+		p2 += a[i + 2] * b[i + 2];                                           \
+		p3 += a[i + 3] * b[i + 3];                                           \
+		p4 += a[i + 4] * b[i + 4];                                           \
+		p5 += a[i + 5] * b[i + 5];                                           \
+		p6 += a[i + 6] * b[i + 6];                                           \
+		p7 += a[i + 7] * b[i + 7];                                           \
+	}                                                                        \
+	T sum = 0;                                                               \
+	sum += (p0 + p4);                                                        \
+	sum += (p1 + p5);                                                        \
+	sum += (p2 + p6);                                                        \
+	sum += (p3 + p7);                                                        \
+	for (; i < d; i++)                                                       \
+		sum += a[i] * b[i];                                                  \
+	return sum;
+
+float orc_dot_f32(const float *a, const float *b, uint64_t d) {
+	ORC_UNROLLED_DOT(float)
+}
+double orc_dot_f64(const double *a, const double *b, uint64_t d) {
+	ORC_UNROLLED_DOT(double)
+}
+
+// ndarray unrolled_fold restatement for (a*a).sum() (vector.rs:238-239,
+// :246-247): same eightfold partials over x*x, combined as
+// acc += ((p0+p4)+(p1+p5)); acc += ((p2+p6)+(p3+p7)); sequential tail.
+// (ndarray's `.sum()` switches to pairwise blocks above an internal
+// threshold; the delta is last-ulp on the norm — see DESIGN.md.)
+#define ORC_UNROLLED_SUMSQ(T)                                                \
+	T p0 = 0, p1 = 0, p2 = 0, p3 = 0, p4 = 0, p5 = 0, p6 = 0, p7 = 0;       \
+	uint64_t i = 0;                                                          \
+	for (; i + 8 <= d; i += 8) {                                             \
+		p0 += a[i + 0] * a[i + 0];                                           \
+		p1 += a[i + 1] * a[i + 1];                                           \
+		p2 += a[i + 2] * a[i + 2];                                           \
+		p3 += a[i + 3] * a[i + 3];                                           \
+		p4 += a[i + 4] * a[i + 4];                                           \
+		p5 += a[i + 5] * a[i + 5];                                           \
+		p6 += a[i + 6] * a[i + 6];                                           \
+		p7 += a[i + 7] * a[i + 7];                                           \
+	}                                                                        \
+	T acc = 0;                                                               \
+	acc += ((p0 + p4) + (p1 + p5));                                          \
+	acc += ((p2 + p6) + (p3 + p7));                                          \
+	for (; i < d; i++)                                                       \
+		acc += a[i] * a[i];                                                  \
+	return acc;
+
+float orc_sumsq_f32(const float *a, uint64_t d) { ORC_UNROLLED_SUMSQ(float) }
+double orc_sumsq_f64(const double *a, uint64_t d) { ORC_UNROLLED_SUMSQ(double) }
+
+// ---------------------------------------------------------------------------
+// Typed distances — Distance::calculate (vector.rs:660-672) on F32/F64.
+// metric codes match include/sdbv.h + extensions used only by tests.
+// ---------------------------------------------------------------------------
+enum {
+	M_COSINE = 0,
+	M_EUCLIDEAN = 1,
+	M_MANHATTAN = 2,
+	M_CHEBYSHEV = 3,
+	M_HAMMING = 4,
+	M_JACCARD = 5,
+	M_MINKOWSKI = 6,
+	M_PEARSON = 7,
+};
+
+// vector.rs:244-249 cosine_distance_f32: f32 dot -> f64; f32 sumsq -> f64 sqrt.
+double orc_cosine_f32(const float *a, const float *b, uint64_t d) {
+	double dot = (double)orc_dot_f32(a, b, d);
+	double na = std::sqrt((double)orc_sumsq_f32(a, d));
+	double nb = std::sqrt((double)orc_sumsq_f32(b, d));
+	return 1.0 - dot / (na * nb);
+}
+
+// vector.rs:236-241 cosine_distance_f64.
+double orc_cosine_f64(const double *a, const double *b, uint64_t d) {
+	double dot = orc_dot_f64(a, b, d);
+	double na = std::sqrt(orc_sumsq_f64(a, d));
+	double nb = std::sqrt(orc_sumsq_f64(b, d));
+	return 1.0 - dot / (na * nb);
+}
+
+// vector.rs:282-283 euclidean via ndarray-stats l2_dist: sequential f32
+// accumulation of (a-b)^2, then f64 sqrt (sq_l2_dist returns A, l2_dist
+// converts to f64 before sqrt).
+double orc_euclidean_f32(const float *a, const float *b, uint64_t d) {
+	float acc = 0.0f;
+	for (uint64_t i = 0; i < d; i++) {
+		float diff = a[i] - b[i];
+		acc += diff * diff;
+	}
+	return std::sqrt((double)acc);
+}
+double orc_euclidean_f64(const double *a, const double *b, uint64_t d) {
+	double acc = 0.0;
+	for (uint64_t i = 0; i < d; i++) {
+		double diff = a[i] - b[i];
+		acc += diff * diff;
+	}
+	return std::sqrt(acc);
+}
+
+// vector.rs:379-382 manhattan via l1_dist (f32 accumulate, then as f64).
+static double orc_manhattan_f32(const float *a, const float *b, uint64_t d) {
+	float acc = 0.0f;
+	for (uint64_t i = 0; i < d; i++)
+		acc += std::fabs(a[i] - b[i]);
+	return (double)acc;
+}
+static double orc_manhattan_f64(const double *a, const double *b, uint64_t d) {
+	double acc = 0.0;
+	for (uint64_t i = 0; i < d; i++)
+		acc += std::fabs(a[i] - b[i]);
+	return acc;
+}
+
+// vector.rs:220-229 chebyshev via linf_dist (max |a-b| in f32, then as f64).
+static double orc_chebyshev_f32(const float *a, const float *b, uint64_t d) {
+	float m = 0.0f;
+	for (uint64_t i = 0; i < d; i++)
+		m = std::fmax(m, std::fabs(a[i] - b[i]));
+	return (double)m;
+}
+static double orc_chebyshev_f64(const double *a, const double *b, uint64_t d) {
+	double m = 0.0;
+	for (uint64_t i = 0; i < d; i++)
+		m = std::fmax(m, std::fabs(a[i] - b[i]));
+	return m;
+}
+
+// vector.rs:292-303 hamming: count of element-wise != (fp compare).
+#define ORC_HAMMING_DEF(NAME, T)                                             \
+	static double NAME(const T *a, const T *b, uint64_t d) {                 \
+		uint64_t acc = 0;                                                    \
+		for (uint64_t i = 0; i < d; i++)                                     \
+			if (a[i] != b[i])                                                \
+				acc++;                                                       \
+		return (double)acc;                                                  \
+	}
+ORC_HAMMING_DEF(orc_hamming_f32, float)
+ORC_HAMMING_DEF(orc_hamming_f64, double)
+
+// vector.rs:317-340 jaccard on bit-pattern sets. NOTE the reference asymmetry
+// restated faithfully: F64 returns 1 - |I|/|U| (:326), F32 returns |I|/|U|
+// (:339).
+static double orc_jaccard_f32(const float *a, const float *b, uint64_t d) {
+	std::unordered_set<uint32_t> u;
+	for (uint64_t i = 0; i < d; i++) {
+		uint32_t bits;
+		std::memcpy(&bits, &a[i], 4);
+		u.insert(bits);
+	}
+	uint64_t inter = 0;
+	for (uint64_t i = 0; i < d; i++) {
+		uint32_t bits;
+		std::memcpy(&bits, &b[i], 4);
+		if (!u.insert(bits).second)
+			inter++;
+	}
+	return (double)inter / (double)u.size();
+}
+static double orc_jaccard_f64(const double *a, const double *b, uint64_t d) {
+	std::unordered_set<uint64_t> u;
+	for (uint64_t i = 0; i < d; i++) {
+		uint64_t bits;
+		std::memcpy(&bits, &a[i], 8);
+		u.insert(bits);
+	}
+	uint64_t inter = 0;
+	for (uint64_t i = 0; i < d; i++) {
+		uint64_t bits;
+		std::memcpy(&bits, &b[i], 8);
+		if (!u.insert(bits).second)
+			inter++;
+	}
+	return 1.0 - (double)inter / (double)u.size();
+}
+
+// vector.rs:389-399 minkowski: per-element to_float() -> f64 diff, abs, powf,
+// sequential sum, final pow(1/order).
+#define ORC_MINKOWSKI_DEF(NAME, T)                                           \
+	static double NAME(const T *a, const T *b, uint64_t d, double order) {   \
+		double acc = 0.0;                                                    \
+		for (uint64_t i = 0; i < d; i++)                                     \
+			acc += std::pow(std::fabs((double)a[i] - (double)b[i]), order);  \
+		return std::pow(acc, 1.0 / order);                                   \
+	}
+ORC_MINKOWSKI_DEF(orc_minkowski_f32, float)
+ORC_MINKOWSKI_DEF(orc_minkowski_f64, double)
+
+// vector.rs:413-440 pearson (similarity used directly as the metric value).
+// mean: ndarray mean = sum/len in the element type, then .to_float().
+static double orc_pearson_f32(const float *a, const float *b, uint64_t d) {
+	float sa = 0.0f, sb = 0.0f;
+	{ // ndarray .mean() -> .sum()/len in f32; restated as unrolled_fold sum
+		const float *x = a;
+		float p0 = 0, p1 = 0, p2 = 0, p3 = 0, p4 = 0, p5 = 0, p6 = 0, p7 = 0;
+		uint64_t i = 0;
+		for (; i + 8 <= d; i += 8) {
+			p0 += x[i]; p1 += x[i + 1]; p2 += x[i + 2]; p3 += x[i + 3];
+			p4 += x[i + 4]; p5 += x[i + 5]; p6 += x[i + 6]; p7 += x[i + 7];
+		}
+		sa += ((p0 + p4) + (p1 + p5));
+		sa += ((p2 + p6) + (p3 + p7));
+		for (; i < d; i++) sa += x[i];
+	}
+	{
+		const float *x = b;
+		float p0 = 0, p1 = 0, p2 = 0, p3 = 0, p4 = 0, p5 = 0, p6 = 0, p7 = 0;
+		uint64_t i = 0;
+		for (; i + 8 <= d; i += 8) {
+			p0 += x[i]; p1 += x[i + 1]; p2 += x[i + 2]; p3 += x[i + 3];
+			p4 += x[i + 4]; p5 += x[i + 5]; p6 += x[i + 6]; p7 += x[i + 7];
+		}
+		sb += ((p0 + p4) + (p1 + p5));
+		sb += ((p2 + p6) + (p3 + p7));
+		for (; i < d; i++) sb += x[i];
+	}
+	double mx = (double)(sa / (float)d);
+	double my = (double)(sb / (float)d);
+	double sum_xy = 0, sum_x2 = 0, sum_y2 = 0;
+	for (uint64_t i = 0; i < d; i++) {
+		double dx = (double)a[i] - mx, dy = (double)b[i] - my;
+		sum_xy += dx * dy;
+		sum_x2 += dx * dx;
+		sum_y2 += dy * dy;
+	}
+	double den = std::sqrt(sum_x2 * sum_y2);
+	if (den == 0.0) return 0.0;
+	return sum_xy / den;
+}
+static double orc_pearson_f64(const double *a, const double *b, uint64_t d) {
+	double sa = 0, sb = 0;
+	for (uint64_t i = 0; i < d; i++) sa += a[i];
+	for (uint64_t i = 0; i < d; i++) sb += b[i];
+	double mx = sa / (double)d, my = sb / (double)d;
+	double sum_xy = 0, sum_x2 = 0, sum_y2 = 0;
+	for (uint64_t i = 0; i < d; i++) {
+		double dx = a[i] - mx, dy = b[i] - my;
+		sum_xy += dx * dy;
+		sum_x2 += dx * dx;
+		sum_y2 += dy * dy;
+	}
+	double den = std::sqrt(sum_x2 * sum_y2);
+	if (den == 0.0) return 0.0;
+	return sum_xy / den;
+}
+
+// Distance::calculate dispatch (vector.rs:660-672) for F32 / F64 vectors.
+double orc_dist_f32(uint8_t metric, double order, const float *a,
+                    const float *b, uint64_t d) {
+	switch (metric) {
+	case M_COSINE: return orc_cosine_f32(a, b, d);
+	case M_EUCLIDEAN: return orc_euclidean_f32(a, b, d);
+	case M_MANHATTAN: return orc_manhattan_f32(a, b, d);
+	case M_CHEBYSHEV: return orc_chebyshev_f32(a, b, d);
+	case M_HAMMING: return orc_hamming_f32(a, b, d);
+	case M_JACCARD: return orc_jaccard_f32(a, b, d);
+	case M_MINKOWSKI: return orc_minkowski_f32(a, b, d, order);
+	case M_PEARSON: return orc_pearson_f32(a, b, d);
+	}
+	return NAN;
+}
+double orc_dist_f64(uint8_t metric, double order, const double *a,
+                    const double *b, uint64_t d) {
+	switch (metric) {
+	case M_COSINE: return orc_cosine_f64(a, b, d);
+	case M_EUCLIDEAN: return orc_euclidean_f64(a, b, d);
+	case M_MANHATTAN: return orc_manhattan_f64(a, b, d);
+	case M_CHEBYSHEV: return orc_chebyshev_f64(a, b, d);
+	case M_HAMMING: return orc_hamming_f64(a, b, d);
+	case M_JACCARD: return orc_jaccard_f64(a, b, d);
+	case M_MINKOWSKI: return orc_minkowski_f64(a, b, d, order);
+	case M_PEARSON: return orc_pearson_f64(a, b, d);
+	}
+	return NAN;
+}
+
+// ---------------------------------------------------------------------------
+// Number-path distances — Distance::compute (catalog/schema/index.rs:287-301
+// -> fnc/util/math/vector.rs) over Vec<Number> where every element is
+// Number::Float(f64): plain sequential f64 arithmetic.
+//  - cosine: dot (fnc vector.rs:279-281, sequential sum of products, the
+//    Number Sum starts at Int(0) and promotes exactly) / magnitudes
+//    (:301-313, sequential sum of squares, f64 sqrt).
+//  - euclidean: (:288-298) sequential sum of (a-b)^2, f64 sqrt.
+// ---------------------------------------------------------------------------
+double orc_dist_number(uint8_t metric, double order, const double *a,
+                       const double *b, uint64_t d) {
+	switch (metric) {
+	case M_COSINE: {
+		double dot = 0, ma = 0, mb = 0;
+		for (uint64_t i = 0; i < d; i++) dot += a[i] * b[i];
+		for (uint64_t i = 0; i < d; i++) ma += a[i] * a[i];
+		for (uint64_t i = 0; i < d; i++) mb += b[i] * b[i];
+		return 1.0 - dot / (std::sqrt(ma) * std::sqrt(mb));
+	}
+	case M_EUCLIDEAN: {
+		double acc = 0;
+		for (uint64_t i = 0; i < d; i++) {
+			double diff = a[i] - b[i];
+			acc += diff * diff;
+		}
+		return std::sqrt(acc);
+	}
+	case M_MANHATTAN: { // fnc vector.rs:153-158
+		double acc = 0;
+		for (uint64_t i = 0; i < d; i++) acc += std::fabs(a[i] - b[i]);
+		return acc;
+	}
+	case M_CHEBYSHEV: { // fnc vector.rs:218-228 (fold starts at f64::MIN)
+		double m = -DBL_MAX;
+		for (uint64_t i = 0; i < d; i++) m = std::fmax(m, std::fabs(a[i] - b[i]));
+		return m;
+	}
+	case M_MINKOWSKI: {
+		double acc = 0;
+		for (uint64_t i = 0; i < d; i++)
+			acc += std::pow(std::fabs(a[i] - b[i]), order);
+		return std::pow(acc, 1.0 / order);
+	}
+	case M_HAMMING: return orc_hamming_f64(a, b, d);
+	case M_JACCARD: { // fnc vector.rs:120-126: Number-equality sets, sim only
+		std::unordered_set<uint64_t> u;
+		for (uint64_t i = 0; i < d; i++) {
+			uint64_t bits; std::memcpy(&bits, &a[i], 8); u.insert(bits);
+		}
+		uint64_t inter = 0;
+		for (uint64_t i = 0; i < d; i++) {
+			uint64_t bits; std::memcpy(&bits, &b[i], 8);
+			if (!u.insert(bits).second) inter++;
+		}
+		return (double)inter / (double)u.size();
+	}
+	case M_PEARSON: { // fnc vector.rs:132-147
+		double sa = 0, sb = 0;
+		for (uint64_t i = 0; i < d; i++) sa += a[i];
+		for (uint64_t i = 0; i < d; i++) sb += b[i];
+		double m1 = sa / (double)d, m2 = sb / (double)d;
+		double covar = 0;
+		for (uint64_t i = 0; i < d; i++) covar += (a[i] - m1) * (b[i] - m2);
+		covar /= (double)d;
+		double v1 = 0, v2 = 0;
+		for (uint64_t i = 0; i < d; i++) v1 += (a[i] - m1) * (a[i] - m1);
+		for (uint64_t i = 0; i < d; i++) v2 += (b[i] - m2) * (b[i] - m2);
+		double s1 = std::sqrt(v1 / (double)d), s2 = std::sqrt(v2 / (double)d);
+		return covar / (s1 * s2);
+	}
+	}
+	return NAN;
+}
+
+// ---------------------------------------------------------------------------
+// f64::total_cmp sort key (knn.rs:128-160 FloatKey): IEEE-754 totalOrder.
+// Monotone u64 key: negative -> ~bits, else bits | sign.
+// ---------------------------------------------------------------------------
+uint64_t orc_total_key(double x) {
+	uint64_t bits;
+	std::memcpy(&bits, &x, 8);
+	if (bits >> 63)
+		return ~bits;
+	return bits | 0x8000000000000000ULL;
+}
+
+// ---------------------------------------------------------------------------
+// Brute-force top-K — KnnTopK semantics (exec/operators/knn_topk.rs:166-267):
+// bounded heap of size k; when full, insert only if STRICTLY closer than the
+// current worst (ties keep the earlier row, :216-227); final order ascending
+// (distance, insertion seq) (:61-73, :230-236). For ordinal ids this equals
+// the KnnResultBuilder ordering (knn.rs:363-437): (dist total_cmp asc, id asc).
+// mode: 0 = typed-F32 distances (Distance::calculate), 1 = Number-path on the
+// same values widened to f64 (Distance::compute).
+// ---------------------------------------------------------------------------
+struct TopK {
+	uint32_t k;
+	// sorted ascending by (total_key(dist), seq)
+	std::vector<std::pair<std::pair<uint64_t, uint64_t>, double>> v;
+	void push(double dist, uint64_t seq) {
+		std::pair<uint64_t, uint64_t> key{orc_total_key(dist), seq};
+		if (v.size() >= k) {
+			// knn_topk.rs:218-219: insert only if dist < worst dist
+			// (strict, distance only — seq never displaces on tie)
+			if (!(dist < v.back().second))
+				return;
+			v.pop_back();
+		}
+		auto it = std::lower_bound(
+		    v.begin(), v.end(), key,
+		    [](const auto &e, const std::pair<uint64_t, uint64_t> &kk) {
+			    return e.first < kk;
+		    });
+		v.insert(it, {key, dist});
+	}
+};
+
+void orc_topk_f32(uint8_t metric, double order, const float *corpus,
+                  uint64_t n, uint32_t d, const float *q, uint32_t k,
+                  uint64_t *out_ids, double *out_dists, uint32_t *out_n) {
+	TopK t{k, {}};
+	for (uint64_t r = 0; r < n; r++)
+		t.push(orc_dist_f32(metric, order, q, corpus + r * d, d), r);
+	*out_n = (uint32_t)t.v.size();
+	for (size_t i = 0; i < t.v.size(); i++) {
+		out_ids[i] = t.v[i].first.second;
+		out_dists[i] = t.v[i].second;
+	}
+}
+
+void orc_topk_number(uint8_t metric, double order, const double *corpus,
+                     uint64_t n, uint32_t d, const double *q, uint32_t k,
+                     uint64_t *out_ids, double *out_dists, uint32_t *out_n) {
+	TopK t{k, {}};
+	for (uint64_t r = 0; r < n; r++)
+		t.push(orc_dist_number(metric, order, q, corpus + r * d, d), r);
+	*out_n = (uint32_t)t.v.size();
+	for (size_t i = 0; i < t.v.size(); i++) {
+		out_ids[i] = t.v[i].first.second;
+		out_dists[i] = t.v[i].second;
+	}
+}
+
+// Multithreaded (OpenMP) variant for bench.py's cpu_baseline leg: row-chunked,
+// per-thread TopK, merged with the same (dist, id) ascending contract.
+// Returns threads used.
+int orc_topk_f32_mt(uint8_t metric, double order, const float *corpus,
+                    uint64_t n, uint32_t d, const float *q, uint32_t k,
+                    uint64_t *out_ids, double *out_dists, uint32_t *out_n,
+                    int nthreads) {
+#if defined(_OPENMP)
+	if (nthreads <= 0)
+		nthreads = omp_get_max_threads();
+	std::vector<TopK> parts((size_t)nthreads, TopK{k, {}});
+#pragma omp parallel num_threads(nthreads)
+	{
+		int tid = omp_get_thread_num();
+		TopK &t = parts[(size_t)tid];
+#pragma omp for schedule(static)
+		for (int64_t r = 0; r < (int64_t)n; r++)
+			t.push(orc_dist_f32(metric, order, q, corpus + (uint64_t)r * d, d),
+			       (uint64_t)r);
+	}
+	std::vector<std::pair<double, uint64_t>> all;
+	for (auto &p : parts)
+		for (auto &e : p.v)
+			all.push_back({e.second, e.first.second});
+	std::sort(all.begin(), all.end(), [](const auto &x, const auto &y) {
+		auto kx = std::make_pair(orc_total_key(x.first), x.second);
+		auto ky = std::make_pair(orc_total_key(y.first), y.second);
+		return kx < ky;
+	});
+	uint32_t m = (uint32_t)std::min<size_t>(k, all.size());
+	*out_n = m;
+	for (uint32_t i = 0; i < m; i++) {
+		out_ids[i] = all[i].second;
+		out_dists[i] = all[i].first;
+	}
+	return nthreads;
+#else
+	orc_topk_f32(metric, order, corpus, n, d, q, k, out_ids, out_dists, out_n);
+	return 1;
+#endif
+}
+
+// ---------------------------------------------------------------------------
+// HNSW restatement — hnsw/mod.rs + layer.rs + heuristic.rs.
+// Element vectors are F32 (the reference's default VectorType, define.rs:1107).
+// DynamicSet is restated as an insertion-ordered vector set (the reference's
+// ArraySet semantics, dynamicset.rs; its AHashSet variant iterates in
+// nondeterministic order, so graph-level bitwise parity is not defined by the
+// reference itself — parity is judged on result sets/recall, matching the
+// reference's own tests hnsw/mod.rs:1104-1184).
+// ---------------------------------------------------------------------------
+
+struct OrcPQ {
+	// DoublePriorityQueue (knn.rs:15-123): BTreeMap<total_cmp(dist), FIFO deque>
+	std::map<uint64_t, std::deque<uint64_t>> m;
+	std::map<uint64_t, double> dist_of; // representative f64 per key
+	size_t n = 0;
+	void push(double d, uint64_t id) {
+		uint64_t k = orc_total_key(d);
+		m[k].push_back(id);
+		dist_of[k] = d;
+		n++;
+	}
+	bool pop_first(double *d, uint64_t *id) {
+		if (m.empty()) return false;
+		auto it = m.begin();
+		*d = dist_of[it->first];
+		*id = it->second.front();
+		it->second.pop_front();
+		if (it->second.empty()) { dist_of.erase(it->first); m.erase(it); }
+		n--;
+		return true;
+	}
+	bool pop_last(double *d, uint64_t *id) {
+		if (m.empty()) return false;
+		auto it = std::prev(m.end());
+		*d = dist_of[it->first];
+		*id = it->second.back();
+		it->second.pop_back();
+		if (it->second.empty()) { dist_of.erase(it->first); m.erase(it); }
+		n--;
+		return true;
+	}
+	bool peek_first(double *d, uint64_t *id) const {
+		if (m.empty()) return false;
+		auto it = m.begin();
+		*d = dist_of.at(it->first);
+		*id = it->second.front();
+		return true;
+	}
+	double peek_last_dist(double fallback) const {
+		if (m.empty()) return fallback;
+		return dist_of.at(std::prev(m.end())->first);
+	}
+	std::vector<std::pair<double, uint64_t>> to_vec() const {
+		std::vector<std::pair<double, uint64_t>> v;
+		for (auto &e : m)
+			for (auto id : e.second)
+				v.push_back({dist_of.at(e.first), id});
+		return v;
+	}
+};
+
+struct OrcLayer {
+	// UndirectedGraph over insertion-ordered sets (graph.rs, dynamicset.rs)
+	std::vector<std::vector<uint32_t>> edges; // indexed by element id
+	uint32_t m_max;
+	bool has(uint64_t id) const {
+		return id < edges.size() && !(edges[id].size() == 1 && edges[id][0] == UINT32_MAX);
+	}
+};
+
+struct orc_hnsw {
+	uint32_t d;
+	uint8_t metric;
+	double order;
+	uint32_t m, m0, efc;
+	bool extend, keep;
+	double ml;
+	uint64_t rng_state;
+	std::vector<float> vecs;          // element id -> row (id * d)
+	std::vector<uint8_t> present;     // per-layer membership: layers[l] set
+	std::vector<OrcLayer> layers;     // [0] = layer0 (m0), [1..] upper (m)
+	int64_t enter_point = -1;
+	uint64_t next_id = 0;
+	// per-element max layer (for membership checks)
+	std::vector<int32_t> top_layer;
+};
+
+static double hdist(const orc_hnsw *h, const float *a, const float *b) {
+	return orc_dist_f32(h->metric, h->order, a, b, h->d);
+}
+static const float *hvec(const orc_hnsw *h, uint64_t id) {
+	return h->vecs.data() + id * h->d;
+}
+
+orc_hnsw *orc_hnsw_new(uint32_t d, uint8_t metric, double order, uint32_t m,
+                       uint32_t m0, uint32_t efc, int extend, int keep,
+                       uint64_t rng_seed, double ml) {
+	auto *h = new orc_hnsw();
+	h->d = d; h->metric = metric; h->order = order;
+	h->m = m; h->m0 = m0; h->efc = efc;
+	h->extend = extend != 0; h->keep = keep != 0;
+	h->ml = ml;
+	h->rng_state = rng_seed;
+	h->layers.push_back(OrcLayer{{}, m0});
+	return h;
+}
+void orc_hnsw_free(orc_hnsw *h) { delete h; }
+
+// hnsw/mod.rs:263-266 get_random_level: floor(-ln(U) * ml), U uniform [0,1).
+// RNG restated with splitmix64 (the reference seeds SmallRng from thread_rng —
+// nondeterministic; level SEQUENCE distribution is what matters).
+static uint32_t next_level(orc_hnsw *h) {
+	h->rng_state = splitmix64(h->rng_state);
+	double u = (double)(h->rng_state >> 11) * 0x1.0p-53;
+	if (u <= 0.0) u = 0x1.0p-53;
+	double lvl = std::floor(-std::log(u) * h->ml);
+	if (lvl < 0) lvl = 0;
+	if (lvl > 30) lvl = 30;
+	return (uint32_t)lvl;
+}
+
+// layer.rs:184-223 HnswLayer::search — the best-first ef-bounded loop.
+static void layer_search(const orc_hnsw *h, const OrcLayer &layer,
+                         const float *q, OrcPQ &candidates,
+                         std::unordered_set<uint64_t> &visited, OrcPQ &w,
+                         uint32_t ef) {
+	double fq_dist = w.peek_last_dist(1.7976931348623157e308);
+	double cq_dist; uint64_t doc;
+	while (candidates.pop_first(&cq_dist, &doc)) {
+		if (cq_dist > fq_dist)
+			break;
+		if (doc < layer.edges.size()) {
+			for (uint32_t e_id : layer.edges[doc]) {
+				if (!visited.insert(e_id).second)
+					continue;
+				double e_dist = hdist(h, hvec(h, e_id), q);
+				if (e_dist < fq_dist || w.n < ef) {
+					candidates.push(e_dist, e_id);
+					w.push(e_dist, e_id);
+					if (w.n > ef) {
+						double dd; uint64_t ii;
+						w.pop_last(&dd, &ii);
+					}
+					fq_dist = w.peek_last_dist(1.7976931348623157e308);
+				}
+			}
+		}
+	}
+}
+
+// heuristic.rs:193-216 is_closer.
+static bool is_closer(const orc_hnsw *h, double e_dist, uint64_t e_id,
+                      std::vector<uint32_t> &r) {
+	const float *ev = hvec(h, e_id);
+	for (uint32_t r_id : r) {
+		double r_dist = hdist(h, ev, hvec(h, r_id));
+		if (e_dist > r_dist)
+			return false;
+	}
+	r.push_back((uint32_t)e_id);
+	return true;
+}
+
+// heuristic.rs:61-82 heuristic / :84-116 heuristic_keep.
+static void heuristic_select(const orc_hnsw *h, const OrcLayer &layer,
+                             OrcPQ c, std::vector<uint32_t> &res, bool keep) {
+	uint32_t m_max = layer.m_max;
+	if (c.n <= m_max) {
+		for (auto &e : c.to_vec())
+			res.push_back((uint32_t)e.second);
+		return;
+	}
+	std::vector<uint64_t> pruned;
+	double e_dist; uint64_t e_id;
+	while (c.pop_first(&e_dist, &e_id)) {
+		if (is_closer(h, e_dist, e_id, res)) {
+			if (res.size() == m_max)
+				break;
+		} else if (keep) {
+			pruned.push_back(e_id);
+		}
+	}
+	if (keep) {
+		size_t nmore = m_max - res.size();
+		for (size_t i = 0; i < nmore && i < pruned.size(); i++)
+			res.push_back((uint32_t)pruned[i]);
+	}
+}
+
+// heuristic.rs:118-157 extend_candidates.
+static void extend_candidates(const orc_hnsw *h, const OrcLayer &layer,
+                              uint64_t q_id, const float *q_pt, OrcPQ &c) {
+	std::unordered_set<uint64_t> ex;
+	for (auto &e : c.to_vec())
+		ex.insert(e.second);
+	std::vector<std::pair<double, uint64_t>> ext;
+	for (auto &e : c.to_vec()) {
+		uint64_t e_id = e.second;
+		if (e_id < layer.edges.size()) {
+			for (uint32_t e_adj : layer.edges[e_id]) {
+				if (e_adj != q_id && ex.insert(e_adj).second) {
+					double dd = hdist(h, q_pt, hvec(h, e_adj));
+					ext.push_back({dd, e_adj});
+				}
+			}
+		}
+	}
+	for (auto &e : ext)
+		c.push(e.first, e.second);
+}
+
+static void select_neighbors(const orc_hnsw *h, const OrcLayer &layer,
+                             uint64_t q_id, const float *q_pt, OrcPQ c,
+                             std::vector<uint32_t> &res) {
+	if (h->extend)
+		extend_candidates(h, layer, q_id, q_pt, c);
+	heuristic_select(h, layer, std::move(c), res, h->keep);
+}
+
+// layer.rs:342-387 HnswLayer::insert (+ graph.rs add_node_and_bidirectional_edges)
+static OrcPQ layer_insert(orc_hnsw *h, OrcLayer &layer, uint32_t q_id,
+                          const float *q_pt, OrcPQ eps) {
+	// search_multi (layer.rs:151-161)
+	OrcPQ w = eps;
+	std::unordered_set<uint64_t> visited;
+	for (auto &e : eps.to_vec())
+		visited.insert(e.second);
+	layer_search(h, layer, q_pt, eps, visited, w, h->efc);
+	OrcPQ out_eps = w;
+
+	std::vector<uint32_t> neighbors;
+	select_neighbors(h, layer, q_id, q_pt, w, neighbors);
+
+	// add node + bidirectional edges
+	if (layer.edges.size() <= q_id)
+		layer.edges.resize(q_id + 1);
+	layer.edges[q_id] = neighbors;
+	for (uint32_t e_id : neighbors)
+		layer.edges[e_id].push_back(q_id);
+
+	// prune over-full neighbors (layer.rs:363-377)
+	for (uint32_t e_id : neighbors) {
+		auto &conn = layer.edges[e_id];
+		if (conn.size() > layer.m_max) {
+			const float *e_pt = hvec(h, e_id);
+			OrcPQ e_c; // build_priority_list (layer.rs:390-404)
+			for (uint32_t n_id : conn)
+				e_c.push(hdist(h, e_pt, hvec(h, n_id)), n_id);
+			std::vector<uint32_t> e_new;
+			select_neighbors(h, layer, e_id, e_pt, std::move(e_c), e_new);
+			layer.edges[e_id] = e_new;
+		}
+	}
+	return out_eps;
+}
+
+void orc_hnsw_insert_level(orc_hnsw *h, const float *pt, uint32_t q_level) {
+	uint32_t q_id = (uint32_t)h->next_id++;
+	h->vecs.insert(h->vecs.end(), pt, pt + h->d);
+	h->top_layer.push_back((int32_t)q_level);
+	uint32_t top_up_layers = (uint32_t)h->layers.size() - 1;
+
+	for (uint32_t i = top_up_layers; i < q_level; i++)
+		h->layers.push_back(OrcLayer{{}, h->m});
+
+	if (h->enter_point < 0) {
+		// insert_first_element (hnsw/mod.rs:272-291)
+		for (uint32_t l = 0; l <= q_level && l < (uint32_t)h->layers.size(); l++) {
+			if (h->layers[l].edges.size() <= q_id)
+				h->layers[l].edges.resize(q_id + 1);
+			h->layers[l].edges[q_id] = {};
+		}
+		h->enter_point = (int64_t)q_id;
+		return;
+	}
+
+	// insert_element (hnsw/mod.rs:296-378)
+	uint64_t ep_id = (uint64_t)h->enter_point;
+	double ep_dist = hdist(h, pt, hvec(h, ep_id));
+	if (q_level < top_up_layers) {
+		for (uint32_t l = top_up_layers; l > q_level; l--) {
+			// upper layers are h->layers[1..]; layer index l == graph layer l
+			OrcPQ cand; cand.push(ep_dist, ep_id);
+			std::unordered_set<uint64_t> visited{ep_id};
+			OrcPQ w = cand;
+			layer_search(h, h->layers[l], pt, cand, visited, w, 1);
+			double dd; uint64_t ii;
+			if (w.peek_first(&dd, &ii)) { ep_dist = dd; ep_id = ii; }
+		}
+	}
+	OrcPQ eps; eps.push(ep_dist, ep_id);
+	uint32_t insert_to = std::min(q_level, top_up_layers);
+	for (uint32_t l = insert_to; l >= 1; l--)
+		eps = layer_insert(h, h->layers[l], q_id, pt, std::move(eps));
+	layer_insert(h, h->layers[0], q_id, pt, std::move(eps));
+
+	for (uint32_t l = top_up_layers + 1; l <= q_level; l++) {
+		if (h->layers[l].edges.size() <= q_id)
+			h->layers[l].edges.resize(q_id + 1);
+		h->layers[l].edges[q_id] = {};
+	}
+	if (q_level > top_up_layers)
+		h->enter_point = (int64_t)q_id;
+}
+
+void orc_hnsw_insert(orc_hnsw *h, const float *pt) {
+	orc_hnsw_insert_level(h, pt, next_level(h));
+}
+
+// hnsw/mod.rs:459-482 knn_search + :521-548 search_ep.
+uint32_t orc_hnsw_search(orc_hnsw *h, const float *q, uint32_t k, uint32_t ef,
+                         uint64_t *out_ids, double *out_dists) {
+	if (h->enter_point < 0)
+		return 0;
+	uint64_t ep_id = (uint64_t)h->enter_point;
+	double ep_dist = hdist(h, q, hvec(h, ep_id));
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		OrcPQ cand; cand.push(ep_dist, ep_id);
+		std::unordered_set<uint64_t> visited{ep_id};
+		OrcPQ w = cand;
+		layer_search(h, h->layers[l], q, cand, visited, w, 1);
+		double dd; uint64_t ii;
+		if (w.peek_first(&dd, &ii)) { ep_dist = dd; ep_id = ii; }
+	}
+	OrcPQ cand; cand.push(ep_dist, ep_id);
+	std::unordered_set<uint64_t> visited{ep_id};
+	OrcPQ w = cand;
+	layer_search(h, h->layers[0], q, cand, visited, w, ef);
+	// to_vec_limit(k) (knn.rs:92-104)
+	auto v = w.to_vec();
+	uint32_t nout = (uint32_t)std::min<size_t>(k, v.size());
+	for (uint32_t i = 0; i < nout; i++) {
+		out_dists[i] = v[i].first;
+		out_ids[i] = v[i].second;
+	}
+	return nout;
+}
+
+uint32_t orc_hnsw_num_layers(orc_hnsw *h) { return (uint32_t)h->layers.size(); }
+uint64_t orc_hnsw_num_elements(orc_hnsw *h) { return h->next_id; }
+int64_t orc_hnsw_entry_point(orc_hnsw *h) { return h->enter_point; }
+
+// check_hnsw_props (hnsw/mod.rs:561-570): edge count <= m_max, no self-edges.
+// Returns 0 if OK, else a negative code.
+int orc_hnsw_check_props(orc_hnsw *h) {
+	for (size_t l = 0; l < h->layers.size(); l++) {
+		const OrcLayer &layer = h->layers[l];
+		for (size_t id = 0; id < layer.edges.size(); id++) {
+			if (layer.edges[id].size() > layer.m_max + 0)
+				return -1;
+			for (uint32_t e : layer.edges[id])
+				if (e == id)
+					return -2;
+		}
+	}
+	return 0;
+}
+
+// Export layer-0 CSR for the GPU HNSW path (offsets: n+1, edges concatenated).
+uint64_t orc_hnsw_l0_edge_count(orc_hnsw *h) {
+	uint64_t c = 0;
+	for (auto &e : h->layers[0].edges)
+		c += e.size();
+	return c;
+}
+void orc_hnsw_l0_export(orc_hnsw *h, uint32_t *offsets, uint32_t *edges) {
+	uint32_t off = 0;
+	for (uint64_t i = 0; i < h->next_id; i++) {
+		offsets[i] = off;
+		if (i < h->layers[0].edges.size())
+			for (uint32_t e : h->layers[0].edges[i])
+				edges[off++] = e;
+	}
+	offsets[h->next_id] = off;
+}
+
+// Host-side upper-layer descent (search_ep, hnsw/mod.rs:521-548) for the GPU
+// path: returns the layer-0 entry element and its distance.
+void orc_hnsw_search_ep(orc_hnsw *h, const float *q, uint64_t *ep_id_out,
+                        double *ep_dist_out) {
+	uint64_t ep_id = (uint64_t)h->enter_point;
+	double ep_dist = hdist(h, q, hvec(h, ep_id));
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		OrcPQ cand; cand.push(ep_dist, ep_id);
+		std::unordered_set<uint64_t> visited{ep_id};
+		OrcPQ w = cand;
+		layer_search(h, h->layers[l], q, cand, visited, w, 1);
+		double dd; uint64_t ii;
+		if (w.peek_first(&dd, &ii)) { ep_dist = dd; ep_id = ii; }
+	}
+	*ep_id_out = ep_id;
+	*ep_dist_out = ep_dist;
+}
+
+const float *orc_hnsw_vec_ptr(orc_hnsw *h, uint64_t id) { return hvec(h, id); }
+
+} // extern "C"

@@ -1,0 +1,194 @@
+"""surrealdb_amd — MI355X-native SurrealDB vector-KNN hot path.
+
+PRODUCT PATH. Loads the in-tree HIP C-ABI library (libsdbv.so, gfx950) and
+mirrors the reference's Index/Knn operator surface (see host.py). On a GPU
+machine this package FAILS LOUDLY if the HIP extension is missing or a GPU
+call fails — there is no CPU fallback here (the parity oracle under oracle/
+is test infrastructure and is never imported by this package).
+"""
+import ctypes
+import os
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libsdbv.so")
+
+METRIC_COSINE = 0
+METRIC_EUCLIDEAN = 1
+METRICS = {"cosine": METRIC_COSINE, "euclidean": METRIC_EUCLIDEAN}
+
+_ERRS = {
+    0: "OK",
+    -1: "HIP runtime error",
+    -2: "table not staged",
+    -3: "bad argument",
+    -4: "device OOM",
+    -5: "unsupported",
+    -6: "ids not strictly increasing",
+}
+
+
+class SdbvError(RuntimeError):
+    pass
+
+
+class _Stats(ctypes.Structure):
+    _fields_ = [
+        ("last_scan_kernel_ms", ctypes.c_double),
+        ("last_merge_kernel_ms", ctypes.c_double),
+        ("last_total_ms", ctypes.c_double),
+        ("bytes_staged", ctypes.c_uint64),
+        ("last_rows_scanned", ctypes.c_uint64),
+    ]
+
+
+_lib = None
+
+
+def lib():
+    """Load libsdbv.so. Raises if absent — no silent fallback."""
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_SO):
+        raise SdbvError(
+            f"HIP extension not built: {_SO} missing. Run "
+            "python surrealdb_amd/build.py (hipcc --offload-arch=gfx950)."
+        )
+    L = ctypes.CDLL(_SO)
+    u64, u32, u8 = ctypes.c_uint64, ctypes.c_uint32, ctypes.c_uint8
+    f32p = ctypes.POINTER(ctypes.c_float)
+    f64p = ctypes.POINTER(ctypes.c_double)
+    u64p = ctypes.POINTER(ctypes.c_uint64)
+    u32p = ctypes.POINTER(ctypes.c_uint32)
+    vp = ctypes.c_void_p
+    L.sdbv_init.argtypes = [ctypes.c_int, ctypes.POINTER(vp)]
+    L.sdbv_shutdown.argtypes = [vp]
+    L.sdbv_last_error.restype = ctypes.c_char_p
+    L.sdbv_last_error.argtypes = [vp]
+    L.sdbv_get_stats.argtypes = [vp, ctypes.POINTER(_Stats)]
+    L.sdbv_stage_corpus.argtypes = [vp, u64, f32p, u64p, u64, u32, u8]
+    L.sdbv_stage_synthetic.argtypes = [vp, u64, u64, u32, u8, u64, u64, u64]
+    L.sdbv_table_rows.restype = u64
+    L.sdbv_table_rows.argtypes = [vp, u64]
+    L.sdbv_drop_table.argtypes = [vp, u64]
+    L.sdbv_knn_bruteforce.argtypes = [vp, u64, f32p, u32, u32, u64p, f64p, u32p]
+    L.sdbv_all_distances.argtypes = [vp, u64, f32p, u32, f64p]
+    L.sdbv_gather_distance.argtypes = [vp, u64, u32p, u32, f32p, u32, f64p]
+    L.sdbv_knn_batch.argtypes = [vp, u64, f32p, u32, u32, u32, u64p, f64p]
+    _lib = L
+    return L
+
+
+def _check(ctx, rc, what):
+    if rc != 0:
+        msg = _ERRS.get(rc, str(rc))
+        detail = ""
+        if ctx is not None:
+            try:
+                detail = lib().sdbv_last_error(ctx).decode()
+            except Exception:
+                pass
+        raise SdbvError(f"{what}: {msg} {detail}")
+
+
+class Context:
+    """Owns the device context (HIP stream + staged tables)."""
+
+    def __init__(self, device=-1):
+        import numpy as np  # noqa: F401  (numpy required by callers)
+        self._ptr = ctypes.c_void_p()
+        _check(None, lib().sdbv_init(device, ctypes.byref(self._ptr)), "sdbv_init")
+
+    def close(self):
+        if self._ptr:
+            lib().sdbv_shutdown(self._ptr)
+            self._ptr = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    def stats(self):
+        s = _Stats()
+        _check(self._ptr, lib().sdbv_get_stats(self._ptr, ctypes.byref(s)),
+               "sdbv_get_stats")
+        return {f: getattr(s, f) for f, _ in s._fields_}
+
+    def stage_corpus(self, table, rows, ids=None, metric="cosine"):
+        import numpy as np
+        rows = np.ascontiguousarray(rows, dtype=np.float32)
+        n, d = rows.shape
+        idp = None
+        if ids is not None:
+            ids = np.ascontiguousarray(ids, dtype=np.uint64)
+            idp = ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
+        _check(self._ptr, lib().sdbv_stage_corpus(
+            self._ptr, table, rows.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            idp, n, d, METRICS[metric]), "sdbv_stage_corpus")
+
+    def stage_synthetic(self, table, n, d, metric="cosine", seed=0x5DB1,
+                        row_offset=0, id_base=0):
+        _check(self._ptr, lib().sdbv_stage_synthetic(
+            self._ptr, table, n, d, METRICS[metric], seed, row_offset, id_base),
+            "sdbv_stage_synthetic")
+
+    def table_rows(self, table):
+        return lib().sdbv_table_rows(self._ptr, table)
+
+    def drop_table(self, table):
+        _check(self._ptr, lib().sdbv_drop_table(self._ptr, table), "drop_table")
+
+    def knn_bruteforce(self, table, q, k):
+        import numpy as np
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        ids = np.empty(k, dtype=np.uint64)
+        dists = np.empty(k, dtype=np.float64)
+        out_n = ctypes.c_uint32(0)
+        _check(self._ptr, lib().sdbv_knn_bruteforce(
+            self._ptr, table, q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            q.size, k,
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.byref(out_n)), "sdbv_knn_bruteforce")
+        m = out_n.value
+        return ids[:m], dists[:m]
+
+    def all_distances(self, table, q):
+        import numpy as np
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        n = self.table_rows(table)
+        out = np.empty(n, dtype=np.float64)
+        _check(self._ptr, lib().sdbv_all_distances(
+            self._ptr, table, q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            q.size, out.ctypes.data_as(ctypes.POINTER(ctypes.c_double))),
+            "sdbv_all_distances")
+        return out
+
+    def gather_distance(self, table, rows, q):
+        import numpy as np
+        rows = np.ascontiguousarray(rows, dtype=np.uint32)
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        out = np.empty(rows.size, dtype=np.float64)
+        _check(self._ptr, lib().sdbv_gather_distance(
+            self._ptr, table,
+            rows.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)), rows.size,
+            q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), q.size,
+            out.ctypes.data_as(ctypes.POINTER(ctypes.c_double))),
+            "sdbv_gather_distance")
+        return out
+
+    def knn_batch(self, table, Q, k):
+        import numpy as np
+        Q = np.ascontiguousarray(Q, dtype=np.float32)
+        b, d = Q.shape
+        ids = np.empty((b, k), dtype=np.uint64)
+        dists = np.empty((b, k), dtype=np.float64)
+        _check(self._ptr, lib().sdbv_knn_batch(
+            self._ptr, table, Q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            b, d, k,
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double))),
+            "sdbv_knn_batch")
+        return ids, dists

@@ -1,0 +1,68 @@
+"""Multi-process sharded-search path on CPU: world_size=2 over gloo.
+
+Covers the exact collective+merge sequence bench.py runs over RCCL on the
+8-GPU box: per-rank local top-K -> all_gather of K (dist,id) pairs -> rank-0
+merge with the reference tie-break order. Distances come from the oracle here
+(CPU box); on GPU the same merge consumes sdbv_knn_bruteforce output.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+import oracle
+from surrealdb_amd.shard import merge_topk, shard_range
+
+N, D, K, WORLD = 4000, 64, 10, 2
+
+
+def _worker(rank, world, port, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        b, e = shard_range(N, rank, world)
+        corpus = oracle.gen_f32(0x5DB1, b, e - b, D)
+        q = oracle.gen_f32(0xBEEF, 0, 1, D)[0]
+        ids, dists = oracle.topk_f32("cosine", corpus, q, K)
+        ids = ids + b
+        # pad to K (a shard can hold < K rows)
+        pad = K - len(ids)
+        if pad:
+            ids = np.concatenate([ids, np.full(pad, np.iinfo(np.uint64).max,
+                                               dtype=np.uint64)])
+            dists = np.concatenate([dists, np.full(pad, np.inf)])
+        local = torch.zeros(K, 2, dtype=torch.float64)
+        local[:, 0] = torch.from_numpy(dists.copy())
+        local[:, 1] = torch.from_numpy(ids.view(np.float64).copy())
+        gathered = [torch.zeros_like(local) for _ in range(world)]
+        dist.all_gather(gathered, local)
+        if rank == 0:
+            ids_list = [g[:, 1].numpy().view(np.uint64) for g in gathered]
+            dists_list = [g[:, 0].numpy() for g in gathered]
+            mids, mdists = merge_topk(ids_list, dists_list, K)
+            out_q.put((mids, mdists))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_sharded_topk_equals_global():
+    ctxm = mp.get_context("spawn")
+    out_q = ctxm.Queue()
+    port = 29511
+    procs = [ctxm.Process(target=_worker, args=(r, WORLD, port, out_q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    mids, mdists = out_q.get(timeout=110)
+    for p in procs:
+        p.join(timeout=30)
+    corpus = oracle.gen_f32(0x5DB1, 0, N, D)
+    q = oracle.gen_f32(0xBEEF, 0, 1, D)[0]
+    gids, gdists = oracle.topk_f32("cosine", corpus, q, K)
+    assert np.array_equal(mids, gids)
+    assert np.array_equal(mdists, gdists)
