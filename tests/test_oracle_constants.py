@@ -108,3 +108,31 @@ def test_dot_unrolled_contract():
         a.ctypes.data_as(__import__("ctypes").POINTER(__import__("ctypes").c_float)),
         b.ctypes.data_as(__import__("ctypes").POINTER(__import__("ctypes").c_float)),
         64) == float(np.dot(a.astype(np.float64), b.astype(np.float64)))
+
+
+def test_dot_chain_bitexact_vs_strict_simulation():
+    """Pins the oracle's fp behaviour against a strict per-op np.float32
+    simulation of the restated ndarray chain — catches any build-flag change
+    (e.g. fp contraction) that would silently fuse mul+add into fma."""
+    import ctypes
+    import math
+    a = oracle.gen_f32(0x5DB1, 0, 1, 768)[0]
+    b = oracle.gen_f32(0xBEEF, 0, 1, 768)[0]
+    p = np.zeros(8, np.float32)
+    for i in range(0, 768, 8):
+        for j in range(8):
+            p[j] = np.float32(p[j] + np.float32(a[i + j] * b[i + j]))
+    s = np.float32(0)
+    s = np.float32(s + np.float32(p[0] + p[4]))
+    s = np.float32(s + np.float32(p[1] + p[5]))
+    s = np.float32(s + np.float32(p[2] + p[6]))
+    s = np.float32(s + np.float32(p[3] + p[7]))
+    f32p = ctypes.POINTER(ctypes.c_float)
+    got = oracle.lib().orc_dot_f32(a.ctypes.data_as(f32p),
+                                   b.ctypes.data_as(f32p), 768)
+    assert np.float32(got).tobytes() == s.tobytes()
+    acc = np.float32(0)
+    for i in range(768):
+        d = np.float32(a[i] - b[i])
+        acc = np.float32(acc + np.float32(d * d))
+    assert math.sqrt(float(acc)) == oracle.dist_f32("euclidean", a, b)
