@@ -10,10 +10,16 @@ _DIR = os.path.dirname(os.path.abspath(__file__))
 SO_PATH = os.path.join(_DIR, "libsdbv.so")
 SRC = os.path.join(_DIR, "csrc", "sdbv.hip")
 
+# -ffp-contract=off is LOAD-BEARING: hipcc contracts a*b+c into v_fma_f32
+# even through __fmul_rn/__fadd_rn (verified on ROCm 7.2/gfx950); the
+# reference (rustc/LLVM) never contracts, and the parity contract requires
+# the exact mul-then-add rounding per element. The scan is HBM-bound, so the
+# extra VALU ops cost nothing.
 CMD = [
     "hipcc",
     "--offload-arch=gfx950",
     "-O3",
+    "-ffp-contract=off",
     "-std=c++17",
     "-fPIC",
     "-shared",
