@@ -37,6 +37,9 @@ def main():
     ap.add_argument("--dim", type=int, default=768)
     ap.add_argument("--k", type=int, default=10)
     ap.add_argument("--metric", default="cosine")
+    ap.add_argument("--batch", type=int, default=0,
+                    help=">0: run BASELINE configs[3] (batched MFMA path); a "
+                         "step is one batch of this many queries")
     ap.add_argument("--seed", type=lambda x: int(x, 0), default=0x5DB1)
     ap.add_argument("--cpu-sample-rows", type=int, default=2_000_000,
                     help="bounded sample for the cpu_baseline leg")
@@ -70,12 +73,40 @@ def main():
                         row_offset=row_offset, id_base=row_offset)
     stage_s = time.perf_counter() - t0
 
-    queries = gen_f32(0xBEEF, 0, args.steps + args.warmup, args.dim)
+    nq = (args.steps + args.warmup) * max(args.batch, 1)
+    queries = gen_f32(0xBEEF, 0, min(nq, 4096), args.dim)
 
     device = torch.device(f"cuda:{local_rank}")
 
+    def one_batch(si):
+        """One step of the batched path: args.batch concurrent queries."""
+        b = args.batch
+        q0 = (si * b) % max(len(queries) - b, 1)
+        ids, dists = ctx.knn_batch(1, queries[q0:q0 + b], args.k)
+        if world > 1:
+            flat = torch.empty(b * args.k, 2, dtype=torch.float64,
+                               device=device)
+            flat[:, 0] = torch.from_numpy(dists.reshape(-1).copy()).to(device)
+            flat[:, 1] = torch.from_numpy(
+                ids.reshape(-1).view(np.float64).copy()).to(device)
+            gathered = [torch.empty_like(flat) for _ in range(world)]
+            dist.all_gather(gathered, flat)
+            if rank == 0:
+                g = [t.cpu().numpy().reshape(b, args.k, 2) for t in gathered]
+                out = []
+                for j in range(b):
+                    out.append(merge_topk(
+                        [x[j, :, 1].copy().view(np.uint64) for x in g],
+                        [x[j, :, 0] for x in g], args.k))
+                return out
+            return None
+        return ids, dists
+
     def one_query(qi):
-        ids, dists = ctx.knn_bruteforce(1, queries[qi], args.k)
+        if args.batch > 0:
+            return one_batch(qi)
+        ids, dists = ctx.knn_bruteforce(1, queries[qi % len(queries)],
+                                        args.k)
         if world > 1:
             pad = args.k - len(ids)
             if pad:
@@ -125,24 +156,40 @@ def main():
     if rank != 0:
         return
 
-    qps = args.steps / t_total
+    queries_per_step = max(args.batch, 1)
+    qps = args.steps * queries_per_step / t_total
     p50_ms = float(np.percentile(np.array(step_times) * 1e3, 50))
     p95_ms = float(np.percentile(np.array(step_times) * 1e3, 95))
 
-    # --- roofline: dominant kernel = the distance scan ---
-    # algorithmic bytes per launch: rows * d * 4 B (corpus read)
     scan_ms_avg = scan_ms_acc / args.steps
-    alg_bytes = rows * args.dim * 4
-    achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
     traffic = os.environ.get("SDBV_TRAFFIC_BYTES_PER_LAUNCH")
-    roofline = {
-        "bound": "hbm",
-        "achieved": round(achieved_gbs, 1),
-        "peak": HBM_PEAK_GBS,
-        "unit": "GB/s",
-        "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-        "traffic": float(traffic) if traffic else None,
-    }
+    if args.batch > 0:
+        # dominant kernel = the f32 MFMA GEMM; stats.last_scan_kernel_ms is
+        # the summed sgemm time of one step's chunks
+        alg_flop = 2.0 * args.batch * rows * args.dim
+        achieved_tf = alg_flop / (scan_ms_avg * 1e-3) / 1e12
+        F32_MFMA_PEAK_TF = 157.3  # gfx950 f32-input MFMA dense peak
+        roofline = {
+            "bound": "mfma",
+            "achieved": round(achieved_tf, 1),
+            "peak": F32_MFMA_PEAK_TF,
+            "unit": "TFLOP/s",
+            "frac": round(achieved_tf / F32_MFMA_PEAK_TF, 4),
+            "traffic": float(traffic) if traffic else None,
+        }
+    else:
+        # dominant kernel = the distance scan; algorithmic bytes per launch =
+        # rows * d * 4 B (corpus read)
+        alg_bytes = rows * args.dim * 4
+        achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved_gbs, 1),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+            "traffic": float(traffic) if traffic else None,
+        }
 
     # --- cpu_baseline: the oracle (kind "port") on host cores, rank0/N=1 ---
     cpu_baseline = None
@@ -185,8 +232,13 @@ def main():
         "dtype": "f32",
         "data": "synthetic",
         "config": {
-            "workload": "brute-force cosine KNN, 10M rows/GPU x 768-dim f32, "
-                        "K=10, single query (BASELINE configs[1])",
+            "workload": (
+                f"brute-force {args.metric} KNN, 10M rows/GPU x 768-dim f32, "
+                f"K=10, batch={args.batch} MFMA path (BASELINE configs[3])"
+                if args.batch > 0 else
+                "brute-force cosine KNN, 10M rows/GPU x 768-dim f32, "
+                "K=10, single query (BASELINE configs[1])"),
+            "batch": args.batch,
             "rows_total": rows * world,
             "rows_per_gpu": rows,
             "dim": args.dim,

@@ -24,6 +24,8 @@ CMD = [
     "-fPIC",
     "-shared",
     SRC,
+    "-L/opt/rocm/lib",
+    "-lrocblas",
     "-o",
     SO_PATH,
 ]

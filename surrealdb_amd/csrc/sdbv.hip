@@ -35,9 +35,12 @@
 // fma — the reference's mul-then-add rounding is the contract).
 
 #include <hip/hip_runtime.h>
+#include <rocblas/rocblas.h>
 
+#include <algorithm>
 #include <chrono>
 #include <cstdint>
+#include <limits>
 #include <cstdio>
 #include <cstring>
 #include <map>
@@ -72,6 +75,8 @@ struct Table {
 	uint8_t metric = 0;
 	float *cm = nullptr;      // [d][n_pad] feature-major
 	double *norms = nullptr;  // per-row f64 norm (cosine only)
+	float *aux = nullptr;     // per-row f32 selection key aux for the batch
+	                          // path: cosine 1/norm, euclidean sum-of-squares
 	uint64_t *ids_dev = nullptr; // per-row id, device (merge kernel maps)
 	uint64_t bytes = 0;
 };
@@ -90,6 +95,19 @@ struct sdbv_ctx {
 	float *q_dev = nullptr;
 	uint32_t q_cap = 0;
 	hipEvent_t ev0, ev1, ev2;
+	// batched path
+	rocblas_handle blas = nullptr;
+	float *S = nullptr; // chunk scores scratch, [chunk][b] col-major
+	uint64_t S_cap = 0;
+	void *bstate = nullptr; // [b][kk] running candidates
+	uint64_t bstate_cap = 0;
+	float *Q_dev = nullptr;
+	uint64_t Q_cap = 0;
+	double *qnorms = nullptr;
+	uint64_t qnorms_cap = 0;
+	double *bdists = nullptr; // [b][kk] exact distances
+	uint64_t bdists_cap = 0;
+	double ms_gemm = 0, ms_select = 0, ms_exact = 0; // last batch timings
 };
 
 struct Cand {
@@ -563,6 +581,244 @@ __global__ void k_gather_dist(const float *__restrict__ cm,
 }
 
 // ---------------------------------------------------------------------------
+// Batched-query path (BASELINE configs[3]) — the genuinely-dense case.
+// scores = corpus x Q^T runs as a plain f32 GEMM on MFMA (rocBLAS; the
+// feature-major store [d][n_pad] IS the column-major corpus matrix with
+// lda = n_pad, so no transform is needed). Selection per query uses a
+// monotone f32 approximation of the distance as the key; the survivors
+// (k + SLACK per query) are then recomputed with the exact restated chain so
+// the final scores are bitwise identical to the single-query path, and
+// re-ranked by the exact (total_cmp dist, id) order.
+// ---------------------------------------------------------------------------
+#define BATCH_SLACK 16
+
+__device__ static inline uint64_t d_total_key32(float x) {
+	uint32_t bits = __float_as_uint(x);
+	uint32_t k = (bits >> 31) ? ~bits : (bits | 0x80000000u);
+	return (uint64_t)k;
+}
+
+struct BCand {
+	float key; // monotone f32 selection key (smaller = better)
+	uint32_t row;
+};
+
+__global__ void k_binit(BCand *state, uint64_t total) {
+	uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+	if (i < total) {
+		state[i].key = __uint_as_float(0x7F800000u); // +inf
+		state[i].row = ~0u;
+	}
+}
+
+// f32 aux for the selection key: cosine -> 1/norm, euclidean -> restated f32
+// sum-of-squares per row.
+__global__ void k_aux(const float *__restrict__ cm,
+                      const double *__restrict__ norms,
+                      float *__restrict__ aux, uint64_t n, uint64_t n_pad,
+                      uint32_t d, int metric) {
+	uint64_t r = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+	if (r >= n)
+		return;
+	if (metric == 0) {
+		aux[r] = (float)(1.0 / norms[r]);
+	} else {
+		float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+		uint32_t k = 0;
+		for (; k + 8 <= d; k += 8) {
+#pragma unroll
+			for (uint32_t j = 0; j < 8; j++) {
+				float x = cm[(uint64_t)(k + j) * n_pad + r];
+				p[j] = __fadd_rn(p[j], __fmul_rn(x, x));
+			}
+		}
+		float acc = 0.0f;
+		acc = __fadd_rn(acc,
+		                __fadd_rn(__fadd_rn(p[0], p[4]), __fadd_rn(p[1], p[5])));
+		acc = __fadd_rn(acc,
+		                __fadd_rn(__fadd_rn(p[2], p[6]), __fadd_rn(p[3], p[7])));
+		for (; k < d; k++) {
+			float x = cm[(uint64_t)k * n_pad + r];
+			acc = __fadd_rn(acc, __fmul_rn(x, x));
+		}
+		aux[r] = acc;
+	}
+}
+
+// Per-query f64 norms of a row-major Q (restated sumsq chain + f64 sqrt).
+__global__ void k_qnorms(const float *__restrict__ Q, uint32_t b, uint32_t d,
+                         double *__restrict__ out) {
+	uint32_t j = blockIdx.x * blockDim.x + threadIdx.x;
+	if (j >= b)
+		return;
+	const float *a = Q + (uint64_t)j * d;
+	float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+	uint32_t i = 0;
+	for (; i + 8 <= d; i += 8)
+#pragma unroll
+		for (uint32_t t = 0; t < 8; t++)
+			p[t] = __fadd_rn(p[t], __fmul_rn(a[i + t], a[i + t]));
+	float acc = 0.0f;
+	acc = __fadd_rn(acc, __fadd_rn(__fadd_rn(p[0], p[4]), __fadd_rn(p[1], p[5])));
+	acc = __fadd_rn(acc, __fadd_rn(__fadd_rn(p[2], p[6]), __fadd_rn(p[3], p[7])));
+	for (; i < d; i++)
+		acc = __fadd_rn(acc, __fmul_rn(a[i], a[i]));
+	out[j] = sqrt((double)acc);
+}
+
+// Per-query running top-(kk) update over one GEMM chunk.
+// S is [n_chunk x b] column-major (column j = scores of query j, contiguous).
+// metric 0 (cosine): key = -(s * inv_norm)   (maximise normalised dot)
+// metric 1 (euclid): key = sumsq_row - 2*s   (minimise ||q-r||^2 - qn2)
+__global__ __launch_bounds__(THREADS) void k_batch_topk(
+    const float *__restrict__ S, const float *__restrict__ aux,
+    uint64_t row0, uint32_t n_chunk, uint64_t n, int metric,
+    BCand *__restrict__ state, int kk) {
+	__shared__ LCand buf[TILE + MAX_K];
+	__shared__ LCand topk[MAX_K];
+	__shared__ int cnt;
+	__shared__ int topk_n;
+	__shared__ uint64_t kth_key;
+	__shared__ uint32_t kth_row;
+
+	uint32_t j = blockIdx.x;
+	const float *col = S + (uint64_t)j * n_chunk;
+	BCand *st = state + (uint64_t)j * kk;
+
+	// load current state into LDS topk (keys already total-ordered u64)
+	if (threadIdx.x == 0) {
+		cnt = 0;
+		topk_n = 0;
+		kth_key = ~0ULL;
+		kth_row = ~0u;
+	}
+	__syncthreads();
+	if (threadIdx.x < (uint32_t)kk) {
+		BCand c = st[threadIdx.x];
+		if (c.row != ~0u) {
+			topk[threadIdx.x].key = d_total_key32(c.key);
+			topk[threadIdx.x].dist = (double)c.key;
+			topk[threadIdx.x].row = c.row;
+			atomicAdd(&topk_n, 1);
+		}
+	}
+	__syncthreads();
+	if (threadIdx.x == 0 && topk_n >= kk) {
+		kth_key = topk[kk - 1].key;
+		kth_row = topk[kk - 1].row;
+	}
+	__syncthreads();
+
+	for (uint32_t tile = 0; tile < n_chunk; tile += TILE) {
+		uint32_t i0 = tile + threadIdx.x * 4;
+		uint64_t kk_key = kth_key;
+		uint32_t kk_row = kth_row;
+		int full = (topk_n >= kk);
+		if (i0 + 3 < n_chunk) {
+			const float4 s4 = *(const float4 *)(col + i0);
+			float sv[4] = {s4.x, s4.y, s4.z, s4.w};
+#pragma unroll
+			for (int c = 0; c < 4; c++) {
+				uint64_t r = row0 + i0 + c;
+				if (r >= n)
+					continue;
+				float key = (metric == 0) ? -(sv[c] * aux[r])
+				                          : (aux[r] - 2.0f * sv[c]);
+				uint64_t tk = d_total_key32(key);
+				uint32_t row32 = (uint32_t)r;
+				bool take =
+				    !full || tk < kk_key || (tk == kk_key && row32 < kk_row);
+				if (take) {
+					int idx = atomicAdd(&cnt, 1);
+					buf[idx].key = tk;
+					buf[idx].dist = (double)key;
+					buf[idx].row = row32;
+				}
+			}
+		}
+		__syncthreads();
+		if (cnt > 0) {
+			int m = cnt;
+			for (int i = threadIdx.x; i < topk_n; i += THREADS)
+				buf[m + i] = topk[i];
+			int total = m + topk_n;
+			__syncthreads();
+			int new_n;
+			block_select_topk(buf, total, topk, kk, &new_n);
+			if (threadIdx.x == 0) {
+				topk_n = new_n;
+				if (new_n >= kk) {
+					kth_key = topk[kk - 1].key;
+					kth_row = topk[kk - 1].row;
+				}
+				cnt = 0;
+			}
+		}
+		__syncthreads();
+	}
+
+	// write back
+	if (threadIdx.x < (uint32_t)kk) {
+		BCand c;
+		if ((int)threadIdx.x < topk_n) {
+			c.key = (float)topk[threadIdx.x].dist;
+			c.row = topk[threadIdx.x].row;
+		} else {
+			c.key = __uint_as_float(0x7F800000u);
+			c.row = ~0u;
+		}
+		st[threadIdx.x] = c;
+	}
+}
+
+// Exact recompute of each surviving candidate with the restated per-row
+// chain (bitwise identical to the single-query scan path).
+__global__ void k_batch_exact(const float *__restrict__ cm,
+                              const double *__restrict__ norms,
+                              uint64_t n_pad, uint32_t d,
+                              const float *__restrict__ Q,
+                              const double *__restrict__ qnorms, int metric,
+                              const BCand *__restrict__ state, int kk,
+                              double *__restrict__ out) {
+	uint32_t j = blockIdx.x;  // query
+	uint32_t c = blockIdx.y;  // candidate slot
+	if (threadIdx.x != 0)
+		return;
+	BCand cand = state[(uint64_t)j * kk + c];
+	double *o = out + (uint64_t)j * kk + c;
+	if (cand.row == ~0u) {
+		*o = __longlong_as_double(0x7FF0000000000000LL);
+		return;
+	}
+	const float *q = Q + (uint64_t)j * d;
+	uint64_t r = cand.row;
+	if (metric == 0) {
+		float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+		uint32_t k = 0;
+		for (; k + 8 <= d; k += 8)
+#pragma unroll
+			for (uint32_t t = 0; t < 8; t++)
+				p[t] = __fadd_rn(
+				    p[t], __fmul_rn(cm[(uint64_t)(k + t) * n_pad + r], q[k + t]));
+		float sum = 0.f;
+		sum = __fadd_rn(sum, __fadd_rn(p[0], p[4]));
+		sum = __fadd_rn(sum, __fadd_rn(p[1], p[5]));
+		sum = __fadd_rn(sum, __fadd_rn(p[2], p[6]));
+		sum = __fadd_rn(sum, __fadd_rn(p[3], p[7]));
+		for (; k < d; k++)
+			sum = __fadd_rn(sum, __fmul_rn(cm[(uint64_t)k * n_pad + r], q[k]));
+		*o = 1.0 - (double)sum / (qnorms[j] * norms[r]);
+	} else {
+		float acc = 0.f;
+		for (uint32_t k = 0; k < d; k++) {
+			float diff = cm[(uint64_t)k * n_pad + r] - q[k];
+			acc = __fadd_rn(acc, __fmul_rn(diff, diff));
+		}
+		*o = sqrt((double)acc);
+	}
+}
+
+// ---------------------------------------------------------------------------
 // Host-side helpers (restated reference arithmetic for the query's own norm —
 // must be bit-identical to oracle orc_sumsq_f32; covered by tests/)
 // ---------------------------------------------------------------------------
@@ -585,6 +841,8 @@ static void free_table(Table &t) {
 		(void)hipFree(t.cm);
 	if (t.norms)
 		(void)hipFree(t.norms);
+	if (t.aux)
+		(void)hipFree(t.aux);
 	if (t.ids_dev)
 		(void)hipFree(t.ids_dev);
 	t = Table{};
@@ -628,6 +886,12 @@ void sdbv_shutdown(sdbv_ctx *ctx) {
 		(void)hipFree(ctx->final_out);
 	if (ctx->q_dev)
 		(void)hipFree(ctx->q_dev);
+	for (void *p : {(void *)ctx->S, ctx->bstate, (void *)ctx->Q_dev,
+	                (void *)ctx->qnorms, (void *)ctx->bdists})
+		if (p)
+			(void)hipFree(p);
+	if (ctx->blas)
+		(void)rocblas_destroy_handle(ctx->blas);
 	(void)hipEventDestroy(ctx->ev0);
 	(void)hipEventDestroy(ctx->ev1);
 	(void)hipEventDestroy(ctx->ev2);
@@ -675,11 +939,16 @@ static int stage_common(sdbv_ctx *ctx, uint64_t table, uint64_t n, uint32_t d,
 }
 
 static int finish_stage(sdbv_ctx *ctx, Table *t) {
+	uint64_t nb = (t->n + THREADS - 1) / THREADS;
 	if (t->metric == SDBV_METRIC_COSINE) {
-		uint64_t nb = (t->n + THREADS - 1) / THREADS;
 		hipLaunchKernelGGL(k_norms, dim3((uint32_t)nb), dim3(THREADS), 0,
 		                   ctx->stream, t->cm, t->norms, t->n, t->n_pad, t->d);
 	}
+	HIP_CHECK(ctx, hipMalloc(&t->aux, t->n_pad * sizeof(float)));
+	t->bytes += t->n_pad * sizeof(float);
+	hipLaunchKernelGGL(k_aux, dim3((uint32_t)nb), dim3(THREADS), 0, ctx->stream,
+	                   t->cm, t->norms, t->aux, t->n, t->n_pad, t->d,
+	                   (int)t->metric);
 	HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
 	HIP_CHECK(ctx, hipGetLastError());
 	uint64_t total = 0;
@@ -921,13 +1190,160 @@ int sdbv_gather_distance(sdbv_ctx *ctx, uint64_t table, const uint32_t *rows,
 	return SDBV_OK;
 }
 
-// Batched and HNSW-search entry points land in the next milestone; declared
-// symbols return UNSUPPORTED so the C-ABI surface is complete and loadable.
-int sdbv_knn_batch(sdbv_ctx *ctx, uint64_t, const float *, uint32_t, uint32_t,
-                   uint32_t, uint64_t *, double *) {
-	if (ctx)
-		ctx->err = "sdbv_knn_batch: not implemented in this revision";
-	return SDBV_ERR_UNSUPPORTED;
+static int ensure_cap(sdbv_ctx *ctx, void **buf, uint64_t *cap, uint64_t need) {
+	if (*cap >= need)
+		return SDBV_OK;
+	if (*buf)
+		(void)hipFree(*buf);
+	*buf = nullptr;
+	*cap = 0;
+	HIP_CHECK(ctx, hipMalloc(buf, need));
+	*cap = need;
+	return SDBV_OK;
+}
+
+int sdbv_knn_batch(sdbv_ctx *ctx, uint64_t table, const float *Q, uint32_t b,
+                   uint32_t d, uint32_t k, uint64_t *out_ids,
+                   double *out_dists) {
+	std::lock_guard<std::mutex> lk(ctx->mu);
+	auto it = ctx->tables.find(table);
+	if (it == ctx->tables.end())
+		return SDBV_ERR_NO_TABLE;
+	Table &t = it->second;
+	if (d != t.d || b == 0 || k == 0 || k > MAX_K - BATCH_SLACK)
+		return SDBV_ERR_BAD_ARG;
+	const int kk = (int)k + BATCH_SLACK;
+
+	if (!ctx->blas) {
+		if (rocblas_create_handle(&ctx->blas) != rocblas_status_success) {
+			ctx->err = "rocblas_create_handle failed";
+			return SDBV_ERR_HIP;
+		}
+		rocblas_set_stream(ctx->blas, ctx->stream);
+	}
+
+	// chunk size: multiple of TILE, bounded scratch (<=1 GiB at b=1024)
+	uint64_t chunk = 262144;
+	if (chunk > t.n_pad)
+		chunk = t.n_pad;
+	int rc;
+	if ((rc = ensure_cap(ctx, (void **)&ctx->S, &ctx->S_cap,
+	                     chunk * b * sizeof(float))))
+		return rc;
+	if ((rc = ensure_cap(ctx, (void **)&ctx->bstate, &ctx->bstate_cap,
+	                     (uint64_t)b * kk * sizeof(BCand))))
+		return rc;
+	if ((rc = ensure_cap(ctx, (void **)&ctx->Q_dev, &ctx->Q_cap,
+	                     (uint64_t)b * d * sizeof(float))))
+		return rc;
+	if ((rc = ensure_cap(ctx, (void **)&ctx->qnorms, &ctx->qnorms_cap,
+	                     (uint64_t)b * sizeof(double))))
+		return rc;
+	if ((rc = ensure_cap(ctx, (void **)&ctx->bdists, &ctx->bdists_cap,
+	                     (uint64_t)b * kk * sizeof(double))))
+		return rc;
+
+	HIP_CHECK(ctx, hipMemcpyAsync(ctx->Q_dev, Q,
+	                              (uint64_t)b * d * sizeof(float),
+	                              hipMemcpyHostToDevice, ctx->stream));
+	hipLaunchKernelGGL(k_qnorms, dim3((b + 255) / 256), dim3(256), 0,
+	                   ctx->stream, ctx->Q_dev, b, d, ctx->qnorms);
+	{
+		uint64_t total = (uint64_t)b * kk;
+		hipLaunchKernelGGL(k_binit, dim3((uint32_t)((total + 255) / 256)),
+		                   dim3(256), 0, ctx->stream, (BCand *)ctx->bstate,
+		                   total);
+	}
+
+	double ms_gemm = 0, ms_select = 0;
+	for (uint64_t row0 = 0; row0 < t.n; row0 += chunk) {
+		uint32_t nc = (uint32_t)std::min<uint64_t>(chunk, t.n_pad - row0);
+		const float alpha = 1.0f, beta = 0.0f;
+		HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
+		// S[nc x b] = corpus[nc x d] (col-major view of cm, lda=n_pad)
+		//           x Q^T[d x b]    (col-major view of row-major Q, ldb=d)
+		if (rocblas_sgemm(ctx->blas, rocblas_operation_none,
+		                  rocblas_operation_none, (rocblas_int)nc,
+		                  (rocblas_int)b, (rocblas_int)d, &alpha, t.cm + row0,
+		                  (rocblas_int)t.n_pad, ctx->Q_dev, (rocblas_int)d,
+		                  &beta, ctx->S, (rocblas_int)nc) !=
+		    rocblas_status_success) {
+			ctx->err = "rocblas_sgemm failed";
+			return SDBV_ERR_HIP;
+		}
+		HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
+		hipLaunchKernelGGL(k_batch_topk, dim3(b), dim3(THREADS), 0,
+		                   ctx->stream, ctx->S, t.aux, row0, nc, t.n,
+		                   (int)t.metric, (BCand *)ctx->bstate, kk);
+		HIP_CHECK(ctx, hipEventRecord(ctx->ev2, ctx->stream));
+		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		float m0 = 0, m1 = 0;
+		(void)hipEventElapsedTime(&m0, ctx->ev0, ctx->ev1);
+		(void)hipEventElapsedTime(&m1, ctx->ev1, ctx->ev2);
+		ms_gemm += m0;
+		ms_select += m1;
+	}
+
+	// exact recompute of survivors with the restated chain
+	HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
+	hipLaunchKernelGGL(k_batch_exact, dim3(b, kk), dim3(64), 0, ctx->stream,
+	                   t.cm, t.norms, t.n_pad, t.d, ctx->Q_dev, ctx->qnorms,
+	                   (int)t.metric, (const BCand *)ctx->bstate, kk,
+	                   ctx->bdists);
+	HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
+
+	std::vector<BCand> h_state((uint64_t)b * kk);
+	std::vector<double> h_dists((uint64_t)b * kk);
+	HIP_CHECK(ctx, hipMemcpyAsync(h_state.data(), ctx->bstate,
+	                              h_state.size() * sizeof(BCand),
+	                              hipMemcpyDeviceToHost, ctx->stream));
+	HIP_CHECK(ctx, hipMemcpyAsync(h_dists.data(), ctx->bdists,
+	                              h_dists.size() * sizeof(double),
+	                              hipMemcpyDeviceToHost, ctx->stream));
+	HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+	HIP_CHECK(ctx, hipGetLastError());
+	float me = 0;
+	(void)hipEventElapsedTime(&me, ctx->ev0, ctx->ev1);
+	ctx->ms_gemm = ms_gemm;
+	ctx->ms_select = ms_select;
+	ctx->ms_exact = me;
+	ctx->stats.last_scan_kernel_ms = ms_gemm; // dominant kernel of the batch
+	ctx->stats.last_merge_kernel_ms = ms_select;
+	ctx->stats.last_rows_scanned = t.n;
+
+	// final exact rank per query: ascending (total_cmp(dist), id)
+	std::vector<uint64_t> ids_host(t.n_pad);
+	HIP_CHECK(ctx, hipMemcpy(ids_host.data(), t.ids_dev,
+	                         t.n_pad * sizeof(uint64_t),
+	                         hipMemcpyDeviceToHost));
+	auto total_key = [](double x) {
+		uint64_t bits;
+		std::memcpy(&bits, &x, 8);
+		return (bits >> 63) ? ~bits : (bits | 0x8000000000000000ULL);
+	};
+	std::vector<std::pair<std::pair<uint64_t, uint64_t>, double>> cand(kk);
+	for (uint32_t j = 0; j < b; j++) {
+		size_t m = 0;
+		for (int c = 0; c < kk; c++) {
+			BCand bc = h_state[(uint64_t)j * kk + c];
+			if (bc.row == ~0u)
+				continue;
+			double dd = h_dists[(uint64_t)j * kk + c];
+			cand[m++] = {{total_key(dd), ids_host[bc.row]}, dd};
+		}
+		std::sort(cand.begin(), cand.begin() + m);
+		uint32_t out_m = (uint32_t)std::min<size_t>(k, m);
+		for (uint32_t c = 0; c < out_m; c++) {
+			out_ids[(uint64_t)j * k + c] = cand[c].first.second;
+			out_dists[(uint64_t)j * k + c] = cand[c].second;
+		}
+		for (uint32_t c = out_m; c < k; c++) {
+			out_ids[(uint64_t)j * k + c] = ~0ULL;
+			out_dists[(uint64_t)j * k + c] =
+			    std::numeric_limits<double>::infinity();
+		}
+	}
+	return SDBV_OK;
 }
 int sdbv_hnsw_upload(sdbv_ctx *ctx, uint64_t, const uint32_t *,
                      const uint32_t *, uint64_t, uint32_t, sdbv_hnsw **) {
