@@ -122,20 +122,47 @@ int sdbv_gather_distance(sdbv_ctx *, uint64_t table, const uint32_t *rows,
                          uint32_t nrows, const float *q, uint32_t d,
                          double *out_dists);
 
-/* Fully-on-GPU HNSW ef-search over a device-resident graph (CSR layer 0 +
- * host-provided upper layers), one query per workgroup. Staged separately
- * from the brute-force store. See DESIGN.md. */
+/* HNSW index — the Hnsw/HnswIndex replacement (hnsw/mod.rs, hnsw/index.rs,
+ * layer.rs, heuristic.rs). Graph topology and upper layers live host-side
+ * (as in the reference engine); layer-0 neighbour expansion runs as the
+ * batched gather+distance kernel over the staged table; doc-id expansion,
+ * pendings merge and cond filtering stay with the caller.
+ *
+ * Element ids are insertion ordinals (the caller maps them to records, as
+ * HnswDocs does, docs.rs:39-145).
+ *
+ * Defaults mirror DEFINE INDEX ... HNSW (syn/parser/stmt/define.rs:1102-1180):
+ * m=12, m0=2m, efc=150, ml=1/ln(m), extend=keep=0, metric EUCLIDEAN, F32. */
 typedef struct sdbv_hnsw sdbv_hnsw;
-int sdbv_hnsw_upload(sdbv_ctx *, uint64_t table,
-                     const uint32_t *l0_offsets /* n+1 */,
-                     const uint32_t *l0_edges /* offsets[n] */, uint64_t n,
-                     uint32_t entry_point, sdbv_hnsw **out);
-int sdbv_hnsw_search(sdbv_hnsw *, const float *Q /* b x d */, uint32_t b,
-                     uint32_t d, uint32_t k, uint32_t ef,
-                     const uint32_t *entry_rows /* b entry rows (from host
-                     upper-layer descent), NULL => uploaded entry_point */,
-                     uint64_t *out_ids, double *out_dists, uint32_t *out_n);
-void sdbv_hnsw_free(sdbv_hnsw *);
+int sdbv_hnsw_create(sdbv_ctx *, uint32_t d, uint8_t metric, uint32_t m,
+                     uint32_t m0, uint32_t efc, int extend_candidates,
+                     int keep_pruned, uint64_t rng_seed, double ml,
+                     sdbv_hnsw **out);
+/* Insert one vector (Hnsw::insert, hnsw/mod.rs:389-394; level drawn from the
+ * committed RNG restatement of :263-266). Sequential build — deterministic,
+ * used for parity. */
+int sdbv_hnsw_insert(sdbv_hnsw *, const float *pt);
+/* Parallel bulk build (bench mode): levels are drawn deterministically per
+ * ordinal, insert order is nondeterministic across threads (striped node
+ * locks) — graph quality validated by the reference's recall bars, like the
+ * reference's own lock-free enqueue + batched apply (index.rs:138-211). */
+int sdbv_hnsw_insert_batch(sdbv_hnsw *, const float *pts, uint64_t n,
+                           int nthreads);
+/* Stage vectors (feature-major + norms) into the device table slot `table`;
+ * required before sdbv_hnsw_knn. */
+int sdbv_hnsw_finalize(sdbv_hnsw *, uint64_t table);
+/* knn_search (hnsw/index.rs:270-335 minus host-kept parts): upper-layer
+ * greedy descent host-side, layer-0 ef-search with GPU batched neighbour
+ * expansion. Results ascending (dist total_cmp, id), truncated to k. */
+int sdbv_hnsw_knn(sdbv_hnsw *, const float *q, uint32_t k, uint32_t ef,
+                  uint64_t *out_ids, double *out_dists, uint32_t *out_n);
+void sdbv_hnsw_destroy(sdbv_hnsw *);
+/* Introspection (parity tests): */
+uint64_t sdbv_hnsw_n(sdbv_hnsw *);
+uint32_t sdbv_hnsw_layers(sdbv_hnsw *);
+uint64_t sdbv_hnsw_l0_edge_count(sdbv_hnsw *);
+void sdbv_hnsw_l0_export(sdbv_hnsw *, uint32_t *offsets /* n+1 */,
+                         uint32_t *edges);
 
 #ifdef __cplusplus
 }

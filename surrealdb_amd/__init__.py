@@ -75,6 +75,21 @@ def lib():
     L.sdbv_all_distances.argtypes = [vp, u64, f32p, u32, f64p]
     L.sdbv_gather_distance.argtypes = [vp, u64, u32p, u32, f32p, u32, f64p]
     L.sdbv_knn_batch.argtypes = [vp, u64, f32p, u32, u32, u32, u64p, f64p]
+    L.sdbv_hnsw_create.argtypes = [vp, u32, u8, u32, u32, u32, ctypes.c_int,
+                                   ctypes.c_int, u64, ctypes.c_double,
+                                   ctypes.POINTER(vp)]
+    L.sdbv_hnsw_insert.argtypes = [vp, f32p]
+    L.sdbv_hnsw_insert_batch.argtypes = [vp, f32p, u64, ctypes.c_int]
+    L.sdbv_hnsw_finalize.argtypes = [vp, u64]
+    L.sdbv_hnsw_knn.argtypes = [vp, f32p, u32, u32, u64p, f64p, u32p]
+    L.sdbv_hnsw_destroy.argtypes = [vp]
+    L.sdbv_hnsw_n.restype = u64
+    L.sdbv_hnsw_n.argtypes = [vp]
+    L.sdbv_hnsw_layers.restype = u32
+    L.sdbv_hnsw_layers.argtypes = [vp]
+    L.sdbv_hnsw_l0_edge_count.restype = u64
+    L.sdbv_hnsw_l0_edge_count.argtypes = [vp]
+    L.sdbv_hnsw_l0_export.argtypes = [vp, u32p, u32p]
     _lib = L
     return L
 
@@ -179,6 +194,19 @@ class Context:
             "sdbv_gather_distance")
         return out
 
+    def hnsw_create(self, d, metric="euclidean", m=12, m0=None, efc=150,
+                    extend=False, keep=False, seed=0x5DB1, ml=None):
+        import math
+        if m0 is None:
+            m0 = 2 * m
+        if ml is None:
+            ml = 1.0 / math.log(m)
+        out = ctypes.c_void_p()
+        _check(self._ptr, lib().sdbv_hnsw_create(
+            self._ptr, d, METRICS[metric], m, m0, efc, int(extend), int(keep),
+            seed, ml, ctypes.byref(out)), "sdbv_hnsw_create")
+        return Hnsw(self, out, d)
+
     def knn_batch(self, table, Q, k):
         import numpy as np
         Q = np.ascontiguousarray(Q, dtype=np.float32)
@@ -192,3 +220,69 @@ class Context:
             dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double))),
             "sdbv_knn_batch")
         return ids, dists
+
+
+class Hnsw:
+    """Product HNSW index (mirrors HnswIndex, hnsw/index.rs): host graph +
+    GPU layer-0 expansion. Element ids are insertion ordinals."""
+
+    def __init__(self, ctx, ptr, d):
+        self._ctx = ctx
+        self._ptr = ptr
+        self.d = d
+
+    def insert(self, pt):
+        import numpy as np
+        pt = np.ascontiguousarray(pt, dtype=np.float32)
+        _check(self._ctx._ptr, lib().sdbv_hnsw_insert(
+            self._ptr, pt.ctypes.data_as(ctypes.POINTER(ctypes.c_float))),
+            "sdbv_hnsw_insert")
+
+    def insert_batch(self, pts, nthreads=0):
+        import numpy as np
+        pts = np.ascontiguousarray(pts, dtype=np.float32)
+        _check(self._ctx._ptr, lib().sdbv_hnsw_insert_batch(
+            self._ptr, pts.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            pts.shape[0], nthreads), "sdbv_hnsw_insert_batch")
+
+    def finalize(self, table):
+        _check(self._ctx._ptr, lib().sdbv_hnsw_finalize(self._ptr, table),
+               "sdbv_hnsw_finalize")
+
+    def knn_search(self, q, k, ef):
+        import numpy as np
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        cap = max(k, ef)
+        ids = np.empty(cap, dtype=np.uint64)
+        dists = np.empty(cap, dtype=np.float64)
+        out_n = ctypes.c_uint32(0)
+        _check(self._ctx._ptr, lib().sdbv_hnsw_knn(
+            self._ptr, q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), k, ef,
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.byref(out_n)), "sdbv_hnsw_knn")
+        n = out_n.value
+        return ids[:n], dists[:n]
+
+    def n(self):
+        return lib().sdbv_hnsw_n(self._ptr)
+
+    def num_layers(self):
+        return lib().sdbv_hnsw_layers(self._ptr)
+
+    def l0_csr(self):
+        import numpy as np
+        n = self.n()
+        ec = lib().sdbv_hnsw_l0_edge_count(self._ptr)
+        offsets = np.empty(n + 1, dtype=np.uint32)
+        edges = np.empty(max(ec, 1), dtype=np.uint32)
+        lib().sdbv_hnsw_l0_export(
+            self._ptr,
+            offsets.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+            edges.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)))
+        return offsets, edges[:ec]
+
+    def destroy(self):
+        if self._ptr:
+            lib().sdbv_hnsw_destroy(self._ptr)
+            self._ptr = None

@@ -38,9 +38,15 @@
 #include <rocblas/rocblas.h>
 
 #include <algorithm>
+#include <atomic>
+#include <cfloat>
 #include <chrono>
+#include <cmath>
 #include <cstdint>
+#include <deque>
 #include <limits>
+#include <thread>
+#include <unordered_set>
 #include <cstdio>
 #include <cstring>
 #include <map>
@@ -1345,17 +1351,648 @@ int sdbv_knn_batch(sdbv_ctx *ctx, uint64_t table, const float *Q, uint32_t b,
 	}
 	return SDBV_OK;
 }
-int sdbv_hnsw_upload(sdbv_ctx *ctx, uint64_t, const uint32_t *,
-                     const uint32_t *, uint64_t, uint32_t, sdbv_hnsw **) {
-	if (ctx)
-		ctx->err = "sdbv_hnsw_upload: not implemented in this revision";
-	return SDBV_ERR_UNSUPPORTED;
+} // extern "C"
+
+// ===========================================================================
+// HNSW — product implementation of the reference graph algorithm
+// (hnsw/mod.rs, layer.rs, heuristic.rs, knn.rs queues). Host-side topology +
+// build (as in the reference engine); layer-0 query expansion = the GPU
+// batched gather+distance kernel. Independent of oracle/ (the oracle is the
+// parity checker for this code, never a dependency).
+// ===========================================================================
+
+namespace hnsw {
+
+// f64 total_cmp key (knn.rs:128-160)
+static inline uint64_t total_key(double x) {
+	uint64_t bits;
+	std::memcpy(&bits, &x, 8);
+	return (bits >> 63) ? ~bits : (bits | 0x8000000000000000ULL);
 }
-int sdbv_hnsw_search(sdbv_hnsw *, const float *, uint32_t, uint32_t, uint32_t,
-                     uint32_t, const uint32_t *, uint64_t *, double *,
-                     uint32_t *) {
-	return SDBV_ERR_UNSUPPORTED;
+
+// DoublePriorityQueue restatement (knn.rs:15-123): ordered by total_cmp(dist),
+// FIFO within equal distance; pop_last removes the LATEST of the max key.
+struct PQ {
+	std::map<uint64_t, std::deque<uint32_t>> m;
+	std::map<uint64_t, double> dval;
+	size_t n = 0;
+	void push(double d, uint32_t id) {
+		uint64_t k = total_key(d);
+		m[k].push_back(id);
+		dval[k] = d;
+		n++;
+	}
+	bool pop_first(double *d, uint32_t *id) {
+		if (m.empty())
+			return false;
+		auto it = m.begin();
+		*d = dval[it->first];
+		*id = it->second.front();
+		it->second.pop_front();
+		if (it->second.empty()) {
+			dval.erase(it->first);
+			m.erase(it);
+		}
+		n--;
+		return true;
+	}
+	void pop_last() {
+		if (m.empty())
+			return;
+		auto it = std::prev(m.end());
+		it->second.pop_back();
+		if (it->second.empty()) {
+			dval.erase(it->first);
+			m.erase(it);
+		}
+		n--;
+	}
+	bool peek_first(double *d, uint32_t *id) const {
+		if (m.empty())
+			return false;
+		auto it = m.begin();
+		*d = dval.at(it->first);
+		*id = it->second.front();
+		return true;
+	}
+	double peek_last_dist(double fb) const {
+		return m.empty() ? fb : dval.at(std::prev(m.end())->first);
+	}
+	std::vector<std::pair<double, uint32_t>> to_vec() const {
+		std::vector<std::pair<double, uint32_t>> v;
+		v.reserve(n);
+		for (auto &e : m)
+			for (uint32_t id : e.second)
+				v.push_back({dval.at(e.first), id});
+		return v;
+	}
+};
+
+// Restated per-row distance chain on the host (same ops as the device
+// kernels; the whole translation unit is compiled -ffp-contract=off).
+static double host_dot_f32(const float *a, const float *b, uint32_t d) {
+	float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+	uint32_t i = 0;
+	for (; i + 8 <= d; i += 8)
+		for (uint32_t t = 0; t < 8; t++)
+			p[t] += a[i + t] * b[i + t];
+	float s = 0;
+	s += (p[0] + p[4]);
+	s += (p[1] + p[5]);
+	s += (p[2] + p[6]);
+	s += (p[3] + p[7]);
+	for (; i < d; i++)
+		s += a[i] * b[i];
+	return (double)s;
 }
-void sdbv_hnsw_free(sdbv_hnsw *) {}
+static double host_sumsq_f32(const float *a, uint32_t d) {
+	float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+	uint32_t i = 0;
+	for (; i + 8 <= d; i += 8)
+		for (uint32_t t = 0; t < 8; t++)
+			p[t] += a[i + t] * a[i + t];
+	float s = 0;
+	s += ((p[0] + p[4]) + (p[1] + p[5]));
+	s += ((p[2] + p[6]) + (p[3] + p[7]));
+	for (; i < d; i++)
+		s += a[i] * a[i];
+	return (double)s;
+}
+
+struct Layer {
+	std::vector<std::vector<uint32_t>> edges;
+	uint32_t m_max;
+};
+
+} // namespace hnsw
+
+struct sdbv_hnsw {
+	sdbv_ctx *ctx;
+	uint32_t d;
+	uint8_t metric;
+	uint32_t m, m0, efc;
+	bool extend, keep;
+	double ml;
+	uint64_t rng_state;
+	std::vector<float> vecs;    // host row-major copy (build + upper layers)
+	std::vector<double> norms;  // per-element f64 norm (cosine)
+	std::vector<hnsw::Layer> layers;
+	int64_t enter_point = -1;
+	uint64_t next_id = 0;
+	// parallel build
+	std::vector<std::mutex> node_locks{4096};
+	std::mutex global_mu;
+	// device side (after finalize)
+	uint64_t table = ~0ULL;
+	bool finalized = false;
+	uint32_t *rows_dev = nullptr;
+	double *dout_dev = nullptr;
+	float *q_dev = nullptr;
+	std::string err;
+};
+
+namespace hnsw {
+
+static inline const float *vec(const sdbv_hnsw *h, uint32_t id) {
+	return h->vecs.data() + (uint64_t)id * h->d;
+}
+
+static double dist(const sdbv_hnsw *h, const float *a, double a_norm,
+                   uint32_t id) {
+	if (h->metric == SDBV_METRIC_COSINE) {
+		double dot = host_dot_f32(a, vec(h, id), h->d);
+		return 1.0 - dot / (a_norm * h->norms[id]);
+	}
+	float acc = 0;
+	const float *b = vec(h, id);
+	for (uint32_t i = 0; i < h->d; i++) {
+		float diff = a[i] - b[i];
+		acc += diff * diff;
+	}
+	return sqrt((double)acc);
+}
+static double dist_ee(const sdbv_hnsw *h, uint32_t a, uint32_t b) {
+	return dist(h, vec(h, a), h->metric == SDBV_METRIC_COSINE ? h->norms[a] : 0,
+	            b);
+}
+
+static std::vector<uint32_t> get_edges(sdbv_hnsw *h, const Layer &layer,
+                                       uint32_t id, bool locked) {
+	if (!locked)
+		return id < layer.edges.size() ? layer.edges[id]
+		                               : std::vector<uint32_t>{};
+	std::lock_guard<std::mutex> lk(h->node_locks[id & 4095]);
+	return id < layer.edges.size() ? layer.edges[id] : std::vector<uint32_t>{};
+}
+
+// layer.rs:184-223 — host-distance variant (build path).
+static void search_layer_host(sdbv_hnsw *h, const Layer &layer, const float *q,
+                              double q_norm, PQ &candidates,
+                              std::unordered_set<uint32_t> &visited, PQ &w,
+                              uint32_t ef, bool locked) {
+	double fq = w.peek_last_dist(DBL_MAX);
+	double cd;
+	uint32_t doc;
+	while (candidates.pop_first(&cd, &doc)) {
+		if (cd > fq)
+			break;
+		for (uint32_t e : get_edges(h, layer, doc, locked)) {
+			if (!visited.insert(e).second)
+				continue;
+			double ed = dist(h, q, q_norm, e);
+			if (ed < fq || w.n < ef) {
+				candidates.push(ed, e);
+				w.push(ed, e);
+				if (w.n > ef)
+					w.pop_last();
+				fq = w.peek_last_dist(DBL_MAX);
+			}
+		}
+	}
+}
+
+// heuristic.rs:193-216
+static bool is_closer(sdbv_hnsw *h, double ed, uint32_t e,
+                      std::vector<uint32_t> &r) {
+	for (uint32_t rid : r)
+		if (ed > dist_ee(h, e, rid))
+			return false;
+	r.push_back(e);
+	return true;
+}
+
+// heuristic.rs:35-116 (+ extend :118-157)
+static void select_neighbors(sdbv_hnsw *h, const Layer &layer, uint32_t q_id,
+                             const float *q_pt, double q_norm, PQ c,
+                             std::vector<uint32_t> &res, bool locked) {
+	if (h->extend) {
+		std::unordered_set<uint32_t> ex;
+		auto base = c.to_vec();
+		for (auto &e : base)
+			ex.insert(e.second);
+		for (auto &e : base)
+			for (uint32_t adj : get_edges(h, layer, e.second, locked))
+				if (adj != q_id && ex.insert(adj).second)
+					c.push(dist(h, q_pt, q_norm, adj), adj);
+	}
+	uint32_t m_max = layer.m_max;
+	if (c.n <= m_max) {
+		for (auto &e : c.to_vec())
+			res.push_back(e.second);
+		return;
+	}
+	std::vector<uint32_t> pruned;
+	double ed;
+	uint32_t e;
+	while (c.pop_first(&ed, &e)) {
+		if (is_closer(h, ed, e, res)) {
+			if (res.size() == m_max)
+				break;
+		} else if (h->keep) {
+			pruned.push_back(e);
+		}
+	}
+	if (h->keep) {
+		size_t nmore = m_max - res.size();
+		for (size_t i = 0; i < nmore && i < pruned.size(); i++)
+			res.push_back(pruned[i]);
+	}
+}
+
+// layer.rs:342-387
+static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
+                       const float *q_pt, double q_norm, PQ eps, bool locked) {
+	PQ w = eps;
+	std::unordered_set<uint32_t> visited;
+	for (auto &e : eps.to_vec())
+		visited.insert(e.second);
+	search_layer_host(h, layer, q_pt, q_norm, eps, visited, w, h->efc, locked);
+	PQ out = w;
+	std::vector<uint32_t> neighbors;
+	select_neighbors(h, layer, q_id, q_pt, q_norm, w, neighbors, locked);
+	{
+		// append (not overwrite): a concurrent inserter may already have
+		// back-linked into q_id; sequential mode this is plain assignment
+		std::lock_guard<std::mutex> lk(h->node_locks[q_id & 4095]);
+		auto &eq = layer.edges[q_id];
+		for (uint32_t e : neighbors)
+			if (e != q_id &&
+			    std::find(eq.begin(), eq.end(), e) == eq.end())
+				eq.push_back(e);
+	}
+	for (uint32_t e : neighbors) {
+		if (e == q_id)
+			continue;
+		std::vector<uint32_t> conn;
+		{
+			std::lock_guard<std::mutex> lk(h->node_locks[e & 4095]);
+			auto &ee = layer.edges[e];
+			if (std::find(ee.begin(), ee.end(), q_id) == ee.end())
+				ee.push_back(q_id);
+			if (ee.size() > layer.m_max)
+				conn = ee;
+		}
+		if (!conn.empty()) {
+			// prune (layer.rs:363-377) — distances computed outside the lock
+			PQ ec;
+			for (uint32_t nid : conn)
+				ec.push(dist_ee(h, e, nid), nid);
+			std::vector<uint32_t> enew;
+			select_neighbors(h, layer, e, vec(h, e),
+			                 h->metric == SDBV_METRIC_COSINE ? h->norms[e] : 0,
+			                 std::move(ec), enew, locked);
+			std::lock_guard<std::mutex> lk(h->node_locks[e & 4095]);
+			layer.edges[e] = enew;
+		}
+	}
+	return out;
+}
+
+// hnsw/mod.rs:263-266 (level RNG restatement — same contract as the oracle)
+static inline uint64_t splitmix_host(uint64_t z) {
+	z += 0x9E3779B97F4A7C15ULL;
+	z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+	z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+	return z ^ (z >> 31);
+}
+static uint32_t next_level(sdbv_hnsw *h) {
+	h->rng_state = splitmix_host(h->rng_state);
+	double u = (double)(h->rng_state >> 11) * 0x1.0p-53;
+	if (u <= 0.0)
+		u = 0x1.0p-53;
+	double lvl = std::floor(-std::log(u) * h->ml);
+	return (uint32_t)std::min(std::max(lvl, 0.0), 30.0);
+}
+
+// hnsw/mod.rs:230-394 insert at a given level (vector already appended)
+static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
+                      bool locked) {
+	const float *q_pt = vec(h, q_id);
+	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+	uint32_t top_up;
+	{
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		top_up = (uint32_t)h->layers.size() - 1;
+		for (uint32_t i = top_up; i < q_level; i++)
+			h->layers.push_back(hnsw::Layer{
+			    std::vector<std::vector<uint32_t>>(h->vecs.size() / h->d),
+			    h->m});
+		for (auto &l : h->layers)
+			if (l.edges.size() <= q_id)
+				l.edges.resize(h->vecs.size() / h->d);
+		if (h->enter_point < 0) {
+			h->enter_point = q_id;
+			return;
+		}
+	}
+	uint64_t ep_id = (uint64_t)h->enter_point;
+	double ep_dist = dist(h, q_pt, q_norm, (uint32_t)ep_id);
+	if (q_level < top_up) {
+		for (uint32_t l = top_up; l > q_level; l--) {
+			PQ cand;
+			cand.push(ep_dist, (uint32_t)ep_id);
+			std::unordered_set<uint32_t> visited{(uint32_t)ep_id};
+			PQ w = cand;
+			search_layer_host(h, h->layers[l], q_pt, q_norm, cand, visited, w,
+			                  1, locked);
+			double dd;
+			uint32_t ii;
+			if (w.peek_first(&dd, &ii)) {
+				ep_dist = dd;
+				ep_id = ii;
+			}
+		}
+	}
+	PQ eps;
+	eps.push(ep_dist, (uint32_t)ep_id);
+	uint32_t ins_to = std::min(q_level, top_up);
+	for (uint32_t l = ins_to; l >= 1; l--)
+		eps = layer_insert(h, h->layers[l], q_id, q_pt, q_norm, std::move(eps),
+		                   locked);
+	layer_insert(h, h->layers[0], q_id, q_pt, q_norm, std::move(eps), locked);
+	if (q_level > top_up) {
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		h->enter_point = q_id;
+	}
+}
+
+} // namespace hnsw
+
+extern "C" {
+
+int sdbv_hnsw_create(sdbv_ctx *ctx, uint32_t d, uint8_t metric, uint32_t m,
+                     uint32_t m0, uint32_t efc, int extend, int keep,
+                     uint64_t seed, double ml, sdbv_hnsw **out) {
+	// ctx may be NULL for a host-only (build/export) index; finalize and
+	// knn require a context and fail loudly without one.
+	if (d == 0 || d > MAX_D || (d % 4) != 0 ||
+	    metric > SDBV_METRIC_EUCLIDEAN || m == 0 || m0 == 0)
+		return SDBV_ERR_BAD_ARG;
+	auto *h = new sdbv_hnsw();
+	h->ctx = ctx;
+	h->d = d;
+	h->metric = metric;
+	h->m = m;
+	h->m0 = m0;
+	h->efc = efc;
+	h->extend = extend != 0;
+	h->keep = keep != 0;
+	h->ml = ml;
+	h->rng_state = seed;
+	h->layers.push_back(hnsw::Layer{{}, m0});
+	*out = h;
+	return SDBV_OK;
+}
+
+void sdbv_hnsw_destroy(sdbv_hnsw *h) {
+	if (!h)
+		return;
+	if (h->rows_dev)
+		(void)hipFree(h->rows_dev);
+	if (h->dout_dev)
+		(void)hipFree(h->dout_dev);
+	if (h->q_dev)
+		(void)hipFree(h->q_dev);
+	delete h;
+}
+
+static void hnsw_append_vec(sdbv_hnsw *h, const float *pt) {
+	h->vecs.insert(h->vecs.end(), pt, pt + h->d);
+	if (h->metric == SDBV_METRIC_COSINE)
+		h->norms.push_back(sqrt(hnsw::host_sumsq_f32(pt, h->d)));
+}
+
+int sdbv_hnsw_insert(sdbv_hnsw *h, const float *pt) {
+	if (!h || h->finalized)
+		return SDBV_ERR_BAD_ARG;
+	uint32_t q_id = (uint32_t)h->next_id++;
+	hnsw_append_vec(h, pt);
+	hnsw::insert_at(h, q_id, hnsw::next_level(h), /*locked=*/false);
+	return SDBV_OK;
+}
+
+int sdbv_hnsw_insert_batch(sdbv_hnsw *h, const float *pts, uint64_t n,
+                           int nthreads) {
+	if (!h || h->finalized)
+		return SDBV_ERR_BAD_ARG;
+	if (nthreads <= 0)
+		nthreads = (int)std::thread::hardware_concurrency();
+	if (nthreads <= 1) {
+		for (uint64_t i = 0; i < n; i++) {
+			int rc = sdbv_hnsw_insert(h, pts + i * h->d);
+			if (rc)
+				return rc;
+		}
+		return SDBV_OK;
+	}
+	uint64_t base = h->next_id;
+	// levels drawn deterministically per ordinal (sequential RNG)
+	std::vector<uint32_t> levels(n);
+	for (uint64_t i = 0; i < n; i++)
+		levels[i] = hnsw::next_level(h);
+	h->vecs.reserve(h->vecs.size() + n * h->d);
+	for (uint64_t i = 0; i < n; i++)
+		hnsw_append_vec(h, pts + i * h->d);
+	h->next_id += n;
+	// pre-size every layer to the final element count (no growth races)
+	{
+		uint32_t maxl = 0;
+		for (auto l : levels)
+			maxl = std::max(maxl, l);
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		while (h->layers.size() <= maxl)
+			h->layers.push_back(hnsw::Layer{{}, h->m});
+		for (auto &l : h->layers)
+			l.edges.resize(h->next_id);
+	}
+	// the first elements (empty/near-empty graph) go in alone so every
+	// worker sees a connected entry point
+	uint64_t start = 0;
+	uint64_t warm = std::min<uint64_t>(n, h->enter_point < 0 ? 64 : 0);
+	for (; start < warm; start++)
+		hnsw::insert_at(h, (uint32_t)(base + start), levels[start], true);
+	std::atomic<uint64_t> next{start};
+	std::vector<std::thread> workers;
+	for (int w = 0; w < nthreads; w++)
+		workers.emplace_back([&] {
+			for (;;) {
+				uint64_t i = next.fetch_add(1);
+				if (i >= n)
+					break;
+				hnsw::insert_at(h, (uint32_t)(base + i), levels[i], true);
+			}
+		});
+	for (auto &w : workers)
+		w.join();
+	return SDBV_OK;
+}
+
+uint64_t sdbv_hnsw_n(sdbv_hnsw *h) { return h ? h->next_id : 0; }
+uint32_t sdbv_hnsw_layers(sdbv_hnsw *h) {
+	return h ? (uint32_t)h->layers.size() : 0;
+}
+uint64_t sdbv_hnsw_l0_edge_count(sdbv_hnsw *h) {
+	uint64_t c = 0;
+	for (auto &e : h->layers[0].edges)
+		c += e.size();
+	return c;
+}
+void sdbv_hnsw_l0_export(sdbv_hnsw *h, uint32_t *offsets, uint32_t *edges) {
+	uint32_t off = 0;
+	for (uint64_t i = 0; i < h->next_id; i++) {
+		offsets[i] = off;
+		if (i < h->layers[0].edges.size())
+			for (uint32_t e : h->layers[0].edges[i])
+				edges[off++] = e;
+	}
+	offsets[h->next_id] = off;
+}
+
+int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
+	if (!h || !h->ctx || h->next_id == 0)
+		return SDBV_ERR_BAD_ARG;
+	int rc = sdbv_stage_corpus(h->ctx, table, h->vecs.data(), nullptr,
+	                           h->next_id, h->d, h->metric);
+	if (rc)
+		return rc;
+	sdbv_ctx *ctx = h->ctx;
+	HIP_CHECK(ctx, hipMalloc(&h->rows_dev, (h->m0 + 1) * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMalloc(&h->dout_dev, (h->m0 + 1) * sizeof(double)));
+	HIP_CHECK(ctx, hipMalloc(&h->q_dev, h->d * sizeof(float)));
+	h->table = table;
+	h->finalized = true;
+	return SDBV_OK;
+}
+
+// knn_search (hnsw/index.rs:270-335 without the host-kept parts): host
+// upper-layer descent, then the layer-0 ef-search where each hop's
+// neighbour distances come from the GPU gather kernel (the north_star's
+// "batched gather + distance" design). Exact queue semantics preserved:
+// distances are queue-independent, so batching them per hop does not change
+// the reference's accept/update order.
+int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
+                  uint64_t *out_ids, double *out_dists, uint32_t *out_n) {
+	using namespace hnsw;
+	if (!h || !h->finalized)
+		return SDBV_ERR_BAD_ARG;
+	if (h->enter_point < 0) {
+		*out_n = 0;
+		return SDBV_OK;
+	}
+	sdbv_ctx *ctx = h->ctx;
+	std::lock_guard<std::mutex> lk(ctx->mu);
+	auto it = ctx->tables.find(h->table);
+	if (it == ctx->tables.end())
+		return SDBV_ERR_NO_TABLE;
+	Table &t = it->second;
+
+	double q_norm_d = 0;
+	float q_sumsq = 0;
+	{
+		float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+		uint32_t i = 0;
+		for (; i + 8 <= h->d; i += 8)
+			for (uint32_t tt = 0; tt < 8; tt++)
+				p[tt] += q[i + tt] * q[i + tt];
+		q_sumsq = 0;
+		q_sumsq += ((p[0] + p[4]) + (p[1] + p[5]));
+		q_sumsq += ((p[2] + p[6]) + (p[3] + p[7]));
+		for (; i < h->d; i++)
+			q_sumsq += q[i] * q[i];
+		q_norm_d = sqrt((double)q_sumsq);
+	}
+
+	// upper layers: greedy descent, host distances (tiny)
+	uint32_t ep_id = (uint32_t)h->enter_point;
+	double ep_dist = dist(h, q, q_norm_d, ep_id);
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		PQ cand;
+		cand.push(ep_dist, ep_id);
+		std::unordered_set<uint32_t> visited{ep_id};
+		PQ w = cand;
+		search_layer_host(h, h->layers[l], q, q_norm_d, cand, visited, w, 1,
+		                  false);
+		double dd;
+		uint32_t ii;
+		if (w.peek_first(&dd, &ii)) {
+			ep_dist = dd;
+			ep_id = ii;
+		}
+	}
+
+	// layer 0: ef-search with GPU batched neighbour expansion
+	HIP_CHECK(ctx, hipMemcpyAsync(h->q_dev, q, h->d * sizeof(float),
+	                              hipMemcpyHostToDevice, ctx->stream));
+	const Layer &l0 = h->layers[0];
+	PQ candidates, w;
+	candidates.push(ep_dist, ep_id);
+	w.push(ep_dist, ep_id);
+	std::vector<bool> visited(h->next_id, false);
+	visited[ep_id] = true;
+	double fq = w.peek_last_dist(DBL_MAX);
+	std::vector<uint32_t> frontier;
+	std::vector<double> fdists(h->m0 + 1);
+	double cd;
+	uint32_t doc;
+	while (candidates.pop_first(&cd, &doc)) {
+		if (cd > fq)
+			break;
+		frontier.clear();
+		for (uint32_t e : l0.edges[doc])
+			if (!visited[e]) {
+				visited[e] = true;
+				frontier.push_back(e);
+			}
+		if (frontier.empty())
+			continue;
+		// ONE gather+distance launch for this hop's neighbours
+		HIP_CHECK(ctx, hipMemcpyAsync(h->rows_dev, frontier.data(),
+		                              frontier.size() * sizeof(uint32_t),
+		                              hipMemcpyHostToDevice, ctx->stream));
+		if (t.metric == SDBV_METRIC_COSINE)
+			hipLaunchKernelGGL(k_gather_dist<0>, dim3((uint32_t)frontier.size()),
+			                   dim3(64), 0, ctx->stream, t.cm, t.norms,
+			                   t.n_pad, t.d, h->rows_dev,
+			                   (uint32_t)frontier.size(), h->q_dev, q_norm_d,
+			                   h->dout_dev);
+		else
+			hipLaunchKernelGGL(k_gather_dist<1>, dim3((uint32_t)frontier.size()),
+			                   dim3(64), 0, ctx->stream, t.cm, t.norms,
+			                   t.n_pad, t.d, h->rows_dev,
+			                   (uint32_t)frontier.size(), h->q_dev, q_norm_d,
+			                   h->dout_dev);
+		HIP_CHECK(ctx, hipMemcpyAsync(fdists.data(), h->dout_dev,
+		                              frontier.size() * sizeof(double),
+		                              hipMemcpyDeviceToHost, ctx->stream));
+		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		// sequential accept/update in edge order (layer.rs:195-218)
+		for (size_t i = 0; i < frontier.size(); i++) {
+			double ed = fdists[i];
+			uint32_t e = frontier[i];
+			if (ed < fq || w.n < ef) {
+				candidates.push(ed, e);
+				w.push(ed, e);
+				if (w.n > ef)
+					w.pop_last();
+				fq = w.peek_last_dist(DBL_MAX);
+			}
+		}
+	}
+
+	// to_vec_limit(k) in (dist, FIFO) order (knn.rs:92-104), then the
+	// KnnResultBuilder (dist, id) final ordering (knn.rs:363)
+	auto v = w.to_vec();
+	size_t m = std::min<size_t>(k, v.size());
+	std::vector<std::pair<std::pair<uint64_t, uint32_t>, double>> fin(m);
+	for (size_t i = 0; i < m; i++)
+		fin[i] = {{total_key(v[i].first), v[i].second}, v[i].first};
+	std::sort(fin.begin(), fin.end());
+	*out_n = (uint32_t)m;
+	for (size_t i = 0; i < m; i++) {
+		out_ids[i] = fin[i].first.second;
+		out_dists[i] = fin[i].second;
+	}
+	return SDBV_OK;
+}
 
 } // extern "C"
