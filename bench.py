@@ -40,6 +40,11 @@ def main():
     ap.add_argument("--batch", type=int, default=0,
                     help=">0: run BASELINE configs[3] (batched MFMA path); a "
                          "step is one batch of this many queries")
+    ap.add_argument("--hnsw", action="store_true",
+                    help="run BASELINE configs[2]: HNSW index (M=16, ef=64) "
+                         "K=10 single query; index built at bench start "
+                         "(parallel host build, outside the timed region)")
+    ap.add_argument("--ef", type=int, default=64)
     ap.add_argument("--seed", type=lambda x: int(x, 0), default=0x5DB1)
     ap.add_argument("--cpu-sample-rows", type=int, default=2_000_000,
                     help="bounded sample for the cpu_baseline leg")
@@ -69,8 +74,21 @@ def main():
     rows = args.rows
     row_offset = rank * rows
     t0 = time.perf_counter()
-    ctx.stage_synthetic(1, rows, args.dim, metric=args.metric, seed=args.seed,
-                        row_offset=row_offset, id_base=row_offset)
+    hnsw_index = None
+    if args.hnsw:
+        # HNSW does not shard (sequential graph traversal):
+        # replicas only — each rank builds/holds a replica of its shard.
+        from surrealdb_amd.synth import gen_f32 as _gen
+        pts = _gen(args.seed, row_offset, rows, args.dim)
+        hnsw_index = ctx.hnsw_create(args.dim, metric=args.metric, m=16,
+                                     m0=32, efc=150, seed=args.seed)
+        hnsw_index.insert_batch(pts, nthreads=0)
+        del pts
+        hnsw_index.finalize(1)
+    else:
+        ctx.stage_synthetic(1, rows, args.dim, metric=args.metric,
+                            seed=args.seed, row_offset=row_offset,
+                            id_base=row_offset)
     stage_s = time.perf_counter() - t0
 
     nq = (args.steps + args.warmup) * max(args.batch, 1)
@@ -105,8 +123,13 @@ def main():
     def one_query(qi):
         if args.batch > 0:
             return one_batch(qi)
-        ids, dists = ctx.knn_bruteforce(1, queries[qi % len(queries)],
-                                        args.k)
+        if hnsw_index is not None:
+            ids, dists = hnsw_index.knn_search(queries[qi % len(queries)],
+                                               args.k, args.ef)
+            ids = ids + row_offset  # shard-local ordinals -> global ids
+        else:
+            ids, dists = ctx.knn_bruteforce(1, queries[qi % len(queries)],
+                                            args.k)
         if world > 1:
             pad = args.k - len(ids)
             if pad:
@@ -136,12 +159,15 @@ def main():
     torch.cuda.synchronize(device) if torch.cuda.is_available() else None
     step_times = []
     scan_ms_acc = 0.0
+    rows_scanned_acc = 0
     t_begin = time.perf_counter()
     for i in range(args.steps):
         ts = time.perf_counter()
         one_query(args.warmup + i)
         step_times.append(time.perf_counter() - ts)
-        scan_ms_acc += ctx.stats()["last_scan_kernel_ms"]
+        st = ctx.stats()
+        scan_ms_acc += st["last_scan_kernel_ms"]
+        rows_scanned_acc += st["last_rows_scanned"]
     torch.cuda.synchronize(device) if torch.cuda.is_available() else None
     if dist:
         dist.barrier()
@@ -175,6 +201,19 @@ def main():
             "peak": F32_MFMA_PEAK_TF,
             "unit": "TFLOP/s",
             "frac": round(achieved_tf / F32_MFMA_PEAK_TF, 4),
+            "traffic": float(traffic) if traffic else None,
+        }
+    elif args.hnsw:
+        # HNSW is latency/gather-bound: achieved = gathered rows
+        # x row bytes over the GPU gather sections (incl. per-hop transfers)
+        alg_bytes = (rows_scanned_acc / args.steps) * args.dim * 4
+        achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved_gbs, 1),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
             "traffic": float(traffic) if traffic else None,
         }
     else:
@@ -236,6 +275,9 @@ def main():
                 f"brute-force {args.metric} KNN, 10M rows/GPU x 768-dim f32, "
                 f"K=10, batch={args.batch} MFMA path (BASELINE configs[3])"
                 if args.batch > 0 else
+                f"HNSW (M=16, ef={args.ef}) {args.metric} K=10 single query "
+                f"(BASELINE configs[2]); latency/gather-bound"
+                if args.hnsw else
                 "brute-force cosine KNN, 10M rows/GPU x 768-dim f32, "
                 "K=10, single query (BASELINE configs[1])"),
             "batch": args.batch,

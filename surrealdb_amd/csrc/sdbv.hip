@@ -1642,6 +1642,12 @@ static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			                 h->metric == SDBV_METRIC_COSINE ? h->norms[e] : 0,
 			                 std::move(ec), enew, locked);
 			std::lock_guard<std::mutex> lk(h->node_locks[e & 4095]);
+			// keep any back-edges another inserter added since the snapshot
+			// (parallel mode only; sequential sees none)
+			for (uint32_t cur : layer.edges[e])
+				if (std::find(conn.begin(), conn.end(), cur) == conn.end() &&
+				    std::find(enew.begin(), enew.end(), cur) == enew.end())
+					enew.push_back(cur);
 			layer.edges[e] = enew;
 		}
 	}
@@ -1934,6 +1940,8 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 	std::vector<double> fdists(h->m0 + 1);
 	double cd;
 	uint32_t doc;
+	double gpu_ms = 0;
+	uint64_t gathered = 0;
 	while (candidates.pop_first(&cd, &doc)) {
 		if (cd > fq)
 			break;
@@ -1945,6 +1953,7 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 			}
 		if (frontier.empty())
 			continue;
+		auto hop_t0 = std::chrono::steady_clock::now();
 		// ONE gather+distance launch for this hop's neighbours
 		HIP_CHECK(ctx, hipMemcpyAsync(h->rows_dev, frontier.data(),
 		                              frontier.size() * sizeof(uint32_t),
@@ -1965,6 +1974,10 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 		                              frontier.size() * sizeof(double),
 		                              hipMemcpyDeviceToHost, ctx->stream));
 		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		gpu_ms += std::chrono::duration<double, std::milli>(
+		              std::chrono::steady_clock::now() - hop_t0)
+		              .count();
+		gathered += frontier.size();
 		// sequential accept/update in edge order (layer.rs:195-218)
 		for (size_t i = 0; i < frontier.size(); i++) {
 			double ed = fdists[i];
@@ -1978,6 +1991,10 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 			}
 		}
 	}
+
+	ctx->stats.last_scan_kernel_ms = gpu_ms; // gather sections (incl. copies)
+	ctx->stats.last_merge_kernel_ms = 0;
+	ctx->stats.last_rows_scanned = gathered;
 
 	// to_vec_limit(k) in (dist, FIFO) order (knn.rs:92-104), then the
 	// KnnResultBuilder (dist, id) final ordering (knn.rs:363)
