@@ -82,7 +82,7 @@ def main():
         pts = _gen(args.seed, row_offset, rows, args.dim)
         hnsw_index = ctx.hnsw_create(args.dim, metric=args.metric, m=16,
                                      m0=32, efc=150, seed=args.seed)
-        hnsw_index.insert_batch(pts, nthreads=0)
+        hnsw_index.insert_batch(pts, nthreads=os.cpu_count())
         del pts
         hnsw_index.finalize(1)
     else:

@@ -97,6 +97,11 @@ int sdbv_stage_synthetic(sdbv_ctx *, uint64_t table, uint64_t n, uint32_t d,
 uint64_t sdbv_table_rows(sdbv_ctx *, uint64_t table);
 int sdbv_drop_table(sdbv_ctx *, uint64_t table);
 
+/* Host-side generator of the committed synthetic-data contract (bench/test
+ * input prep; bit-identical to the device staging generator). */
+void sdbv_gen_f32(uint64_t seed, uint64_t row0, uint64_t nrows, uint32_t d,
+                  float *out);
+
 /* Brute-force exact K-nearest-neighbour scan of a staged table.
  * Results sorted ascending by (distance f64-total_cmp, id).
  * Distances follow Distance::calculate F32 semantics (vector.rs:244-249 /

@@ -1488,6 +1488,8 @@ struct sdbv_hnsw {
 	uint32_t *rows_dev = nullptr;
 	double *dout_dev = nullptr;
 	float *q_dev = nullptr;
+	uint32_t *rows_pinned = nullptr; // pinned host staging (per-hop latency)
+	double *dists_pinned = nullptr;
 	std::string err;
 };
 
@@ -1726,6 +1728,36 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 
 extern "C" {
 
+// Host-side synthetic generator (same committed bit contract as k_gen_cm and
+// the oracle); parallel over rows. Input prep for benches/tests — no GPU.
+void sdbv_gen_f32(uint64_t seed, uint64_t row0, uint64_t nrows, uint32_t d,
+                  float *out) {
+	int nthreads = (int)std::thread::hardware_concurrency();
+	if (nthreads < 1)
+		nthreads = 1;
+	if (nrows < 4096)
+		nthreads = 1;
+	std::vector<std::thread> ws;
+	std::atomic<uint64_t> next{0};
+	auto work = [&] {
+		for (;;) {
+			uint64_t i = next.fetch_add(4096);
+			if (i >= nrows)
+				break;
+			uint64_t end = std::min(nrows, i + 4096);
+			for (uint64_t r = i; r < end; r++)
+				for (uint32_t j = 0; j < d; j++)
+					out[r * d + j] =
+					    d_gen_elem(seed, (row0 + r) * (uint64_t)d + j);
+		}
+	};
+	for (int w = 0; w < nthreads - 1; w++)
+		ws.emplace_back(work);
+	work();
+	for (auto &w : ws)
+		w.join();
+}
+
 int sdbv_hnsw_create(sdbv_ctx *ctx, uint32_t d, uint8_t metric, uint32_t m,
                      uint32_t m0, uint32_t efc, int extend, int keep,
                      uint64_t seed, double ml, sdbv_hnsw **out) {
@@ -1759,6 +1791,10 @@ void sdbv_hnsw_destroy(sdbv_hnsw *h) {
 		(void)hipFree(h->dout_dev);
 	if (h->q_dev)
 		(void)hipFree(h->q_dev);
+	if (h->rows_pinned)
+		(void)hipHostFree(h->rows_pinned);
+	if (h->dists_pinned)
+		(void)hipHostFree(h->dists_pinned);
 	delete h;
 }
 
@@ -1865,6 +1901,10 @@ int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
 	HIP_CHECK(ctx, hipMalloc(&h->rows_dev, (h->m0 + 1) * sizeof(uint32_t)));
 	HIP_CHECK(ctx, hipMalloc(&h->dout_dev, (h->m0 + 1) * sizeof(double)));
 	HIP_CHECK(ctx, hipMalloc(&h->q_dev, h->d * sizeof(float)));
+	HIP_CHECK(ctx, hipHostMalloc(&h->rows_pinned,
+	                             (h->m0 + 1) * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipHostMalloc(&h->dists_pinned,
+	                             (h->m0 + 1) * sizeof(double)));
 	h->table = table;
 	h->finalized = true;
 	return SDBV_OK;
@@ -1955,7 +1995,9 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 			continue;
 		auto hop_t0 = std::chrono::steady_clock::now();
 		// ONE gather+distance launch for this hop's neighbours
-		HIP_CHECK(ctx, hipMemcpyAsync(h->rows_dev, frontier.data(),
+		std::memcpy(h->rows_pinned, frontier.data(),
+		            frontier.size() * sizeof(uint32_t));
+		HIP_CHECK(ctx, hipMemcpyAsync(h->rows_dev, h->rows_pinned,
 		                              frontier.size() * sizeof(uint32_t),
 		                              hipMemcpyHostToDevice, ctx->stream));
 		if (t.metric == SDBV_METRIC_COSINE)
@@ -1970,10 +2012,12 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 			                   t.n_pad, t.d, h->rows_dev,
 			                   (uint32_t)frontier.size(), h->q_dev, q_norm_d,
 			                   h->dout_dev);
-		HIP_CHECK(ctx, hipMemcpyAsync(fdists.data(), h->dout_dev,
+		HIP_CHECK(ctx, hipMemcpyAsync(h->dists_pinned, h->dout_dev,
 		                              frontier.size() * sizeof(double),
 		                              hipMemcpyDeviceToHost, ctx->stream));
 		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		std::memcpy(fdists.data(), h->dists_pinned,
+		            frontier.size() * sizeof(double));
 		gpu_ms += std::chrono::duration<double, std::milli>(
 		              std::chrono::steady_clock::now() - hop_t0)
 		              .count();

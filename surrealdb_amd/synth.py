@@ -1,14 +1,31 @@
-"""Vectorised host-side generator of the committed synthetic-data contract.
+"""Host-side generator of the committed synthetic-data contract.
 
-Bit-identical to the device generator (csrc/sdbv.hip k_gen_cm) and the oracle
-(orc_gen_f32): element(seed, row*d+j) = splitmix64 -> f64 [0,1) -> [-20,20) f32.
-Product-side (bench/input prep) — no oracle dependency.
+Primary path: the product C library's parallel generator (sdbv_gen_f32,
+bit-identical to the device staging generator and the oracle).
+`gen_f32_numpy` is a pure-numpy restatement kept for cross-checking the
+contract in tests (numpy's uint64 ops are too slow for bulk use).
 """
+import ctypes
+
 import numpy as np
 
 _C1 = np.uint64(0x9E3779B97F4A7C15)
 _C2 = np.uint64(0xBF58476D1CE4E5B9)
 _C3 = np.uint64(0x94D049BB133111EB)
+
+
+def gen_f32(seed, row0, nrows, d):
+    from . import lib
+    L = lib()
+    if not hasattr(L.sdbv_gen_f32, "_typed"):
+        L.sdbv_gen_f32.argtypes = [ctypes.c_uint64, ctypes.c_uint64,
+                                   ctypes.c_uint64, ctypes.c_uint32,
+                                   ctypes.POINTER(ctypes.c_float)]
+        L.sdbv_gen_f32._typed = True
+    out = np.empty((nrows, d), dtype=np.float32)
+    L.sdbv_gen_f32(seed, row0, nrows, d,
+                   out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)))
+    return out
 
 
 def _splitmix64(z):
@@ -18,10 +35,10 @@ def _splitmix64(z):
     return z ^ (z >> np.uint64(31))
 
 
-def gen_f32(seed, row0, nrows, d):
+def gen_f32_numpy(seed, row0, nrows, d):
     with np.errstate(over="ignore"):
-        gidx = (np.uint64(row0) + np.arange(nrows, dtype=np.uint64)[:, None] ) * np.uint64(d) \
-            + np.arange(d, dtype=np.uint64)[None, :]
+        gidx = (np.uint64(row0) + np.arange(nrows, dtype=np.uint64)[:, None]) \
+            * np.uint64(d) + np.arange(d, dtype=np.uint64)[None, :]
         x = _splitmix64(np.uint64(seed) + gidx.ravel())
     u = (x >> np.uint64(11)).astype(np.float64) * (2.0 ** -53)
     return (-20.0 + 40.0 * u).astype(np.float32).reshape(nrows, d)

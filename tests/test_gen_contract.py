@@ -41,7 +41,9 @@ def test_gen_row_offset_consistency():
 def test_synth_numpy_matches_oracle():
     """The product-side numpy generator (surrealdb_amd.synth) must be
     bit-identical to the oracle C generator."""
-    from surrealdb_amd.synth import gen_f32 as np_gen
+    from surrealdb_amd.synth import gen_f32 as np_gen, gen_f32_numpy
     a = oracle.gen_f32(0x5DB1, 999_983, 64, 768)
-    b = np_gen(0x5DB1, 999_983, 64, 768)
+    b = np_gen(0x5DB1, 999_983, 64, 768)          # product C generator
+    c = gen_f32_numpy(0x5DB1, 999_983, 64, 768)   # numpy restatement
     assert np.array_equal(a, b)
+    assert np.array_equal(a, c)
