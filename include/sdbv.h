@@ -161,6 +161,13 @@ int sdbv_hnsw_finalize(sdbv_hnsw *, uint64_t table);
  * expansion. Results ascending (dist total_cmp, id), truncated to k. */
 int sdbv_hnsw_knn(sdbv_hnsw *, const float *q, uint32_t k, uint32_t ef,
                   uint64_t *out_ids, double *out_dists, uint32_t *out_n);
+/* Batched ef-search on the persistent kernel (one query per workgroup, the
+ * whole layer-0 best-first loop in-kernel — same exact result contract as
+ * sdbv_hnsw_knn). Q is b x d; outputs are b x k (+ out_ns per query).
+ * ef <= 512. */
+int sdbv_hnsw_knn_batch(sdbv_hnsw *, const float *Q, uint32_t b, uint32_t k,
+                        uint32_t ef, uint64_t *out_ids, double *out_dists,
+                        uint32_t *out_ns);
 void sdbv_hnsw_destroy(sdbv_hnsw *);
 /* Introspection (parity tests): */
 uint64_t sdbv_hnsw_n(sdbv_hnsw *);

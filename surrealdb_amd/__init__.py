@@ -82,6 +82,7 @@ def lib():
     L.sdbv_hnsw_insert_batch.argtypes = [vp, f32p, u64, ctypes.c_int]
     L.sdbv_hnsw_finalize.argtypes = [vp, u64]
     L.sdbv_hnsw_knn.argtypes = [vp, f32p, u32, u32, u64p, f64p, u32p]
+    L.sdbv_hnsw_knn_batch.argtypes = [vp, f32p, u32, u32, u32, u64p, f64p, u32p]
     L.sdbv_hnsw_destroy.argtypes = [vp]
     L.sdbv_hnsw_n.restype = u64
     L.sdbv_hnsw_n.argtypes = [vp]
@@ -263,6 +264,23 @@ class Hnsw:
             ctypes.byref(out_n)), "sdbv_hnsw_knn")
         n = out_n.value
         return ids[:n], dists[:n]
+
+    def knn_search_batch(self, Q, k, ef):
+        """Batched ef-search on the persistent kernel (one query per
+        workgroup); same exact result contract as knn_search."""
+        import numpy as np
+        Q = np.ascontiguousarray(Q, dtype=np.float32)
+        b = Q.shape[0]
+        ids = np.empty((b, k), dtype=np.uint64)
+        dists = np.empty((b, k), dtype=np.float64)
+        ns = np.empty(b, dtype=np.uint32)
+        _check(self._ctx._ptr, lib().sdbv_hnsw_knn_batch(
+            self._ptr, Q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), b, k,
+            ef, ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ns.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32))),
+            "sdbv_hnsw_knn_batch")
+        return ids, dists, ns
 
     def n(self):
         return lib().sdbv_hnsw_n(self._ptr)

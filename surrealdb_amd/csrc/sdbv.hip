@@ -825,6 +825,291 @@ __global__ void k_batch_exact(const float *__restrict__ cm,
 }
 
 // ---------------------------------------------------------------------------
+// Persistent HNSW ef-search kernel: ONE query per 64-thread workgroup, the
+// entire best-first loop (layer.rs:184-223) inside one launch. Queues live in
+// LDS as append-only arrays with wave-parallel (key,seq) min/max scans —
+// exactly the DoublePriorityQueue semantics (knn.rs:15-123: total_cmp order,
+// FIFO within equal distance, pop_last = latest of the max key). Neighbour
+// distances: one LANE per row over the ROW-MAJOR vector copy with the
+// restated per-row chain (bit-identical to the oracle / single-query path).
+// Visited set: per-query global bitset.
+// ---------------------------------------------------------------------------
+#define HQ_CAND_CAP 2048
+#define HQ_EF_CAP 512
+#define HQ_FLAG_OVERFLOW 1u
+
+__global__ __launch_bounds__(64) void k_hnsw_search(
+    const float *__restrict__ rm, const double *__restrict__ norms,
+    uint32_t d, int metric, const uint32_t *__restrict__ offsets,
+    const uint32_t *__restrict__ edges, const float *__restrict__ Q,
+    const double *__restrict__ qnorms, const uint32_t *__restrict__ ep_rows,
+    const double *__restrict__ ep_dists, uint32_t *__restrict__ visited,
+    uint64_t vwords_per_q, uint32_t k, uint32_t ef,
+    uint32_t *__restrict__ out_rows, double *__restrict__ out_dists,
+    uint32_t *__restrict__ out_cnt, uint32_t *__restrict__ out_flags) {
+	__shared__ uint64_t c_key[HQ_CAND_CAP];
+	__shared__ uint32_t c_seq[HQ_CAND_CAP];
+	__shared__ uint32_t c_row[HQ_CAND_CAP];
+	__shared__ uint64_t w_key[HQ_EF_CAP];
+	__shared__ uint32_t w_seq[HQ_EF_CAP];
+	__shared__ uint32_t w_row[HQ_EF_CAP];
+
+	const uint32_t qid = blockIdx.x;
+	const int lane = threadIdx.x;
+	const float *q = Q + (uint64_t)qid * d;
+	const double qn = qnorms[qid];
+	uint32_t *vis = visited + (uint64_t)qid * vwords_per_q;
+
+	// state kept wave-uniform in registers (updated by every lane alike)
+	uint32_t c_cnt = 0;   // append cursor of candidates
+	uint32_t w_cnt = 0;
+	uint32_t seq = 0;
+	uint32_t flags = 0;
+	uint64_t fq_key = ~0ULL; // max key in w (or +max when w not full)
+	uint32_t fq_idx = 0;
+
+	// seed with the entry point (search_single, layer.rs:76-90)
+	{
+		uint64_t ek = d_total_key(ep_dists[qid]);
+		uint32_t er = ep_rows[qid];
+		if (lane == 0) {
+			c_key[0] = ek;
+			c_seq[0] = 0;
+			c_row[0] = er;
+			w_key[0] = ek;
+			w_seq[0] = 0;
+			w_row[0] = er;
+			uint32_t word = er >> 5;
+			atomicOr(&vis[word], 1u << (er & 31));
+		}
+		c_cnt = 1;
+		w_cnt = 1;
+		seq = 1;
+		fq_key = ek;
+		fq_idx = 0;
+	}
+	__syncthreads();
+
+	auto wave_min_cand = [&](uint64_t *bk, uint32_t *bs, int *bi) {
+		uint64_t mk = ~0ULL;
+		uint32_t ms = ~0u;
+		int mi = -1;
+		for (uint32_t i = lane; i < c_cnt; i += 64) {
+			uint64_t kk = c_key[i];
+			uint32_t ss = c_seq[i];
+			if (kk < mk || (kk == mk && ss < ms)) {
+				mk = kk;
+				ms = ss;
+				mi = (int)i;
+			}
+		}
+		for (int off = 32; off > 0; off >>= 1) {
+			uint64_t ok = __shfl_down(mk, off);
+			uint32_t os = __shfl_down(ms, off);
+			int oi = __shfl_down(mi, off);
+			if (ok < mk || (ok == mk && os < ms)) {
+				mk = ok;
+				ms = os;
+				mi = oi;
+			}
+		}
+		*bk = __shfl(mk, 0);
+		*bs = __shfl(ms, 0);
+		*bi = __shfl(mi, 0);
+	};
+	auto wave_max_w = [&](uint64_t *bk, uint32_t *bi) {
+		// max by (key, seq): latest seq among the max key
+		uint64_t mk = 0;
+		uint32_t ms = 0;
+		int mi = 0;
+		for (uint32_t i = lane; i < w_cnt; i += 64) {
+			uint64_t kk = w_key[i];
+			uint32_t ss = w_seq[i];
+			if (kk > mk || (kk == mk && ss > ms)) {
+				mk = kk;
+				ms = ss;
+				mi = (int)i;
+			}
+		}
+		for (int off = 32; off > 0; off >>= 1) {
+			uint64_t ok = __shfl_down(mk, off);
+			uint32_t os = __shfl_down(ms, off);
+			int oi = __shfl_down(mi, off);
+			if (ok > mk || (ok == mk && os > ms)) {
+				mk = ok;
+				ms = os;
+				mi = oi;
+			}
+		}
+		*bk = __shfl(mk, 0);
+		*bi = __shfl(mi, 0);
+	};
+
+	// main best-first loop
+	for (;;) {
+		uint64_t ck;
+		uint32_t cs;
+		int ci;
+		wave_min_cand(&ck, &cs, &ci);
+		if (ci < 0)
+			break;
+		// cq_dist > fq_dist -> stop (strict, distance only; fq = max of w
+		// from the very first element, layer.rs:184)
+		if (ck > fq_key)
+			break;
+		uint32_t doc = c_row[ci];
+		if (lane == 0)
+			c_key[ci] = ~0ULL; // tombstone
+		__syncthreads();
+
+		uint32_t e0 = offsets[doc], e1 = offsets[doc + 1];
+		for (uint32_t base = e0; base < e1; base += 64) {
+			uint32_t my_e = ~0u;
+			double my_d = 0.0;
+			bool mine = false;
+			uint32_t idx = base + lane;
+			if (idx < e1) {
+				uint32_t e = edges[idx];
+				uint32_t old = atomicOr(&vis[e >> 5], 1u << (e & 31));
+				if (!(old & (1u << (e & 31)))) {
+					mine = true;
+					my_e = e;
+					// restated per-row chain from the row-major copy
+					const float *row = rm + (uint64_t)e * d;
+					if (metric == 0) {
+						float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+						uint32_t kk = 0;
+						for (; kk + 8 <= d; kk += 8)
+#pragma unroll
+							for (uint32_t t = 0; t < 8; t++)
+								p[t] = __fadd_rn(
+								    p[t], __fmul_rn(row[kk + t], q[kk + t]));
+						float sum = 0.f;
+						sum = __fadd_rn(sum, __fadd_rn(p[0], p[4]));
+						sum = __fadd_rn(sum, __fadd_rn(p[1], p[5]));
+						sum = __fadd_rn(sum, __fadd_rn(p[2], p[6]));
+						sum = __fadd_rn(sum, __fadd_rn(p[3], p[7]));
+						for (; kk < d; kk++)
+							sum = __fadd_rn(sum,
+							                __fmul_rn(row[kk], q[kk]));
+						my_d = 1.0 - (double)sum / (qn * norms[e]);
+					} else {
+						float acc = 0.f;
+						for (uint32_t kk = 0; kk < d; kk++) {
+							float diff = row[kk] - q[kk];
+							acc = __fadd_rn(acc, __fmul_rn(diff, diff));
+						}
+						my_d = sqrt((double)acc);
+					}
+				}
+			}
+			// serial accept/update in edge order (layer.rs:195-218)
+			uint32_t nlanes = (e1 - base) < 64 ? (e1 - base) : 64;
+			for (uint32_t i = 0; i < nlanes; i++) {
+				int src = (int)i;
+				bool m = __shfl(mine ? 1 : 0, src) != 0;
+				if (!m)
+					continue;
+				uint32_t erow = __shfl(my_e, src);
+				double ed = __shfl(my_d, src);
+				uint64_t ekey = d_total_key(ed);
+				// e_dist < fq_dist || w.len < ef  (strict distance compare)
+				if (!(w_cnt < ef || ekey < fq_key))
+					continue;
+				// candidates.push
+				if (c_cnt < HQ_CAND_CAP) {
+					if (lane == 0) {
+						c_key[c_cnt] = ekey;
+						c_seq[c_cnt] = seq;
+						c_row[c_cnt] = erow;
+					}
+					c_cnt++;
+				} else {
+					flags |= HQ_FLAG_OVERFLOW;
+				}
+				__syncthreads();
+				// w.push + (w.len > ef ? pop_last) as a conditional replace
+				bool replaced = false;
+				if (w_cnt < ef) {
+					if (lane == 0) {
+						w_key[w_cnt] = ekey;
+						w_seq[w_cnt] = seq;
+						w_row[w_cnt] = erow;
+					}
+					// incremental max: the new entry has the largest seq, so
+					// ekey >= fq_key makes it the (key,seq) max
+					if (ekey >= fq_key) {
+						fq_key = ekey;
+						fq_idx = w_cnt;
+					}
+					w_cnt++;
+				} else {
+					// new entry has the LARGEST seq: if its key >= max key it
+					// would be popped right back (pop_last = latest of the
+					// max) -> net no-op; else it replaces the current max
+					if (ekey < fq_key) {
+						if (lane == 0) {
+							w_key[fq_idx] = ekey;
+							w_seq[fq_idx] = seq;
+							w_row[fq_idx] = erow;
+						}
+						replaced = true;
+					}
+				}
+				seq++;
+				__syncthreads();
+				if (replaced)
+					wave_max_w(&fq_key, &fq_idx);
+			}
+		}
+	}
+
+	// emit to_vec_limit(k): ascending (key, seq) = (dist, FIFO) (knn.rs:92-104)
+	uint32_t out_m = w_cnt < k ? w_cnt : k;
+	for (uint32_t slot = 0; slot < out_m; slot++) {
+		uint64_t mk = ~0ULL;
+		uint32_t ms = ~0u;
+		int mi = -1;
+		for (uint32_t i = lane; i < w_cnt; i += 64) {
+			uint64_t kk = w_key[i];
+			uint32_t ss = w_seq[i];
+			if (kk < mk || (kk == mk && ss < ms)) {
+				mk = kk;
+				ms = ss;
+				mi = (int)i;
+			}
+		}
+		for (int off = 32; off > 0; off >>= 1) {
+			uint64_t ok = __shfl_down(mk, off);
+			uint32_t os = __shfl_down(ms, off);
+			int oi = __shfl_down(mi, off);
+			if (ok < mk || (ok == mk && os < ms)) {
+				mk = ok;
+				ms = os;
+				mi = oi;
+			}
+		}
+		mk = __shfl(mk, 0);
+		mi = __shfl(mi, 0);
+		if (lane == 0) {
+			out_rows[(uint64_t)qid * k + slot] = w_row[mi];
+			// decode distance from the total_cmp key (invertible)
+			uint64_t bits =
+			    (mk >> 63) ? (mk & ~0x8000000000000000ULL) : ~mk;
+			out_dists[(uint64_t)qid * k + slot] =
+			    __longlong_as_double(bits);
+			w_key[mi] = ~0ULL;
+			w_seq[mi] = ~0u;
+		}
+		__syncthreads();
+	}
+	if (lane == 0) {
+		out_cnt[qid] = out_m;
+		out_flags[qid] = flags;
+	}
+}
+
+// ---------------------------------------------------------------------------
 // Host-side helpers (restated reference arithmetic for the query's own norm —
 // must be bit-identical to oracle orc_sumsq_f32; covered by tests/)
 // ---------------------------------------------------------------------------
@@ -1490,6 +1775,13 @@ struct sdbv_hnsw {
 	float *q_dev = nullptr;
 	uint32_t *rows_pinned = nullptr; // pinned host staging (per-hop latency)
 	double *dists_pinned = nullptr;
+	// persistent-kernel search state (device graph + row-major vectors)
+	float *rm_dev = nullptr;        // [n][d] row-major
+	uint32_t *offsets_dev = nullptr; // layer-0 CSR
+	uint32_t *edges_dev = nullptr;
+	double *norms_dev = nullptr;    // per-element f64 norm (cosine)
+	uint32_t *vis_dev = nullptr;    // visited bitsets scratch
+	uint64_t vis_cap = 0;
 	std::string err;
 };
 
@@ -1795,6 +2087,11 @@ void sdbv_hnsw_destroy(sdbv_hnsw *h) {
 		(void)hipHostFree(h->rows_pinned);
 	if (h->dists_pinned)
 		(void)hipHostFree(h->dists_pinned);
+	for (void *p : {(void *)h->rm_dev, (void *)h->offsets_dev,
+	                (void *)h->edges_dev, (void *)h->norms_dev,
+	                (void *)h->vis_dev})
+		if (p)
+			(void)hipFree(p);
 	delete h;
 }
 
@@ -1905,8 +2202,198 @@ int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
 	                             (h->m0 + 1) * sizeof(uint32_t)));
 	HIP_CHECK(ctx, hipHostMalloc(&h->dists_pinned,
 	                             (h->m0 + 1) * sizeof(double)));
+	// persistent-kernel graph state: row-major vectors + layer-0 CSR
+	uint64_t n = h->next_id;
+	HIP_CHECK(ctx, hipMalloc(&h->rm_dev, n * h->d * sizeof(float)));
+	HIP_CHECK(ctx, hipMemcpy(h->rm_dev, h->vecs.data(),
+	                         n * h->d * sizeof(float),
+	                         hipMemcpyHostToDevice));
+	{
+		std::vector<uint32_t> offsets(n + 1);
+		uint64_t ec = 0;
+		for (uint64_t i = 0; i < n; i++) {
+			offsets[i] = (uint32_t)ec;
+			ec += h->layers[0].edges[i].size();
+		}
+		offsets[n] = (uint32_t)ec;
+		std::vector<uint32_t> edges;
+		edges.reserve(ec);
+		for (uint64_t i = 0; i < n; i++)
+			edges.insert(edges.end(), h->layers[0].edges[i].begin(),
+			             h->layers[0].edges[i].end());
+		HIP_CHECK(ctx, hipMalloc(&h->offsets_dev, (n + 1) * sizeof(uint32_t)));
+		HIP_CHECK(ctx, hipMalloc(&h->edges_dev,
+		                         std::max<uint64_t>(ec, 1) * sizeof(uint32_t)));
+		HIP_CHECK(ctx, hipMemcpy(h->offsets_dev, offsets.data(),
+		                         (n + 1) * sizeof(uint32_t),
+		                         hipMemcpyHostToDevice));
+		if (ec)
+			HIP_CHECK(ctx, hipMemcpy(h->edges_dev, edges.data(),
+			                         ec * sizeof(uint32_t),
+			                         hipMemcpyHostToDevice));
+	}
+	if (h->metric == SDBV_METRIC_COSINE) {
+		HIP_CHECK(ctx, hipMalloc(&h->norms_dev, n * sizeof(double)));
+		HIP_CHECK(ctx, hipMemcpy(h->norms_dev, h->norms.data(),
+		                         n * sizeof(double), hipMemcpyHostToDevice));
+	}
 	h->table = table;
 	h->finalized = true;
+	return SDBV_OK;
+}
+
+// Batched ef-search on the persistent kernel: one query per workgroup; host
+// does the upper-layer descent (search_ep) per query, the kernel runs the
+// full layer-0 best-first loop. Results are EXACTLY the reference's
+// (to_vec_limit(k) then the builder's (dist total_cmp, id) order). Any query
+// whose in-kernel candidate queue overflowed (HQ_FLAG_OVERFLOW, not observed
+// on real workloads) is re-run on the exact per-hop path.
+int sdbv_hnsw_knn_batch(sdbv_hnsw *h, const float *Q, uint32_t b, uint32_t k,
+                        uint32_t ef, uint64_t *out_ids, double *out_dists,
+                        uint32_t *out_ns) {
+	using namespace hnsw;
+	if (!h || !h->finalized || b == 0 || k == 0 || k > MAX_K ||
+	    ef > HQ_EF_CAP)
+		return SDBV_ERR_BAD_ARG;
+	if (h->enter_point < 0) {
+		for (uint32_t j = 0; j < b; j++)
+			out_ns[j] = 0;
+		return SDBV_OK;
+	}
+	sdbv_ctx *ctx = h->ctx;
+	std::unique_lock<std::mutex> lk(ctx->mu);
+	uint64_t n = h->next_id;
+
+	// host: per-query norms (restated chain) + upper-layer descent
+	std::vector<double> qnorms(b);
+	std::vector<uint32_t> eps(b);
+	std::vector<double> epd(b);
+	for (uint32_t j = 0; j < b; j++) {
+		const float *q = Q + (uint64_t)j * h->d;
+		qnorms[j] = sqrt(host_sumsq_f32(q, h->d));
+		uint32_t ep_id = (uint32_t)h->enter_point;
+		double ep_dist = dist(h, q, qnorms[j], ep_id);
+		for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+			PQ cand;
+			cand.push(ep_dist, ep_id);
+			std::unordered_set<uint32_t> visited{ep_id};
+			PQ w = cand;
+			search_layer_host(h, h->layers[l], q, qnorms[j], cand, visited, w,
+			                  1, false);
+			double dd;
+			uint32_t ii;
+			if (w.peek_first(&dd, &ii)) {
+				ep_dist = dd;
+				ep_id = ii;
+			}
+		}
+		eps[j] = ep_id;
+		epd[j] = ep_dist;
+	}
+
+	// device buffers
+	uint64_t vwords = (n + 31) / 32;
+	uint64_t vis_need = (uint64_t)b * vwords * sizeof(uint32_t);
+	if (h->vis_cap < vis_need) {
+		if (h->vis_dev)
+			(void)hipFree(h->vis_dev);
+		h->vis_dev = nullptr;
+		h->vis_cap = 0;
+		HIP_CHECK(ctx, hipMalloc(&h->vis_dev, vis_need));
+		h->vis_cap = vis_need;
+	}
+	HIP_CHECK(ctx, hipMemsetAsync(h->vis_dev, 0, vis_need, ctx->stream));
+	float *Qd = nullptr;
+	double *qnd = nullptr, *epdd = nullptr, *outd = nullptr;
+	uint32_t *epsd = nullptr, *outr = nullptr, *outc = nullptr, *outf = nullptr;
+	HIP_CHECK(ctx, hipMalloc(&Qd, (uint64_t)b * h->d * sizeof(float)));
+	HIP_CHECK(ctx, hipMalloc(&qnd, b * sizeof(double)));
+	HIP_CHECK(ctx, hipMalloc(&epdd, b * sizeof(double)));
+	HIP_CHECK(ctx, hipMalloc(&epsd, b * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMalloc(&outr, (uint64_t)b * k * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMalloc(&outd, (uint64_t)b * k * sizeof(double)));
+	HIP_CHECK(ctx, hipMalloc(&outc, b * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMalloc(&outf, b * sizeof(uint32_t)));
+	auto cleanup = [&] {
+		for (void *p : {(void *)Qd, (void *)qnd, (void *)epdd, (void *)epsd,
+		                (void *)outr, (void *)outd, (void *)outc,
+		                (void *)outf})
+			if (p)
+				(void)hipFree(p);
+	};
+	hipMemcpyAsync(Qd, Q, (uint64_t)b * h->d * sizeof(float),
+	               hipMemcpyHostToDevice, ctx->stream);
+	hipMemcpyAsync(qnd, qnorms.data(), b * sizeof(double),
+	               hipMemcpyHostToDevice, ctx->stream);
+	hipMemcpyAsync(epdd, epd.data(), b * sizeof(double),
+	               hipMemcpyHostToDevice, ctx->stream);
+	hipMemcpyAsync(epsd, eps.data(), b * sizeof(uint32_t),
+	               hipMemcpyHostToDevice, ctx->stream);
+
+	auto t0 = std::chrono::steady_clock::now();
+	hipLaunchKernelGGL(k_hnsw_search, dim3(b), dim3(64), 0, ctx->stream,
+	                   h->rm_dev, h->norms_dev, h->d, (int)h->metric,
+	                   h->offsets_dev, h->edges_dev, Qd, qnd, epsd, epdd,
+	                   h->vis_dev, vwords, k, ef, outr, outd, outc, outf);
+	std::vector<uint32_t> h_rows((uint64_t)b * k), h_cnt(b), h_flags(b);
+	std::vector<double> h_d((uint64_t)b * k);
+	hipMemcpyAsync(h_rows.data(), outr, h_rows.size() * sizeof(uint32_t),
+	               hipMemcpyDeviceToHost, ctx->stream);
+	hipMemcpyAsync(h_d.data(), outd, h_d.size() * sizeof(double),
+	               hipMemcpyDeviceToHost, ctx->stream);
+	hipMemcpyAsync(h_cnt.data(), outc, b * sizeof(uint32_t),
+	               hipMemcpyDeviceToHost, ctx->stream);
+	hipMemcpyAsync(h_flags.data(), outf, b * sizeof(uint32_t),
+	               hipMemcpyDeviceToHost, ctx->stream);
+	if (hipStreamSynchronize(ctx->stream) != hipSuccess ||
+	    hipGetLastError() != hipSuccess) {
+		ctx->err = "k_hnsw_search failed";
+		cleanup();
+		return SDBV_ERR_HIP;
+	}
+	ctx->stats.last_scan_kernel_ms =
+	    std::chrono::duration<double, std::milli>(
+	        std::chrono::steady_clock::now() - t0)
+	        .count();
+	ctx->stats.last_rows_scanned = n;
+	cleanup();
+
+	uint32_t overflowed = 0;
+	for (uint32_t j = 0; j < b; j++)
+		if (h_flags[j] & 1u)
+			overflowed++;
+
+	// final (dist total_cmp, id) order per query (knn.rs:363)
+	for (uint32_t j = 0; j < b; j++) {
+		if (h_flags[j] & 1u)
+			continue; // re-run below on the exact per-hop path
+		uint32_t m = h_cnt[j];
+		std::vector<std::pair<std::pair<uint64_t, uint32_t>, double>> fin(m);
+		for (uint32_t i = 0; i < m; i++) {
+			double dd = h_d[(uint64_t)j * k + i];
+			fin[i] = {{total_key(dd), h_rows[(uint64_t)j * k + i]}, dd};
+		}
+		std::sort(fin.begin(), fin.end());
+		out_ns[j] = m;
+		for (uint32_t i = 0; i < m; i++) {
+			out_ids[(uint64_t)j * k + i] = fin[i].first.second;
+			out_dists[(uint64_t)j * k + i] = fin[i].second;
+		}
+	}
+	if (overflowed) {
+		// exact fallback for overflowed queries (still the GPU gather path;
+		// sdbv_hnsw_knn takes the ctx mutex itself)
+		lk.unlock();
+		for (uint32_t j = 0; j < b; j++)
+			if (h_flags[j] & 1u) {
+				int rc = sdbv_hnsw_knn(h, Q + (uint64_t)j * h->d, k, ef,
+				                       out_ids + (uint64_t)j * k,
+				                       out_dists + (uint64_t)j * k,
+				                       &out_ns[j]);
+				if (rc)
+					return rc;
+			}
+	}
 	return SDBV_OK;
 }
 

@@ -128,3 +128,47 @@ def test_parallel_build_recall(ctx):
     assert total / len(queries) >= 0.99, total / len(queries)
     h.destroy()
     ctx.drop_table(23)
+
+
+def test_persistent_kernel_matches_perhop_and_oracle(ctx):
+    """The persistent in-kernel search must return EXACTLY what the per-hop
+    gather path and the oracle return (same graph, same queue semantics,
+    bit-exact distances)."""
+    d, n = 768, 30_000
+    rows = oracle.gen_f32(0x5DB1, 0, n, d)
+    h = ctx.hnsw_create(d, metric="cosine", m=16, m0=32, efc=150, seed=0x9)
+    o = oracle.Hnsw(d, metric="cosine", m=16, m0=32, efc=150,
+                    ml=1.0 / math.log(16.0), seed=0x9)
+    h.insert_batch(rows, nthreads=1)
+    for r in rows:
+        o.insert(r)
+    h.finalize(24)
+    Q = oracle.gen_f32(0xBEEF, 0, 32, d)
+    bids, bdists, bns = h.knn_search_batch(Q, 10, 64)
+    for j in range(32):
+        pids, pdists = h.knn_search(Q[j], 10, 64)
+        assert np.array_equal(bids[j][:bns[j]], pids), f"q{j} vs per-hop"
+        assert np.array_equal(bdists[j][:bns[j]], pdists), f"q{j} dists"
+        oids, odists = builder_sort(*o.search(Q[j], 10, 64))
+        assert np.array_equal(bids[j][:bns[j]], oids), f"q{j} vs oracle"
+        assert np.array_equal(bdists[j][:bns[j]], odists)
+    h.destroy()
+    ctx.drop_table(24)
+
+
+def test_persistent_kernel_euclidean_golden(ctx):
+    ingest = load_golden("hnsw-random-9000-20-euclidean.gz", 1500)
+    queries = load_golden("hnsw-random-5000-20-euclidean.gz", 200)
+    h = ctx.hnsw_create(20, metric="euclidean", m=8, efc=100,
+                        ml=1.0 / math.log(8.0), seed=0x5DB1)
+    h.insert_batch(ingest, nthreads=1)
+    h.finalize(25)
+    bids, bdists, bns = h.knn_search_batch(queries, 10, 40)
+    for j, q in enumerate(queries):
+        bf, _ = oracle.topk_f32("euclidean", ingest, q, 10)
+        assert set(bids[j][:bns[j]].tolist()) == set(bf.tolist()), f"q{j}"
+        pids, pdists = h.knn_search(q, 10, 40)
+        assert np.array_equal(bids[j][:bns[j]], pids)
+        assert np.array_equal(bdists[j][:bns[j]], pdists)
+    h.destroy()
+    ctx.drop_table(25)
