@@ -223,12 +223,33 @@ class Context:
         return ids, dists
 
 
+def hnsw_create_host(d, metric="euclidean", m=12, m0=None, efc=150,
+                     extend=False, keep=False, seed=0x5DB1, ml=None):
+    """Host-only HNSW index (ctx = NULL): graph build + CSR export work on
+    any machine; finalize()/knn_search() fail loudly (no GPU context).
+    Lets the pure-CPU graph-build path be tested without a GPU."""
+    import math
+    if m0 is None:
+        m0 = 2 * m
+    if ml is None:
+        ml = 1.0 / math.log(m)
+    out = ctypes.c_void_p()
+    _check(None, lib().sdbv_hnsw_create(
+        None, d, METRICS[metric], m, m0, efc, int(extend), int(keep),
+        seed, ml, ctypes.byref(out)), "sdbv_hnsw_create")
+    return Hnsw(None, out, d)
+
+
+class _NullCtx:
+    _ptr = None
+
+
 class Hnsw:
     """Product HNSW index (mirrors HnswIndex, hnsw/index.rs): host graph +
     GPU layer-0 expansion. Element ids are insertion ordinals."""
 
     def __init__(self, ctx, ptr, d):
-        self._ctx = ctx
+        self._ctx = ctx if ctx is not None else _NullCtx()
         self._ptr = ptr
         self.d = d
 

@@ -14,19 +14,10 @@ import oracle
 
 @pytest.fixture(scope="module")
 def ctxless_hnsw_pair():
+    # Context creation needs a GPU; the host-only index builds anywhere.
     import surrealdb_amd
-    # Context creation needs a GPU; hnsw_create only stores params — build a
-    # raw handle without a Context (host-only path).
-    import ctypes
-    out = ctypes.c_void_p()
-    rc = surrealdb_amd.lib().sdbv_hnsw_create(
-        None, 20, surrealdb_amd.METRICS["euclidean"], 8, 16, 100, 0, 0,
-        0x5DB1, 1.0 / math.log(8.0), ctypes.byref(out))
-    assert rc == 0
-    h = surrealdb_amd.Hnsw.__new__(surrealdb_amd.Hnsw)
-    h._ctx = type("X", (), {"_ptr": None})()
-    h._ptr = out
-    h.d = 20
+    h = surrealdb_amd.hnsw_create_host(20, metric="euclidean", m=8, m0=16,
+                                       efc=100, seed=0x5DB1)
     o = oracle.Hnsw(20, metric="euclidean", m=8, m0=16, efc=100,
                     ml=1.0 / math.log(8.0), seed=0x5DB1)
     yield h, o
@@ -51,17 +42,9 @@ def test_parallel_build_quality():
     """Parallel (bench-mode) build: nondeterministic graph, validated by the
     reference's recall bar (==1.0 @ efs=40 would need GPU search; here check
     structure invariants only — the GPU recall test covers quality)."""
-    import ctypes
     import surrealdb_amd
-    out = ctypes.c_void_p()
-    rc = surrealdb_amd.lib().sdbv_hnsw_create(
-        None, 32, surrealdb_amd.METRICS["euclidean"], 8, 16, 100, 0, 0,
-        7, 1.0 / math.log(8.0), ctypes.byref(out))
-    assert rc == 0
-    h = surrealdb_amd.Hnsw.__new__(surrealdb_amd.Hnsw)
-    h._ctx = type("X", (), {"_ptr": None})()
-    h._ptr = out
-    h.d = 32
+    h = surrealdb_amd.hnsw_create_host(32, metric="euclidean", m=8, m0=16,
+                                       efc=100, seed=7)
     rows = oracle.gen_f32(0x321, 0, 3000, 32)
     h.insert_batch(rows, nthreads=4)
     assert h.n() == 3000
@@ -75,4 +58,26 @@ def test_parallel_build_quality():
         assert not (edges[offsets[i]:offsets[i + 1]] == i).any()
     # connectivity proxy: nearly every node has at least one edge
     assert (deg > 0).mean() > 0.999
+    h.destroy()
+
+
+def test_host_build_matches_committed_fixture():
+    """The product host build must reproduce the committed oracle fixture
+    graph (tests/golden/hnsw_fix_seq128_cos.npz, built by
+    make_hnsw_fixtures.py) — catches fixture/build drift without a GPU."""
+    import os
+    import numpy as np
+    import surrealdb_amd
+    fx = np.load(os.path.join(os.path.dirname(__file__), "golden",
+                              "hnsw_fix_seq128_cos.npz"))
+    d, n, m = int(fx["d"]), int(fx["n"]), int(fx["m"])
+    rows = oracle.gen_f32(int(fx["data_seed"]), 0, n, d)
+    h = surrealdb_amd.hnsw_create_host(
+        d, metric=str(fx["metric"]), m=m, m0=int(fx["m0"]),
+        efc=int(fx["efc"]), ml=1.0 / math.log(m), seed=int(fx["seed"]))
+    h.insert_batch(rows, nthreads=1)
+    assert h.num_layers() == int(fx["num_layers"])
+    po, pe = h.l0_csr()
+    assert np.array_equal(po, fx["l0_offsets"])
+    assert np.array_equal(pe, fx["l0_edges"])
     h.destroy()
