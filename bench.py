@@ -122,6 +122,11 @@ def main():
 
     def one_query(qi):
         if args.batch > 0:
+            if hnsw_index is not None:
+                raise SystemExit(
+                    "--hnsw with --batch is not a bench mode yet: the "
+                    "batched path would silently run the brute-force GEMM, "
+                    "not the HNSW persistent kernel")
             return one_batch(qi)
         if hnsw_index is not None:
             ids, dists = hnsw_index.knn_search(queries[qi % len(queries)],
@@ -272,14 +277,17 @@ def main():
         "data": "synthetic",
         "config": {
             "workload": (
-                f"brute-force {args.metric} KNN, 10M rows/GPU x 768-dim f32, "
-                f"K=10, batch={args.batch} MFMA path (BASELINE configs[3])"
+                f"brute-force {args.metric} KNN, {rows} rows/GPU x "
+                f"{args.dim}-dim f32, K={args.k}, batch={args.batch} "
+                f"MFMA path (BASELINE configs[3])"
                 if args.batch > 0 else
-                f"HNSW (M=16, ef={args.ef}) {args.metric} K=10 single query "
-                f"(BASELINE configs[2]); latency/gather-bound"
+                f"HNSW (M=16, ef={args.ef}) {args.metric} K={args.k} single "
+                f"query, {rows} rows/GPU (BASELINE configs[2] shape; "
+                f"latency/gather-bound)"
                 if args.hnsw else
-                "brute-force cosine KNN, 10M rows/GPU x 768-dim f32, "
-                "K=10, single query (BASELINE configs[1])"),
+                f"brute-force {args.metric} KNN, {rows} rows/GPU x "
+                f"{args.dim}-dim f32, K={args.k}, single query "
+                "(BASELINE configs[1])"),
             "batch": args.batch,
             "rows_total": rows * world,
             "rows_per_gpu": rows,

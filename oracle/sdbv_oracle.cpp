@@ -611,11 +611,22 @@ struct OrcPQ {
 };
 
 struct OrcLayer {
-	// UndirectedGraph over insertion-ordered sets (graph.rs, dynamicset.rs)
+	// UndirectedGraph over insertion-ordered sets (graph.rs, dynamicset.rs).
+	// `in_layer` mirrors the reference's explicit nodes map (graph.rs:
+	// get_edges returns None for absent nodes; remove_node_and_bidirectional_
+	// edges needs membership, not just an empty edge list).
 	std::vector<std::vector<uint32_t>> edges; // indexed by element id
 	uint32_t m_max;
+	std::vector<uint8_t> in_layer;
 	bool has(uint64_t id) const {
-		return id < edges.size() && !(edges[id].size() == 1 && edges[id][0] == UINT32_MAX);
+		return id < in_layer.size() && in_layer[id];
+	}
+	void add_node(uint64_t id) {
+		if (edges.size() <= id)
+			edges.resize(id + 1);
+		if (in_layer.size() <= id)
+			in_layer.resize(id + 1, 0);
+		in_layer[id] = 1;
 	}
 };
 
@@ -634,6 +645,19 @@ struct orc_hnsw {
 	uint64_t next_id = 0;
 	// per-element max layer (for membership checks)
 	std::vector<int32_t> top_layer;
+	// elements map membership (hnsw/elements.rs: remove() deletes the entry;
+	// vector storage is retained, the slot just becomes dead)
+	std::vector<uint8_t> elem_present;
+};
+
+// are_all_docs_in_pending (layer.rs:320-338) needs the index's vec_docs;
+// searches below thread this through. `pending` = the DocId bitmap returned
+// by search_pendings; `docs_of_elem` resolves an element to its doc set.
+struct OrcPend {
+	const std::set<uint64_t> *pending;
+	const void *idx; // orc_index*
+	bool (*all_docs_pending)(const void *idx, uint64_t e_id,
+	                         const std::set<uint64_t> *pending);
 };
 
 static double hdist(const orc_hnsw *h, const float *a, const float *b) {
@@ -671,22 +695,28 @@ static uint32_t next_level(orc_hnsw *h) {
 }
 
 // layer.rs:184-223 HnswLayer::search — the best-first ef-bounded loop.
+// `pend` restates the pending_docs parameter: an element whose docs are ALL
+// pending is excluded from `candidates` (the expansion frontier) but still
+// pushed into `w` (layer.rs:209-212 — the reference pushes to w outside the
+// exclusion check; restated as-is).
 static void layer_search(const orc_hnsw *h, const OrcLayer &layer,
                          const float *q, OrcPQ &candidates,
                          std::unordered_set<uint64_t> &visited, OrcPQ &w,
-                         uint32_t ef) {
+                         uint32_t ef, const OrcPend *pend = nullptr) {
 	double fq_dist = w.peek_last_dist(1.7976931348623157e308);
 	double cq_dist; uint64_t doc;
 	while (candidates.pop_first(&cq_dist, &doc)) {
 		if (cq_dist > fq_dist)
 			break;
-		if (doc < layer.edges.size()) {
+		if (layer.has(doc)) {
 			for (uint32_t e_id : layer.edges[doc]) {
 				if (!visited.insert(e_id).second)
 					continue;
 				double e_dist = hdist(h, hvec(h, e_id), q);
 				if (e_dist < fq_dist || w.n < ef) {
-					candidates.push(e_dist, e_id);
+					if (!pend || !pend->all_docs_pending(pend->idx, e_id,
+					                                     pend->pending))
+						candidates.push(e_dist, e_id);
 					w.push(e_dist, e_id);
 					if (w.n > ef) {
 						double dd; uint64_t ii;
