@@ -88,6 +88,31 @@ def _load():
     lib.orc_hnsw_search_ep.argtypes = [ctypes.c_void_p, f32p, u64p, f64p]
     lib.orc_hnsw_entry_point.restype = ctypes.c_int64
     lib.orc_hnsw_entry_point.argtypes = [ctypes.c_void_p]
+    lib.orc_hnsw_remove.restype = ctypes.c_int
+    lib.orc_hnsw_remove.argtypes = [ctypes.c_void_p, u64]
+    # index layer (hnsw/index.rs + docs.rs + knn.rs Ids64)
+    u8p = ctypes.POINTER(u8)
+    lib.orc_index_new.restype = ctypes.c_void_p
+    lib.orc_index_new.argtypes = [u32, u8, ctypes.c_double, u32, u32, u32,
+                                  ctypes.c_int, ctypes.c_int, u64,
+                                  ctypes.c_double]
+    lib.orc_index_free.argtypes = [ctypes.c_void_p]
+    lib.orc_index_hnsw.restype = ctypes.c_void_p
+    lib.orc_index_hnsw.argtypes = [ctypes.c_void_p]
+    lib.orc_index_doc_count.restype = u64
+    lib.orc_index_doc_count.argtypes = [ctypes.c_void_p]
+    lib.orc_index_pending_count.restype = u64
+    lib.orc_index_pending_count.argtypes = [ctypes.c_void_p]
+    lib.orc_index_enqueue.restype = ctypes.c_int
+    lib.orc_index_enqueue.argtypes = [ctypes.c_void_p, u64, f32p, u32, f32p,
+                                      u32]
+    lib.orc_index_apply.restype = u64
+    lib.orc_index_apply.argtypes = [ctypes.c_void_p]
+    lib.orc_index_knn.restype = u32
+    lib.orc_index_knn.argtypes = [ctypes.c_void_p, f32p, u32, u32, u8p, u64p,
+                                  f64p]
+    lib.orc_index_check_props.restype = ctypes.c_int
+    lib.orc_index_check_props.argtypes = [ctypes.c_void_p, u64]
     return lib
 
 
@@ -231,8 +256,88 @@ class Hnsw:
     def entry_point(self):
         return lib().orc_hnsw_entry_point(self._h)
 
+    def remove(self, e_id):
+        """Hnsw::remove (hnsw/mod.rs:398-455). True if removed."""
+        return bool(lib().orc_hnsw_remove(self._h, e_id))
+
     def __del__(self):
         try:
             lib().orc_hnsw_free(self._h)
+        except Exception:
+            pass
+
+
+class _HnswView(Hnsw):
+    """Non-owning Hnsw view over an index's internal graph."""
+
+    def __init__(self, ptr, d):
+        self._h = ptr
+        self.d = d
+
+    def __del__(self):
+        pass  # owned by the Index
+
+
+class Index:
+    """Oracle HnswIndex (hnsw/index.rs): the pendings queue, VecDocs/Ids64
+    doc expansion and the pendings-merged knn_search over the restated graph.
+    Record keys are opaque u64 handles (the host's RecordIdKey mapping)."""
+
+    def __init__(self, d, metric="euclidean", order=0.0, m=12, m0=None,
+                 efc=150, extend=False, keep=False, seed=0x5DB1, ml=None):
+        import math
+        if m0 is None:
+            m0 = 2 * m
+        if ml is None:
+            ml = 1.0 / math.log(m)
+        self._ix = lib().orc_index_new(d, METRICS[metric], order, m, m0, efc,
+                                       int(extend), int(keep), seed, ml)
+        self.d = d
+
+    def hnsw(self):
+        return _HnswView(lib().orc_index_hnsw(self._ix), self.d)
+
+    def enqueue(self, record_key, old_vectors=None, new_vectors=None):
+        """HnswIndex::index (index.rs:138-186): one pending update. Vectors
+        are (n, d) f32 arrays (None == no values of that kind)."""
+        def flat(a):
+            if a is None:
+                return np.empty((0, self.d), dtype=np.float32)
+            a = np.ascontiguousarray(a, dtype=np.float32).reshape(-1, self.d)
+            return a
+        o, nw = flat(old_vectors), flat(new_vectors)
+        rc = lib().orc_index_enqueue(self._ix, record_key, _f32p(o),
+                                     o.shape[0], _f32p(nw), nw.shape[0])
+        assert rc == 0
+
+    def apply_pendings(self):
+        return lib().orc_index_apply(self._ix)
+
+    def knn_search(self, q, k, ef):
+        """index.rs:270-335 without the record materialisation: returns
+        (kinds u8 [0=DocId, 1=RecordKey], ids u64, dists f64) ascending."""
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        kinds = np.empty(k, dtype=np.uint8)
+        ids = np.empty(k, dtype=np.uint64)
+        dists = np.empty(k, dtype=np.float64)
+        n = lib().orc_index_knn(
+            self._ix, _f32p(q), k, ef,
+            kinds.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        return kinds[:n], ids[:n], dists[:n]
+
+    def doc_count(self):
+        return lib().orc_index_doc_count(self._ix)
+
+    def pending_count(self):
+        return lib().orc_index_pending_count(self._ix)
+
+    def check_props(self, expected_count):
+        return lib().orc_index_check_props(self._ix, expected_count)
+
+    def __del__(self):
+        try:
+            lib().orc_index_free(self._ix)
         except Exception:
             pass

@@ -35,6 +35,8 @@
 #include <deque>
 #include <map>
 #include <set>
+#include <string>
+#include <unordered_map>
 #include <unordered_set>
 #include <algorithm>
 
@@ -768,12 +770,16 @@ static void heuristic_select(const orc_hnsw *h, const OrcLayer &layer,
 	}
 }
 
-// heuristic.rs:118-157 extend_candidates.
+// heuristic.rs:118-157 extend_candidates (`ignore`: heuristic.rs:130-134 —
+// the element being removed is excluded from the extension set).
 static void extend_candidates(const orc_hnsw *h, const OrcLayer &layer,
-                              uint64_t q_id, const float *q_pt, OrcPQ &c) {
+                              uint64_t q_id, const float *q_pt, OrcPQ &c,
+                              int64_t ignore = -1) {
 	std::unordered_set<uint64_t> ex;
 	for (auto &e : c.to_vec())
 		ex.insert(e.second);
+	if (ignore >= 0)
+		ex.insert((uint64_t)ignore);
 	std::vector<std::pair<double, uint64_t>> ext;
 	for (auto &e : c.to_vec()) {
 		uint64_t e_id = e.second;
@@ -792,9 +798,9 @@ static void extend_candidates(const orc_hnsw *h, const OrcLayer &layer,
 
 static void select_neighbors(const orc_hnsw *h, const OrcLayer &layer,
                              uint64_t q_id, const float *q_pt, OrcPQ c,
-                             std::vector<uint32_t> &res) {
+                             std::vector<uint32_t> &res, int64_t ignore = -1) {
 	if (h->extend)
-		extend_candidates(h, layer, q_id, q_pt, c);
+		extend_candidates(h, layer, q_id, q_pt, c, ignore);
 	heuristic_select(h, layer, std::move(c), res, h->keep);
 }
 
@@ -813,8 +819,7 @@ static OrcPQ layer_insert(orc_hnsw *h, OrcLayer &layer, uint32_t q_id,
 	select_neighbors(h, layer, q_id, q_pt, w, neighbors);
 
 	// add node + bidirectional edges
-	if (layer.edges.size() <= q_id)
-		layer.edges.resize(q_id + 1);
+	layer.add_node(q_id);
 	layer.edges[q_id] = neighbors;
 	for (uint32_t e_id : neighbors)
 		layer.edges[e_id].push_back(q_id);
@@ -839,6 +844,7 @@ void orc_hnsw_insert_level(orc_hnsw *h, const float *pt, uint32_t q_level) {
 	uint32_t q_id = (uint32_t)h->next_id++;
 	h->vecs.insert(h->vecs.end(), pt, pt + h->d);
 	h->top_layer.push_back((int32_t)q_level);
+	h->elem_present.push_back(1);
 	uint32_t top_up_layers = (uint32_t)h->layers.size() - 1;
 
 	for (uint32_t i = top_up_layers; i < q_level; i++)
@@ -847,8 +853,7 @@ void orc_hnsw_insert_level(orc_hnsw *h, const float *pt, uint32_t q_level) {
 	if (h->enter_point < 0) {
 		// insert_first_element (hnsw/mod.rs:272-291)
 		for (uint32_t l = 0; l <= q_level && l < (uint32_t)h->layers.size(); l++) {
-			if (h->layers[l].edges.size() <= q_id)
-				h->layers[l].edges.resize(q_id + 1);
+			h->layers[l].add_node(q_id);
 			h->layers[l].edges[q_id] = {};
 		}
 		h->enter_point = (int64_t)q_id;
@@ -876,8 +881,7 @@ void orc_hnsw_insert_level(orc_hnsw *h, const float *pt, uint32_t q_level) {
 	layer_insert(h, h->layers[0], q_id, pt, std::move(eps));
 
 	for (uint32_t l = top_up_layers + 1; l <= q_level; l++) {
-		if (h->layers[l].edges.size() <= q_id)
-			h->layers[l].edges.resize(q_id + 1);
+		h->layers[l].add_node(q_id);
 		h->layers[l].edges[q_id] = {};
 	}
 	if (q_level > top_up_layers)
@@ -888,9 +892,98 @@ void orc_hnsw_insert(orc_hnsw *h, const float *pt) {
 	orc_hnsw_insert_level(h, pt, next_level(h));
 }
 
-// hnsw/mod.rs:459-482 knn_search + :521-548 search_ep.
-uint32_t orc_hnsw_search(orc_hnsw *h, const float *q, uint32_t k, uint32_t ef,
-                         uint64_t *out_ids, double *out_dists) {
+// layer.rs:92-108 search_single_with_ignore — seeds the search FROM the
+// ignored element itself (its distance enters candidates but visited blocks
+// it from w); returns the closest found element or -1 (None).
+static int64_t search_single_with_ignore(const orc_hnsw *h,
+                                         const OrcLayer &layer,
+                                         const float *pt, uint64_t ignore_id,
+                                         uint32_t ef) {
+	std::unordered_set<uint64_t> visited{ignore_id};
+	OrcPQ candidates;
+	candidates.push(hdist(h, pt, hvec(h, ignore_id)), ignore_id);
+	OrcPQ w;
+	layer_search(h, layer, pt, candidates, visited, w, ef);
+	double dd; uint64_t ii;
+	if (w.peek_first(&dd, &ii))
+		return (int64_t)ii;
+	return -1;
+}
+
+// layer.rs:164-181 search_multi_with_ignore.
+static OrcPQ search_multi_with_ignore(const orc_hnsw *h, const OrcLayer &layer,
+                                      const float *pt,
+                                      const std::vector<uint64_t> &ignore_ids,
+                                      uint32_t efc) {
+	OrcPQ candidates;
+	for (uint64_t id : ignore_ids)
+		candidates.push(hdist(h, pt, hvec(h, id)), id);
+	std::unordered_set<uint64_t> visited(ignore_ids.begin(),
+	                                     ignore_ids.end());
+	OrcPQ w;
+	layer_search(h, layer, pt, candidates, visited, w, efc);
+	return w;
+}
+
+// layer.rs:408-460 HnswLayer::remove: drop the node and its back-edges
+// (graph.rs remove_node_and_bidirectional_edges), then repair each former
+// neighbour with an efc-search (ignoring itself and the removed element) and
+// a heuristic re-selection (ignore = the removed element).
+static bool layer_remove(orc_hnsw *h, OrcLayer &layer, uint64_t e_id) {
+	if (!layer.has(e_id))
+		return false;
+	std::vector<uint32_t> f_ids = std::move(layer.edges[e_id]);
+	layer.edges[e_id].clear();
+	layer.in_layer[e_id] = 0;
+	for (uint32_t f : f_ids) {
+		auto &fe = layer.edges[f];
+		fe.erase(std::remove(fe.begin(), fe.end(), (uint32_t)e_id),
+		         fe.end());
+	}
+	for (uint32_t q_id : f_ids) {
+		const float *q_pt = hvec(h, q_id);
+		OrcPQ c = search_multi_with_ignore(h, layer, q_pt, {q_id, e_id},
+		                                   h->efc);
+		std::vector<uint32_t> q_new_conn;
+		select_neighbors(h, layer, q_id, q_pt, std::move(c), q_new_conn,
+		                 (int64_t)e_id);
+		layer.edges[q_id] = q_new_conn; // graph.set_node
+	}
+	return true;
+}
+
+// hnsw/mod.rs:398-455 Hnsw::remove. Returns 1 if the element was removed.
+int orc_hnsw_remove(orc_hnsw *h, uint64_t e_id) {
+	if (e_id >= h->next_id || !h->elem_present[e_id])
+		return 0; // elements.get_vector -> None (mod.rs:401)
+	bool removed = false;
+	const float *e_pt = hvec(h, e_id);
+	int64_t new_enter_point =
+	    ((int64_t)e_id == h->enter_point) ? -1 : h->enter_point;
+	// upper layers, top-down (mod.rs:411-422)
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		if (new_enter_point < 0)
+			new_enter_point = search_single_with_ignore(
+			    h, h->layers[l], e_pt, e_id, h->efc);
+		if (layer_remove(h, h->layers[l], e_id))
+			removed = true;
+	}
+	// possible new enter_point at layer0 (mod.rs:424-429)
+	if (new_enter_point < 0)
+		new_enter_point =
+		    search_single_with_ignore(h, h->layers[0], e_pt, e_id, h->efc);
+	if (layer_remove(h, h->layers[0], e_id))
+		removed = true;
+	h->elem_present[e_id] = 0; // elements.remove (mod.rs:448)
+	h->enter_point = new_enter_point;
+	return removed ? 1 : 0;
+}
+
+// hnsw/mod.rs:459-482 knn_search + :521-548 search_ep, with the optional
+// pending_docs parameter threaded through every layer (index.rs knn path).
+static uint32_t hnsw_search_core(orc_hnsw *h, const float *q, uint32_t k,
+                                 uint32_t ef, const OrcPend *pend,
+                                 uint64_t *out_ids, double *out_dists) {
 	if (h->enter_point < 0)
 		return 0;
 	uint64_t ep_id = (uint64_t)h->enter_point;
@@ -899,14 +992,14 @@ uint32_t orc_hnsw_search(orc_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 		OrcPQ cand; cand.push(ep_dist, ep_id);
 		std::unordered_set<uint64_t> visited{ep_id};
 		OrcPQ w = cand;
-		layer_search(h, h->layers[l], q, cand, visited, w, 1);
+		layer_search(h, h->layers[l], q, cand, visited, w, 1, pend);
 		double dd; uint64_t ii;
 		if (w.peek_first(&dd, &ii)) { ep_dist = dd; ep_id = ii; }
 	}
 	OrcPQ cand; cand.push(ep_dist, ep_id);
 	std::unordered_set<uint64_t> visited{ep_id};
 	OrcPQ w = cand;
-	layer_search(h, h->layers[0], q, cand, visited, w, ef);
+	layer_search(h, h->layers[0], q, cand, visited, w, ef, pend);
 	// to_vec_limit(k) (knn.rs:92-104)
 	auto v = w.to_vec();
 	uint32_t nout = (uint32_t)std::min<size_t>(k, v.size());
@@ -915,6 +1008,11 @@ uint32_t orc_hnsw_search(orc_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 		out_ids[i] = v[i].second;
 	}
 	return nout;
+}
+
+uint32_t orc_hnsw_search(orc_hnsw *h, const float *q, uint32_t k, uint32_t ef,
+                         uint64_t *out_ids, double *out_dists) {
+	return hnsw_search_core(h, q, k, ef, nullptr, out_ids, out_dists);
 }
 
 uint32_t orc_hnsw_num_layers(orc_hnsw *h) { return (uint32_t)h->layers.size(); }
@@ -974,5 +1072,447 @@ void orc_hnsw_search_ep(orc_hnsw *h, const float *q, uint64_t *ep_id_out,
 }
 
 const float *orc_hnsw_vec_ptr(orc_hnsw *h, uint64_t id) { return hvec(h, id); }
+
+// ===========================================================================
+// Index layer — HnswIndex (hnsw/index.rs) + VecDocs/Ids64 (docs.rs, knn.rs)
+// + the pendings queue (VectorPendingUpdate over Hp keys), restated as an
+// in-memory structure: everything the reference keeps behind the KV
+// transaction (hi/hd record-key maps, Hp pendings stream, Hv vector->docs
+// entries) is held in maps here; the record-key itself is an opaque u64
+// handle supplied by the host (INTEGRATION.md "record-key handles").
+// ===========================================================================
+
+// knn.rs:163-326 Ids64 — doc-id set with size-dependent representation.
+// Vec1..Vec8 keep INSERTION order; the 9th insert collapses to Bits
+// (RoaringTreemap — iteration ascending); dropping back to exactly 8 keeps
+// ascending order. insert()/remove() return "a new variant was produced"
+// exactly like the reference, because VecDocs only persists on Some — Bits
+// in-place mutations are DROPPED by the caller (docs.rs:374-382, :437-447;
+// restated as-is, quirks included — see test_index_oracle.py).
+struct OrcIds64 {
+	std::vector<uint64_t> v;
+	bool bits = false;
+	size_t len() const { return v.size(); }
+	bool empty() const { return v.empty(); }
+	bool contains(uint64_t d) const {
+		return std::find(v.begin(), v.end(), d) != v.end();
+	}
+	// knn.rs:234-256: None on duplicate; Bits insert mutates in place and
+	// returns None (=> false here).
+	bool insert_ret_variant(uint64_t d) {
+		if (contains(d))
+			return false;
+		if (!bits) {
+			v.push_back(d);
+			if (v.size() > 8) { // Vec8 -> Bits: RoaringTreemap::from(sorted)
+				std::sort(v.begin(), v.end());
+				bits = true;
+			}
+			return true; // Vec*->Vec* and Vec8->Bits both produce Some
+		}
+		v.insert(std::lower_bound(v.begin(), v.end(), d), d);
+		return false; // Bits: in-place, None
+	}
+	// knn.rs:258-326: per-variant removal. Returns true and writes *out if a
+	// NEW variant was produced; Bits removals mutate in place (and only
+	// produce a variant when dropping to exactly 8). The Vec2 non-member
+	// case reproduces the reference's `find(|i| i != d).map(One)` exactly.
+	bool remove_ret_variant(uint64_t d, OrcIds64 *out) {
+		if (bits) {
+			auto it = std::lower_bound(v.begin(), v.end(), d);
+			bool had = (it != v.end() && *it == d);
+			if (had)
+				v.erase(it); // RoaringTreemap::remove, in place
+			if (!had || v.size() != 8)
+				return false;
+			out->v = v; // Bits -> Vec8 (ascending, b.iter() order)
+			out->bits = false;
+			return true;
+		}
+		switch (v.size()) {
+		case 0:
+			return false; // Empty -> None
+		case 1:
+			if (v[0] == d) { // One -> Empty
+				out->v.clear();
+				out->bits = false;
+				return true;
+			}
+			return false;
+		case 2:
+			// Vec2: first element != d becomes One (knn.rs:268 — for a
+			// non-member d this DROPS the second element; restated as-is)
+			for (uint64_t x : v)
+				if (x != d) {
+					out->v = {x};
+					out->bits = false;
+					return true;
+				}
+			return false;
+		default: {
+			// Vec3..Vec8: filter; a variant is produced only when exactly
+			// one element was removed
+			std::vector<uint64_t> f;
+			for (uint64_t x : v)
+				if (x != d)
+					f.push_back(x);
+			if (f.size() == v.size() - 1) {
+				out->v = std::move(f);
+				out->bits = false;
+				return true;
+			}
+			return false;
+		}
+		}
+	}
+};
+
+// knn.rs:363-437 KnnResultBuilder: BTreeSet<(FloatKey, VectorId)> +
+// per-VectorId multiplicity. VectorId (hnsw/mod.rs VectorId enum, derived
+// Ord): kind 0 = DocId(u64) < kind 1 = RecordKey (host-ordered u64 handle).
+struct OrcVid {
+	uint8_t kind;
+	uint64_t id;
+	bool operator<(const OrcVid &o) const {
+		if (kind != o.kind)
+			return kind < o.kind;
+		return id < o.id;
+	}
+};
+struct OrcBuilder {
+	size_t knn;
+	struct Ent {
+		uint64_t key; // orc_total_key(dist)
+		OrcVid vid;
+		double dist;
+		bool operator<(const Ent &o) const {
+			if (key != o.key)
+				return key < o.key;
+			return vid < o.vid;
+		}
+	};
+	std::set<Ent> pl;
+	std::map<OrcVid, size_t> count;
+	explicit OrcBuilder(size_t k) : knn(k) {}
+	// knn.rs:386-394 check_add: plain f64 `>` against the current worst
+	// (NOT total_cmp — NaN submitted passes).
+	bool check_add(double submitted) const {
+		if (pl.size() >= knn && !pl.empty() &&
+		    submitted > std::prev(pl.end())->dist)
+			return false;
+		return true;
+	}
+	// knn.rs:410-433 add_vector_id_result. Returns true + *evicted when an
+	// id fell out of the result entirely (its count hit 0).
+	bool add(double dist, OrcVid vid, OrcVid *evicted) {
+		pl.insert(Ent{orc_total_key(dist), vid, dist});
+		count[vid]++; // incremented even when the set insert was a dup
+		if (pl.size() <= knn)
+			return false;
+		auto last = std::prev(pl.end());
+		OrcVid ev = last->vid;
+		pl.erase(last);
+		auto it = count.find(ev);
+		if (it != count.end()) {
+			if (it->second <= 1) {
+				count.erase(it);
+				if (evicted)
+					*evicted = ev;
+				return true;
+			}
+			it->second--;
+		}
+		return false;
+	}
+	// knn.rs:398-407 add_graph_result: one add per doc in the Ids64.
+	void add_graph(double dist, const OrcIds64 &docs) {
+		for (uint64_t doc : docs.v)
+			add(dist, OrcVid{0, doc}, nullptr);
+	}
+};
+
+// docs.rs:281-450 VecDocs (non-hashed Hv path — the default; the hashed Hh
+// variant differs only in KV key layout, not in observable results) plus
+// docs.rs:20-135 HnswDocs (record-key <-> doc-id maps, recycled allocation).
+struct OrcED {
+	uint64_t e_id;
+	OrcIds64 docs;
+};
+struct orc_index {
+	orc_hnsw *h;
+	// Hv entries: serialized vector bytes -> (element, docs)
+	std::unordered_map<std::string, OrcED> vd;
+	std::unordered_map<uint64_t, const std::string *> by_elem;
+	// HnswDocs (hi/hd keys + HnswDocsState)
+	std::map<uint64_t, uint64_t> key2doc, doc2key;
+	std::set<uint64_t> available;
+	uint64_t next_doc_id = 0;
+	// Hp pendings, appending order (index.rs:131-174)
+	struct Pending {
+		uint8_t kind; // 0 DocId, 1 RecordKey
+		uint64_t id;
+		std::vector<float> olds, news; // n*d each
+	};
+	std::vector<Pending> pendings;
+};
+
+static std::string vec_key(const orc_index *ix, const float *v) {
+	return std::string((const char *)v, (size_t)ix->h->d * 4);
+}
+
+orc_index *orc_index_new(uint32_t d, uint8_t metric, double order, uint32_t m,
+                         uint32_t m0, uint32_t efc, int extend, int keep,
+                         uint64_t seed, double ml) {
+	auto *ix = new orc_index();
+	ix->h = orc_hnsw_new(d, metric, order, m, m0, efc, extend, keep, seed, ml);
+	return ix;
+}
+void orc_index_free(orc_index *ix) {
+	if (!ix)
+		return;
+	orc_hnsw_free(ix->h);
+	delete ix;
+}
+orc_hnsw *orc_index_hnsw(orc_index *ix) { return ix->h; }
+uint64_t orc_index_doc_count(orc_index *ix) { return ix->doc2key.size(); }
+uint64_t orc_index_pending_count(orc_index *ix) {
+	return ix->pendings.size();
+}
+
+// HnswIndex::index (index.rs:138-186): resolve the id kind via the hi map
+// (HnswDocs::get_doc_id) and append one VectorPendingUpdate. old/new vectors
+// are n*d f32 each (the host's content_to_vectors output).
+int orc_index_enqueue(orc_index *ix, uint64_t record_key, const float *olds,
+                      uint32_t n_old, const float *news, uint32_t n_new) {
+	orc_index::Pending p;
+	auto it = ix->key2doc.find(record_key);
+	if (it != ix->key2doc.end()) {
+		p.kind = 0;
+		p.id = it->second;
+	} else {
+		p.kind = 1;
+		p.id = record_key;
+	}
+	uint32_t d = ix->h->d;
+	p.olds.assign(olds, olds + (size_t)n_old * d);
+	p.news.assign(news, news + (size_t)n_new * d);
+	ix->pendings.push_back(std::move(p));
+	return 0;
+}
+
+// docs.rs:64-76 resolve + :78-90 next_doc_id (smallest recycled id first).
+static uint64_t docs_resolve(orc_index *ix, uint64_t record_key) {
+	auto it = ix->key2doc.find(record_key);
+	if (it != ix->key2doc.end())
+		return it->second;
+	uint64_t doc_id;
+	if (!ix->available.empty()) {
+		doc_id = *ix->available.begin();
+		ix->available.erase(ix->available.begin());
+	} else {
+		doc_id = ix->next_doc_id++;
+	}
+	ix->key2doc[record_key] = doc_id;
+	ix->doc2key[doc_id] = record_key;
+	return doc_id;
+}
+
+// docs.rs:113-135 HnswDocs::remove (recycle the id).
+static void docs_remove(orc_index *ix, uint64_t doc_id) {
+	auto it = ix->doc2key.find(doc_id);
+	if (it == ix->doc2key.end())
+		return;
+	ix->key2doc.erase(it->second);
+	ix->doc2key.erase(it);
+	ix->available.insert(doc_id);
+}
+
+// docs.rs:363-393 VecDocs::insert.
+static void vd_insert(orc_index *ix, const float *vec, uint64_t doc_id) {
+	std::string key = vec_key(ix, vec);
+	auto it = ix->vd.find(key);
+	if (it == ix->vd.end()) {
+		// new vector: insert into the graph, ElementDocs::new
+		uint32_t e_id = (uint32_t)ix->h->next_id;
+		orc_hnsw_insert(ix->h, vec);
+		auto r = ix->vd.emplace(std::move(key), OrcED{e_id, {}});
+		r.first->second.docs.v = {doc_id};
+		ix->by_elem[e_id] = &r.first->first;
+	} else {
+		// existing vector: persist only if a new Ids64 variant was
+		// produced (docs.rs:376-381 — Bits in-place adds are dropped)
+		OrcED ed = it->second; // owned copy, like tx.get
+		if (ed.docs.insert_ret_variant(doc_id))
+			it->second = ed;
+	}
+}
+
+// docs.rs:424-449 VecDocs::remove.
+static void vd_remove(orc_index *ix, const float *vec, uint64_t doc_id) {
+	std::string key = vec_key(ix, vec);
+	auto it = ix->vd.find(key);
+	if (it == ix->vd.end())
+		return;
+	OrcED ed = it->second; // owned copy, like tx.get
+	OrcIds64 new_docs;
+	if (ed.docs.remove_ret_variant(doc_id, &new_docs)) {
+		if (new_docs.empty()) {
+			uint64_t e_id = ed.e_id;
+			ix->by_elem.erase(e_id);
+			ix->vd.erase(it);
+			orc_hnsw_remove(ix->h, e_id);
+		} else {
+			ed.docs = new_docs;
+			it->second = ed;
+		}
+	}
+	// else: no variant produced — mutation dropped (docs.rs:437-447)
+}
+
+// HnswIndex::index_pendings + index_pending (index.rs:188-257): drain the
+// queue in appending order; old-vector removals only for DocId pendings;
+// empty new_vectors deletes the doc mapping; RecordKey ids resolve (and
+// allocate) at apply time.
+uint64_t orc_index_apply(orc_index *ix) {
+	uint64_t count = 0;
+	uint32_t d = ix->h->d;
+	for (auto &p : ix->pendings) {
+		if (p.kind == 0) {
+			for (size_t i = 0; i * d < p.olds.size(); i++)
+				vd_remove(ix, p.olds.data() + i * d, p.id);
+			if (p.news.empty())
+				docs_remove(ix, p.id);
+		}
+		if (!p.news.empty()) {
+			uint64_t doc_id = (p.kind == 0) ? p.id : docs_resolve(ix, p.id);
+			for (size_t i = 0; i * d < p.news.size(); i++)
+				vd_insert(ix, p.news.data() + i * d, doc_id);
+		}
+		count++;
+	}
+	ix->pendings.clear();
+	return count;
+}
+
+// layer.rs:320-338 are_all_docs_in_pending: an element whose vector has no
+// VecDocs entry, or whose every doc is pending, counts as all-pending.
+static bool orc_all_docs_pending(const void *vix, uint64_t e_id,
+                                 const std::set<uint64_t> *pending) {
+	if (!pending || pending->empty())
+		return false;
+	auto *ix = (const orc_index *)vix;
+	auto it = ix->by_elem.find(e_id);
+	if (it != ix->by_elem.end()) {
+		const OrcIds64 &docs = ix->vd.at(*it->second).docs;
+		for (uint64_t doc : docs.v)
+			if (!pending->count(doc))
+				return false;
+	}
+	return true;
+}
+
+// HnswIndex::knn_search (index.rs:270-335): search_pendings overlay, then
+// the graph search excluding all-pending elements (the layer.rs:209-212
+// candidates-only exclusion), doc expansion via VecDocs, one KnnResultBuilder
+// across both. Out arrays sized k; returns the entry count. No cond_filter
+// here (hnsw/filter.rs is SURVEY $8f rank 2).
+uint32_t orc_index_knn(orc_index *ix, const float *q, uint32_t k, uint32_t ef,
+                       uint8_t *out_kinds, uint64_t *out_ids,
+                       double *out_dists) {
+	orc_hnsw *h = ix->h;
+	uint32_t d = h->d;
+	OrcBuilder builder(k);
+	// search_pendings (index.rs:366-421): two passes over the queue
+	std::set<uint64_t> all_existing;
+	std::map<OrcVid, const std::vector<float> *> non_deleted;
+	for (auto &p : ix->pendings) {
+		if (p.kind == 0)
+			all_existing.insert(p.id);
+		OrcVid vid{p.kind, p.id};
+		if (p.news.empty())
+			non_deleted.erase(vid);
+		else
+			non_deleted[vid] = &p.news;
+	}
+	if (!(all_existing.empty() && non_deleted.empty())) {
+		for (auto &e : non_deleted) {
+			const std::vector<float> &vecs = *e.second;
+			for (size_t i = 0; i * d < vecs.size(); i++) {
+				// Distance::calculate on the typed F32 path
+				// (idx/trees/vector.rs:660-672)
+				double dd = orc_dist_f32(h->metric, h->order, q,
+				                         vecs.data() + i * d, d);
+				if (builder.check_add(dd))
+					builder.add(dd, e.first, nullptr);
+			}
+		}
+	}
+	// graph search with the pending_docs bitmap (None when no DocId pending)
+	OrcPend pend{&all_existing, ix, &orc_all_docs_pending};
+	const OrcPend *pp = all_existing.empty() ? nullptr : &pend;
+	std::vector<uint64_t> gids(std::max<uint32_t>(k, ef));
+	std::vector<double> gdists(std::max<uint32_t>(k, ef));
+	uint32_t ng = hnsw_search_core(h, q, k, ef, pp, gids.data(),
+	                               gdists.data());
+	// add_graph_results (index.rs:454-483)
+	for (uint32_t i = 0; i < ng; i++) {
+		uint64_t e_id = gids[i];
+		if (!builder.check_add(gdists[i]))
+			continue;
+		auto it = ix->by_elem.find(e_id);
+		if (it == ix->by_elem.end())
+			continue; // get_vector/get_docs -> None
+		builder.add_graph(gdists[i], ix->vd.at(*it->second).docs);
+	}
+	// collect() -> ascending (FloatKey, VectorId)
+	uint32_t n = 0;
+	for (const auto &e : builder.pl) {
+		out_kinds[n] = e.vid.kind;
+		out_ids[n] = e.vid.id;
+		out_dists[n] = e.dist;
+		n++;
+	}
+	return n;
+}
+
+// Test hook: drive OrcIds64 directly so tests can restate the reference's
+// own test_ids sequence (knn.rs:669-717) — variant transitions (Some/None)
+// and contents/order, bit for bit.
+OrcIds64 *orc_ids64_new() { return new OrcIds64(); }
+void orc_ids64_free(OrcIds64 *s) { delete s; }
+// returns 1 if a new variant was produced (the reference's Some)
+int orc_ids64_insert(OrcIds64 *s, uint64_t d) {
+	return s->insert_ret_variant(d) ? 1 : 0;
+}
+// returns 1 if a new variant was produced; the set is REPLACED by it (the
+// caller-persists model) — matching how VecDocs uses the return value
+int orc_ids64_remove(OrcIds64 *s, uint64_t d) {
+	OrcIds64 out;
+	if (s->remove_ret_variant(d, &out)) {
+		*s = out;
+		return 1;
+	}
+	return 0;
+}
+uint32_t orc_ids64_export(OrcIds64 *s, uint64_t *out, int *is_bits) {
+	for (size_t i = 0; i < s->v.size(); i++)
+		out[i] = s->v[i];
+	*is_bits = s->bits ? 1 : 0;
+	return (uint32_t)s->v.size();
+}
+
+// check_hnsw_properties for the index (mod.rs:561-570 + the element count
+// from the reference's index tests): elements present == expected, plus the
+// layer invariants.
+int orc_index_check_props(orc_index *ix, uint64_t expected_count) {
+	uint64_t present = 0;
+	for (uint8_t p : ix->h->elem_present)
+		present += p;
+	if (present != expected_count)
+		return -10;
+	if (present != ix->vd.size())
+		return -11; // every element is exactly one Hv entry
+	return orc_hnsw_check_props(ix->h);
+}
 
 } // extern "C"
