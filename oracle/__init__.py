@@ -113,6 +113,15 @@ def _load():
                                   f64p]
     lib.orc_index_check_props.restype = ctypes.c_int
     lib.orc_index_check_props.argtypes = [ctypes.c_void_p, u64]
+    lib.orc_ids64_new.restype = ctypes.c_void_p
+    lib.orc_ids64_free.argtypes = [ctypes.c_void_p]
+    lib.orc_ids64_insert.restype = ctypes.c_int
+    lib.orc_ids64_insert.argtypes = [ctypes.c_void_p, u64]
+    lib.orc_ids64_remove.restype = ctypes.c_int
+    lib.orc_ids64_remove.argtypes = [ctypes.c_void_p, u64]
+    lib.orc_ids64_export.restype = u32
+    lib.orc_ids64_export.argtypes = [ctypes.c_void_p, u64p,
+                                     ctypes.POINTER(ctypes.c_int)]
     return lib
 
 
@@ -276,6 +285,35 @@ class _HnswView(Hnsw):
 
     def __del__(self):
         pass  # owned by the Index
+
+
+class Ids64:
+    """Test hook over the restated Ids64 (knn.rs:163-326). insert()/remove()
+    return True when a new variant was produced (the reference's Some) —
+    the contract VecDocs persists on."""
+
+    def __init__(self):
+        self._s = lib().orc_ids64_new()
+
+    def insert(self, d):
+        return bool(lib().orc_ids64_insert(self._s, d))
+
+    def remove(self, d):
+        return bool(lib().orc_ids64_remove(self._s, d))
+
+    def export(self):
+        out = np.empty(4096, dtype=np.uint64)
+        is_bits = ctypes.c_int(0)
+        n = lib().orc_ids64_export(
+            self._s, out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            ctypes.byref(is_bits))
+        return out[:n].tolist(), bool(is_bits.value)
+
+    def __del__(self):
+        try:
+            lib().orc_ids64_free(self._s)
+        except Exception:
+            pass
 
 
 class Index:

@@ -714,6 +714,13 @@ static void layer_search(const orc_hnsw *h, const OrcLayer &layer,
 			for (uint32_t e_id : layer.edges[doc]) {
 				if (!visited.insert(e_id).second)
 					continue;
+				// elements.get_vector -> None for removed elements
+				// (layer.rs:206: dangling edges — left behind by
+				// insert-time pruning asymmetry — are skipped AFTER the
+				// visited mark)
+				if (e_id < h->elem_present.size() &&
+				    !h->elem_present[e_id])
+					continue;
 				double e_dist = hdist(h, hvec(h, e_id), q);
 				if (e_dist < fq_dist || w.n < ef) {
 					if (!pend || !pend->all_docs_pending(pend->idx, e_id,
@@ -786,6 +793,11 @@ static void extend_candidates(const orc_hnsw *h, const OrcLayer &layer,
 		if (e_id < layer.edges.size()) {
 			for (uint32_t e_adj : layer.edges[e_id]) {
 				if (e_adj != q_id && ex.insert(e_adj).second) {
+					// get_distance -> None for removed elements
+					// (heuristic.rs:140-143: ex is marked first)
+					if (e_adj < h->elem_present.size() &&
+					    !h->elem_present[e_adj])
+						continue;
 					double dd = hdist(h, q_pt, hvec(h, e_adj));
 					ext.push_back({dd, e_adj});
 				}
