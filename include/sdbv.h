@@ -169,12 +169,57 @@ int sdbv_hnsw_knn_batch(sdbv_hnsw *, const float *Q, uint32_t b, uint32_t k,
                         uint32_t ef, uint64_t *out_ids, double *out_dists,
                         uint32_t *out_ns);
 void sdbv_hnsw_destroy(sdbv_hnsw *);
+/* Remove one graph element with neighbour repair (Hnsw::remove,
+ * hnsw/mod.rs:398-455 + layer.rs:408-460). Pre-finalize host graphs only;
+ * a finalized index mutates through sdbv_index_* (which re-finalizes).
+ * Returns 1 if removed, 0 if the element did not exist. */
+int sdbv_hnsw_remove(sdbv_hnsw *, uint64_t e_id);
 /* Introspection (parity tests): */
 uint64_t sdbv_hnsw_n(sdbv_hnsw *);
 uint32_t sdbv_hnsw_layers(sdbv_hnsw *);
 uint64_t sdbv_hnsw_l0_edge_count(sdbv_hnsw *);
 void sdbv_hnsw_l0_export(sdbv_hnsw *, uint32_t *offsets /* n+1 */,
                          uint32_t *edges);
+
+/* ------------------------------------------------------------------------
+ * Index layer — the HnswIndex operator surface (hnsw/index.rs) with the
+ * KV-backed parts (Hp pendings queue, Hv vector->docs entries, hi/hd
+ * record-key<->doc-id maps) held in host memory. The host binding passes
+ * record keys as opaque u64 handles ordered like its RecordIdKey ordering
+ * (INTEGRATION.md "record-key handles"); value->vector extraction
+ * (content_to_vectors, index.rs:118-129) stays on the host side of this
+ * boundary. ctx may be NULL for a host-only (CPU-testable) index; with a
+ * ctx, searches run the GPU per-hop path, re-finalizing into device table
+ * slot `table` after writes. */
+typedef struct sdbv_index sdbv_index;
+int sdbv_index_create(sdbv_ctx * /* nullable */, uint64_t table, uint32_t d,
+                      uint8_t metric, uint32_t m, uint32_t m0, uint32_t efc,
+                      int extend_candidates, int keep_pruned,
+                      uint64_t rng_seed, double ml, sdbv_index **out);
+void sdbv_index_destroy(sdbv_index *);
+/* HnswIndex::index (index.rs:138-186): append one pending update (old/new
+ * vectors, n*d f32 each; the id kind — DocId vs RecordKey — resolves via
+ * the record-key map exactly as HnswDocs::get_doc_id does). */
+int sdbv_index_enqueue(sdbv_index *, uint64_t record_key, const float *olds,
+                       uint32_t n_old, const float *news, uint32_t n_new);
+/* HnswIndex::index_pendings (index.rs:188-257): drain + apply the queue in
+ * appending order through VecDocs (insert/remove, graph element removal
+ * with repair, doc-id recycling). */
+int sdbv_index_apply_pendings(sdbv_index *, uint64_t *out_count);
+/* HnswIndex::knn_search (index.rs:270-335 minus record materialisation):
+ * pendings overlay + graph search (pending docs excluded from the
+ * expansion frontier, layer.rs:320-338) + Ids64 doc expansion through one
+ * KnnResultBuilder. Out arrays sized k; entries ascending
+ * (dist total_cmp, VectorId); kind 0 = DocId, 1 = RecordKey handle. */
+int sdbv_index_knn(sdbv_index *, const float *q, uint32_t k, uint32_t ef,
+                   uint8_t *out_kinds, uint64_t *out_ids, double *out_dists,
+                   uint32_t *out_n);
+uint64_t sdbv_index_doc_count(sdbv_index *);
+uint64_t sdbv_index_pending_count(sdbv_index *);
+/* check_hnsw_properties (hnsw/mod.rs:561-570): 0 == OK. */
+int sdbv_index_check_props(sdbv_index *, uint64_t expected_count);
+/* Test access to the underlying graph (CSR parity vs the oracle). */
+sdbv_hnsw *sdbv_index_hnsw(sdbv_index *);
 
 #ifdef __cplusplus
 }

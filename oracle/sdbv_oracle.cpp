@@ -830,11 +830,15 @@ static OrcPQ layer_insert(orc_hnsw *h, OrcLayer &layer, uint32_t q_id,
 	std::vector<uint32_t> neighbors;
 	select_neighbors(h, layer, q_id, q_pt, w, neighbors);
 
-	// add node + bidirectional edges
+	// add node + bidirectional edges (graph.rs:52-64 — the back-edge
+	// `nodes.entry(e).or_insert_with(..)` IMPLICITLY creates a missing
+	// target node, e.g. an upper-layer seed that was never inserted here)
 	layer.add_node(q_id);
 	layer.edges[q_id] = neighbors;
-	for (uint32_t e_id : neighbors)
+	for (uint32_t e_id : neighbors) {
+		layer.add_node(e_id);
 		layer.edges[e_id].push_back(q_id);
+	}
 
 	// prune over-full neighbors (layer.rs:363-377)
 	for (uint32_t e_id : neighbors) {

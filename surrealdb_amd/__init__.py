@@ -91,6 +91,24 @@ def lib():
     L.sdbv_hnsw_l0_edge_count.restype = u64
     L.sdbv_hnsw_l0_edge_count.argtypes = [vp]
     L.sdbv_hnsw_l0_export.argtypes = [vp, u32p, u32p]
+    L.sdbv_hnsw_remove.restype = ctypes.c_int
+    L.sdbv_hnsw_remove.argtypes = [vp, u64]
+    u8p = ctypes.POINTER(u8)
+    L.sdbv_index_create.argtypes = [vp, u64, u32, u8, u32, u32, u32,
+                                    ctypes.c_int, ctypes.c_int, u64,
+                                    ctypes.c_double, ctypes.POINTER(vp)]
+    L.sdbv_index_destroy.argtypes = [vp]
+    L.sdbv_index_enqueue.argtypes = [vp, u64, f32p, u32, f32p, u32]
+    L.sdbv_index_apply_pendings.argtypes = [vp, u64p]
+    L.sdbv_index_knn.argtypes = [vp, f32p, u32, u32, u8p, u64p, f64p, u32p]
+    L.sdbv_index_doc_count.restype = u64
+    L.sdbv_index_doc_count.argtypes = [vp]
+    L.sdbv_index_pending_count.restype = u64
+    L.sdbv_index_pending_count.argtypes = [vp]
+    L.sdbv_index_check_props.restype = ctypes.c_int
+    L.sdbv_index_check_props.argtypes = [vp, u64]
+    L.sdbv_index_hnsw.restype = vp
+    L.sdbv_index_hnsw.argtypes = [vp]
     _lib = L
     return L
 
@@ -321,7 +339,111 @@ class Hnsw:
             edges.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)))
         return offsets, edges[:ec]
 
+    def remove(self, e_id):
+        """Hnsw::remove (hnsw/mod.rs:398-455). True if removed. Host graphs
+        only (pre-finalize); finalized indexes mutate through Index."""
+        rc = lib().sdbv_hnsw_remove(self._ptr, e_id)
+        if rc < 0:
+            raise SdbvError(f"sdbv_hnsw_remove: {_ERRS.get(rc, rc)}")
+        return bool(rc)
+
     def destroy(self):
+        if not getattr(self, "_owned", True):
+            self._ptr = None  # view over an Index-owned graph
+            return
         if self._ptr:
             lib().sdbv_hnsw_destroy(self._ptr)
+            self._ptr = None
+
+
+def index_create_host(d, metric="euclidean", m=12, m0=None, efc=150,
+                      extend=False, keep=False, seed=0x5DB1, ml=None):
+    """Host-only Index (ctx = NULL): the full write path + pendings-merged
+    search on the host graph — the CPU-testable configuration."""
+    return Index(None, 0, d, metric, m, m0, efc, extend, keep, seed, ml)
+
+
+class Index:
+    """Product HnswIndex (hnsw/index.rs operator surface): pendings queue,
+    VecDocs/Ids64 doc expansion, doc-id allocation, pendings-merged
+    knn_search. Record keys are opaque u64 handles (host's RecordIdKey
+    mapping — INTEGRATION.md). With a Context, graph searches run the GPU
+    per-hop path, auto-re-finalizing into `table` after writes."""
+
+    def __init__(self, ctx, table, d, metric="euclidean", m=12, m0=None,
+                 efc=150, extend=False, keep=False, seed=0x5DB1, ml=None):
+        import math
+        if m0 is None:
+            m0 = 2 * m
+        if ml is None:
+            ml = 1.0 / math.log(m)
+        out = ctypes.c_void_p()
+        cptr = ctx._ptr if ctx is not None else None
+        _check(cptr, lib().sdbv_index_create(
+            cptr, table, d, METRICS[metric], m, m0, efc, int(extend),
+            int(keep), seed, ml, ctypes.byref(out)), "sdbv_index_create")
+        self._ctx = ctx
+        self._ptr = out
+        self.d = d
+
+    def enqueue(self, record_key, old_vectors=None, new_vectors=None):
+        """HnswIndex::index (index.rs:138-186). Vectors: (n, d) f32 arrays
+        (None == no values of that kind)."""
+        import numpy as np
+
+        def flat(a):
+            if a is None:
+                return np.empty((0, self.d), dtype=np.float32)
+            return np.ascontiguousarray(a, dtype=np.float32).reshape(
+                -1, self.d)
+        o, nw = flat(old_vectors), flat(new_vectors)
+        _check(None, lib().sdbv_index_enqueue(
+            self._ptr, record_key,
+            o.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), o.shape[0],
+            nw.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), nw.shape[0]),
+            "sdbv_index_enqueue")
+
+    def apply_pendings(self):
+        n = ctypes.c_uint64(0)
+        _check(None, lib().sdbv_index_apply_pendings(
+            self._ptr, ctypes.byref(n)), "sdbv_index_apply_pendings")
+        return n.value
+
+    def knn_search(self, q, k, ef):
+        """index.rs:270-335 minus record materialisation: (kinds u8
+        [0=DocId, 1=RecordKey], ids u64, dists f64), ascending."""
+        import numpy as np
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        kinds = np.empty(k, dtype=np.uint8)
+        ids = np.empty(k, dtype=np.uint64)
+        dists = np.empty(k, dtype=np.float64)
+        out_n = ctypes.c_uint32(0)
+        _check(self._ctx._ptr if self._ctx else None, lib().sdbv_index_knn(
+            self._ptr, q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+            k, ef, kinds.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.byref(out_n)), "sdbv_index_knn")
+        n = out_n.value
+        return kinds[:n], ids[:n], dists[:n]
+
+    def doc_count(self):
+        return lib().sdbv_index_doc_count(self._ptr)
+
+    def pending_count(self):
+        return lib().sdbv_index_pending_count(self._ptr)
+
+    def check_props(self, expected_count):
+        return lib().sdbv_index_check_props(self._ptr, expected_count)
+
+    def hnsw(self):
+        """Non-owning view of the underlying graph (parity introspection)."""
+        h = Hnsw(self._ctx, ctypes.c_void_p(lib().sdbv_index_hnsw(self._ptr)),
+                 self.d)
+        h._owned = False
+        return h
+
+    def destroy(self):
+        if self._ptr:
+            lib().sdbv_index_destroy(self._ptr)
             self._ptr = None

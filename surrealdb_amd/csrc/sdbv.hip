@@ -51,7 +51,9 @@
 #include <cstring>
 #include <map>
 #include <mutex>
+#include <set>
 #include <string>
+#include <unordered_map>
 #include <vector>
 
 #include "../../include/sdbv.h"
@@ -1747,6 +1749,13 @@ static double host_sumsq_f32(const float *a, uint32_t d) {
 struct Layer {
 	std::vector<std::vector<uint32_t>> edges;
 	uint32_t m_max;
+	// graph.rs nodes-map membership: get_edges is None for absent nodes and
+	// layer.remove is a no-op on them. Kept alongside `edges` (which is
+	// sized to the element count for lock-striped parallel inserts).
+	std::vector<uint8_t> in_layer;
+	bool has(uint32_t id) const {
+		return id < in_layer.size() && in_layer[id];
+	}
 };
 
 } // namespace hnsw
@@ -1762,6 +1771,10 @@ struct sdbv_hnsw {
 	std::vector<float> vecs;    // host row-major copy (build + upper layers)
 	std::vector<double> norms;  // per-element f64 norm (cosine)
 	std::vector<hnsw::Layer> layers;
+	// elements-map membership (hnsw/elements.rs): remove() keeps the vector
+	// slot but the element no longer exists for searches
+	std::vector<uint8_t> elem_present;
+	bool dirty = false; // host graph changed since finalize (device stale)
 	int64_t enter_point = -1;
 	uint64_t next_id = 0;
 	// parallel build
@@ -1819,11 +1832,27 @@ static std::vector<uint32_t> get_edges(sdbv_hnsw *h, const Layer &layer,
 	return id < layer.edges.size() ? layer.edges[id] : std::vector<uint32_t>{};
 }
 
-// layer.rs:184-223 — host-distance variant (build path).
+} // namespace hnsw
+struct sdbv_index; // defined below (index layer)
+namespace hnsw {
+
+// Pending-docs context for index searches (hnsw/index.rs knn path): the
+// DocId bitmap from search_pendings plus the element->docs accessor.
+struct IdxPend {
+	const std::set<uint64_t> *pending;
+	const ::sdbv_index *ix;
+};
+static bool idx_all_docs_pending(const IdxPend *p, uint32_t e_id);
+
+// layer.rs:184-223 — host-distance variant (build path + index host path).
+// `pend`: an element whose docs are ALL pending is excluded from
+// `candidates` only; it still enters `w` (layer.rs:209-212, the reference
+// pushes to w outside the exclusion check — restated as-is).
 static void search_layer_host(sdbv_hnsw *h, const Layer &layer, const float *q,
                               double q_norm, PQ &candidates,
                               std::unordered_set<uint32_t> &visited, PQ &w,
-                              uint32_t ef, bool locked) {
+                              uint32_t ef, bool locked,
+                              const IdxPend *pend = nullptr) {
 	double fq = w.peek_last_dist(DBL_MAX);
 	double cd;
 	uint32_t doc;
@@ -1833,9 +1862,15 @@ static void search_layer_host(sdbv_hnsw *h, const Layer &layer, const float *q,
 		for (uint32_t e : get_edges(h, layer, doc, locked)) {
 			if (!visited.insert(e).second)
 				continue;
+			// elements.get_vector -> None for removed elements
+			// (layer.rs:206): dangling edges left by insert-time pruning
+			// asymmetry are skipped after the visited mark
+			if (e < h->elem_present.size() && !h->elem_present[e])
+				continue;
 			double ed = dist(h, q, q_norm, e);
 			if (ed < fq || w.n < ef) {
-				candidates.push(ed, e);
+				if (!pend || !idx_all_docs_pending(pend, e))
+					candidates.push(ed, e);
 				w.push(ed, e);
 				if (w.n > ef)
 					w.pop_last();
@@ -1855,19 +1890,28 @@ static bool is_closer(sdbv_hnsw *h, double ed, uint32_t e,
 	return true;
 }
 
-// heuristic.rs:35-116 (+ extend :118-157)
+// heuristic.rs:35-116 (+ extend :118-157; `ignore` = heuristic.rs:130-134,
+// the element being removed is excluded from the extension set)
 static void select_neighbors(sdbv_hnsw *h, const Layer &layer, uint32_t q_id,
                              const float *q_pt, double q_norm, PQ c,
-                             std::vector<uint32_t> &res, bool locked) {
+                             std::vector<uint32_t> &res, bool locked,
+                             int64_t ignore = -1) {
 	if (h->extend) {
 		std::unordered_set<uint32_t> ex;
 		auto base = c.to_vec();
 		for (auto &e : base)
 			ex.insert(e.second);
+		if (ignore >= 0)
+			ex.insert((uint32_t)ignore);
 		for (auto &e : base)
 			for (uint32_t adj : get_edges(h, layer, e.second, locked))
-				if (adj != q_id && ex.insert(adj).second)
+				if (adj != q_id && ex.insert(adj).second) {
+					// get_distance -> None for removed elements
+					if (adj < h->elem_present.size() &&
+					    !h->elem_present[adj])
+						continue;
 					c.push(dist(h, q_pt, q_norm, adj), adj);
+				}
 	}
 	uint32_t m_max = layer.m_max;
 	if (c.n <= m_max) {
@@ -1920,6 +1964,10 @@ static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 		std::vector<uint32_t> conn;
 		{
 			std::lock_guard<std::mutex> lk(h->node_locks[e & 4095]);
+			// graph.rs:52-64: the back-edge entry().or_insert IMPLICITLY
+			// creates a missing target node (e.g. an upper-layer seed)
+			if (e < layer.in_layer.size())
+				layer.in_layer[e] = 1;
 			auto &ee = layer.edges[e];
 			if (std::find(ee.begin(), ee.end(), q_id) == ee.end())
 				ee.push_back(q_id);
@@ -1977,9 +2025,19 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 			h->layers.push_back(hnsw::Layer{
 			    std::vector<std::vector<uint32_t>>(h->vecs.size() / h->d),
 			    h->m});
-		for (auto &l : h->layers)
+		uint64_t nelem = h->vecs.size() / h->d;
+		for (auto &l : h->layers) {
 			if (l.edges.size() <= q_id)
-				l.edges.resize(h->vecs.size() / h->d);
+				l.edges.resize(nelem);
+			if (l.in_layer.size() < nelem)
+				l.in_layer.resize(nelem, 0);
+		}
+		// node membership on layers 0..=q_level (graph.rs add_node /
+		// add_empty_node in insert_first_element and insert_element)
+		for (uint32_t l = 0;
+		     l < (uint32_t)h->layers.size() && l <= q_level; l++)
+			h->layers[l].in_layer[q_id] = 1;
+		h->dirty = true;
 		if (h->enter_point < 0) {
 			h->enter_point = q_id;
 			return;
@@ -2014,6 +2072,96 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 		std::lock_guard<std::mutex> lk(h->global_mu);
 		h->enter_point = q_id;
 	}
+}
+
+// ---- graph element removal (sequential only: apply_pendings holds the
+// reference's write lock; the parallel bench-mode build never removes) ----
+
+// layer.rs:92-108 search_single_with_ignore: seeded FROM the ignored
+// element; returns the closest found element or -1 (None).
+static int64_t search_single_with_ignore(sdbv_hnsw *h, const Layer &layer,
+                                         const float *pt, double pt_norm,
+                                         uint32_t ignore_id, uint32_t ef) {
+	std::unordered_set<uint32_t> visited{ignore_id};
+	PQ candidates;
+	candidates.push(dist(h, pt, pt_norm, ignore_id), ignore_id);
+	PQ w;
+	search_layer_host(h, layer, pt, pt_norm, candidates, visited, w, ef,
+	                  false);
+	double dd;
+	uint32_t ii;
+	if (w.peek_first(&dd, &ii))
+		return (int64_t)ii;
+	return -1;
+}
+
+// layer.rs:164-181 search_multi_with_ignore.
+static PQ search_multi_with_ignore(sdbv_hnsw *h, const Layer &layer,
+                                   const float *pt, double pt_norm,
+                                   const std::vector<uint32_t> &ignore_ids,
+                                   uint32_t efc) {
+	PQ candidates;
+	for (uint32_t id : ignore_ids)
+		candidates.push(dist(h, pt, pt_norm, id), id);
+	std::unordered_set<uint32_t> visited(ignore_ids.begin(),
+	                                     ignore_ids.end());
+	PQ w;
+	search_layer_host(h, layer, pt, pt_norm, candidates, visited, w, efc,
+	                  false);
+	return w;
+}
+
+// layer.rs:408-460 HnswLayer::remove: drop node + back-edges, then repair
+// each former neighbour (efc-search ignoring {q_id, e_id}, heuristic
+// re-selection with ignore=e_id, one-directional set_node).
+static bool layer_remove(sdbv_hnsw *h, Layer &layer, uint32_t e_id) {
+	if (!layer.has(e_id))
+		return false;
+	std::vector<uint32_t> f_ids = std::move(layer.edges[e_id]);
+	layer.edges[e_id].clear();
+	layer.in_layer[e_id] = 0;
+	for (uint32_t f : f_ids) {
+		auto &fe = layer.edges[f];
+		fe.erase(std::remove(fe.begin(), fe.end(), e_id), fe.end());
+	}
+	for (uint32_t q_id : f_ids) {
+		const float *q_pt = vec(h, q_id);
+		double q_norm =
+		    h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+		PQ c = search_multi_with_ignore(h, layer, q_pt, q_norm,
+		                                {q_id, e_id}, h->efc);
+		std::vector<uint32_t> q_new_conn;
+		select_neighbors(h, layer, q_id, q_pt, q_norm, std::move(c),
+		                 q_new_conn, false, (int64_t)e_id);
+		layer.edges[q_id] = q_new_conn; // graph.set_node
+	}
+	return true;
+}
+
+// hnsw/mod.rs:398-455 Hnsw::remove.
+static bool hnsw_remove(sdbv_hnsw *h, uint32_t e_id) {
+	if (e_id >= h->next_id || !h->elem_present[e_id])
+		return false; // elements.get_vector -> None
+	bool removed = false;
+	const float *e_pt = vec(h, e_id);
+	double e_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[e_id] : 0;
+	int64_t new_ep = ((int64_t)e_id == h->enter_point) ? -1 : h->enter_point;
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		if (new_ep < 0)
+			new_ep = search_single_with_ignore(h, h->layers[l], e_pt,
+			                                   e_norm, e_id, h->efc);
+		if (layer_remove(h, h->layers[l], e_id))
+			removed = true;
+	}
+	if (new_ep < 0)
+		new_ep = search_single_with_ignore(h, h->layers[0], e_pt, e_norm,
+		                                   e_id, h->efc);
+	if (layer_remove(h, h->layers[0], e_id))
+		removed = true;
+	h->elem_present[e_id] = 0; // elements.remove
+	h->enter_point = new_ep;
+	h->dirty = true;
+	return removed;
 }
 
 } // namespace hnsw
@@ -2097,6 +2245,7 @@ void sdbv_hnsw_destroy(sdbv_hnsw *h) {
 
 static void hnsw_append_vec(sdbv_hnsw *h, const float *pt) {
 	h->vecs.insert(h->vecs.end(), pt, pt + h->d);
+	h->elem_present.push_back(1);
 	if (h->metric == SDBV_METRIC_COSINE)
 		h->norms.push_back(sqrt(hnsw::host_sumsq_f32(pt, h->d)));
 }
@@ -2187,9 +2336,33 @@ void sdbv_hnsw_l0_export(sdbv_hnsw *h, uint32_t *offsets, uint32_t *edges) {
 	offsets[h->next_id] = off;
 }
 
+// Frees the per-index device state so finalize can be called again after
+// host-graph mutations (apply_pendings re-finalizes a dirty index).
+static void hnsw_free_device_state(sdbv_hnsw *h) {
+	for (void **p : {(void **)&h->rows_dev, (void **)&h->dout_dev,
+	                 (void **)&h->q_dev, (void **)&h->rm_dev,
+	                 (void **)&h->offsets_dev, (void **)&h->edges_dev,
+	                 (void **)&h->norms_dev, (void **)&h->vis_dev}) {
+		if (*p)
+			(void)hipFree(*p);
+		*p = nullptr;
+	}
+	if (h->rows_pinned) {
+		(void)hipHostFree(h->rows_pinned);
+		h->rows_pinned = nullptr;
+	}
+	if (h->dists_pinned) {
+		(void)hipHostFree(h->dists_pinned);
+		h->dists_pinned = nullptr;
+	}
+	h->vis_cap = 0;
+	h->finalized = false;
+}
+
 int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
 	if (!h || !h->ctx || h->next_id == 0)
 		return SDBV_ERR_BAD_ARG;
+	hnsw_free_device_state(h);
 	int rc = sdbv_stage_corpus(h->ctx, table, h->vecs.data(), nullptr,
 	                           h->next_id, h->d, h->metric);
 	if (rc)
@@ -2209,18 +2382,22 @@ int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
 	                         n * h->d * sizeof(float),
 	                         hipMemcpyHostToDevice));
 	{
+		// Scrubbed CSR: dangling edges to removed elements are dropped at
+		// export, so the device kernels (which have no elements map) see
+		// exactly the live graph — equivalent to the reference's
+		// get_vector None gate (layer.rs:206).
 		std::vector<uint32_t> offsets(n + 1);
+		std::vector<uint32_t> edges;
 		uint64_t ec = 0;
 		for (uint64_t i = 0; i < n; i++) {
 			offsets[i] = (uint32_t)ec;
-			ec += h->layers[0].edges[i].size();
+			for (uint32_t e : h->layers[0].edges[i])
+				if (h->elem_present[e]) {
+					edges.push_back(e);
+					ec++;
+				}
 		}
 		offsets[n] = (uint32_t)ec;
-		std::vector<uint32_t> edges;
-		edges.reserve(ec);
-		for (uint64_t i = 0; i < n; i++)
-			edges.insert(edges.end(), h->layers[0].edges[i].begin(),
-			             h->layers[0].edges[i].end());
 		HIP_CHECK(ctx, hipMalloc(&h->offsets_dev, (n + 1) * sizeof(uint32_t)));
 		HIP_CHECK(ctx, hipMalloc(&h->edges_dev,
 		                         std::max<uint64_t>(ec, 1) * sizeof(uint32_t)));
@@ -2239,6 +2416,7 @@ int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
 	}
 	h->table = table;
 	h->finalized = true;
+	h->dirty = false;
 	return SDBV_OK;
 }
 
@@ -2476,6 +2654,10 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 		for (uint32_t e : l0.edges[doc])
 			if (!visited[e]) {
 				visited[e] = true;
+				// elements.get_vector -> None (dangling edge to a
+				// removed element): visited-marked then skipped
+				if (!h->elem_present[e])
+					continue;
 				frontier.push_back(e);
 			}
 		if (frontier.empty())
@@ -2541,6 +2723,644 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 		out_dists[i] = fin[i].second;
 	}
 	return SDBV_OK;
+}
+
+int sdbv_hnsw_remove(sdbv_hnsw *h, uint64_t e_id) {
+	if (!h || h->finalized)
+		return SDBV_ERR_BAD_ARG;
+	return hnsw::hnsw_remove(h, (uint32_t)e_id) ? 1 : 0;
+}
+
+} // extern "C"
+
+// ===========================================================================
+// Index layer — the operator surface of HnswIndex (hnsw/index.rs) with the
+// parts the reference keeps behind the KV transaction held in host memory:
+// the Hp pendings queue, the Hv vector->docs entries (VecDocs + Ids64), and
+// the hi/hd record-key<->doc-id maps (HnswDocs). Record keys are opaque u64
+// handles supplied by the host binding (INTEGRATION.md "record-key
+// handles"). Searches run the pendings overlay on the host and the graph
+// search on the GPU per-hop path once the index is finalized (re-finalizing
+// automatically after writes); a host-only index (ctx == NULL) searches the
+// host graph — that is the CPU-testable path, not a product fallback: a
+// GPU-bound index always has a ctx.
+// ===========================================================================
+
+namespace vdocs {
+
+// knn.rs:163-326 Ids64 — doc-id set with size-dependent representation.
+// Vec1..Vec8 keep INSERTION order; the 9th insert collapses to Bits
+// (iteration ascending); dropping back to exactly 8 keeps ascending order.
+// insert/remove return "a new variant was produced" — the contract VecDocs
+// persists on (Bits in-place mutations are dropped by the caller,
+// docs.rs:374-382/:437-447; restated as-is).
+struct Ids64 {
+	std::vector<uint64_t> v;
+	bool bits = false;
+	size_t len() const { return v.size(); }
+	bool empty() const { return v.empty(); }
+	bool contains(uint64_t d) const {
+		return std::find(v.begin(), v.end(), d) != v.end();
+	}
+	bool insert_ret_variant(uint64_t d) {
+		if (contains(d))
+			return false;
+		if (!bits) {
+			v.push_back(d);
+			if (v.size() > 8) {
+				std::sort(v.begin(), v.end());
+				bits = true;
+			}
+			return true;
+		}
+		v.insert(std::lower_bound(v.begin(), v.end(), d), d);
+		return false;
+	}
+	bool remove_ret_variant(uint64_t d, Ids64 *out) {
+		if (bits) {
+			auto it = std::lower_bound(v.begin(), v.end(), d);
+			bool had = (it != v.end() && *it == d);
+			if (had)
+				v.erase(it);
+			if (!had || v.size() != 8)
+				return false;
+			out->v = v;
+			out->bits = false;
+			return true;
+		}
+		switch (v.size()) {
+		case 0:
+			return false;
+		case 1:
+			if (v[0] == d) {
+				out->v.clear();
+				out->bits = false;
+				return true;
+			}
+			return false;
+		case 2:
+			// knn.rs:266-268: first element != d becomes One — for a
+			// non-member d this drops the second element (restated as-is)
+			for (uint64_t x : v)
+				if (x != d) {
+					out->v = {x};
+					out->bits = false;
+					return true;
+				}
+			return false;
+		default: {
+			std::vector<uint64_t> f;
+			for (uint64_t x : v)
+				if (x != d)
+					f.push_back(x);
+			if (f.size() == v.size() - 1) {
+				out->v = std::move(f);
+				out->bits = false;
+				return true;
+			}
+			return false;
+		}
+		}
+	}
+};
+
+// knn.rs:363-437 KnnResultBuilder over VectorId (kind 0 = DocId < kind 1 =
+// RecordKey, the enum's derived Ord).
+struct Vid {
+	uint8_t kind;
+	uint64_t id;
+	bool operator<(const Vid &o) const {
+		if (kind != o.kind)
+			return kind < o.kind;
+		return id < o.id;
+	}
+};
+struct Builder {
+	size_t knn;
+	struct Ent {
+		uint64_t key;
+		Vid vid;
+		double dist;
+		bool operator<(const Ent &o) const {
+			if (key != o.key)
+				return key < o.key;
+			return vid < o.vid;
+		}
+	};
+	std::set<Ent> pl;
+	std::map<Vid, size_t> count;
+	explicit Builder(size_t k) : knn(k) {}
+	// knn.rs:386-394: plain f64 `>` against the current worst
+	bool check_add(double submitted) const {
+		if (pl.size() >= knn && !pl.empty() &&
+		    submitted > std::prev(pl.end())->dist)
+			return false;
+		return true;
+	}
+	void add(double dist, Vid vid) {
+		pl.insert(Ent{hnsw::total_key(dist), vid, dist});
+		count[vid]++; // incremented even on duplicate set inserts
+		if (pl.size() <= knn)
+			return;
+		auto last = std::prev(pl.end());
+		Vid ev = last->vid;
+		pl.erase(last);
+		auto it = count.find(ev);
+		if (it != count.end()) {
+			if (it->second <= 1)
+				count.erase(it);
+			else
+				it->second--;
+		}
+	}
+	void add_graph(double dist, const Ids64 &docs) {
+		for (uint64_t doc : docs.v)
+			add(dist, Vid{0, doc});
+	}
+};
+
+} // namespace vdocs
+
+struct sdbv_index {
+	sdbv_hnsw *h;
+	uint64_t table;
+	struct ED {
+		uint32_t e_id;
+		vdocs::Ids64 docs;
+	};
+	// Hv entries: serialized vector bytes -> (element, docs)
+	std::unordered_map<std::string, ED> vd;
+	std::unordered_map<uint32_t, const std::string *> by_elem;
+	// HnswDocs: hi/hd maps + recycled allocation (docs.rs:20-135)
+	std::map<uint64_t, uint64_t> key2doc, doc2key;
+	std::set<uint64_t> available;
+	uint64_t next_doc_id = 0;
+	// Hp pendings in appending order (index.rs:131-174)
+	struct Pending {
+		uint8_t kind; // 0 DocId, 1 RecordKey
+		uint64_t id;
+		std::vector<float> olds, news;
+	};
+	std::vector<Pending> pendings;
+	// the reference's RwLock<HnswFlavor> (index.rs:55): one mutex here
+	std::mutex mu;
+};
+
+namespace hnsw {
+// layer.rs:320-338 are_all_docs_in_pending: an element with no VecDocs
+// entry, or whose every doc is pending, counts as all-pending.
+static bool idx_all_docs_pending(const IdxPend *p, uint32_t e_id) {
+	if (!p->pending || p->pending->empty())
+		return false;
+	auto it = p->ix->by_elem.find(e_id);
+	if (it != p->ix->by_elem.end()) {
+		for (uint64_t doc : p->ix->vd.at(*it->second).docs.v)
+			if (!p->pending->count(doc))
+				return false;
+	}
+	return true;
+}
+} // namespace hnsw
+
+// Distance of the query against a raw (not yet indexed) vector — the
+// pendings overlay's Distance::calculate (idx/trees/vector.rs:660-672),
+// same restated chain as the element path.
+static double idx_dist_raw(const sdbv_hnsw *h, const float *q, double q_norm,
+                           const float *v) {
+	if (h->metric == SDBV_METRIC_COSINE) {
+		double dot = hnsw::host_dot_f32(q, v, h->d);
+		double v_norm = sqrt(hnsw::host_sumsq_f32(v, h->d));
+		return 1.0 - dot / (q_norm * v_norm);
+	}
+	float acc = 0;
+	for (uint32_t i = 0; i < h->d; i++) {
+		float diff = q[i] - v[i];
+		acc += diff * diff;
+	}
+	return sqrt((double)acc);
+}
+
+// docs.rs:64-90 resolve / :78-90 next_doc_id (smallest recycled id first).
+static uint64_t idx_docs_resolve(sdbv_index *ix, uint64_t record_key) {
+	auto it = ix->key2doc.find(record_key);
+	if (it != ix->key2doc.end())
+		return it->second;
+	uint64_t doc_id;
+	if (!ix->available.empty()) {
+		doc_id = *ix->available.begin();
+		ix->available.erase(ix->available.begin());
+	} else {
+		doc_id = ix->next_doc_id++;
+	}
+	ix->key2doc[record_key] = doc_id;
+	ix->doc2key[doc_id] = record_key;
+	return doc_id;
+}
+
+// docs.rs:113-135 HnswDocs::remove (recycle the id).
+static void idx_docs_remove(sdbv_index *ix, uint64_t doc_id) {
+	auto it = ix->doc2key.find(doc_id);
+	if (it == ix->doc2key.end())
+		return;
+	ix->key2doc.erase(it->second);
+	ix->doc2key.erase(it);
+	ix->available.insert(doc_id);
+}
+
+// docs.rs:363-393 VecDocs::insert.
+static void idx_vd_insert(sdbv_index *ix, const float *v, uint64_t doc_id) {
+	std::string key((const char *)v, (size_t)ix->h->d * 4);
+	auto it = ix->vd.find(key);
+	if (it == ix->vd.end()) {
+		uint32_t e_id = (uint32_t)ix->h->next_id;
+		sdbv_hnsw_insert(ix->h, v);
+		auto r = ix->vd.emplace(std::move(key), sdbv_index::ED{e_id, {}});
+		r.first->second.docs.v = {doc_id};
+		ix->by_elem[e_id] = &r.first->first;
+	} else {
+		sdbv_index::ED ed = it->second; // owned copy, like tx.get
+		if (ed.docs.insert_ret_variant(doc_id))
+			it->second = ed; // persisted only on a new variant
+	}
+}
+
+// docs.rs:424-449 VecDocs::remove.
+static void idx_vd_remove(sdbv_index *ix, const float *v, uint64_t doc_id) {
+	std::string key((const char *)v, (size_t)ix->h->d * 4);
+	auto it = ix->vd.find(key);
+	if (it == ix->vd.end())
+		return;
+	sdbv_index::ED ed = it->second;
+	vdocs::Ids64 new_docs;
+	if (ed.docs.remove_ret_variant(doc_id, &new_docs)) {
+		if (new_docs.empty()) {
+			uint32_t e_id = ed.e_id;
+			ix->by_elem.erase(e_id);
+			ix->vd.erase(it);
+			hnsw::hnsw_remove(ix->h, e_id);
+		} else {
+			ed.docs = new_docs;
+			it->second = ed;
+		}
+	}
+	// else: no new variant — the mutation is dropped (docs.rs:437-447)
+}
+
+// Host full graph search with pendings (knn_search, mod.rs:459-482 +
+// search_ep :521-548): used for a host-only index (CPU tests).
+static uint32_t idx_graph_search_host(sdbv_index *ix, const float *q,
+                                      uint32_t k, uint32_t ef,
+                                      const hnsw::IdxPend *pend,
+                                      std::vector<std::pair<double, uint32_t>>
+                                          &out) {
+	using namespace hnsw;
+	sdbv_hnsw *h = ix->h;
+	if (h->enter_point < 0)
+		return 0;
+	double q_norm = h->metric == SDBV_METRIC_COSINE
+	                    ? sqrt(host_sumsq_f32(q, h->d))
+	                    : 0;
+	uint32_t ep_id = (uint32_t)h->enter_point;
+	double ep_dist = dist(h, q, q_norm, ep_id);
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		PQ cand;
+		cand.push(ep_dist, ep_id);
+		std::unordered_set<uint32_t> visited{ep_id};
+		PQ w = cand;
+		search_layer_host(h, h->layers[l], q, q_norm, cand, visited, w, 1,
+		                  false, pend);
+		double dd;
+		uint32_t ii;
+		if (w.peek_first(&dd, &ii)) {
+			ep_dist = dd;
+			ep_id = ii;
+		}
+	}
+	PQ cand;
+	cand.push(ep_dist, ep_id);
+	std::unordered_set<uint32_t> visited{ep_id};
+	PQ w = cand;
+	search_layer_host(h, h->layers[0], q, q_norm, cand, visited, w, ef,
+	                  false, pend);
+	auto v = w.to_vec(); // to_vec_limit(k), knn.rs:92-104
+	uint32_t n = (uint32_t)std::min<size_t>(k, v.size());
+	for (uint32_t i = 0; i < n; i++)
+		out.push_back({v[i].first, v[i].second});
+	return n;
+}
+
+// GPU per-hop graph search with pendings: the sdbv_hnsw_knn layer-0 loop
+// (host queue + k_gather_dist batched expansion) with the pending gate on
+// the candidates push. Caller holds ix->mu; takes ctx->mu itself.
+static int idx_graph_search_gpu(sdbv_index *ix, const float *q, uint32_t k,
+                                uint32_t ef, const hnsw::IdxPend *pend,
+                                std::vector<std::pair<double, uint32_t>> &out,
+                                uint32_t *out_n) {
+	using namespace hnsw;
+	sdbv_hnsw *h = ix->h;
+	*out_n = 0;
+	if (h->enter_point < 0)
+		return SDBV_OK;
+	sdbv_ctx *ctx = h->ctx;
+	std::lock_guard<std::mutex> lk(ctx->mu);
+	auto it = ctx->tables.find(h->table);
+	if (it == ctx->tables.end())
+		return SDBV_ERR_NO_TABLE;
+	Table &t = it->second;
+	double q_norm = h->metric == SDBV_METRIC_COSINE
+	                    ? sqrt(host_sumsq_f32(q, h->d))
+	                    : 0;
+	// upper layers: host descent (tiny)
+	uint32_t ep_id = (uint32_t)h->enter_point;
+	double ep_dist = dist(h, q, q_norm, ep_id);
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		PQ cand;
+		cand.push(ep_dist, ep_id);
+		std::unordered_set<uint32_t> visited{ep_id};
+		PQ w = cand;
+		search_layer_host(h, h->layers[l], q, q_norm, cand, visited, w, 1,
+		                  false, pend);
+		double dd;
+		uint32_t ii;
+		if (w.peek_first(&dd, &ii)) {
+			ep_dist = dd;
+			ep_id = ii;
+		}
+	}
+	// layer 0: ef-search, one gather+distance launch per hop
+	HIP_CHECK(ctx, hipMemcpyAsync(h->q_dev, q, h->d * sizeof(float),
+	                              hipMemcpyHostToDevice, ctx->stream));
+	const Layer &l0 = h->layers[0];
+	PQ candidates, w;
+	candidates.push(ep_dist, ep_id);
+	w.push(ep_dist, ep_id);
+	std::vector<bool> visited(h->next_id, false);
+	visited[ep_id] = true;
+	double fq = w.peek_last_dist(DBL_MAX);
+	std::vector<uint32_t> frontier;
+	std::vector<double> fdists(h->m0 + 1);
+	double cd;
+	uint32_t doc;
+	while (candidates.pop_first(&cd, &doc)) {
+		if (cd > fq)
+			break;
+		frontier.clear();
+		for (uint32_t e : l0.edges[doc])
+			if (!visited[e]) {
+				visited[e] = true;
+				if (!h->elem_present[e])
+					continue;
+				frontier.push_back(e);
+			}
+		if (frontier.empty())
+			continue;
+		std::memcpy(h->rows_pinned, frontier.data(),
+		            frontier.size() * sizeof(uint32_t));
+		HIP_CHECK(ctx, hipMemcpyAsync(h->rows_dev, h->rows_pinned,
+		                              frontier.size() * sizeof(uint32_t),
+		                              hipMemcpyHostToDevice, ctx->stream));
+		if (t.metric == SDBV_METRIC_COSINE)
+			hipLaunchKernelGGL(k_gather_dist<0>,
+			                   dim3((uint32_t)frontier.size()), dim3(64), 0,
+			                   ctx->stream, t.cm, t.norms, t.n_pad, t.d,
+			                   h->rows_dev, (uint32_t)frontier.size(),
+			                   h->q_dev, q_norm, h->dout_dev);
+		else
+			hipLaunchKernelGGL(k_gather_dist<1>,
+			                   dim3((uint32_t)frontier.size()), dim3(64), 0,
+			                   ctx->stream, t.cm, t.norms, t.n_pad, t.d,
+			                   h->rows_dev, (uint32_t)frontier.size(),
+			                   h->q_dev, q_norm, h->dout_dev);
+		HIP_CHECK(ctx, hipMemcpyAsync(h->dists_pinned, h->dout_dev,
+		                              frontier.size() * sizeof(double),
+		                              hipMemcpyDeviceToHost, ctx->stream));
+		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		std::memcpy(fdists.data(), h->dists_pinned,
+		            frontier.size() * sizeof(double));
+		// sequential accept in edge order (layer.rs:195-218) with the
+		// pending gate on candidates only (:209-212)
+		for (size_t i = 0; i < frontier.size(); i++) {
+			double ed = fdists[i];
+			uint32_t e = frontier[i];
+			if (ed < fq || w.n < ef) {
+				if (!pend || !idx_all_docs_pending(pend, e))
+					candidates.push(ed, e);
+				w.push(ed, e);
+				if (w.n > ef)
+					w.pop_last();
+				fq = w.peek_last_dist(DBL_MAX);
+			}
+		}
+	}
+	auto v = w.to_vec();
+	uint32_t n = (uint32_t)std::min<size_t>(k, v.size());
+	for (uint32_t i = 0; i < n; i++)
+		out.push_back({v[i].first, v[i].second});
+	*out_n = n;
+	return SDBV_OK;
+}
+
+extern "C" {
+
+int sdbv_index_create(sdbv_ctx *ctx, uint64_t table, uint32_t d,
+                      uint8_t metric, uint32_t m, uint32_t m0, uint32_t efc,
+                      int extend, int keep, uint64_t seed, double ml,
+                      sdbv_index **out) {
+	sdbv_hnsw *h = nullptr;
+	int rc = sdbv_hnsw_create(ctx, d, metric, m, m0, efc, extend, keep, seed,
+	                          ml, &h);
+	if (rc)
+		return rc;
+	auto *ix = new sdbv_index();
+	ix->h = h;
+	ix->table = table;
+	*out = ix;
+	return SDBV_OK;
+}
+
+void sdbv_index_destroy(sdbv_index *ix) {
+	if (!ix)
+		return;
+	sdbv_hnsw_destroy(ix->h);
+	delete ix;
+}
+
+sdbv_hnsw *sdbv_index_hnsw(sdbv_index *ix) { return ix ? ix->h : nullptr; }
+uint64_t sdbv_index_doc_count(sdbv_index *ix) { return ix->doc2key.size(); }
+uint64_t sdbv_index_pending_count(sdbv_index *ix) {
+	return ix->pendings.size();
+}
+
+// HnswIndex::index (index.rs:138-186): resolve the id kind via the hi map
+// and append one VectorPendingUpdate (old/new = n*d f32 each).
+int sdbv_index_enqueue(sdbv_index *ix, uint64_t record_key, const float *olds,
+                       uint32_t n_old, const float *news, uint32_t n_new) {
+	if (!ix)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	sdbv_index::Pending p;
+	auto it = ix->key2doc.find(record_key);
+	if (it != ix->key2doc.end()) {
+		p.kind = 0;
+		p.id = it->second;
+	} else {
+		p.kind = 1;
+		p.id = record_key;
+	}
+	uint32_t d = ix->h->d;
+	p.olds.assign(olds, olds + (size_t)n_old * d);
+	p.news.assign(news, news + (size_t)n_new * d);
+	ix->pendings.push_back(std::move(p));
+	return SDBV_OK;
+}
+
+// index_pendings + index_pending (index.rs:188-257): drain in appending
+// order; old-vector removals only for DocId pendings; empty new_vectors
+// deletes the doc mapping; RecordKey ids resolve (allocate) at apply time.
+int sdbv_index_apply_pendings(sdbv_index *ix, uint64_t *out_count) {
+	if (!ix)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	uint64_t count = 0;
+	uint32_t d = ix->h->d;
+	for (auto &p : ix->pendings) {
+		if (p.kind == 0) {
+			for (size_t i = 0; i * d < p.olds.size(); i++)
+				idx_vd_remove(ix, p.olds.data() + i * d, p.id);
+			if (p.news.empty())
+				idx_docs_remove(ix, p.id);
+		}
+		if (!p.news.empty()) {
+			uint64_t doc_id =
+			    (p.kind == 0) ? p.id : idx_docs_resolve(ix, p.id);
+			for (size_t i = 0; i * d < p.news.size(); i++)
+				idx_vd_insert(ix, p.news.data() + i * d, doc_id);
+		}
+		count++;
+	}
+	ix->pendings.clear();
+	if (out_count)
+		*out_count = count;
+	return SDBV_OK;
+}
+
+// knn_search (index.rs:270-335 without record materialisation):
+// search_pendings overlay + graph search (GPU per-hop once finalized,
+// auto-refinalizing after writes) + VecDocs doc expansion through one
+// KnnResultBuilder. Out arrays sized k; *out_n = entries returned. Entries
+// ascending (dist total_cmp, VectorId); kind 0 = DocId, 1 = RecordKey.
+int sdbv_index_knn(sdbv_index *ix, const float *q, uint32_t k, uint32_t ef,
+                   uint8_t *out_kinds, uint64_t *out_ids, double *out_dists,
+                   uint32_t *out_n) {
+	if (!ix || !q || k == 0)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	sdbv_hnsw *h = ix->h;
+	vdocs::Builder builder(k);
+	// search_pendings (index.rs:366-421)
+	std::set<uint64_t> all_existing;
+	std::map<vdocs::Vid, const std::vector<float> *> non_deleted;
+	for (auto &p : ix->pendings) {
+		if (p.kind == 0)
+			all_existing.insert(p.id);
+		vdocs::Vid vid{p.kind, p.id};
+		if (p.news.empty())
+			non_deleted.erase(vid);
+		else
+			non_deleted[vid] = &p.news;
+	}
+	if (!(all_existing.empty() && non_deleted.empty())) {
+		double q_norm = h->metric == SDBV_METRIC_COSINE
+		                    ? sqrt(hnsw::host_sumsq_f32(q, h->d))
+		                    : 0;
+		for (auto &e : non_deleted) {
+			const std::vector<float> &vecs = *e.second;
+			for (size_t i = 0; i * h->d < vecs.size(); i++) {
+				double dd =
+				    idx_dist_raw(h, q, q_norm, vecs.data() + i * h->d);
+				if (builder.check_add(dd))
+					builder.add(dd, e.first);
+			}
+		}
+	}
+	hnsw::IdxPend pend{&all_existing, ix};
+	const hnsw::IdxPend *pp = all_existing.empty() ? nullptr : &pend;
+	// graph search: GPU per-hop when a device context exists (finalizing
+	// on first use / after writes), host path otherwise
+	std::vector<std::pair<double, uint32_t>> neighbors;
+	if (h->ctx && h->next_id > 0) {
+		if (h->dirty || !h->finalized) {
+			int rc = sdbv_hnsw_finalize(h, ix->table);
+			if (rc)
+				return rc;
+		}
+		uint32_t ng = 0;
+		int rc = idx_graph_search_gpu(ix, q, k, ef, pp, neighbors, &ng);
+		if (rc)
+			return rc;
+	} else {
+		idx_graph_search_host(ix, q, k, ef, pp, neighbors);
+	}
+	// add_graph_results (index.rs:454-483)
+	for (auto &nb : neighbors) {
+		if (!builder.check_add(nb.first))
+			continue;
+		auto it = ix->by_elem.find(nb.second);
+		if (it == ix->by_elem.end())
+			continue; // get_vector/get_docs -> None
+		builder.add_graph(nb.first, ix->vd.at(*it->second).docs);
+	}
+	uint32_t n = 0;
+	for (const auto &e : builder.pl) {
+		out_kinds[n] = e.vid.kind;
+		out_ids[n] = e.vid.id;
+		out_dists[n] = e.dist;
+		n++;
+	}
+	*out_n = n;
+	return SDBV_OK;
+}
+
+// check_hnsw_properties (mod.rs:561-570 + the reference index tests'
+// element count): present elements == expected == Hv entries, layer edge
+// invariants (counts, no self-edges, in-range targets).
+int sdbv_index_check_props(sdbv_index *ix, uint64_t expected_count) {
+	if (!ix)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	sdbv_hnsw *h = ix->h;
+	uint64_t present = 0;
+	for (uint8_t p : h->elem_present)
+		present += p;
+	if (present != expected_count)
+		return -10;
+	if (present != ix->vd.size())
+		return -11;
+	for (size_t li = 0; li < h->layers.size(); li++) {
+		auto &layer = h->layers[li];
+		for (size_t id = 0; id < layer.edges.size(); id++) {
+			if (!layer.has((uint32_t)id)) {
+				if (!layer.edges[id].empty()) {
+					if (getenv("SDBV_DEBUG"))
+						fprintf(stderr,
+						        "check_props: layer %zu node %zu not "
+						        "member, %zu edges (present=%d)\n",
+						        li, id, layer.edges[id].size(),
+						        id < h->elem_present.size()
+						            ? (int)h->elem_present[id]
+						            : -1);
+					return -12;
+				}
+				continue;
+			}
+			if (layer.edges[id].size() > layer.m_max)
+				return -13;
+			for (uint32_t e : layer.edges[id])
+				if (e == id || e >= h->next_id)
+					return -14;
+		}
+	}
+	return 0;
 }
 
 } // extern "C"
