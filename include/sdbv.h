@@ -214,6 +214,21 @@ int sdbv_index_apply_pendings(sdbv_index *, uint64_t *out_count);
 int sdbv_index_knn(sdbv_index *, const float *q, uint32_t k, uint32_t ef,
                    uint8_t *out_kinds, uint64_t *out_ids, double *out_dists,
                    uint32_t *out_n);
+/* Filtered KNN (cond_filter pushdown — hnsw/filter.rs + layer.rs:110-318):
+ * the host evaluates the WHERE condition per VectorId (is_record_truthy,
+ * filter.rs:111-138 — KV fetch + expression compute stay host-side); the
+ * library keeps the FilterCache semantics (one evaluation per id while
+ * cached; `expire` fires when the result builder evicts an id,
+ * filter.rs:141-151) and all accept/expand gating (add_if_truthy,
+ * layer.rs:278-306). `truthy` must be deterministic within one call;
+ * `expire` may be NULL. */
+typedef int (*sdbv_truthy_cb)(void *user, uint8_t kind, uint64_t id);
+typedef void (*sdbv_expire_cb)(void *user, uint8_t kind, uint64_t id);
+int sdbv_index_knn_filtered(sdbv_index *, const float *q, uint32_t k,
+                            uint32_t ef, sdbv_truthy_cb truthy,
+                            sdbv_expire_cb expire, void *user,
+                            uint8_t *out_kinds, uint64_t *out_ids,
+                            double *out_dists, uint32_t *out_n);
 uint64_t sdbv_index_doc_count(sdbv_index *);
 uint64_t sdbv_index_pending_count(sdbv_index *);
 /* check_hnsw_properties (hnsw/mod.rs:561-570): 0 == OK. */

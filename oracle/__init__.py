@@ -12,6 +12,9 @@ import numpy as np
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "liborcl.so")
 
+TRUTHY_CB = None  # ctypes callback types, set when lib() loads
+EXPIRE_CB = None
+
 METRICS = {
     "cosine": 0,
     "euclidean": 1,
@@ -111,6 +114,13 @@ def _load():
     lib.orc_index_knn.restype = u32
     lib.orc_index_knn.argtypes = [ctypes.c_void_p, f32p, u32, u32, u8p, u64p,
                                   f64p]
+    global TRUTHY_CB, EXPIRE_CB
+    TRUTHY_CB = ctypes.CFUNCTYPE(ctypes.c_int, ctypes.c_void_p, u8, u64)
+    EXPIRE_CB = ctypes.CFUNCTYPE(None, ctypes.c_void_p, u8, u64)
+    lib.orc_index_knn_filtered.restype = u32
+    lib.orc_index_knn_filtered.argtypes = [ctypes.c_void_p, f32p, u32, u32,
+                                           TRUTHY_CB, EXPIRE_CB,
+                                           ctypes.c_void_p, u8p, u64p, f64p]
     lib.orc_index_check_props.restype = ctypes.c_int
     lib.orc_index_check_props.argtypes = [ctypes.c_void_p, u64]
     lib.orc_ids64_new.restype = ctypes.c_void_p
@@ -360,6 +370,24 @@ class Index:
         dists = np.empty(k, dtype=np.float64)
         n = lib().orc_index_knn(
             self._ix, _f32p(q), k, ef,
+            kinds.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        return kinds[:n], ids[:n], dists[:n]
+
+    def knn_search_filtered(self, q, k, ef, truthy, expire=None):
+        """Filtered knn (index.rs:270-335 with cond_filter): truthy(kind,
+        id) -> bool is the host WHERE evaluation; expire mirrors the
+        filter-cache eviction signal."""
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        kinds = np.empty(k, dtype=np.uint8)
+        ids = np.empty(k, dtype=np.uint64)
+        dists = np.empty(k, dtype=np.float64)
+        cb = TRUTHY_CB(lambda u, kind, i: 1 if truthy(kind, i) else 0)
+        ex = EXPIRE_CB((lambda u, kind, i: expire(kind, i)) if expire
+                       else (lambda u, kind, i: None))
+        n = lib().orc_index_knn_filtered(
+            self._ix, _f32p(q), k, ef, cb, ex, None,
             kinds.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
             ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
             dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))

@@ -1491,6 +1491,221 @@ uint32_t orc_index_knn(orc_index *ix, const float *q, uint32_t k, uint32_t ef,
 	return n;
 }
 
+// ---------------------------------------------------------------------------
+// Filtered KNN (hnsw/filter.rs + layer.rs:110-318 + the index.rs filtered
+// flow). The WHERE-condition evaluation itself (is_record_truthy:
+// KV record fetch + expression compute, filter.rs:111-138) stays on the
+// host side of the boundary as a callback; the library keeps the
+// reference's FilterCache semantics (one evaluation per VectorId while
+// cached, entries expired when the builder evicts the id) and all the
+// accept/expand gating. The callback must be deterministic within one call.
+// ---------------------------------------------------------------------------
+
+typedef int (*orc_truthy_cb)(void *user, uint8_t kind, uint64_t id);
+typedef void (*orc_expire_cb)(void *user, uint8_t kind, uint64_t id);
+
+struct OrcFilter {
+	orc_truthy_cb cb;
+	orc_expire_cb ex;
+	void *user;
+	// FilterCache (filter.rs:22): VectorId -> truthy
+	std::map<std::pair<uint8_t, uint64_t>, bool> cache;
+	bool truthy(uint8_t kind, uint64_t id) {
+		auto key = std::make_pair(kind, id);
+		auto it = cache.find(key);
+		if (it != cache.end())
+			return it->second; // filter.rs:77-78 cached
+		bool t = cb(user, kind, id) != 0;
+		cache[key] = t;
+		return t;
+	}
+	void expire(uint8_t kind, uint64_t id) { // filter.rs:141-144
+		cache.erase({kind, id});
+		if (ex)
+			ex(user, kind, id);
+	}
+	// filter.rs:53-66 check_any_doc_truthy (Ids64 iteration order)
+	bool any_doc_truthy(const OrcIds64 &docs) {
+		for (uint64_t d : docs.v)
+			if (truthy(0, d))
+				return true;
+		return false;
+	}
+};
+
+// layer.rs:308-318 check_all_docs_in_pending (the plain Ids64 variant used
+// by add_if_truthy — distinct from are_all_docs_in_pending).
+static bool check_all_docs_in_pending(const OrcIds64 &docs,
+                                      const std::set<uint64_t> *pending) {
+	if (!pending || pending->empty())
+		return false;
+	for (uint64_t d : docs.v)
+		if (!pending->count(d))
+			return false;
+	return true;
+}
+
+// layer.rs:278-306 add_if_truthy: w gets the element only if its vector has
+// a VecDocs entry, the docs are not all-pending, and >=1 doc is truthy.
+// Looks docs up BY VECTOR (vec_docs.get_docs(e_pt)).
+static bool add_if_truthy(orc_index *ix, uint32_t ef, OrcPQ &w,
+                          const float *e_pt, double e_dist, uint64_t e_id,
+                          OrcFilter &filter,
+                          const std::set<uint64_t> *pending) {
+	auto it = ix->vd.find(vec_key(ix, e_pt));
+	if (it == ix->vd.end())
+		return false;
+	const OrcIds64 &docs = it->second.docs;
+	if (check_all_docs_in_pending(docs, pending))
+		return false;
+	if (filter.any_doc_truthy(docs)) {
+		w.push(e_dist, e_id);
+		if (w.n > ef) {
+			double dd;
+			uint64_t ii;
+			w.pop_last(&dd, &ii);
+		}
+		return true;
+	}
+	return false;
+}
+
+// layer.rs:226-275 search_with_filter: candidates expand unconditionally
+// (within the distance gate); w is gated by add_if_truthy.
+static void layer_search_with_filter(orc_index *ix, const OrcLayer &layer,
+                                     const float *q, OrcPQ &candidates,
+                                     std::unordered_set<uint64_t> &visited,
+                                     OrcPQ &w, uint32_t ef,
+                                     OrcFilter &filter,
+                                     const std::set<uint64_t> *pending) {
+	orc_hnsw *h = ix->h;
+	double f_dist = w.peek_last_dist(1.7976931348623157e308);
+	double cq_dist;
+	uint64_t doc;
+	while (candidates.pop_first(&cq_dist, &doc)) {
+		if (cq_dist > f_dist)
+			break;
+		if (!layer.has(doc))
+			continue;
+		for (uint32_t e_id : layer.edges[doc]) {
+			if (!visited.insert(e_id).second)
+				continue;
+			if (e_id < h->elem_present.size() && !h->elem_present[e_id])
+				continue; // get_vector -> None
+			double e_dist = hdist(h, hvec(h, e_id), q);
+			if (e_dist < f_dist || w.n < ef) {
+				candidates.push(e_dist, e_id);
+				if (add_if_truthy(ix, ef, w, hvec(h, e_id), e_dist, e_id,
+				                  filter, pending))
+					f_dist = w.peek_last_dist(1.7976931348623157e308);
+			}
+		}
+	}
+}
+
+// HnswIndex::knn_search with cond_filter (index.rs:270-335 +
+// knn_search_with_filter mod.rs:484-515 + search_single_with_filter
+// layer.rs:110-149). Returns entry count (<= k).
+uint32_t orc_index_knn_filtered(orc_index *ix, const float *q, uint32_t k,
+                                uint32_t ef, orc_truthy_cb truthy,
+                                orc_expire_cb expire, void *user,
+                                uint8_t *out_kinds, uint64_t *out_ids,
+                                double *out_dists) {
+	orc_hnsw *h = ix->h;
+	uint32_t d = h->d;
+	OrcBuilder builder(k);
+	OrcFilter filter{truthy, expire, user, {}};
+	// search_pendings with the filter (index.rs:400-404): non-truthy
+	// pending ids are skipped before their vectors are scored
+	std::set<uint64_t> all_existing;
+	std::map<OrcVid, const std::vector<float> *> non_deleted;
+	for (auto &p : ix->pendings) {
+		if (p.kind == 0)
+			all_existing.insert(p.id);
+		OrcVid vid{p.kind, p.id};
+		if (p.news.empty())
+			non_deleted.erase(vid);
+		else
+			non_deleted[vid] = &p.news;
+	}
+	if (!(all_existing.empty() && non_deleted.empty())) {
+		for (auto &e : non_deleted) {
+			if (!filter.truthy(e.first.kind, e.first.id))
+				continue;
+			const std::vector<float> &vecs = *e.second;
+			for (size_t i = 0; i * d < vecs.size(); i++) {
+				double dd = orc_dist_f32(h->metric, h->order, q,
+				                         vecs.data() + i * d, d);
+				if (builder.check_add(dd)) {
+					OrcVid ev;
+					if (builder.add(dd, e.first, &ev))
+						filter.expire(ev.kind, ev.id);
+				}
+			}
+		}
+	}
+	const std::set<uint64_t> *pp =
+	    all_existing.empty() ? nullptr : &all_existing;
+	OrcPend pend{&all_existing, ix, &orc_all_docs_pending};
+	const OrcPend *ep_pend = all_existing.empty() ? nullptr : &pend;
+	// knn_search_with_filter (mod.rs:484-515)
+	std::vector<std::pair<double, uint64_t>> neighbors;
+	if (h->enter_point >= 0) {
+		// search_ep: plain upper-layer descent with pending (mod.rs:520-548)
+		uint64_t ep_id = (uint64_t)h->enter_point;
+		double ep_dist = hdist(h, q, hvec(h, ep_id));
+		for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+			OrcPQ cand;
+			cand.push(ep_dist, ep_id);
+			std::unordered_set<uint64_t> visited{ep_id};
+			OrcPQ w = cand;
+			layer_search(h, h->layers[l], q, cand, visited, w, 1, ep_pend);
+			double dd;
+			uint64_t ii;
+			if (w.peek_first(&dd, &ii)) {
+				ep_dist = dd;
+				ep_id = ii;
+			}
+		}
+		// search_single_with_filter (layer.rs:110-149). NOTE the seed
+		// passes SEARCH.PT as e_pt (layer.rs:125-135): the entry point
+		// enters w only if the QUERY VECTOR itself has a VecDocs entry
+		// with a truthy, not-all-pending doc — restated as-is.
+		OrcPQ candidates;
+		candidates.push(ep_dist, ep_id);
+		std::unordered_set<uint64_t> visited{ep_id};
+		OrcPQ w;
+		add_if_truthy(ix, ef, w, q, ep_dist, ep_id, filter, pp);
+		layer_search_with_filter(ix, h->layers[0], q, candidates, visited,
+		                         w, ef, filter, pp);
+		auto v = w.to_vec();
+		size_t m = std::min<size_t>(k, v.size());
+		for (size_t i = 0; i < m; i++)
+			neighbors.push_back({v[i].first, v[i].second});
+	}
+	// add_graph_results with eviction expiry (index.rs:353-359, :454-483)
+	for (auto &nb : neighbors) {
+		if (!builder.check_add(nb.first))
+			continue;
+		auto it = ix->by_elem.find(nb.second);
+		if (it == ix->by_elem.end())
+			continue;
+		for (uint64_t docid : ix->vd.at(*it->second).docs.v) {
+			OrcVid ev;
+			if (builder.add(nb.first, OrcVid{0, docid}, &ev))
+				filter.expire(ev.kind, ev.id);
+		}
+	}
+	uint32_t n = 0;
+	for (const auto &e : builder.pl) {
+		out_kinds[n] = e.vid.kind;
+		out_ids[n] = e.vid.id;
+		out_dists[n] = e.dist;
+		n++;
+	}
+	return n;
+}
+
 // Test hook: drive OrcIds64 directly so tests can restate the reference's
 // own test_ids sequence (knn.rs:669-717) — variant transitions (Some/None)
 // and contents/order, bit for bit.

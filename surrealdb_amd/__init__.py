@@ -16,6 +16,9 @@ METRIC_COSINE = 0
 METRIC_EUCLIDEAN = 1
 METRICS = {"cosine": METRIC_COSINE, "euclidean": METRIC_EUCLIDEAN}
 
+TRUTHY_CB = None  # ctypes callback types, set when lib() loads
+EXPIRE_CB = None
+
 _ERRS = {
     0: "OK",
     -1: "HIP runtime error",
@@ -101,6 +104,12 @@ def lib():
     L.sdbv_index_enqueue.argtypes = [vp, u64, f32p, u32, f32p, u32]
     L.sdbv_index_apply_pendings.argtypes = [vp, u64p]
     L.sdbv_index_knn.argtypes = [vp, f32p, u32, u32, u8p, u64p, f64p, u32p]
+    global TRUTHY_CB, EXPIRE_CB
+    TRUTHY_CB = ctypes.CFUNCTYPE(ctypes.c_int, vp, u8, u64)
+    EXPIRE_CB = ctypes.CFUNCTYPE(None, vp, u8, u64)
+    L.sdbv_index_knn_filtered.argtypes = [vp, f32p, u32, u32, TRUTHY_CB,
+                                          EXPIRE_CB, vp, u8p, u64p, f64p,
+                                          u32p]
     L.sdbv_index_doc_count.restype = u64
     L.sdbv_index_doc_count.argtypes = [vp]
     L.sdbv_index_pending_count.restype = u64
@@ -424,6 +433,31 @@ class Index:
             ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
             dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
             ctypes.byref(out_n)), "sdbv_index_knn")
+        n = out_n.value
+        return kinds[:n], ids[:n], dists[:n]
+
+    def knn_search_filtered(self, q, k, ef, truthy, expire=None):
+        """Filtered knn (cond_filter pushdown): `truthy(kind, id) -> bool`
+        is the host-side WHERE evaluation (filter.rs is_record_truthy);
+        `expire(kind, id)` mirrors the filter-cache eviction signal."""
+        import numpy as np
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        kinds = np.empty(k, dtype=np.uint8)
+        ids = np.empty(k, dtype=np.uint64)
+        dists = np.empty(k, dtype=np.float64)
+        out_n = ctypes.c_uint32(0)
+        cb = TRUTHY_CB(lambda u, kind, i: 1 if truthy(kind, i) else 0)
+        ex = EXPIRE_CB((lambda u, kind, i: expire(kind, i)) if expire
+                       else (lambda u, kind, i: None))
+        _check(self._ctx._ptr if self._ctx else None,
+               lib().sdbv_index_knn_filtered(
+                   self._ptr,
+                   q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), k, ef,
+                   cb, ex, None,
+                   kinds.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+                   ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+                   dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+                   ctypes.byref(out_n)), "sdbv_index_knn_filtered")
         n = out_n.value
         return kinds[:n], ids[:n], dists[:n]
 

@@ -2857,22 +2857,29 @@ struct Builder {
 			return false;
 		return true;
 	}
-	void add(double dist, Vid vid) {
+	// knn.rs:410-433: returns true + *evicted when an id fell out of the
+	// result entirely (count hit 0) — the filter-cache expiry signal.
+	bool add_ret_evicted(double dist, Vid vid, Vid *evicted) {
 		pl.insert(Ent{hnsw::total_key(dist), vid, dist});
 		count[vid]++; // incremented even on duplicate set inserts
 		if (pl.size() <= knn)
-			return;
+			return false;
 		auto last = std::prev(pl.end());
 		Vid ev = last->vid;
 		pl.erase(last);
 		auto it = count.find(ev);
 		if (it != count.end()) {
-			if (it->second <= 1)
+			if (it->second <= 1) {
 				count.erase(it);
-			else
-				it->second--;
+				if (evicted)
+					*evicted = ev;
+				return true;
+			}
+			it->second--;
 		}
+		return false;
 	}
+	void add(double dist, Vid vid) { add_ret_evicted(dist, vid, nullptr); }
 	void add_graph(double dist, const Ids64 &docs) {
 		for (uint64_t doc : docs.v)
 			add(dist, Vid{0, doc});
@@ -3160,6 +3167,185 @@ static int idx_graph_search_gpu(sdbv_index *ix, const float *q, uint32_t k,
 	return SDBV_OK;
 }
 
+// ---------------------------------------------------------------------------
+// Filtered KNN (hnsw/filter.rs + layer.rs:110-318): the WHERE-condition
+// evaluation (is_record_truthy — KV fetch + expression compute) stays on
+// the host side as a callback; the library keeps the FilterCache semantics
+// (one evaluation per VectorId while cached, expired on builder eviction)
+// and the accept/expand gating. Callback must be deterministic per call.
+// ---------------------------------------------------------------------------
+
+typedef int (*sdbv_truthy_cb)(void *user, uint8_t kind, uint64_t id);
+typedef void (*sdbv_expire_cb)(void *user, uint8_t kind, uint64_t id);
+
+namespace vdocs {
+
+struct Filter {
+	sdbv_truthy_cb cb;
+	sdbv_expire_cb ex;
+	void *user;
+	std::map<std::pair<uint8_t, uint64_t>, bool> cache; // filter.rs:22
+	bool truthy(uint8_t kind, uint64_t id) {
+		auto key = std::make_pair(kind, id);
+		auto it = cache.find(key);
+		if (it != cache.end())
+			return it->second;
+		bool t = cb(user, kind, id) != 0;
+		cache[key] = t;
+		return t;
+	}
+	void expire(uint8_t kind, uint64_t id) { // filter.rs:141-144
+		cache.erase({kind, id});
+		if (ex)
+			ex(user, kind, id);
+	}
+	bool any_doc_truthy(const Ids64 &docs) { // filter.rs:53-66
+		for (uint64_t d : docs.v)
+			if (truthy(0, d))
+				return true;
+		return false;
+	}
+};
+
+} // namespace vdocs
+
+// layer.rs:308-318 check_all_docs_in_pending (plain Ids64 variant).
+static bool idx_check_all_docs_in_pending(const vdocs::Ids64 &docs,
+                                          const std::set<uint64_t> *pending) {
+	if (!pending || pending->empty())
+		return false;
+	for (uint64_t d : docs.v)
+		if (!pending->count(d))
+			return false;
+	return true;
+}
+
+// layer.rs:278-306 add_if_truthy: docs looked up BY VECTOR.
+static bool idx_add_if_truthy(sdbv_index *ix, uint32_t ef, hnsw::PQ &w,
+                              const float *e_pt, double e_dist, uint32_t e_id,
+                              vdocs::Filter &filter,
+                              const std::set<uint64_t> *pending) {
+	std::string key((const char *)e_pt, (size_t)ix->h->d * 4);
+	auto it = ix->vd.find(key);
+	if (it == ix->vd.end())
+		return false;
+	const vdocs::Ids64 &docs = it->second.docs;
+	if (idx_check_all_docs_in_pending(docs, pending))
+		return false;
+	if (filter.any_doc_truthy(docs)) {
+		w.push(e_dist, e_id);
+		if (w.n > ef)
+			w.pop_last();
+		return true;
+	}
+	return false;
+}
+
+// layer.rs:226-275 search_with_filter, host distances (host-only index).
+static void idx_search_l0_filter_host(sdbv_index *ix, const float *q,
+                                      double q_norm, hnsw::PQ &candidates,
+                                      std::vector<bool> &visited, hnsw::PQ &w,
+                                      uint32_t ef, vdocs::Filter &filter,
+                                      const std::set<uint64_t> *pending) {
+	using namespace hnsw;
+	sdbv_hnsw *h = ix->h;
+	const Layer &l0 = h->layers[0];
+	double f_dist = w.peek_last_dist(DBL_MAX);
+	double cd;
+	uint32_t doc;
+	while (candidates.pop_first(&cd, &doc)) {
+		if (cd > f_dist)
+			break;
+		for (uint32_t e : l0.edges[doc]) {
+			if (visited[e])
+				continue;
+			visited[e] = true;
+			if (!h->elem_present[e])
+				continue;
+			double ed = dist(h, q, q_norm, e);
+			if (ed < f_dist || w.n < ef) {
+				candidates.push(ed, e);
+				if (idx_add_if_truthy(ix, ef, w, vec(h, e), ed, e, filter,
+				                      pending))
+					f_dist = w.peek_last_dist(DBL_MAX);
+			}
+		}
+	}
+}
+
+// Same loop with the per-hop GPU gather for distances (finalized index).
+// Caller holds ix->mu; takes ctx->mu itself.
+static int idx_search_l0_filter_gpu(sdbv_index *ix, const float *q,
+                                    double q_norm, hnsw::PQ &candidates,
+                                    std::vector<bool> &visited, hnsw::PQ &w,
+                                    uint32_t ef, vdocs::Filter &filter,
+                                    const std::set<uint64_t> *pending) {
+	using namespace hnsw;
+	sdbv_hnsw *h = ix->h;
+	sdbv_ctx *ctx = h->ctx;
+	std::lock_guard<std::mutex> lk(ctx->mu);
+	auto it = ctx->tables.find(h->table);
+	if (it == ctx->tables.end())
+		return SDBV_ERR_NO_TABLE;
+	Table &t = it->second;
+	HIP_CHECK(ctx, hipMemcpyAsync(h->q_dev, q, h->d * sizeof(float),
+	                              hipMemcpyHostToDevice, ctx->stream));
+	const Layer &l0 = h->layers[0];
+	double f_dist = w.peek_last_dist(DBL_MAX);
+	std::vector<uint32_t> frontier;
+	std::vector<double> fdists(h->m0 + 1);
+	double cd;
+	uint32_t doc;
+	while (candidates.pop_first(&cd, &doc)) {
+		if (cd > f_dist)
+			break;
+		frontier.clear();
+		for (uint32_t e : l0.edges[doc])
+			if (!visited[e]) {
+				visited[e] = true;
+				if (!h->elem_present[e])
+					continue;
+				frontier.push_back(e);
+			}
+		if (frontier.empty())
+			continue;
+		std::memcpy(h->rows_pinned, frontier.data(),
+		            frontier.size() * sizeof(uint32_t));
+		HIP_CHECK(ctx, hipMemcpyAsync(h->rows_dev, h->rows_pinned,
+		                              frontier.size() * sizeof(uint32_t),
+		                              hipMemcpyHostToDevice, ctx->stream));
+		if (t.metric == SDBV_METRIC_COSINE)
+			hipLaunchKernelGGL(k_gather_dist<0>,
+			                   dim3((uint32_t)frontier.size()), dim3(64), 0,
+			                   ctx->stream, t.cm, t.norms, t.n_pad, t.d,
+			                   h->rows_dev, (uint32_t)frontier.size(),
+			                   h->q_dev, q_norm, h->dout_dev);
+		else
+			hipLaunchKernelGGL(k_gather_dist<1>,
+			                   dim3((uint32_t)frontier.size()), dim3(64), 0,
+			                   ctx->stream, t.cm, t.norms, t.n_pad, t.d,
+			                   h->rows_dev, (uint32_t)frontier.size(),
+			                   h->q_dev, q_norm, h->dout_dev);
+		HIP_CHECK(ctx, hipMemcpyAsync(h->dists_pinned, h->dout_dev,
+		                              frontier.size() * sizeof(double),
+		                              hipMemcpyDeviceToHost, ctx->stream));
+		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		std::memcpy(fdists.data(), h->dists_pinned,
+		            frontier.size() * sizeof(double));
+		for (size_t i = 0; i < frontier.size(); i++) {
+			double ed = fdists[i];
+			uint32_t e = frontier[i];
+			if (ed < f_dist || w.n < ef) {
+				candidates.push(ed, e);
+				if (idx_add_if_truthy(ix, ef, w, vec(h, e), ed, e, filter,
+				                      pending))
+					f_dist = w.peek_last_dist(DBL_MAX);
+			}
+		}
+	}
+	return SDBV_OK;
+}
+
 extern "C" {
 
 int sdbv_index_create(sdbv_ctx *ctx, uint64_t table, uint32_t d,
@@ -3309,6 +3495,133 @@ int sdbv_index_knn(sdbv_index *ix, const float *q, uint32_t k, uint32_t ef,
 		if (it == ix->by_elem.end())
 			continue; // get_vector/get_docs -> None
 		builder.add_graph(nb.first, ix->vd.at(*it->second).docs);
+	}
+	uint32_t n = 0;
+	for (const auto &e : builder.pl) {
+		out_kinds[n] = e.vid.kind;
+		out_ids[n] = e.vid.id;
+		out_dists[n] = e.dist;
+		n++;
+	}
+	*out_n = n;
+	return SDBV_OK;
+}
+
+// knn_search with cond_filter (index.rs:270-335 + knn_search_with_filter
+// mod.rs:484-515 + search_single_with_filter layer.rs:110-149): same flow
+// as sdbv_index_knn but candidates expand unconditionally and w is gated
+// by add_if_truthy (>=1 truthy doc, not all-pending). NOTE the reference
+// seeds add_if_truthy with SEARCH.PT as the entry point's vector
+// (layer.rs:125-135) — the entry point enters w only if the QUERY VECTOR
+// itself is an indexed vector with a truthy doc; restated as-is.
+int sdbv_index_knn_filtered(sdbv_index *ix, const float *q, uint32_t k,
+                            uint32_t ef, sdbv_truthy_cb truthy,
+                            sdbv_expire_cb expire, void *user,
+                            uint8_t *out_kinds, uint64_t *out_ids,
+                            double *out_dists, uint32_t *out_n) {
+	using namespace hnsw;
+	if (!ix || !q || k == 0 || !truthy)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	sdbv_hnsw *h = ix->h;
+	vdocs::Builder builder(k);
+	vdocs::Filter filter{truthy, expire, user, {}};
+	// search_pendings with the filter (index.rs:400-421)
+	std::set<uint64_t> all_existing;
+	std::map<vdocs::Vid, const std::vector<float> *> non_deleted;
+	for (auto &p : ix->pendings) {
+		if (p.kind == 0)
+			all_existing.insert(p.id);
+		vdocs::Vid vid{p.kind, p.id};
+		if (p.news.empty())
+			non_deleted.erase(vid);
+		else
+			non_deleted[vid] = &p.news;
+	}
+	if (!(all_existing.empty() && non_deleted.empty())) {
+		double q_norm = h->metric == SDBV_METRIC_COSINE
+		                    ? sqrt(host_sumsq_f32(q, h->d))
+		                    : 0;
+		for (auto &e : non_deleted) {
+			if (!filter.truthy(e.first.kind, e.first.id))
+				continue;
+			const std::vector<float> &vecs = *e.second;
+			for (size_t i = 0; i * h->d < vecs.size(); i++) {
+				double dd =
+				    idx_dist_raw(h, q, q_norm, vecs.data() + i * h->d);
+				if (builder.check_add(dd)) {
+					vdocs::Vid ev;
+					if (builder.add_ret_evicted(dd, e.first, &ev))
+						filter.expire(ev.kind, ev.id);
+				}
+			}
+		}
+	}
+	const std::set<uint64_t> *pp =
+	    all_existing.empty() ? nullptr : &all_existing;
+	IdxPend pend{&all_existing, ix};
+	const IdxPend *ep_pend = all_existing.empty() ? nullptr : &pend;
+	// knn_search_with_filter (mod.rs:484-515)
+	std::vector<std::pair<double, uint32_t>> neighbors;
+	if (h->next_id > 0 && h->enter_point >= 0) {
+		bool use_gpu = h->ctx != nullptr;
+		if (use_gpu && (h->dirty || !h->finalized)) {
+			int rc = sdbv_hnsw_finalize(h, ix->table);
+			if (rc)
+				return rc;
+		}
+		double q_norm = h->metric == SDBV_METRIC_COSINE
+		                    ? sqrt(host_sumsq_f32(q, h->d))
+		                    : 0;
+		// search_ep: plain upper-layer descent with pending (mod.rs:520-548)
+		uint32_t ep_id = (uint32_t)h->enter_point;
+		double ep_dist = dist(h, q, q_norm, ep_id);
+		for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+			PQ cand;
+			cand.push(ep_dist, ep_id);
+			std::unordered_set<uint32_t> visited{ep_id};
+			PQ w = cand;
+			search_layer_host(h, h->layers[l], q, q_norm, cand, visited, w,
+			                  1, false, ep_pend);
+			double dd;
+			uint32_t ii;
+			if (w.peek_first(&dd, &ii)) {
+				ep_dist = dd;
+				ep_id = ii;
+			}
+		}
+		// search_single_with_filter (layer.rs:110-149; the search.pt seed)
+		PQ candidates, w;
+		candidates.push(ep_dist, ep_id);
+		std::vector<bool> visited(h->next_id, false);
+		visited[ep_id] = true;
+		idx_add_if_truthy(ix, ef, w, q, ep_dist, ep_id, filter, pp);
+		if (use_gpu) {
+			int rc = idx_search_l0_filter_gpu(ix, q, q_norm, candidates,
+			                                  visited, w, ef, filter, pp);
+			if (rc)
+				return rc;
+		} else {
+			idx_search_l0_filter_host(ix, q, q_norm, candidates, visited, w,
+			                          ef, filter, pp);
+		}
+		auto v = w.to_vec();
+		size_t m = std::min<size_t>(k, v.size());
+		for (size_t i = 0; i < m; i++)
+			neighbors.push_back({v[i].first, v[i].second});
+	}
+	// add_graph_results with eviction expiry (index.rs:353-359, :454-483)
+	for (auto &nb : neighbors) {
+		if (!builder.check_add(nb.first))
+			continue;
+		auto it = ix->by_elem.find(nb.second);
+		if (it == ix->by_elem.end())
+			continue;
+		for (uint64_t docid : ix->vd.at(*it->second).docs.v) {
+			vdocs::Vid ev;
+			if (builder.add_ret_evicted(nb.first, vdocs::Vid{0, docid}, &ev))
+				filter.expire(ev.kind, ev.id);
+		}
 	}
 	uint32_t n = 0;
 	for (const auto &e : builder.pl) {

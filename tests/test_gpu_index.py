@@ -59,3 +59,25 @@ def test_gpu_index_matches_oracle_through_writes(ctx, metric):
     assert p.doc_count() == o.doc_count() == n - 100
     p.destroy()
     ctx.drop_table(40)
+
+
+def test_gpu_index_filtered_matches_oracle(ctx):
+    d, n = 64, 2000
+    rows = oracle.gen_f32(0xB1, 0, n, d)
+    p = sa.Index(ctx, 41, d, metric="cosine", m=8, m0=16, efc=60, seed=0x3)
+    o = oracle.Index(d, metric="cosine", m=8, m0=16, efc=60, seed=0x3)
+    for i, r in enumerate(rows):
+        p.enqueue(i, None, r)
+        o.enqueue(i, None, r)
+    p.apply_pendings()
+    o.apply_pendings()
+    pred = lambda kind, i: i % 5 != 0
+    for j, q in enumerate(oracle.gen_f32(0xBEEF, 0, 8, d)):
+        pk, pi, pd = p.knn_search_filtered(q, 10, 40, pred)
+        ok, oi, od = o.knn_search_filtered(q, 10, 40, pred)
+        assert np.array_equal(pk, ok), f"q{j}: kinds"
+        assert np.array_equal(pi, oi), f"q{j}: ids"
+        assert np.array_equal(pd, od), f"q{j}: dist bits"
+        assert all(int(i) % 5 != 0 for i in pi)
+    p.destroy()
+    ctx.drop_table(41)
