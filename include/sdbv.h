@@ -236,6 +236,39 @@ int sdbv_index_check_props(sdbv_index *, uint64_t expected_count);
 /* Test access to the underlying graph (CSR parity vs the oracle). */
 sdbv_hnsw *sdbv_index_hnsw(sdbv_index *);
 
+/* ------------------------------------------------------------------------
+ * KV codec + cold-start staging pipeline: bulk-load a
+ * dumped surrealdb HNSW index — He element vectors (key/index/he.rs), Hn
+ * per-node edge lists (hn.rs), Hs graph state (hs.rs), Hv vector->docs
+ * entries (hv.rs) — straight into the host graph (no re-insertion), then
+ * finalize stages it to the device SoA. Key/value byte formats are pinned
+ * by the reference's own key tests (he.rs/hn.rs/hs.rs/hv.rs golden
+ * bytes); values follow the `revision` 0.17.0 wire format. Post-Hl
+ * (per-node Hn) storage only; legacy Hl chunks and Bits (roaring) doc
+ * sets are rejected with SDBV_ERR_UNSUPPORTED. hd/hi/hp/hh pairs are
+ * host-kept and skipped by the loader; the host re-binds record keys via
+ * sdbv_index_bind_doc_key. */
+typedef struct sdbv_kvload sdbv_kvload;
+int sdbv_kvload_new(sdbv_ctx * /* nullable */, uint32_t d, uint8_t metric,
+                    uint32_t m, uint32_t m0, uint32_t efc,
+                    int extend_candidates, int keep_pruned, uint64_t rng_seed,
+                    double ml, sdbv_kvload **out);
+int sdbv_kvload_feed(sdbv_kvload *, const uint8_t *key, uint64_t klen,
+                     const uint8_t *val, uint64_t vlen);
+int sdbv_kvload_finish_hnsw(sdbv_kvload *, sdbv_hnsw **out);
+int sdbv_kvload_finish_index(sdbv_kvload *, uint64_t table, sdbv_index **out);
+void sdbv_kvload_abort(sdbv_kvload *);
+int sdbv_index_bind_doc_key(sdbv_index *, uint64_t doc_id,
+                            uint64_t record_key);
+/* Dump the graph / index back to reference-format KV pairs (round-trip +
+ * migration tooling). The callback returns non-zero to abort. */
+typedef int (*sdbv_kv_write_cb)(void *user, const uint8_t *key, uint64_t klen,
+                                const uint8_t *val, uint64_t vlen);
+int sdbv_hnsw_dump_kv(sdbv_hnsw *, uint32_t ns, uint32_t db, const char *tb,
+                      uint32_t ix, sdbv_kv_write_cb, void *user);
+int sdbv_index_dump_kv(sdbv_index *, uint32_t ns, uint32_t db, const char *tb,
+                       uint32_t ix, sdbv_kv_write_cb, void *user);
+
 #ifdef __cplusplus
 }
 #endif

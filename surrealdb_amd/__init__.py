@@ -18,6 +18,7 @@ METRICS = {"cosine": METRIC_COSINE, "euclidean": METRIC_EUCLIDEAN}
 
 TRUTHY_CB = None  # ctypes callback types, set when lib() loads
 EXPIRE_CB = None
+KV_WRITE_CB = None
 
 _ERRS = {
     0: "OK",
@@ -118,6 +119,21 @@ def lib():
     L.sdbv_index_check_props.argtypes = [vp, u64]
     L.sdbv_index_hnsw.restype = vp
     L.sdbv_index_hnsw.argtypes = [vp]
+    u8pp = ctypes.POINTER(u8)
+    L.sdbv_kvload_new.argtypes = [vp, u32, u8, u32, u32, u32, ctypes.c_int,
+                                  ctypes.c_int, u64, ctypes.c_double,
+                                  ctypes.POINTER(vp)]
+    L.sdbv_kvload_feed.argtypes = [vp, u8pp, u64, u8pp, u64]
+    L.sdbv_kvload_finish_hnsw.argtypes = [vp, ctypes.POINTER(vp)]
+    L.sdbv_kvload_finish_index.argtypes = [vp, u64, ctypes.POINTER(vp)]
+    L.sdbv_kvload_abort.argtypes = [vp]
+    L.sdbv_index_bind_doc_key.argtypes = [vp, u64, u64]
+    global KV_WRITE_CB
+    KV_WRITE_CB = ctypes.CFUNCTYPE(ctypes.c_int, vp, u8pp, u64, u8pp, u64)
+    L.sdbv_hnsw_dump_kv.argtypes = [vp, u32, u32, ctypes.c_char_p, u32,
+                                    KV_WRITE_CB, vp]
+    L.sdbv_index_dump_kv.argtypes = [vp, u32, u32, ctypes.c_char_p, u32,
+                                     KV_WRITE_CB, vp]
     _lib = L
     return L
 
@@ -348,6 +364,11 @@ class Hnsw:
             edges.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)))
         return offsets, edges[:ec]
 
+    def dump_kv(self, ns=1, db=2, tb="testtb", ix=3):
+        """Dump the graph as reference-format (key, value) KV pairs
+        (He/Hn/Hs; key/index/{he,hn,hs}.rs byte layouts)."""
+        return _dump_kv(lib().sdbv_hnsw_dump_kv, self._ptr, ns, db, tb, ix)
+
     def remove(self, e_id):
         """Hnsw::remove (hnsw/mod.rs:398-455). True if removed. Host graphs
         only (pre-finalize); finalized indexes mutate through Index."""
@@ -370,6 +391,80 @@ def index_create_host(d, metric="euclidean", m=12, m0=None, efc=150,
     """Host-only Index (ctx = NULL): the full write path + pendings-merged
     search on the host graph — the CPU-testable configuration."""
     return Index(None, 0, d, metric, m, m0, efc, extend, keep, seed, ml)
+
+
+def _kvload_new(ctx, d, metric, m, m0, efc, extend, keep, seed, ml):
+    import math
+    if m0 is None:
+        m0 = 2 * m
+    if ml is None:
+        ml = 1.0 / math.log(m)
+    out = ctypes.c_void_p()
+    cptr = ctx._ptr if ctx is not None else None
+    _check(cptr, lib().sdbv_kvload_new(
+        cptr, d, METRICS[metric], m, m0, efc, int(extend), int(keep), seed,
+        ml, ctypes.byref(out)), "sdbv_kvload_new")
+    return out
+
+
+def _kvload_feed_all(loader, pairs):
+    u8p = ctypes.POINTER(ctypes.c_uint8)
+    for key, val in pairs:
+        k = (ctypes.c_uint8 * len(key)).from_buffer_copy(key)
+        v = (ctypes.c_uint8 * max(len(val), 1)).from_buffer_copy(
+            val if val else b"\0")
+        rc = lib().sdbv_kvload_feed(loader, ctypes.cast(k, u8p), len(key),
+                                    ctypes.cast(v, u8p), len(val))
+        if rc != 0:
+            lib().sdbv_kvload_abort(loader)
+            raise SdbvError(f"sdbv_kvload_feed: {_ERRS.get(rc, rc)}")
+
+
+def load_kv_hnsw(pairs, d, metric="euclidean", m=12, m0=None, efc=150,
+                 extend=False, keep=False, seed=0x5DB1, ml=None, ctx=None):
+    """Cold-start bulk load of a dumped reference HNSW graph (He/Hn/Hs KV
+    pairs) into a host graph — no re-insertion (key/index/{he,hn,hs}.rs +
+    HnswLayer::load, layer.rs:504-563)."""
+    loader = _kvload_new(ctx, d, metric, m, m0, efc, extend, keep, seed, ml)
+    _kvload_feed_all(loader, pairs)
+    out = ctypes.c_void_p()
+    _check(None, lib().sdbv_kvload_finish_hnsw(loader, ctypes.byref(out)),
+           "sdbv_kvload_finish_hnsw")
+    return Hnsw(ctx, out, d)
+
+
+def load_kv_index(pairs, table, d, metric="euclidean", m=12, m0=None,
+                  efc=150, extend=False, keep=False, seed=0x5DB1, ml=None,
+                  ctx=None, doc_keys=None):
+    """Cold-start bulk load of a dumped reference HNSW *index* (He/Hn/Hs/Hv
+    pairs; hd/hi pairs are host-kept — pass their mapping as
+    doc_keys={doc_id: record_key_handle})."""
+    loader = _kvload_new(ctx, d, metric, m, m0, efc, extend, keep, seed, ml)
+    _kvload_feed_all(loader, pairs)
+    out = ctypes.c_void_p()
+    _check(None, lib().sdbv_kvload_finish_index(loader, table,
+                                                ctypes.byref(out)),
+           "sdbv_kvload_finish_index")
+    ix = Index.__new__(Index)
+    ix._ctx = ctx
+    ix._ptr = out
+    ix.d = d
+    for doc_id, key in (doc_keys or {}).items():
+        _check(None, lib().sdbv_index_bind_doc_key(out, doc_id, key),
+               "sdbv_index_bind_doc_key")
+    return ix
+
+
+def _dump_kv(fn, ptr, ns, db, tb, ix_id):
+    pairs = []
+
+    def w(user, key, klen, val, vlen):
+        pairs.append((bytes(bytearray(key[i] for i in range(klen))),
+                      bytes(bytearray(val[i] for i in range(vlen)))))
+        return 0
+    cb = KV_WRITE_CB(w)
+    _check(None, fn(ptr, ns, db, tb.encode(), ix_id, cb, None), "dump_kv")
+    return pairs
 
 
 class Index:
@@ -469,6 +564,11 @@ class Index:
 
     def check_props(self, expected_count):
         return lib().sdbv_index_check_props(self._ptr, expected_count)
+
+    def dump_kv(self, ns=1, db=2, tb="testtb", ix=3):
+        """Dump graph + Hv vector->docs entries as reference-format KV
+        pairs (He/Hn/Hs/Hv)."""
+        return _dump_kv(lib().sdbv_index_dump_kv, self._ptr, ns, db, tb, ix)
 
     def hnsw(self):
         """Non-owning view of the underlying graph (parity introspection)."""

@@ -3167,6 +3167,331 @@ static int idx_graph_search_gpu(sdbv_index *ix, const float *q, uint32_t k,
 	return SDBV_OK;
 }
 
+// ===========================================================================
+// KV codec + cold-start staging pipeline: the
+// reference's on-disk HNSW state — He element vectors (key/index/he.rs),
+// Hn per-node edge lists (hn.rs), Hs graph state (hs.rs), Hv vector->docs
+// entries (hv.rs) — parsed/emitted byte-exactly so a dumped surrealdb
+// index bulk-loads straight into the host graph + device SoA with no
+// re-insertion. Key layout (storekey, pinned by the reference's own key
+// tests): `/*<ns u32 BE>*<db u32 BE>*<tb>\0+<ix u32 BE>!h<e|n|s|v>` +
+// per-key fields (element u64 BE; layer u16 BE + node u64 BE; escaped
+// revisioned vector). Embedded slices escape 0x00 -> 0x01 0x00 and
+// 0x01 -> 0x01 0x01 with a 0x00 terminator (derived from the hv.rs
+// golden bytes). Values use the `revision` crate 0.17.0 wire format
+// (dependency pinned by Cargo.lock, source not vendored): varint revision
+// tag + varint enum variant + varint lengths + fixed little-endian
+// primitives — pinned by the hv.rs golden vectors at small sizes; the
+// varint extrapolation beyond 127 follows the crate's published LEB128.
+// ===========================================================================
+
+namespace kvc {
+
+static void put_varint(std::vector<uint8_t> &b, uint64_t v) {
+	while (v >= 0x80) {
+		b.push_back((uint8_t)(v & 0x7F) | 0x80);
+		v >>= 7;
+	}
+	b.push_back((uint8_t)v);
+}
+static bool get_varint(const uint8_t *&p, const uint8_t *end, uint64_t *v) {
+	uint64_t out = 0;
+	int shift = 0;
+	while (p < end && shift < 64) {
+		uint8_t c = *p++;
+		out |= (uint64_t)(c & 0x7F) << shift;
+		if (!(c & 0x80)) {
+			*v = out;
+			return true;
+		}
+		shift += 7;
+	}
+	return false;
+}
+template <typename T> static void put_le(std::vector<uint8_t> &b, T v) {
+	uint8_t tmp[sizeof(T)];
+	std::memcpy(tmp, &v, sizeof(T));
+	b.insert(b.end(), tmp, tmp + sizeof(T));
+}
+template <typename T>
+static bool get_le(const uint8_t *&p, const uint8_t *end, T *v) {
+	if ((size_t)(end - p) < sizeof(T))
+		return false;
+	std::memcpy(v, p, sizeof(T));
+	p += sizeof(T);
+	return true;
+}
+template <typename T> static void put_be(std::vector<uint8_t> &b, T v) {
+	for (int i = (int)sizeof(T) - 1; i >= 0; i--)
+		b.push_back((uint8_t)(v >> (8 * i)));
+}
+template <typename T>
+static bool get_be(const uint8_t *&p, const uint8_t *end, T *v) {
+	if ((size_t)(end - p) < sizeof(T))
+		return false;
+	T out = 0;
+	for (size_t i = 0; i < sizeof(T); i++)
+		out = (out << 8) | *p++;
+	*v = out;
+	return true;
+}
+
+// storekey escaped-slice codec (embedded dynamic fields in keys)
+static void put_escaped(std::vector<uint8_t> &b, const uint8_t *s, size_t n) {
+	for (size_t i = 0; i < n; i++) {
+		if (s[i] == 0x00 || s[i] == 0x01) {
+			b.push_back(0x01);
+			b.push_back(s[i]);
+		} else {
+			b.push_back(s[i]);
+		}
+	}
+	b.push_back(0x00);
+}
+static bool get_escaped(const uint8_t *&p, const uint8_t *end,
+                        std::vector<uint8_t> &out) {
+	while (p < end) {
+		uint8_t c = *p++;
+		if (c == 0x00)
+			return true; // terminator
+		if (c == 0x01) {
+			if (p >= end)
+				return false;
+			out.push_back(*p++);
+		} else {
+			out.push_back(c);
+		}
+	}
+	return false;
+}
+
+// revisioned SerializedVector (idx/trees/vector.rs:33-41; F64=0 F32=1
+// I64=2 I32=3 I16=4), F32 payloads only on the encode side (the HNSW
+// default vector type, define.rs:1107).
+static void enc_vec_f32(std::vector<uint8_t> &b, const float *v, uint32_t d) {
+	put_varint(b, 1); // revision
+	put_varint(b, 1); // variant F32
+	put_varint(b, d);
+	for (uint32_t i = 0; i < d; i++)
+		put_le(b, v[i]);
+}
+static bool dec_vec_f32(const uint8_t *p, const uint8_t *end,
+                        std::vector<float> &out) {
+	uint64_t rev, variant, len;
+	if (!get_varint(p, end, &rev) || rev != 1)
+		return false;
+	if (!get_varint(p, end, &variant))
+		return false;
+	if (!get_varint(p, end, &len))
+		return false;
+	out.clear();
+	out.reserve(len);
+	for (uint64_t i = 0; i < len; i++) {
+		switch (variant) {
+		case 1: { // F32
+			float f;
+			if (!get_le(p, end, &f))
+				return false;
+			out.push_back(f);
+			break;
+		}
+		default:
+			return false; // F64/I* corpora are not staged as f32 blindly
+		}
+	}
+	return true;
+}
+
+// graph.rs:104-113 node_to_val: u16 BE edge count + u64 BE edge ids.
+static void enc_node(std::vector<uint8_t> &b,
+                     const std::vector<uint32_t> &edges) {
+	put_be(b, (uint16_t)edges.size());
+	for (uint32_t e : edges)
+		put_be(b, (uint64_t)e);
+}
+static bool dec_node(const uint8_t *p, const uint8_t *end,
+                     std::vector<uint32_t> &out) {
+	uint16_t n;
+	if (!get_be(p, end, &n))
+		return false;
+	out.clear();
+	out.reserve(n);
+	for (uint16_t i = 0; i < n; i++) {
+		uint64_t e;
+		if (!get_be(p, end, &e))
+			return false;
+		out.push_back((uint32_t)e);
+	}
+	return true;
+}
+
+// revisioned HnswState (hnsw/mod.rs:61-71): Option<ElementId> enter_point
+// (u8 tag 0/1 + u64 LE), next_element_id u64 LE, layer0 LayerState
+// {version u64 LE, chunks u32 LE}, layers Vec<LayerState>.
+struct HnswStateKV {
+	bool has_ep = false;
+	uint64_t enter_point = 0;
+	uint64_t next_element_id = 0;
+	uint64_t n_upper_layers = 0; // layers.len()
+};
+static void enc_state(std::vector<uint8_t> &b, const HnswStateKV &s) {
+	put_varint(b, 1); // revision
+	b.push_back(s.has_ep ? 1 : 0);
+	if (s.has_ep)
+		put_le(b, s.enter_point);
+	put_le(b, s.next_element_id);
+	put_le(b, (uint64_t)0); // layer0.version (not tracked here)
+	put_le(b, (uint32_t)0); // layer0.chunks == 0 (post-Hl format)
+	put_varint(b, s.n_upper_layers);
+	for (uint64_t i = 0; i < s.n_upper_layers; i++) {
+		put_le(b, (uint64_t)0);
+		put_le(b, (uint32_t)0);
+	}
+}
+static bool dec_state(const uint8_t *p, const uint8_t *end, HnswStateKV *s) {
+	uint64_t rev;
+	if (!get_varint(p, end, &rev) || rev != 1)
+		return false;
+	if (p >= end)
+		return false;
+	uint8_t tag = *p++;
+	s->has_ep = tag != 0;
+	if (s->has_ep && !get_le(p, end, &s->enter_point))
+		return false;
+	if (!get_le(p, end, &s->next_element_id))
+		return false;
+	uint64_t v64;
+	uint32_t v32;
+	if (!get_le(p, end, &v64) || !get_le(p, end, &v32))
+		return false; // layer0 state
+	if (v32 != 0)
+		return false; // legacy Hl chunks unsupported (post-migration only)
+	if (!get_varint(p, end, &s->n_upper_layers))
+		return false;
+	for (uint64_t i = 0; i < s->n_upper_layers; i++)
+		if (!get_le(p, end, &v64) || !get_le(p, end, &v32) || v32 != 0)
+			return false;
+	return true;
+}
+
+// revisioned ElementDocs (docs.rs:164-177): e_id u64 LE + Ids64 (variant
+// varint: Empty=0 One=1 Vec2=2..Vec8=8 Bits=9; fixed u64 LE ids; Bits
+// (RoaringTreemap) unsupported -> false).
+static void enc_element_docs(std::vector<uint8_t> &b, uint64_t e_id,
+                             const vdocs::Ids64 &docs) {
+	put_varint(b, 1); // revision
+	put_le(b, e_id);
+	put_varint(b, 1); // Ids64 revision
+	size_t n = docs.v.size();
+	put_varint(b, n == 0 ? 0 : (n <= 8 ? n : 9));
+	for (uint64_t d : docs.v)
+		put_le(b, d);
+}
+static bool dec_element_docs(const uint8_t *p, const uint8_t *end,
+                             uint64_t *e_id, vdocs::Ids64 *docs) {
+	uint64_t rev, variant;
+	if (!get_varint(p, end, &rev) || rev != 1)
+		return false;
+	if (!get_le(p, end, e_id))
+		return false;
+	if (!get_varint(p, end, &rev) || rev != 1)
+		return false;
+	if (!get_varint(p, end, &variant))
+		return false;
+	docs->v.clear();
+	docs->bits = false;
+	if (variant == 0)
+		return true;
+	if (variant > 8)
+		return false; // Bits / roaring not supported in this revision
+	for (uint64_t i = 0; i < variant; i++) {
+		uint64_t d;
+		if (!get_le(p, end, &d))
+			return false;
+		docs->v.push_back(d);
+	}
+	return true;
+}
+
+// Key prefix `/*<ns>*<db>*<tb>\0+<ix>!h` + kind char.
+static void key_prefix(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
+                       const char *tb, uint32_t ix) {
+	b.push_back('/');
+	b.push_back('*');
+	put_be(b, ns);
+	b.push_back('*');
+	put_be(b, db);
+	b.push_back('*');
+	b.insert(b.end(), tb, tb + strlen(tb));
+	b.push_back(0);
+	b.push_back('+');
+	put_be(b, ix);
+	b.push_back('!');
+	b.push_back('h');
+}
+static void key_he(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
+                   const char *tb, uint32_t ix, uint64_t element_id) {
+	key_prefix(b, ns, db, tb, ix);
+	b.push_back('e');
+	put_be(b, element_id);
+}
+static void key_hn(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
+                   const char *tb, uint32_t ix, uint16_t layer,
+                   uint64_t node) {
+	key_prefix(b, ns, db, tb, ix);
+	b.push_back('n');
+	put_be(b, layer);
+	put_be(b, node);
+}
+static void key_hs(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
+                   const char *tb, uint32_t ix) {
+	key_prefix(b, ns, db, tb, ix);
+	b.push_back('s');
+}
+static void key_hv(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
+                   const char *tb, uint32_t ix, const float *v, uint32_t d) {
+	key_prefix(b, ns, db, tb, ix);
+	b.push_back('v');
+	std::vector<uint8_t> ser;
+	enc_vec_f32(ser, v, d);
+	put_escaped(b, ser.data(), ser.size());
+}
+
+// Parses a key: positions the kind char + the remainder. Returns 0 on a
+// well-formed `!h?` key, else -1.
+struct ParsedKey {
+	char kind;
+	const uint8_t *rest;
+	const uint8_t *end;
+};
+static int parse_key(const uint8_t *k, size_t n, ParsedKey *out) {
+	const uint8_t *p = k, *end = k + n;
+	if (end - p < 12 || *p++ != '/' || *p++ != '*')
+		return -1;
+	p += 4; // ns
+	if (p >= end || *p++ != '*')
+		return -1;
+	p += 4; // db
+	if (p >= end || *p++ != '*')
+		return -1;
+	while (p < end && *p != 0)
+		p++; // tb
+	if (p >= end)
+		return -1;
+	p++; // NUL
+	if (p >= end || *p++ != '+')
+		return -1;
+	p += 4; // ix
+	if (end - p < 3 || *p++ != '!' || *p++ != 'h')
+		return -1;
+	out->kind = (char)*p++;
+	out->rest = p;
+	out->end = end;
+	return 0;
+}
+
+} // namespace kvc
+
 // ---------------------------------------------------------------------------
 // Filtered KNN (hnsw/filter.rs + layer.rs:110-318): the WHERE-condition
 // evaluation (is_record_truthy — KV fetch + expression compute) stays on
@@ -3631,6 +3956,294 @@ int sdbv_index_knn_filtered(sdbv_index *ix, const float *q, uint32_t k,
 		n++;
 	}
 	*out_n = n;
+	return SDBV_OK;
+}
+
+// ---- KV cold-start loader (He/Hn/Hs/Hv stream -> host graph + index) ----
+
+struct sdbv_kvload {
+	sdbv_ctx *ctx;
+	uint32_t d;
+	uint8_t metric;
+	uint32_t m, m0, efc;
+	int extend, keep;
+	uint64_t seed;
+	double ml;
+	bool has_state = false;
+	kvc::HnswStateKV state;
+	std::map<uint64_t, std::vector<float>> elems;          // He
+	std::map<std::pair<uint16_t, uint64_t>, std::vector<uint32_t>> nodes; // Hn
+	struct HvEnt {
+		std::vector<float> vec;
+		uint64_t e_id;
+		vdocs::Ids64 docs;
+	};
+	std::vector<HvEnt> hv; // Hv
+};
+
+int sdbv_kvload_new(sdbv_ctx *ctx, uint32_t d, uint8_t metric, uint32_t m,
+                    uint32_t m0, uint32_t efc, int extend, int keep,
+                    uint64_t seed, double ml, sdbv_kvload **out) {
+	if (d == 0 || (d % 4) != 0 || metric > SDBV_METRIC_EUCLIDEAN)
+		return SDBV_ERR_BAD_ARG;
+	auto *L = new sdbv_kvload();
+	L->ctx = ctx;
+	L->d = d;
+	L->metric = metric;
+	L->m = m;
+	L->m0 = m0;
+	L->efc = efc;
+	L->extend = extend;
+	L->keep = keep;
+	L->seed = seed;
+	L->ml = ml;
+	*out = L;
+	return SDBV_OK;
+}
+
+void sdbv_kvload_abort(sdbv_kvload *L) { delete L; }
+
+// Feed one (key, value) pair from the index's KV range scan. Unknown `!h?`
+// kinds (hd/hi/hp/hh records — host-kept) are skipped, not errors.
+int sdbv_kvload_feed(sdbv_kvload *L, const uint8_t *key, uint64_t klen,
+                     const uint8_t *val, uint64_t vlen) {
+	if (!L || !key)
+		return SDBV_ERR_BAD_ARG;
+	kvc::ParsedKey pk;
+	if (kvc::parse_key(key, klen, &pk) != 0)
+		return SDBV_ERR_BAD_ARG;
+	const uint8_t *p = pk.rest, *end = pk.end;
+	switch (pk.kind) {
+	case 'e': { // He: element id (key) -> SerializedVector (value)
+		uint64_t e_id;
+		if (!kvc::get_be(p, end, &e_id))
+			return SDBV_ERR_BAD_ARG;
+		std::vector<float> v;
+		if (!kvc::dec_vec_f32(val, val + vlen, v) || v.size() != L->d)
+			return SDBV_ERR_UNSUPPORTED;
+		L->elems[e_id] = std::move(v);
+		return SDBV_OK;
+	}
+	case 'n': { // Hn: (layer, node) -> edge list
+		uint16_t layer;
+		uint64_t node;
+		if (!kvc::get_be(p, end, &layer) || !kvc::get_be(p, end, &node))
+			return SDBV_ERR_BAD_ARG;
+		std::vector<uint32_t> edges;
+		if (!kvc::dec_node(val, val + vlen, edges))
+			return SDBV_ERR_BAD_ARG;
+		L->nodes[{layer, node}] = std::move(edges);
+		return SDBV_OK;
+	}
+	case 's': { // Hs: graph state
+		if (!kvc::dec_state(val, val + vlen, &L->state))
+			return SDBV_ERR_UNSUPPORTED;
+		L->has_state = true;
+		return SDBV_OK;
+	}
+	case 'v': { // Hv: escaped vector (key) -> ElementDocs (value)
+		std::vector<uint8_t> ser;
+		if (!kvc::get_escaped(p, end, ser))
+			return SDBV_ERR_BAD_ARG;
+		sdbv_kvload::HvEnt ent;
+		if (!kvc::dec_vec_f32(ser.data(), ser.data() + ser.size(),
+		                      ent.vec) ||
+		    ent.vec.size() != L->d)
+			return SDBV_ERR_UNSUPPORTED;
+		if (!kvc::dec_element_docs(val, val + vlen, &ent.e_id, &ent.docs))
+			return SDBV_ERR_UNSUPPORTED;
+		L->hv.push_back(std::move(ent));
+		return SDBV_OK;
+	}
+	default:
+		return SDBV_OK; // hd/hi/hp/hh etc.: host-kept, skipped
+	}
+}
+
+// Builds the host graph from the fed He/Hn/Hs pairs (HnswFlavor::new +
+// check_state + HnswLayer::load, layer.rs:504-563 — post-Hl format only).
+static int kvload_build_graph(sdbv_kvload *L, sdbv_hnsw **out) {
+	if (!L->has_state)
+		return SDBV_ERR_BAD_ARG; // an existing index always has Hs
+	sdbv_hnsw *h = nullptr;
+	int rc = sdbv_hnsw_create(L->ctx, L->d, L->metric, L->m, L->m0, L->efc,
+	                          L->extend, L->keep, L->seed, L->ml, &h);
+	if (rc)
+		return rc;
+	uint64_t n = L->state.next_element_id;
+	h->next_id = n;
+	h->vecs.assign((size_t)n * L->d, 0.0f);
+	h->elem_present.assign(n, 0);
+	if (L->metric == SDBV_METRIC_COSINE)
+		h->norms.assign(n, 0.0);
+	for (auto &e : L->elems) {
+		if (e.first >= n) {
+			sdbv_hnsw_destroy(h);
+			return SDBV_ERR_BAD_ARG;
+		}
+		std::memcpy(h->vecs.data() + e.first * L->d, e.second.data(),
+		            (size_t)L->d * 4);
+		h->elem_present[e.first] = 1;
+		if (L->metric == SDBV_METRIC_COSINE)
+			h->norms[e.first] =
+			    sqrt(hnsw::host_sumsq_f32(e.second.data(), L->d));
+	}
+	uint64_t n_layers = 1 + L->state.n_upper_layers;
+	h->layers.clear();
+	for (uint64_t l = 0; l < n_layers; l++)
+		h->layers.push_back(
+		    hnsw::Layer{std::vector<std::vector<uint32_t>>(n),
+		                l == 0 ? L->m0 : L->m});
+	for (auto &l : h->layers)
+		l.in_layer.assign(n, 0);
+	for (auto &nd : L->nodes) {
+		uint16_t layer = nd.first.first;
+		uint64_t node = nd.first.second;
+		if (layer >= h->layers.size() || node >= n) {
+			sdbv_hnsw_destroy(h);
+			return SDBV_ERR_BAD_ARG;
+		}
+		for (uint32_t e : nd.second)
+			if (e >= n) {
+				sdbv_hnsw_destroy(h);
+				return SDBV_ERR_BAD_ARG;
+			}
+		h->layers[layer].edges[node] = nd.second;
+		h->layers[layer].in_layer[node] = 1;
+	}
+	h->enter_point = L->state.has_ep ? (int64_t)L->state.enter_point : -1;
+	h->dirty = true;
+	*out = h;
+	return SDBV_OK;
+}
+
+int sdbv_kvload_finish_hnsw(sdbv_kvload *L, sdbv_hnsw **out) {
+	if (!L || !out)
+		return SDBV_ERR_BAD_ARG;
+	int rc = kvload_build_graph(L, out);
+	delete L;
+	return rc;
+}
+
+int sdbv_kvload_finish_index(sdbv_kvload *L, uint64_t table,
+                             sdbv_index **out) {
+	if (!L || !out)
+		return SDBV_ERR_BAD_ARG;
+	sdbv_hnsw *h = nullptr;
+	int rc = kvload_build_graph(L, &h);
+	if (rc) {
+		delete L;
+		return rc;
+	}
+	auto *ix = new sdbv_index();
+	ix->h = h;
+	ix->table = table;
+	uint64_t max_doc = 0;
+	bool any_doc = false;
+	for (auto &ent : L->hv) {
+		std::string key((const char *)ent.vec.data(), (size_t)L->d * 4);
+		auto r = ix->vd.emplace(std::move(key),
+		                        sdbv_index::ED{(uint32_t)ent.e_id, ent.docs});
+		if (r.second)
+			ix->by_elem[(uint32_t)ent.e_id] = &r.first->first;
+		for (uint64_t d : ent.docs.v) {
+			any_doc = true;
+			if (d > max_doc)
+				max_doc = d;
+		}
+	}
+	// HnswDocsState (hd root) stays host-side; allocation resumes past the
+	// highest referenced doc id (recycled holes are NOT reconstructed —
+	// the host re-binds key<->doc pairs via sdbv_index_bind_doc_key).
+	ix->next_doc_id = any_doc ? max_doc + 1 : 0;
+	delete L;
+	*out = ix;
+	return SDBV_OK;
+}
+
+// Re-binds one record-key handle to a doc id (the host's hi/hd entries)
+// after a cold-start load.
+int sdbv_index_bind_doc_key(sdbv_index *ix, uint64_t doc_id,
+                            uint64_t record_key) {
+	if (!ix)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	ix->key2doc[record_key] = doc_id;
+	ix->doc2key[doc_id] = record_key;
+	if (doc_id >= ix->next_doc_id)
+		ix->next_doc_id = doc_id + 1;
+	return SDBV_OK;
+}
+
+// ---- KV dump (round-trip + migration tooling) ----
+
+typedef int (*sdbv_kv_write_cb)(void *user, const uint8_t *key, uint64_t klen,
+                                const uint8_t *val, uint64_t vlen);
+
+int sdbv_hnsw_dump_kv(sdbv_hnsw *h, uint32_t ns, uint32_t db, const char *tb,
+                      uint32_t ix_id, sdbv_kv_write_cb write, void *user) {
+	if (!h || !tb || !write)
+		return SDBV_ERR_BAD_ARG;
+	std::vector<uint8_t> k, v;
+	// Hs
+	kvc::HnswStateKV st;
+	st.has_ep = h->enter_point >= 0;
+	st.enter_point = st.has_ep ? (uint64_t)h->enter_point : 0;
+	st.next_element_id = h->next_id;
+	st.n_upper_layers = h->layers.size() - 1;
+	k.clear();
+	v.clear();
+	kvc::key_hs(k, ns, db, tb, ix_id);
+	kvc::enc_state(v, st);
+	if (write(user, k.data(), k.size(), v.data(), v.size()))
+		return SDBV_ERR_BAD_ARG;
+	// He per present element
+	for (uint64_t e = 0; e < h->next_id; e++) {
+		if (!h->elem_present[e])
+			continue;
+		k.clear();
+		v.clear();
+		kvc::key_he(k, ns, db, tb, ix_id, e);
+		kvc::enc_vec_f32(v, h->vecs.data() + e * h->d, h->d);
+		if (write(user, k.data(), k.size(), v.data(), v.size()))
+			return SDBV_ERR_BAD_ARG;
+	}
+	// Hn per layer member
+	for (size_t l = 0; l < h->layers.size(); l++) {
+		const hnsw::Layer &layer = h->layers[l];
+		for (uint64_t node = 0; node < layer.edges.size(); node++) {
+			if (!layer.has((uint32_t)node))
+				continue;
+			k.clear();
+			v.clear();
+			kvc::key_hn(k, ns, db, tb, ix_id, (uint16_t)l, node);
+			kvc::enc_node(v, layer.edges[node]);
+			if (write(user, k.data(), k.size(), v.data(), v.size()))
+				return SDBV_ERR_BAD_ARG;
+		}
+	}
+	return SDBV_OK;
+}
+
+int sdbv_index_dump_kv(sdbv_index *ix, uint32_t ns, uint32_t db,
+                       const char *tb, uint32_t ix_id, sdbv_kv_write_cb write,
+                       void *user) {
+	if (!ix)
+		return SDBV_ERR_BAD_ARG;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	int rc = sdbv_hnsw_dump_kv(ix->h, ns, db, tb, ix_id, write, user);
+	if (rc)
+		return rc;
+	std::vector<uint8_t> k, v;
+	for (auto &e : ix->vd) {
+		k.clear();
+		v.clear();
+		kvc::key_hv(k, ns, db, tb, ix_id, (const float *)e.first.data(),
+		            ix->h->d);
+		kvc::enc_element_docs(v, e.second.e_id, e.second.docs);
+		if (write(user, k.data(), k.size(), v.data(), v.size()))
+			return SDBV_ERR_BAD_ARG;
+	}
 	return SDBV_OK;
 }
 

@@ -1,0 +1,170 @@
+"""KV codec + cold-start staging: key/value byte
+formats pinned against the reference's OWN key-test constants
+(key/index/{he,hn,hs,hv}.rs), plus round-trips through the bulk loader
+(no re-insertion) and an independently hand-encoded fixture."""
+import struct
+
+import numpy as np
+import pytest
+
+import oracle
+import surrealdb_amd as sa
+
+PREFIX = b"/*\x00\x00\x00\x01*\x00\x00\x00\x02*testtb\0+\0\0\0\x03!h"
+
+
+def test_he_key_and_value_pinned():
+    """he.rs:66-73 golden key bytes (ns=1, db=2, tb=testtb, ix=3,
+    element 7); value = revisioned SerializedVector::F32."""
+    h = sa.hnsw_create_host(4, metric="euclidean", m=4, m0=8, efc=20, seed=1)
+    for i in range(8):
+        h.insert(np.full(4, float(i + 1), dtype=np.float32))
+    d = dict(h.dump_kv(ns=1, db=2, tb="testtb", ix=3))
+    key7 = PREFIX + b"e" + (7).to_bytes(8, "big")
+    assert key7 == (b"/*\x00\x00\x00\x01*\x00\x00\x00\x02*testtb\0+\0\0\0"
+                    b"\x03!he\0\0\0\0\0\0\0\x07")  # he.rs test, verbatim
+    assert key7 in d
+    # revisioned F32: rev=1, variant=1, len=4, f32 LE payload
+    assert d[key7] == bytes([1, 1, 4]) + struct.pack("<4f", 8, 8, 8, 8)
+    h.destroy()
+
+
+def test_hn_key_and_value_pinned():
+    """hn.rs:84-97 golden key bytes (layer 7, node 8 -> layer u16 BE +
+    node u64 BE); value = u16 BE edge count + u64 BE edges
+    (graph.rs:104-113 node_to_val)."""
+    expected = (b"/*\x00\x00\x00\x01*\x00\x00\x00\x02*testtb\0+\0\0\0\x03"
+                b"!hn\0\x07\0\0\0\0\0\0\0\x08")
+    built = PREFIX + b"n" + (7).to_bytes(2, "big") + (8).to_bytes(8, "big")
+    assert built == expected  # hn.rs test, verbatim
+    # value codec via a dumped 2-node graph
+    h = sa.hnsw_create_host(4, metric="euclidean", m=4, m0=8, efc=20, seed=1)
+    h.insert(np.array([0, 0, 0, 0], dtype=np.float32))
+    h.insert(np.array([1, 1, 1, 1], dtype=np.float32))
+    d = dict(h.dump_kv())
+    v = d[PREFIX + b"n" + (0).to_bytes(2, "big") + (0).to_bytes(8, "big")]
+    assert v == (1).to_bytes(2, "big") + (1).to_bytes(8, "big")
+    h.destroy()
+
+
+def test_hs_key_pinned():
+    """hs.rs golden key bytes."""
+    h = sa.hnsw_create_host(4, metric="euclidean", m=4, m0=8, efc=20, seed=1)
+    h.insert(np.zeros(4, dtype=np.float32))
+    d = dict(h.dump_kv())
+    assert (b"/*\x00\x00\x00\x01*\x00\x00\x00\x02*testtb\0+\0\0\0\x03!hs"
+            in d)  # hs.rs test, verbatim
+    h.destroy()
+
+
+def test_hv_key_pinned_f32_golden():
+    """hv.rs:95-99 golden: the F32 [1,2,3] vector embedded (storekey
+    escaping 0x00 -> 0x01 0x00, 0x01 -> 0x01 0x01, 0x00 terminator)."""
+    ix = sa.index_create_host(4, metric="euclidean", m=4, m0=8, efc=20,
+                              seed=1)
+    # hv.rs pins d=3; our index requires d%4==0, so pin the escaping rule
+    # itself on the d=3 encoding constructed byte-for-byte:
+    raw = bytes([1, 1, 3]) + struct.pack("<3f", 1.0, 2.0, 3.0)
+    esc = bytearray()
+    for b in raw:
+        if b in (0, 1):
+            esc += bytes([1, b])
+        else:
+            esc.append(b)
+    esc.append(0)
+    built = PREFIX + b"v" + bytes(esc)
+    assert built == (b"/*\x00\x00\x00\x01*\x00\x00\x00\x02*testtb\0+\0\0\0"
+                     b"\x03!hv\x01\x01\x01\x01\x03\x01\0\x01\0\x80\x3F\x01"
+                     b"\0\x01\0\x01\0\x40\x01\0\x01\0\x40\x40\0")
+    # and the product emits the same construction for its own vectors
+    ix.enqueue(50, None, np.array([1, 2, 3, 4], dtype=np.float32))
+    ix.apply_pendings()
+    d = dict(ix.dump_kv())
+    raw4 = bytes([1, 1, 4]) + struct.pack("<4f", 1, 2, 3, 4)
+    esc4 = bytearray()
+    for b in raw4:
+        if b in (0, 1):
+            esc4 += bytes([1, b])
+        else:
+            esc4.append(b)
+    esc4.append(0)
+    kv_key = PREFIX + b"v" + bytes(esc4)
+    assert kv_key in d, list(k for k in d if b"!hv" in k)
+    # ElementDocs value: rev=1 + e_id u64 LE + Ids64 rev=1 + variant One +
+    # doc u64 LE
+    assert d[kv_key] == (bytes([1]) + (0).to_bytes(8, "little") +
+                         bytes([1, 1]) + (0).to_bytes(8, "little"))
+    ix.destroy()
+
+
+def test_graph_round_trip_with_removals():
+    d, n = 20, 500
+    rows = oracle.gen_f32(0x77, 0, n, d)
+    g = sa.hnsw_create_host(d, metric="cosine", m=8, m0=16, efc=60, seed=5)
+    g.insert_batch(rows, nthreads=1)
+    for e in (3, 77, 401):
+        assert g.remove(e)
+    pairs = g.dump_kv()
+    g2 = sa.load_kv_hnsw(pairs, d, metric="cosine", m=8, m0=16, efc=60,
+                         seed=5)
+    a, b = g.l0_csr(), g2.l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    assert g.num_layers() == g2.num_layers()
+    # a removed element stays removed through the round trip
+    assert not g2.remove(3)
+    g.destroy()
+    g2.destroy()
+
+
+def test_index_round_trip_searches_and_writes():
+    d = 16
+    ix = sa.index_create_host(d, metric="euclidean", m=8, m0=16, efc=50,
+                              seed=9)
+    rows = oracle.gen_f32(0x9, 0, 300, d)
+    for i, r in enumerate(rows):
+        ix.enqueue(100 + i, None, r)
+    ix.apply_pendings()
+    ix.enqueue(105, rows[5], None)
+    ix.apply_pendings()
+    pairs = ix.dump_kv()
+    ix2 = sa.load_kv_index(
+        pairs, 0, d, metric="euclidean", m=8, m0=16, efc=50, seed=9,
+        doc_keys={i: 100 + i for i in range(300) if i != 5})
+    assert ix2.doc_count() == 299
+    for q in oracle.gen_f32(0xB, 0, 10, d):
+        k1, i1, d1 = ix.knn_search(q, 10, 40)
+        k2, i2, d2_ = ix2.knn_search(q, 10, 40)
+        assert np.array_equal(i1, i2) and np.array_equal(d1, d2_)
+    # writes continue working after a cold start (doc-id allocation resumes
+    # past the highest loaded id; key resolution via the re-bound handles)
+    newv = oracle.gen_f32(0xC, 0, 1, d)[0]
+    ix2.enqueue(104, rows[4], newv)  # update via re-bound key
+    ix2.enqueue(999, None, rows[5])  # fresh doc
+    assert ix2.apply_pendings() == 2
+    k, i, dd = ix2.knn_search(newv, 1, 20)
+    assert (k[0], i[0], dd[0]) == (0, 4, 0.0)
+    k, i, dd = ix2.knn_search(rows[5], 1, 20)
+    assert k[0] == 0 and dd[0] == 0.0 and i[0] >= 300  # fresh allocation
+    ix.destroy()
+    ix2.destroy()
+
+
+def test_loader_rejects_garbage_and_skips_host_keys():
+    d = 8
+    g = sa.hnsw_create_host(d, metric="euclidean", m=4, m0=8, efc=20, seed=2)
+    g.insert(np.zeros(d, dtype=np.float32))
+    pairs = g.dump_kv()
+    # hd/hi/hp-style keys (host-kept kinds) are skipped, not errors
+    pairs.append((PREFIX + b"d" + b"\0" * 8, b"\x01\x02\x03"))
+    pairs.append((PREFIX + b"p" + b"\0" * 8, b"junk"))
+    g2 = sa.load_kv_hnsw(pairs, d, metric="euclidean", m=4, m0=8, efc=20,
+                         seed=2)
+    a, b = g.l0_csr(), g2.l0_csr()
+    assert np.array_equal(a[0], b[0])
+    # a malformed He value is an error
+    bad = [(PREFIX + b"e" + (0).to_bytes(8, "big"), b"\xff\xff")]
+    with pytest.raises(sa.SdbvError):
+        sa.load_kv_hnsw(pairs + bad, d, metric="euclidean", m=4, m0=8,
+                        efc=20, seed=2)
+    g.destroy()
+    g2.destroy()
