@@ -20,6 +20,12 @@ CMD = [
     "--offload-arch=gfx950",
     "-O3",
     "-ffp-contract=off",
+    # AVX2 for the HOST build path (graph construction distance chains).
+    # Bit-exactness holds: the restated unrolled-8 chain keeps 8 independent
+    # sequential partials, which vectorize to one 8-lane register without
+    # reassociation, and -ffp-contract=off still blocks mul+add fusion.
+    "-Xarch_host",
+    "-mavx2",
     "-std=c++17",
     "-fPIC",
     "-shared",

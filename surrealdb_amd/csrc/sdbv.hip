@@ -1658,60 +1658,61 @@ static inline uint64_t total_key(double x) {
 }
 
 // DoublePriorityQueue restatement (knn.rs:15-123): ordered by total_cmp(dist),
-// FIFO within equal distance; pop_last removes the LATEST of the max key.
+// FIFO within equal distance (push order); pop_last removes the LATEST of
+// the max key. Implemented as a sorted (key, seq) vector with a lazy head —
+// exactly the BTreeMap<FloatKey, VecDeque> observable order, without the
+// per-node allocations (this queue is the host build/search hot path).
 struct PQ {
-	std::map<uint64_t, std::deque<uint32_t>> m;
-	std::map<uint64_t, double> dval;
+	struct E {
+		uint64_t key;
+		uint32_t seq;
+		uint32_t id;
+		double d;
+	};
+	std::vector<E> v; // ascending (key, seq); v[head..] is the live queue
+	size_t head = 0;
+	uint32_t next_seq = 0;
 	size_t n = 0;
 	void push(double d, uint32_t id) {
-		uint64_t k = total_key(d);
-		m[k].push_back(id);
-		dval[k] = d;
+		E e{total_key(d), next_seq++, id, d};
+		auto it = std::upper_bound(
+		    v.begin() + head, v.end(), e, [](const E &a, const E &b) {
+			    return a.key != b.key ? a.key < b.key : a.seq < b.seq;
+		    });
+		v.insert(it, e);
 		n++;
 	}
 	bool pop_first(double *d, uint32_t *id) {
-		if (m.empty())
+		if (n == 0)
 			return false;
-		auto it = m.begin();
-		*d = dval[it->first];
-		*id = it->second.front();
-		it->second.pop_front();
-		if (it->second.empty()) {
-			dval.erase(it->first);
-			m.erase(it);
-		}
+		*d = v[head].d;
+		*id = v[head].id;
+		head++;
 		n--;
 		return true;
 	}
-	void pop_last() {
-		if (m.empty())
+	void pop_last() { // latest push of the max key == max (key, seq)
+		if (n == 0)
 			return;
-		auto it = std::prev(m.end());
-		it->second.pop_back();
-		if (it->second.empty()) {
-			dval.erase(it->first);
-			m.erase(it);
-		}
+		v.pop_back();
 		n--;
 	}
 	bool peek_first(double *d, uint32_t *id) const {
-		if (m.empty())
+		if (n == 0)
 			return false;
-		auto it = m.begin();
-		*d = dval.at(it->first);
-		*id = it->second.front();
+		*d = v[head].d;
+		*id = v[head].id;
 		return true;
 	}
 	double peek_last_dist(double fb) const {
-		return m.empty() ? fb : dval.at(std::prev(m.end())->first);
+		return n == 0 ? fb : v.back().d;
 	}
 	std::vector<std::pair<double, uint32_t>> to_vec() const {
-		std::vector<std::pair<double, uint32_t>> v;
-		v.reserve(n);
-		for (auto &e : m)
-			for (uint32_t id : e.second)
-				v.push_back({dval.at(e.first), id});
-		return v;
+		std::vector<std::pair<double, uint32_t>> out;
+		out.reserve(n);
+		for (size_t i = head; i < v.size(); i++)
+			out.push_back({v[i].d, v[i].id});
+		return out;
 	}
 };
 
@@ -1844,22 +1845,62 @@ struct IdxPend {
 };
 static bool idx_all_docs_pending(const IdxPend *p, uint32_t e_id);
 
+// Epoch-stamped visited set (build hot path): O(1) insert, no per-search
+// allocation or clearing. Drop-in for unordered_set<uint32_t> in
+// search_layer_host (template).
+struct VisitSet {
+	std::vector<uint32_t> stamp;
+	uint32_t epoch = 0;
+	void begin(size_t n) {
+		if (stamp.size() < n)
+			stamp.resize(n, 0);
+		if (++epoch == 0) {
+			std::fill(stamp.begin(), stamp.end(), 0);
+			epoch = 1;
+		}
+	}
+	// mimics unordered_set::insert().second
+	struct R {
+		bool second;
+	};
+	R insert(uint32_t id) {
+		if (id >= stamp.size())
+			stamp.resize(id + 1, 0);
+		if (stamp[id] == epoch)
+			return {false};
+		stamp[id] = epoch;
+		return {true};
+	}
+};
+
 // layer.rs:184-223 — host-distance variant (build path + index host path).
 // `pend`: an element whose docs are ALL pending is excluded from
 // `candidates` only; it still enters `w` (layer.rs:209-212, the reference
 // pushes to w outside the exclusion check — restated as-is).
+template <class VS>
 static void search_layer_host(sdbv_hnsw *h, const Layer &layer, const float *q,
-                              double q_norm, PQ &candidates,
-                              std::unordered_set<uint32_t> &visited, PQ &w,
-                              uint32_t ef, bool locked,
+                              double q_norm, PQ &candidates, VS &visited,
+                              PQ &w, uint32_t ef, bool locked,
                               const IdxPend *pend = nullptr) {
 	double fq = w.peek_last_dist(DBL_MAX);
 	double cd;
 	uint32_t doc;
+	std::vector<uint32_t> scratch; // locked-mode edge snapshot (reused)
 	while (candidates.pop_first(&cd, &doc)) {
 		if (cd > fq)
 			break;
-		for (uint32_t e : get_edges(h, layer, doc, locked)) {
+		const std::vector<uint32_t> *edges_p;
+		if (!locked) {
+			static const std::vector<uint32_t> kEmpty;
+			edges_p = doc < layer.edges.size() ? &layer.edges[doc] : &kEmpty;
+		} else {
+			scratch.clear();
+			std::lock_guard<std::mutex> lk(h->node_locks[doc & 4095]);
+			if (doc < layer.edges.size())
+				scratch = layer.edges[doc];
+			edges_p = &scratch;
+		}
+		for (uint32_t e : *edges_p) {
 			if (!visited.insert(e).second)
 				continue;
 			// elements.get_vector -> None for removed elements
@@ -1941,7 +1982,8 @@ static void select_neighbors(sdbv_hnsw *h, const Layer &layer, uint32_t q_id,
 static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
                        const float *q_pt, double q_norm, PQ eps, bool locked) {
 	PQ w = eps;
-	std::unordered_set<uint32_t> visited;
+	static thread_local VisitSet visited;
+	visited.begin(h->vecs.size() / h->d);
 	for (auto &e : eps.to_vec())
 		visited.insert(e.second);
 	search_layer_host(h, layer, q_pt, q_norm, eps, visited, w, h->efc, locked);
