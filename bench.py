@@ -237,7 +237,12 @@ def main():
 
     # --- cpu_baseline: the oracle (kind "port") on host cores, rank0/N=1 ---
     cpu_baseline = None
-    if world == 1 and not args.no_cpu_baseline:
+    if args.hnsw:
+        # the oracle's HNSW search would need the oracle-built graph at
+        # bench scale (~19 ms/insert) — not affordable per run; the
+        # brute-force oracle scan is the wrong comparison for this mode
+        pass
+    elif world == 1 and not args.no_cpu_baseline:
         import oracle
         srows = min(args.cpu_sample_rows, rows)
         sample = oracle.gen_f32(args.seed, 0, srows, args.dim)
