@@ -66,3 +66,60 @@ def test_sharded_topk_equals_global():
     gids, gdists = oracle.topk_f32("cosine", corpus, q, K)
     assert np.array_equal(mids, gids)
     assert np.array_equal(mdists, gdists)
+
+
+def _batch_worker(rank, world, port, out_q):
+    """Batched-query variant: the exact per-query gather+merge sequence
+    bench.py's one_batch runs over RCCL (b x K pairs per rank)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        B = 6
+        b0, e0 = shard_range(N, rank, world)
+        corpus = oracle.gen_f32(0x5DB1, b0, e0 - b0, D)
+        Q = oracle.gen_f32(0xBEEF, 0, B, D)
+        flat = torch.zeros(B * K, 2, dtype=torch.float64)
+        for j in range(B):
+            ids, dists = oracle.topk_f32("cosine", corpus, Q[j], K)
+            ids = ids + b0
+            pad = K - len(ids)
+            if pad:
+                ids = np.concatenate(
+                    [ids, np.full(pad, np.iinfo(np.uint64).max, np.uint64)])
+                dists = np.concatenate([dists, np.full(pad, np.inf)])
+            flat[j * K:(j + 1) * K, 0] = torch.from_numpy(dists.copy())
+            flat[j * K:(j + 1) * K, 1] = torch.from_numpy(
+                ids.view(np.float64).copy())
+        gathered = [torch.zeros_like(flat) for _ in range(world)]
+        dist.all_gather(gathered, flat)
+        if rank == 0:
+            out = []
+            g = [t.numpy().reshape(B, K, 2) for t in gathered]
+            for j in range(B):
+                out.append(merge_topk(
+                    [x[j, :, 1].copy().view(np.uint64) for x in g],
+                    [x[j, :, 0] for x in g], K))
+            out_q.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_sharded_batch_topk_equals_global():
+    ctxm = mp.get_context("spawn")
+    out_q = ctxm.Queue()
+    port = 29513
+    procs = [ctxm.Process(target=_batch_worker, args=(r, WORLD, port, out_q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    merged = out_q.get(timeout=110)
+    for p in procs:
+        p.join(timeout=30)
+    corpus = oracle.gen_f32(0x5DB1, 0, N, D)
+    Q = oracle.gen_f32(0xBEEF, 0, 6, D)
+    for j in range(6):
+        gids, gdists = oracle.topk_f32("cosine", corpus, Q[j], K)
+        assert np.array_equal(merged[j][0], gids), f"q{j}"
+        assert np.array_equal(merged[j][1], gdists), f"q{j}"
