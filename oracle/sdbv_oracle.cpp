@@ -1706,6 +1706,34 @@ uint32_t orc_index_knn_filtered(orc_index *ix, const float *q, uint32_t k,
 	return n;
 }
 
+// Test hooks: drive OrcPQ directly so tests can replay the reference's own
+// test_double_priority_queue sequence (knn.rs:735-790).
+void *orc_test_pq_new() { return new OrcPQ(); }
+void orc_test_pq_free(void *q) { delete (OrcPQ *)q; }
+uint64_t orc_test_pq_len(void *q) { return ((OrcPQ *)q)->n; }
+void orc_test_pq_push(void *q, double d, uint64_t id) {
+	((OrcPQ *)q)->push(d, id);
+}
+int orc_test_pq_peek_first(void *q, double *d, uint64_t *id) {
+	return ((OrcPQ *)q)->peek_first(d, id) ? 1 : 0;
+}
+int orc_test_pq_peek_last_dist(void *q, double *d) {
+	auto *pq = (OrcPQ *)q;
+	if (pq->n == 0)
+		return 0;
+	*d = pq->peek_last_dist(0);
+	return 1;
+}
+int orc_test_pq_pop_first(void *q, double *d, uint64_t *id) {
+	return ((OrcPQ *)q)->pop_first(d, id) ? 1 : 0;
+}
+int orc_test_pq_pop_last(void *q, double *d, uint64_t *id) {
+	auto *pq = (OrcPQ *)q;
+	if (pq->n == 0)
+		return 0;
+	return pq->pop_last(d, id) ? 1 : 0;
+}
+
 // Test hook: drive OrcIds64 directly so tests can restate the reference's
 // own test_ids sequence (knn.rs:669-717) — variant transitions (Some/None)
 // and contents/order, bit for bit.

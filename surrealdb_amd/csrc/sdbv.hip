@@ -2773,6 +2773,46 @@ int sdbv_hnsw_remove(sdbv_hnsw *h, uint64_t e_id) {
 	return hnsw::hnsw_remove(h, (uint32_t)e_id) ? 1 : 0;
 }
 
+// Test hooks: drive the DoublePriorityQueue restatement directly so tests
+// can replay the reference's own test_double_priority_queue sequence
+// (knn.rs:735-790) against this implementation.
+void *sdbv_test_pq_new() { return new hnsw::PQ(); }
+void sdbv_test_pq_free(void *q) { delete (hnsw::PQ *)q; }
+uint64_t sdbv_test_pq_len(void *q) { return ((hnsw::PQ *)q)->n; }
+void sdbv_test_pq_push(void *q, double d, uint64_t id) {
+	((hnsw::PQ *)q)->push(d, (uint32_t)id);
+}
+int sdbv_test_pq_peek_first(void *q, double *d, uint64_t *id) {
+	uint32_t i;
+	if (!((hnsw::PQ *)q)->peek_first(d, &i))
+		return 0;
+	*id = i;
+	return 1;
+}
+int sdbv_test_pq_peek_last_dist(void *q, double *d) {
+	auto *pq = (hnsw::PQ *)q;
+	if (pq->n == 0)
+		return 0;
+	*d = pq->peek_last_dist(0);
+	return 1;
+}
+int sdbv_test_pq_pop_first(void *q, double *d, uint64_t *id) {
+	uint32_t i;
+	if (!((hnsw::PQ *)q)->pop_first(d, &i))
+		return 0;
+	*id = i;
+	return 1;
+}
+int sdbv_test_pq_pop_last(void *q, double *d, uint64_t *id) {
+	auto *pq = (hnsw::PQ *)q;
+	if (pq->n == 0)
+		return 0;
+	*d = pq->v.back().d;
+	*id = pq->v.back().id;
+	pq->pop_last();
+	return 1;
+}
+
 } // extern "C"
 
 // ===========================================================================
