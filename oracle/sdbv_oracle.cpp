@@ -1706,6 +1706,79 @@ uint32_t orc_index_knn_filtered(orc_index *ix, const float *q, uint32_t k,
 	return n;
 }
 
+// Test hooks: a standalone OrcLayer driven by the reference's
+// UndirectedGraph operations (graph.rs:43-130), so tests can replay
+// graph.rs:166-233 test_undirected_graph verbatim. The internal build /
+// remove paths use the same primitives (add_node, bidirectional add,
+// set_node on members, remove+back-edge cleanup).
+void *orc_test_graph_new(uint32_t m_max) {
+	auto *l = new OrcLayer();
+	l->m_max = m_max;
+	return l;
+}
+void orc_test_graph_free(void *g) { delete (OrcLayer *)g; }
+// graph.rs:43-50 add_empty_node: 1 if inserted, 0 if it already existed
+int orc_test_graph_add_empty_node(void *g, uint64_t id) {
+	auto *l = (OrcLayer *)g;
+	if (l->has(id))
+		return 0;
+	l->add_node(id);
+	l->edges[id].clear();
+	return 1;
+}
+// graph.rs:52-64 add_node_and_bidirectional_edges (implicit target create)
+void orc_test_graph_add_bidir(void *g, uint64_t id, const uint64_t *edges,
+                              uint32_t n) {
+	auto *l = (OrcLayer *)g;
+	for (uint32_t i = 0; i < n; i++) {
+		l->add_node(edges[i]);
+		l->edges[edges[i]].push_back((uint32_t)id);
+	}
+	l->add_node(id);
+	l->edges[id].clear();
+	for (uint32_t i = 0; i < n; i++)
+		l->edges[id].push_back((uint32_t)edges[i]);
+}
+// graph.rs:66-68 set_node (creates a missing node)
+void orc_test_graph_set_node(void *g, uint64_t id, const uint64_t *edges,
+                             uint32_t n) {
+	auto *l = (OrcLayer *)g;
+	l->add_node(id);
+	l->edges[id].clear();
+	for (uint32_t i = 0; i < n; i++)
+		l->edges[id].push_back((uint32_t)edges[i]);
+}
+// graph.rs:70-81 remove_node_and_bidirectional_edges: returns the removed
+// node's edge count (into out) or -1 (None)
+int orc_test_graph_remove(void *g, uint64_t id, uint64_t *out,
+                          uint32_t cap) {
+	auto *l = (OrcLayer *)g;
+	if (!l->has(id))
+		return -1;
+	std::vector<uint32_t> f = l->edges[id];
+	l->edges[id].clear();
+	l->in_layer[id] = 0;
+	for (uint32_t e : f) {
+		auto &fe = l->edges[e];
+		fe.erase(std::remove(fe.begin(), fe.end(), (uint32_t)id), fe.end());
+	}
+	uint32_t m = std::min<uint32_t>((uint32_t)f.size(), cap);
+	for (uint32_t i = 0; i < m; i++)
+		out[i] = f[i];
+	return (int)f.size();
+}
+// edge readout: -1 if the node is absent (get_edges None), else count
+int orc_test_graph_edges(void *g, uint64_t id, uint64_t *out, uint32_t cap) {
+	auto *l = (OrcLayer *)g;
+	if (!l->has(id))
+		return -1;
+	const auto &e = l->edges[id];
+	uint32_t m = std::min<uint32_t>((uint32_t)e.size(), cap);
+	for (uint32_t i = 0; i < m; i++)
+		out[i] = e[i];
+	return (int)e.size();
+}
+
 // Test hooks: drive OrcPQ directly so tests can replay the reference's own
 // test_double_priority_queue sequence (knn.rs:735-790).
 void *orc_test_pq_new() { return new OrcPQ(); }
