@@ -120,13 +120,40 @@ def main():
             return None
         return ids, dists
 
+    def one_hnsw_batch(si):
+        """One step of the batched HNSW path: args.batch concurrent queries
+        on the persistent in-kernel search (one workgroup per query)."""
+        b = args.batch
+        q0 = (si * b) % max(len(queries) - b, 1)
+        ids, dists, ns = hnsw_index.knn_search_batch(queries[q0:q0 + b],
+                                                     args.k, args.ef)
+        if world > 1:
+            # per-query merge across shard replicas (same protocol as the
+            # brute-force batch path; HNSW shards queries over replicas of
+            # row shards)
+            flat = torch.empty(b * args.k, 2, dtype=torch.float64,
+                               device=device)
+            gids = ids.astype(np.uint64) + row_offset
+            for j in range(b):
+                gids[j, ns[j]:] = np.iinfo(np.uint64).max
+                dists[j, ns[j]:] = np.inf
+            flat[:, 0] = torch.from_numpy(dists.reshape(-1).copy()).to(device)
+            flat[:, 1] = torch.from_numpy(
+                gids.reshape(-1).view(np.float64).copy()).to(device)
+            gathered = [torch.empty_like(flat) for _ in range(world)]
+            dist.all_gather(gathered, flat)
+            if rank == 0:
+                g = [t.cpu().numpy().reshape(b, args.k, 2) for t in gathered]
+                return [merge_topk(
+                    [x[j, :, 1].copy().view(np.uint64) for x in g],
+                    [x[j, :, 0] for x in g], args.k) for j in range(b)]
+            return None
+        return ids, dists
+
     def one_query(qi):
         if args.batch > 0:
             if hnsw_index is not None:
-                raise SystemExit(
-                    "--hnsw with --batch is not a bench mode yet: the "
-                    "batched path would silently run the brute-force GEMM, "
-                    "not the HNSW persistent kernel")
+                return one_hnsw_batch(qi)
             return one_batch(qi)
         if hnsw_index is not None:
             ids, dists = hnsw_index.knn_search(queries[qi % len(queries)],
@@ -157,6 +184,19 @@ def main():
     # --- warmup ---
     for i in range(args.warmup):
         one_query(i)
+
+    # hnsw-batch roofline calibration: the persistent kernel does not count
+    # its gathered rows, but it visits exactly the per-hop path's set (same
+    # best-first queue), so sample the per-hop gather count per query here,
+    # outside the timed region.
+    hnsw_gather_avg = None
+    if args.hnsw and args.batch > 0:
+        tot = 0
+        nsample = min(32, len(queries))
+        for j in range(nsample):
+            hnsw_index.knn_search(queries[j], args.k, args.ef)
+            tot += ctx.stats()["last_rows_scanned"]
+        hnsw_gather_avg = tot / nsample
 
     # --- timed region: EXACTLY args.steps steps ---
     if dist:
@@ -194,7 +234,20 @@ def main():
 
     scan_ms_avg = scan_ms_acc / args.steps
     traffic = os.environ.get("SDBV_TRAFFIC_BYTES_PER_LAUNCH")
-    if args.batch > 0:
+    if args.hnsw and args.batch > 0:
+        # persistent-kernel batch: gather-bound; algorithmic bytes per
+        # launch = (sampled per-hop gather rows) x batch x row bytes
+        alg_bytes = hnsw_gather_avg * args.batch * args.dim * 4
+        achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved_gbs, 1),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+            "traffic": float(traffic) if traffic else None,
+        }
+    elif args.batch > 0:
         # dominant kernel = the f32 MFMA GEMM; stats.last_scan_kernel_ms is
         # the summed sgemm time of one step's chunks
         alg_flop = 2.0 * args.batch * rows * args.dim
@@ -282,6 +335,10 @@ def main():
         "data": "synthetic",
         "config": {
             "workload": (
+                f"HNSW (M=16, ef={args.ef}) {args.metric} K={args.k}, "
+                f"batch={args.batch} persistent-kernel path, {rows} "
+                f"rows/GPU (configs[2] shape, batched)"
+                if args.hnsw and args.batch > 0 else
                 f"brute-force {args.metric} KNN, {rows} rows/GPU x "
                 f"{args.dim}-dim f32, K={args.k}, batch={args.batch} "
                 f"MFMA path (BASELINE configs[3])"
