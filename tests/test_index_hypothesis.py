@@ -1,0 +1,77 @@
+"""Property-based differential testing of the index layer: hypothesis
+generates insert/update/delete/apply/search programs and every observable
+(result kinds/ids/distance bits, doc counts, graph CSR) must be identical
+between the product (host-only C++) and the oracle — two independent
+restatements of hnsw/index.rs + docs.rs + knn.rs."""
+import numpy as np
+from hypothesis import given, settings, HealthCheck
+from hypothesis import strategies as st
+
+import oracle
+import surrealdb_amd as sa
+
+D = 12
+N_KEYS = 24
+ROWS = oracle.gen_f32(0xF00D, 0, 256, D)
+
+
+ops = st.lists(
+    st.tuples(
+        st.sampled_from(["write", "delete", "apply", "search", "fsearch",
+                         "dup_write"]),
+        st.integers(min_value=0, max_value=N_KEYS - 1),
+        st.integers(min_value=0, max_value=255),
+    ),
+    min_size=4, max_size=60,
+)
+
+
+@settings(max_examples=150, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(ops=ops, metric=st.sampled_from(["euclidean", "cosine"]))
+def test_program_equivalence(ops, metric):
+    p = sa.index_create_host(D, metric=metric, m=4, m0=8, efc=24, seed=3)
+    o = oracle.Index(D, metric=metric, m=4, m0=8, efc=24, seed=3)
+    try:
+        live = {}
+        for op, key, ridx in ops:
+            if op == "write":
+                v = ROWS[ridx]
+                old = live.get(key)
+                p.enqueue(key, old, v)
+                o.enqueue(key, old, v)
+                live[key] = v
+            elif op == "dup_write":
+                # multiple keys sharing ONE vector (Ids64/VecDocs churn)
+                v = ROWS[0]
+                old = live.get(key)
+                p.enqueue(key, old, v)
+                o.enqueue(key, old, v)
+                live[key] = v
+            elif op == "delete":
+                if key in live:
+                    p.enqueue(key, live[key], None)
+                    o.enqueue(key, live[key], None)
+                    del live[key]
+            elif op == "apply":
+                assert p.apply_pendings() == o.apply_pendings()
+                assert p.doc_count() == o.doc_count()
+            elif op in ("search", "fsearch"):
+                q = ROWS[ridx] + np.float32(0.05)
+                if op == "search":
+                    pk, pi, pd = p.knn_search(q, 5, 12)
+                    ok, oi, od = o.knn_search(q, 5, 12)
+                else:
+                    pred = lambda kind, i: (i + key) % 3 != 0
+                    pk, pi, pd = p.knn_search_filtered(q, 5, 12, pred)
+                    ok, oi, od = o.knn_search_filtered(q, 5, 12, pred)
+                assert np.array_equal(pk, ok)
+                assert np.array_equal(pi, oi)
+                assert np.array_equal(pd, od)
+        assert p.apply_pendings() == o.apply_pendings()
+        ph, oh = p.hnsw(), o.hnsw()
+        po, pe = ph.l0_csr()
+        oo, oe = oh.l0_csr()
+        assert np.array_equal(po, oo) and np.array_equal(pe, oe)
+    finally:
+        p.destroy()
