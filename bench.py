@@ -45,6 +45,9 @@ def main():
                          "K=10 single query; index built at bench start "
                          "(parallel host build, outside the timed region)")
     ap.add_argument("--ef", type=int, default=64)
+    ap.add_argument("--hnsw-perhop", action="store_true",
+                    help="HNSW mode: use the per-hop gather path instead of "
+                         "the persistent kernel for single queries")
     ap.add_argument("--seed", type=lambda x: int(x, 0), default=0x5DB1)
     ap.add_argument("--cpu-sample-rows", type=int, default=2_000_000,
                     help="bounded sample for the cpu_baseline leg")
@@ -156,8 +159,15 @@ def main():
                 return one_hnsw_batch(qi)
             return one_batch(qi)
         if hnsw_index is not None:
-            ids, dists = hnsw_index.knn_search(queries[qi % len(queries)],
-                                               args.k, args.ef)
+            q = queries[qi % len(queries)]
+            if args.hnsw_perhop:
+                ids, dists = hnsw_index.knn_search(q, args.k, args.ef)
+            else:
+                # persistent kernel (whole best-first loop in ONE launch —
+                # same exact results, validated in tests/test_gpu_hnsw.py)
+                bids, bdists, bns = hnsw_index.knn_search_batch(
+                    q.reshape(1, -1), args.k, args.ef)
+                ids, dists = bids[0][:bns[0]], bdists[0][:bns[0]]
             ids = ids + row_offset  # shard-local ordinals -> global ids
         else:
             ids, dists = ctx.knn_bruteforce(1, queries[qi % len(queries)],
@@ -190,7 +200,7 @@ def main():
     # best-first queue), so sample the per-hop gather count per query here,
     # outside the timed region.
     hnsw_gather_avg = None
-    if args.hnsw and args.batch > 0:
+    if args.hnsw and not args.hnsw_perhop:
         tot = 0
         nsample = min(32, len(queries))
         for j in range(nsample):
@@ -234,20 +244,7 @@ def main():
 
     scan_ms_avg = scan_ms_acc / args.steps
     traffic = os.environ.get("SDBV_TRAFFIC_BYTES_PER_LAUNCH")
-    if args.hnsw and args.batch > 0:
-        # persistent-kernel batch: gather-bound; algorithmic bytes per
-        # launch = (sampled per-hop gather rows) x batch x row bytes
-        alg_bytes = hnsw_gather_avg * args.batch * args.dim * 4
-        achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
-        roofline = {
-            "bound": "hbm",
-            "achieved": round(achieved_gbs, 1),
-            "peak": HBM_PEAK_GBS,
-            "unit": "GB/s",
-            "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-            "traffic": float(traffic) if traffic else None,
-        }
-    elif args.batch > 0:
+    if args.batch > 0 and not args.hnsw:
         # dominant kernel = the f32 MFMA GEMM; stats.last_scan_kernel_ms is
         # the summed sgemm time of one step's chunks
         alg_flop = 2.0 * args.batch * rows * args.dim
@@ -263,8 +260,13 @@ def main():
         }
     elif args.hnsw:
         # HNSW is latency/gather-bound: achieved = gathered rows
-        # x row bytes over the GPU gather sections (incl. per-hop transfers)
-        alg_bytes = (rows_scanned_acc / args.steps) * args.dim * 4
+        # x row bytes over the GPU search sections. Per-hop mode counts its
+        # gathers live; the persistent kernel visits the identical set, so
+        # its count comes from the pre-timed-region per-hop calibration.
+        if hnsw_gather_avg is not None:
+            alg_bytes = hnsw_gather_avg * max(args.batch, 1) * args.dim * 4
+        else:
+            alg_bytes = (rows_scanned_acc / args.steps) * args.dim * 4
         achieved_gbs = alg_bytes / (scan_ms_avg * 1e-3) / 1e9
         roofline = {
             "bound": "hbm",
@@ -345,7 +347,7 @@ def main():
                 if args.batch > 0 else
                 f"HNSW (M=16, ef={args.ef}) {args.metric} K={args.k} single "
                 f"query, {rows} rows/GPU (BASELINE configs[2] shape; "
-                f"latency/gather-bound)"
+                f"{'per-hop gather' if args.hnsw_perhop else 'persistent kernel'})"
                 if args.hnsw else
                 f"brute-force {args.metric} KNN, {rows} rows/GPU x "
                 f"{args.dim}-dim f32, K={args.k}, single query "
