@@ -1041,11 +1041,19 @@ int orc_hnsw_check_props(orc_hnsw *h) {
 	for (size_t l = 0; l < h->layers.size(); l++) {
 		const OrcLayer &layer = h->layers[l];
 		for (size_t id = 0; id < layer.edges.size(); id++) {
+			if (!layer.has(id)) {
+				if (!layer.edges[id].empty())
+					return -3; // absent nodes have no edge list
+				continue;
+			}
 			if (layer.edges[id].size() > layer.m_max + 0)
 				return -1;
 			for (uint32_t e : layer.edges[id])
 				if (e == id)
 					return -2;
+			// layer.rs check_props: every graph node is a LIVE element
+			if (id >= h->elem_present.size() || !h->elem_present[id])
+				return -4;
 		}
 	}
 	return 0;

@@ -4363,6 +4363,10 @@ int sdbv_index_check_props(sdbv_index *ix, uint64_t expected_count) {
 			}
 			if (layer.edges[id].size() > layer.m_max)
 				return -13;
+			// layer.rs check_props: every graph node is a LIVE element
+			// (edge targets may dangle, nodes may not)
+			if (id >= h->elem_present.size() || !h->elem_present[id])
+				return -15;
 			for (uint32_t e : layer.edges[id])
 				if (e == id || e >= h->next_id)
 					return -14;
