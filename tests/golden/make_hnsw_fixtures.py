@@ -45,6 +45,9 @@ FIXTURES = [
     # cosine d=128 test
     ("seq128_cos", 128, 5000, "cosine", 8, 16, 100, 0x11, 0x77,
      10, 0x88, 10, (40,)),
+    # deeper-graph coverage (round-2 scale; ~5 min oracle build)
+    ("seq768_cos_16k", 768, 16000, "cosine", 16, 32, 150, 0x10, 0x5DB1,
+     48, 0xBEEF, 10, (64,)),
 ]
 
 

@@ -22,7 +22,8 @@ pytestmark = pytest.mark.gpu
 
 GOLDEN_DIR = os.path.join(os.path.dirname(__file__), "golden")
 
-FIXTURE_NAMES = ["seq768_cos", "seq768_euc", "seq128_cos"]
+FIXTURE_NAMES = ["seq768_cos", "seq768_euc", "seq128_cos",
+                 "seq768_cos_16k"]
 
 
 def load_golden(name, limit):
