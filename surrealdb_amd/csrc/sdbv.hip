@@ -3269,26 +3269,49 @@ static int idx_graph_search_gpu(sdbv_index *ix, const float *q, uint32_t k,
 
 namespace kvc {
 
+// bincode-style unsigned varint (revision 0.17.0): < 251 one byte;
+// 0xfb + u16 LE; 0xfc + u32 LE; 0xfd + u64 LE.
 static void put_varint(std::vector<uint8_t> &b, uint64_t v) {
-	while (v >= 0x80) {
-		b.push_back((uint8_t)(v & 0x7F) | 0x80);
-		v >>= 7;
+	if (v < 251) {
+		b.push_back((uint8_t)v);
+	} else if (v <= 0xFFFF) {
+		b.push_back(0xFB);
+		b.push_back((uint8_t)v);
+		b.push_back((uint8_t)(v >> 8));
+	} else if (v <= 0xFFFFFFFFull) {
+		b.push_back(0xFC);
+		for (int i = 0; i < 4; i++)
+			b.push_back((uint8_t)(v >> (8 * i)));
+	} else {
+		b.push_back(0xFD);
+		for (int i = 0; i < 8; i++)
+			b.push_back((uint8_t)(v >> (8 * i)));
 	}
-	b.push_back((uint8_t)v);
 }
 static bool get_varint(const uint8_t *&p, const uint8_t *end, uint64_t *v) {
-	uint64_t out = 0;
-	int shift = 0;
-	while (p < end && shift < 64) {
-		uint8_t c = *p++;
-		out |= (uint64_t)(c & 0x7F) << shift;
-		if (!(c & 0x80)) {
-			*v = out;
-			return true;
-		}
-		shift += 7;
+	if (p >= end)
+		return false;
+	uint8_t c = *p++;
+	int extra;
+	if (c < 251) {
+		*v = c;
+		return true;
+	} else if (c == 0xFB) {
+		extra = 2;
+	} else if (c == 0xFC) {
+		extra = 4;
+	} else if (c == 0xFD) {
+		extra = 8;
+	} else {
+		return false; // 0xFE (u128) unsupported
 	}
-	return false;
+	if (end - p < extra)
+		return false;
+	uint64_t out = 0;
+	for (int i = 0; i < extra; i++)
+		out |= (uint64_t)(*p++) << (8 * i);
+	*v = out;
+	return true;
 }
 template <typename T> static void put_le(std::vector<uint8_t> &b, T v) {
 	uint8_t tmp[sizeof(T)];
@@ -3418,16 +3441,18 @@ struct HnswStateKV {
 };
 static void enc_state(std::vector<uint8_t> &b, const HnswStateKV &s) {
 	put_varint(b, 1); // revision
-	b.push_back(s.has_ep ? 1 : 0);
+	b.push_back(s.has_ep ? 1 : 0); // Option tag
 	if (s.has_ep)
-		put_le(b, s.enter_point);
-	put_le(b, s.next_element_id);
-	put_le(b, (uint64_t)0); // layer0.version (not tracked here)
-	put_le(b, (uint32_t)0); // layer0.chunks == 0 (post-Hl format)
+		put_varint(b, s.enter_point); // ElementId: unsigned -> varint
+	put_varint(b, s.next_element_id);
+	put_varint(b, 1);            // layer0 LayerState revision
+	put_varint(b, 0);            // layer0.version (not tracked here)
+	put_varint(b, 0);            // layer0.chunks == 0 (post-Hl format)
 	put_varint(b, s.n_upper_layers);
 	for (uint64_t i = 0; i < s.n_upper_layers; i++) {
-		put_le(b, (uint64_t)0);
-		put_le(b, (uint32_t)0);
+		put_varint(b, 1); // LayerState revision
+		put_varint(b, 0);
+		put_varint(b, 0);
 	}
 }
 static bool dec_state(const uint8_t *p, const uint8_t *end, HnswStateKV *s) {
@@ -3438,20 +3463,22 @@ static bool dec_state(const uint8_t *p, const uint8_t *end, HnswStateKV *s) {
 		return false;
 	uint8_t tag = *p++;
 	s->has_ep = tag != 0;
-	if (s->has_ep && !get_le(p, end, &s->enter_point))
+	if (s->has_ep && !get_varint(p, end, &s->enter_point))
 		return false;
-	if (!get_le(p, end, &s->next_element_id))
+	if (!get_varint(p, end, &s->next_element_id))
 		return false;
-	uint64_t v64;
-	uint32_t v32;
-	if (!get_le(p, end, &v64) || !get_le(p, end, &v32))
+	uint64_t lrev, version, chunks;
+	if (!get_varint(p, end, &lrev) || lrev != 1 ||
+	    !get_varint(p, end, &version) || !get_varint(p, end, &chunks))
 		return false; // layer0 state
-	if (v32 != 0)
+	if (chunks != 0)
 		return false; // legacy Hl chunks unsupported (post-migration only)
 	if (!get_varint(p, end, &s->n_upper_layers))
 		return false;
 	for (uint64_t i = 0; i < s->n_upper_layers; i++)
-		if (!get_le(p, end, &v64) || !get_le(p, end, &v32) || v32 != 0)
+		if (!get_varint(p, end, &lrev) || lrev != 1 ||
+		    !get_varint(p, end, &version) ||
+		    !get_varint(p, end, &chunks) || chunks != 0)
 			return false;
 	return true;
 }
@@ -3461,20 +3488,20 @@ static bool dec_state(const uint8_t *p, const uint8_t *end, HnswStateKV *s) {
 // (RoaringTreemap) unsupported -> false).
 static void enc_element_docs(std::vector<uint8_t> &b, uint64_t e_id,
                              const vdocs::Ids64 &docs) {
-	put_varint(b, 1); // revision
-	put_le(b, e_id);
-	put_varint(b, 1); // Ids64 revision
+	put_varint(b, 1);     // revision
+	put_varint(b, e_id);  // ElementId: unsigned -> varint
+	put_varint(b, 1);     // Ids64 revision
 	size_t n = docs.v.size();
 	put_varint(b, n == 0 ? 0 : (n <= 8 ? n : 9));
 	for (uint64_t d : docs.v)
-		put_le(b, d);
+		put_varint(b, d); // DocId: unsigned -> varint
 }
 static bool dec_element_docs(const uint8_t *p, const uint8_t *end,
                              uint64_t *e_id, vdocs::Ids64 *docs) {
 	uint64_t rev, variant;
 	if (!get_varint(p, end, &rev) || rev != 1)
 		return false;
-	if (!get_le(p, end, e_id))
+	if (!get_varint(p, end, e_id))
 		return false;
 	if (!get_varint(p, end, &rev) || rev != 1)
 		return false;
@@ -3488,7 +3515,7 @@ static bool dec_element_docs(const uint8_t *p, const uint8_t *end,
 		return false; // Bits / roaring not supported in this revision
 	for (uint64_t i = 0; i < variant; i++) {
 		uint64_t d;
-		if (!get_le(p, end, &d))
+		if (!get_varint(p, end, &d))
 			return false;
 		docs->v.push_back(d);
 	}

@@ -90,10 +90,10 @@ def test_hv_key_pinned_f32_golden():
     esc4.append(0)
     kv_key = PREFIX + b"v" + bytes(esc4)
     assert kv_key in d, list(k for k in d if b"!hv" in k)
-    # ElementDocs value: rev=1 + e_id u64 LE + Ids64 rev=1 + variant One +
-    # doc u64 LE
-    assert d[kv_key] == (bytes([1]) + (0).to_bytes(8, "little") +
-                         bytes([1, 1]) + (0).to_bytes(8, "little"))
+    # ElementDocs value: rev=1 + e_id varint + Ids64 rev=1 + variant One +
+    # doc varint (unsigned ints are bincode-style varints in the revision
+    # wire format — see the catalog compat fixtures)
+    assert d[kv_key] == bytes([1, 0, 1, 1, 0])
     ix.destroy()
 
 
@@ -168,3 +168,50 @@ def test_loader_rejects_garbage_and_skips_host_keys():
                         efc=20, seed=2)
     g.destroy()
     g2.destroy()
+
+
+def test_varint_markers_above_250():
+    """revision 0.17.0 unsigned varint: one byte < 251; 0xfb + u16 LE above
+    (catalog/compat/v3_0_0.rs golden durations: 900 -> fb 84 03,
+    3600 -> fb 10 0e, 86400 -> fc 80 51 01 00). A d=768 vector length and
+    large doc ids must take the marker form, and round-trip."""
+    d = 768
+    h = sa.hnsw_create_host(d, metric="euclidean", m=4, m0=8, efc=20, seed=1)
+    v = oracle.gen_f32(1, 0, 1, d)[0]
+    h.insert(v)
+    pairs = dict(h.dump_kv())
+    he_key = PREFIX + b"e" + (0).to_bytes(8, "big")
+    # rev=1, variant F32=1, len 768 = fb 00 03, then payload
+    assert pairs[he_key][:5] == bytes([1, 1, 0xFB, 0x00, 0x03])
+    assert pairs[he_key][5:] == v.tobytes()
+    g2 = sa.load_kv_hnsw(list(pairs.items()), d, metric="euclidean", m=4,
+                         m0=8, efc=20, seed=1)
+    a, b = h.l0_csr(), g2.l0_csr()
+    assert np.array_equal(a[0], b[0])
+    h.destroy()
+    g2.destroy()
+    # doc ids above 250 in ElementDocs
+    ix = sa.index_create_host(8, metric="euclidean", m=4, m0=8, efc=20)
+    w = oracle.gen_f32(2, 0, 1, 8)[0]
+    ix.enqueue(7, None, w)
+    ix.apply_pendings()
+    # force a large doc id via the host re-bind path
+    pairs = ix.dump_kv()
+    ix2 = sa.load_kv_index(pairs, 0, 8, metric="euclidean", m=4, m0=8,
+                           efc=20, doc_keys={0: 7})
+    ix2.enqueue(900, None, oracle.gen_f32(3, 0, 1, 8)[0])
+    for j in range(300):  # push next_doc_id past 251
+        ix2.enqueue(1000 + j, None, oracle.gen_f32(4, j, 1, 8)[0])
+    ix2.apply_pendings()
+    pairs2 = ix2.dump_kv()
+    ix3 = sa.load_kv_index(pairs2, 0, 8, metric="euclidean", m=4, m0=8,
+                           efc=20)
+    assert ix3.doc_count() == 0  # doc keys host-bound; docs live in Hv
+    k3, i3, d3 = ix3.knn_search(w, 1, 16)
+    assert (k3[0], i3[0], d3[0]) == (0, 0, 0.0)
+    # a doc id >= 251 survives the round trip through the varint form
+    k3, i3, d3 = ix3.knn_search(oracle.gen_f32(4, 299, 1, 8)[0], 1, 16)
+    assert k3[0] == 0 and i3[0] == 301 and d3[0] == 0.0
+    ix.destroy()
+    ix2.destroy()
+    ix3.destroy()
