@@ -3380,8 +3380,8 @@ static void enc_vec_f32(std::vector<uint8_t> &b, const float *v, uint32_t d) {
 	for (uint32_t i = 0; i < d; i++)
 		put_le(b, v[i]);
 }
-static bool dec_vec_f32(const uint8_t *p, const uint8_t *end,
-                        std::vector<float> &out) {
+static bool dec_vec_f32_cursor(const uint8_t *&p, const uint8_t *end,
+                               std::vector<float> &out) {
 	uint64_t rev, variant, len;
 	if (!get_varint(p, end, &rev) || rev != 1)
 		return false;
@@ -3405,6 +3405,10 @@ static bool dec_vec_f32(const uint8_t *p, const uint8_t *end,
 		}
 	}
 	return true;
+}
+static bool dec_vec_f32(const uint8_t *p, const uint8_t *end,
+                        std::vector<float> &out) {
+	return dec_vec_f32_cursor(p, end, out);
 }
 
 // graph.rs:104-113 node_to_val: u16 BE edge count + u64 BE edge ids.
@@ -3522,6 +3526,85 @@ static bool dec_element_docs(const uint8_t *p, const uint8_t *end,
 	return true;
 }
 
+// revisioned VectorPendingUpdate (hnsw/mod.rs:93-116): VectorId enum
+// (DocId(u64 varint) = 0 | RecordKey(RecordIdKey) = 1; RecordIdKey
+// revisioned enum — only Number(i64, fixed LE) maps onto the u64
+// record-key handles of this boundary, other key kinds are the host's to
+// apply) + old/new Vec<SerializedVector>.
+struct PendingKV {
+	uint8_t kind; // 0 DocId, 1 RecordKey(Number handle)
+	uint64_t id;
+	std::vector<float> olds, news; // n*d concatenated
+};
+static bool dec_vec_list(const uint8_t *&p, const uint8_t *end, uint32_t d,
+                         std::vector<float> &out) {
+	uint64_t n;
+	if (!get_varint(p, end, &n))
+		return false;
+	out.clear();
+	std::vector<float> one;
+	for (uint64_t i = 0; i < n; i++) {
+		if (!dec_vec_f32_cursor(p, end, one) || one.size() != d)
+			return false;
+		out.insert(out.end(), one.begin(), one.end());
+	}
+	return true;
+}
+static bool dec_pending(const uint8_t *p, const uint8_t *end, uint32_t d,
+                        PendingKV *out) {
+	uint64_t rev, variant;
+	if (!get_varint(p, end, &rev) || rev != 1)
+		return false; // VectorPendingUpdate revision
+	if (!get_varint(p, end, &rev) || rev != 1)
+		return false; // VectorId revision
+	if (!get_varint(p, end, &variant))
+		return false;
+	if (variant == 0) { // DocId(u64)
+		out->kind = 0;
+		if (!get_varint(p, end, &out->id))
+			return false;
+	} else if (variant == 1) { // RecordKey(RecordIdKey)
+		uint64_t krev, kvar;
+		if (!get_varint(p, end, &krev) || krev != 1)
+			return false;
+		if (!get_varint(p, end, &kvar))
+			return false;
+		if (kvar != 0)
+			return false; // only Number keys map to u64 handles here
+		int64_t num;
+		if (!get_le(p, end, &num))
+			return false; // i64: signed -> fixed LE
+		out->kind = 1;
+		out->id = (uint64_t)num;
+	} else {
+		return false;
+	}
+	if (!dec_vec_list(p, end, d, out->olds))
+		return false;
+	if (!dec_vec_list(p, end, d, out->news))
+		return false;
+	return true;
+}
+static void enc_pending(std::vector<uint8_t> &b, uint32_t d,
+                        const PendingKV &pk) {
+	put_varint(b, 1); // VectorPendingUpdate revision
+	put_varint(b, 1); // VectorId revision
+	put_varint(b, pk.kind);
+	if (pk.kind == 0) {
+		put_varint(b, pk.id);
+	} else {
+		put_varint(b, 1);          // RecordIdKey revision
+		put_varint(b, 0);          // Number variant
+		put_le(b, (int64_t)pk.id); // signed fixed LE
+	}
+	put_varint(b, pk.olds.size() / d);
+	for (size_t i = 0; i * d < pk.olds.size(); i++)
+		enc_vec_f32(b, pk.olds.data() + i * d, d);
+	put_varint(b, pk.news.size() / d);
+	for (size_t i = 0; i * d < pk.news.size(); i++)
+		enc_vec_f32(b, pk.news.data() + i * d, d);
+}
+
 // Key prefix `/*<ns>*<db>*<tb>\0+<ix>!h` + kind char.
 static void key_prefix(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
                        const char *tb, uint32_t ix) {
@@ -3556,6 +3639,12 @@ static void key_hs(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
                    const char *tb, uint32_t ix) {
 	key_prefix(b, ns, db, tb, ix);
 	b.push_back('s');
+}
+static void key_hp(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
+                   const char *tb, uint32_t ix, uint64_t appending_id) {
+	key_prefix(b, ns, db, tb, ix);
+	b.push_back('p');
+	put_be(b, appending_id);
 }
 static void key_hv(std::vector<uint8_t> &b, uint32_t ns, uint32_t db,
                    const char *tb, uint32_t ix, const float *v, uint32_t d) {
@@ -4088,6 +4177,7 @@ struct sdbv_kvload {
 		vdocs::Ids64 docs;
 	};
 	std::vector<HvEnt> hv; // Hv
+	std::map<uint64_t, kvc::PendingKV> hp; // Hp by appending id
 };
 
 int sdbv_kvload_new(sdbv_ctx *ctx, uint32_t d, uint8_t metric, uint32_t m,
@@ -4164,8 +4254,18 @@ int sdbv_kvload_feed(sdbv_kvload *L, const uint8_t *key, uint64_t klen,
 		L->hv.push_back(std::move(ent));
 		return SDBV_OK;
 	}
+	case 'p': { // Hp: appending id (key) -> VectorPendingUpdate (value)
+		uint64_t aid;
+		if (!kvc::get_be(p, end, &aid))
+			return SDBV_ERR_BAD_ARG;
+		kvc::PendingKV pk;
+		if (!kvc::dec_pending(val, val + vlen, L->d, &pk))
+			return SDBV_ERR_UNSUPPORTED;
+		L->hp[aid] = std::move(pk);
+		return SDBV_OK;
+	}
 	default:
-		return SDBV_OK; // hd/hi/hp/hh etc.: host-kept, skipped
+		return SDBV_OK; // hd/hi/hh etc.: host-kept, skipped
 	}
 }
 
@@ -4265,6 +4365,16 @@ int sdbv_kvload_finish_index(sdbv_kvload *L, uint64_t table,
 	// highest referenced doc id (recycled holes are NOT reconstructed —
 	// the host re-binds key<->doc pairs via sdbv_index_bind_doc_key).
 	ix->next_doc_id = any_doc ? max_doc + 1 : 0;
+	// outstanding Hp pendings, in appending order (the reference drains
+	// the Hp range in key order, index.rs:195-205)
+	for (auto &e : L->hp) {
+		sdbv_index::Pending p;
+		p.kind = e.second.kind;
+		p.id = e.second.id;
+		p.olds = std::move(e.second.olds);
+		p.news = std::move(e.second.news);
+		ix->pendings.push_back(std::move(p));
+	}
 	delete L;
 	*out = ix;
 	return SDBV_OK;
@@ -4350,6 +4460,22 @@ int sdbv_index_dump_kv(sdbv_index *ix, uint32_t ns, uint32_t db,
 		kvc::key_hv(k, ns, db, tb, ix_id, (const float *)e.first.data(),
 		            ix->h->d);
 		kvc::enc_element_docs(v, e.second.e_id, e.second.docs);
+		if (write(user, k.data(), k.size(), v.data(), v.size()))
+			return SDBV_ERR_BAD_ARG;
+	}
+	// outstanding pendings as Hp pairs (RecordKey handles dump as
+	// RecordIdKey::Number — see INTEGRATION.md "record-key handles")
+	uint64_t aid = 0;
+	for (auto &p : ix->pendings) {
+		kvc::PendingKV pk;
+		pk.kind = p.kind;
+		pk.id = p.id;
+		pk.olds = p.olds;
+		pk.news = p.news;
+		k.clear();
+		v.clear();
+		kvc::key_hp(k, ns, db, tb, ix_id, aid++);
+		kvc::enc_pending(v, ix->h->d, pk);
 		if (write(user, k.data(), k.size(), v.data(), v.size()))
 			return SDBV_ERR_BAD_ARG;
 	}

@@ -154,9 +154,9 @@ def test_loader_rejects_garbage_and_skips_host_keys():
     g = sa.hnsw_create_host(d, metric="euclidean", m=4, m0=8, efc=20, seed=2)
     g.insert(np.zeros(d, dtype=np.float32))
     pairs = g.dump_kv()
-    # hd/hi/hp-style keys (host-kept kinds) are skipped, not errors
+    # hd/hi-style keys (host-kept kinds) are skipped, not errors
     pairs.append((PREFIX + b"d" + b"\0" * 8, b"\x01\x02\x03"))
-    pairs.append((PREFIX + b"p" + b"\0" * 8, b"junk"))
+    pairs.append((PREFIX + b"i" + b"\0" * 8, b"junk"))
     g2 = sa.load_kv_hnsw(pairs, d, metric="euclidean", m=4, m0=8, efc=20,
                          seed=2)
     a, b = g.l0_csr(), g2.l0_csr()
@@ -215,3 +215,42 @@ def test_varint_markers_above_250():
     ix.destroy()
     ix2.destroy()
     ix3.destroy()
+
+
+def test_pendings_survive_cold_start():
+    """Outstanding Hp pendings live in the KV store in the reference
+    (VectorPendingUpdate values over HnswPending keys, hp.rs) — a cold
+    start must carry them: dumped pendings reload in appending order and
+    produce identical searches and identical post-apply state."""
+    d = 8
+    rows = oracle.gen_f32(0x31, 0, 40, d)
+    ix = sa.index_create_host(d, metric="euclidean", m=4, m0=8, efc=20,
+                              seed=6)
+    for i in range(20):
+        ix.enqueue(i, None, rows[i])
+    ix.apply_pendings()
+    # outstanding: one update, one delete, one fresh RecordKey insert
+    ix.enqueue(3, rows[3], rows[30])
+    ix.enqueue(5, rows[5], None)
+    ix.enqueue(777, None, rows[31])
+    pairs = ix.dump_kv()
+    assert sum(1 for k, _ in pairs if b"!hp" in k) == 3
+    ix2 = sa.load_kv_index(pairs, 0, d, metric="euclidean", m=4, m0=8,
+                           efc=20, seed=6,
+                           doc_keys={i: i for i in range(20)})
+    assert ix2.pending_count() == 3
+    # searches with the pendings outstanding agree
+    for q in (rows[30], rows[5], rows[31]):
+        k1, i1, d1 = ix.knn_search(q, 5, 16)
+        k2, i2, d2_ = ix2.knn_search(q, 5, 16)
+        assert np.array_equal(k1, k2)
+        assert np.array_equal(i1, i2) and np.array_equal(d1, d2_)
+    # apply on both: identical graphs and results
+    assert ix.apply_pendings() == ix2.apply_pendings() == 3
+    a, b = ix.hnsw().l0_csr(), ix2.hnsw().l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    k1, i1, d1 = ix.knn_search(rows[31], 1, 16)
+    k2, i2, d2_ = ix2.knn_search(rows[31], 1, 16)
+    assert np.array_equal(i1, i2) and d1[0] == d2_[0] == 0.0
+    ix.destroy()
+    ix2.destroy()
