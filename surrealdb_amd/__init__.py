@@ -97,6 +97,7 @@ def lib():
     L.sdbv_hnsw_l0_export.argtypes = [vp, u32p, u32p]
     L.sdbv_hnsw_remove.restype = ctypes.c_int
     L.sdbv_hnsw_remove.argtypes = [vp, u64]
+    L.sdbv_hnsw_knn_host.argtypes = [vp, f32p, u32, u32, u64p, f64p, u32p]
     u8p = ctypes.POINTER(u8)
     L.sdbv_index_create.argtypes = [vp, u64, u32, u8, u32, u32, u32,
                                     ctypes.c_int, ctypes.c_int, u64,
@@ -313,6 +314,22 @@ class Hnsw:
     def finalize(self, table):
         _check(self._ctx._ptr, lib().sdbv_hnsw_finalize(self._ptr, table),
                "sdbv_hnsw_finalize")
+
+    def knn_search_host(self, q, k, ef):
+        """Host-side search (no device): the build path's algorithm over
+        the host graph — for CPU tests and quality audits."""
+        import numpy as np
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        ids = np.empty(k, dtype=np.uint64)
+        dists = np.empty(k, dtype=np.float64)
+        out_n = ctypes.c_uint32(0)
+        _check(None, lib().sdbv_hnsw_knn_host(
+            self._ptr, q.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), k,
+            ef, ids.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            dists.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.byref(out_n)), "sdbv_hnsw_knn_host")
+        n = out_n.value
+        return ids[:n], dists[:n]
 
     def knn_search(self, q, k, ef):
         import numpy as np

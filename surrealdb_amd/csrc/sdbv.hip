@@ -2777,6 +2777,58 @@ int sdbv_hnsw_remove(sdbv_hnsw *h, uint64_t e_id) {
 	return hnsw::hnsw_remove(h, (uint32_t)e_id) ? 1 : 0;
 }
 
+// Host-side knn_search (same algorithm as the GPU per-hop path with host
+// distances — the code path the build's efc-searches exercise). No device
+// needed: lets CPU tests and quality audits search any host graph.
+int sdbv_hnsw_knn_host(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
+                       uint64_t *out_ids, double *out_dists,
+                       uint32_t *out_n) {
+	using namespace hnsw;
+	if (!h || !q || k == 0)
+		return SDBV_ERR_BAD_ARG;
+	if (h->enter_point < 0) {
+		*out_n = 0;
+		return SDBV_OK;
+	}
+	double q_norm = h->metric == SDBV_METRIC_COSINE
+	                    ? sqrt(host_sumsq_f32(q, h->d))
+	                    : 0;
+	uint32_t ep_id = (uint32_t)h->enter_point;
+	double ep_dist = dist(h, q, q_norm, ep_id);
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		PQ cand;
+		cand.push(ep_dist, ep_id);
+		std::unordered_set<uint32_t> visited{ep_id};
+		PQ w = cand;
+		search_layer_host(h, h->layers[l], q, q_norm, cand, visited, w, 1,
+		                  false);
+		double dd;
+		uint32_t ii;
+		if (w.peek_first(&dd, &ii)) {
+			ep_dist = dd;
+			ep_id = ii;
+		}
+	}
+	PQ cand, w;
+	cand.push(ep_dist, ep_id);
+	w.push(ep_dist, ep_id);
+	std::unordered_set<uint32_t> visited{ep_id};
+	search_layer_host(h, h->layers[0], q, q_norm, cand, visited, w, ef,
+	                  false);
+	auto v = w.to_vec();
+	size_t m = std::min<size_t>(k, v.size());
+	std::vector<std::pair<std::pair<uint64_t, uint32_t>, double>> fin(m);
+	for (size_t i = 0; i < m; i++)
+		fin[i] = {{total_key(v[i].first), v[i].second}, v[i].first};
+	std::sort(fin.begin(), fin.end());
+	*out_n = (uint32_t)m;
+	for (size_t i = 0; i < m; i++) {
+		out_ids[i] = fin[i].first.second;
+		out_dists[i] = fin[i].second;
+	}
+	return SDBV_OK;
+}
+
 // Test hooks: drive the DoublePriorityQueue restatement directly so tests
 // can replay the reference's own test_double_priority_queue sequence
 // (knn.rs:735-790) against this implementation.

@@ -81,3 +81,28 @@ def test_host_build_matches_committed_fixture():
     assert np.array_equal(po, fx["l0_offsets"])
     assert np.array_equal(pe, fx["l0_edges"])
     h.destroy()
+
+
+def test_host_search_matches_oracle_bitexact():
+    """sdbv_hnsw_knn_host (the host-distance search path) must return
+    exactly the oracle's builder-sorted results on the identical
+    (sequentially built) graph — the CPU twin of the GPU search parity."""
+    import numpy as np
+    import surrealdb_amd
+    d, n = 20, 800
+    rows = oracle.gen_f32(0x123, 0, n, d)
+    h = surrealdb_amd.hnsw_create_host(d, metric="euclidean", m=8, m0=16,
+                                       efc=100, seed=0x5DB1)
+    o = oracle.Hnsw(d, metric="euclidean", m=8, m0=16, efc=100,
+                    ml=math.log(8.0) ** -1, seed=0x5DB1)
+    h.insert_batch(rows, nthreads=1)
+    for r in rows:
+        o.insert(r)
+    from surrealdb_amd.shard import total_key
+    for q in oracle.gen_f32(0x99, 0, 20, d):
+        gids, gdists = h.knn_search_host(q, 10, 40)
+        oids, odists = o.search(q, 10, 40)
+        order = np.lexsort((oids, total_key(odists)))
+        assert np.array_equal(gids, oids[order])
+        assert np.array_equal(gdists, odists[order])
+    h.destroy()
