@@ -117,3 +117,30 @@ def test_negative_distance_ordering():
     k = total_key(d)
     order = np.argsort(k)
     assert list(order) == [0, 3, 1, 2]  # -1e-16 < -0.0 < 0.0 < 1e-16
+
+
+def test_integer_vectors_number_path_exact():
+    """a5: the Number path keeps Int x Int in i64 (val/number.rs:926-1009).
+    For integer-valued vectors within f64-exact range the restated f64
+    accumulation produces the identical values (every partial sum is an
+    integer < 2^53, exactly representable), so the oracle's Number path is
+    exact for Int corpora too — asserted against exact Python ints — and
+    the f32-staged scan agrees in ranks for ints within f32-exact range."""
+    rng = np.random.default_rng(11)
+    corpus_i = rng.integers(-1000, 1000, size=(500, 64))
+    q_i = rng.integers(-1000, 1000, size=64)
+    ids, dists = oracle.topk_number("euclidean",
+                                    corpus_i.astype(np.float64),
+                                    q_i.astype(np.float64), 5)
+    import math as _m
+    exact = sorted(
+        ( _m.sqrt(int(sum((int(a) - int(b)) ** 2
+                          for a, b in zip(q_i, row)))), i)
+        for i, row in enumerate(corpus_i))
+    for rank, (ed, ei) in enumerate(exact[:5]):
+        assert ids[rank] == ei
+        assert dists[rank] == ed  # bit-exact: f64 sqrt of the exact i64 sum
+    ids32, d32 = oracle.topk_f32("euclidean",
+                                 corpus_i.astype(np.float32),
+                                 q_i.astype(np.float32), 5)
+    assert np.array_equal(ids, ids32)  # ints < 2^24: f32 staging exact
