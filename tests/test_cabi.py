@@ -4,7 +4,6 @@ import ctypes
 import os
 import re
 
-import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SO = os.path.join(REPO, "surrealdb_amd", "libsdbv.so")

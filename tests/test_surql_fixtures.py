@@ -8,7 +8,6 @@ EFC 150, M 12, M0 2*M, ML 1/ln(M).
 Run on both the product (host-only mode) and the oracle — both must hit the
 fixtures' exact expected distances and id order."""
 import numpy as np
-import pytest
 
 import oracle
 import surrealdb_amd as sa
@@ -19,11 +18,6 @@ def both(d, metric="euclidean", m=12, efc=150):
                              seed=0x5DB1)
     o = oracle.Index(d, metric=metric, m=m, m0=2 * m, efc=efc, seed=0x5DB1)
     return p, o
-
-
-def run_both(p, o, fn):
-    fn(p)
-    fn(o)
 
 
 def test_hnsw_knn_surql():

@@ -6,7 +6,6 @@ Anchors:
    insertion-order tie-break.
 """
 import numpy as np
-import pytest
 
 import oracle
 from surrealdb_amd.shard import merge_topk, shard_range, total_key
