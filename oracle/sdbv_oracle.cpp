@@ -562,53 +562,62 @@ int orc_topk_f32_mt(uint8_t metric, double order, const float *corpus,
 // ---------------------------------------------------------------------------
 
 struct OrcPQ {
-	// DoublePriorityQueue (knn.rs:15-123): BTreeMap<total_cmp(dist), FIFO deque>
-	std::map<uint64_t, std::deque<uint64_t>> m;
-	std::map<uint64_t, double> dist_of; // representative f64 per key
+	// DoublePriorityQueue (knn.rs:15-123): BTreeMap<total_cmp(dist), FIFO
+	// deque>. Restated as a sorted (total_key, push-seq) vector with a
+	// lazy head — the exact observable order (FIFO within equal distance
+	// on pop_first; latest-of-max-key on pop_last), pinned verbatim by the
+	// reference's own test_double_priority_queue in
+	// tests/test_pq_semantics.py.
+	struct E {
+		uint64_t key;
+		uint32_t seq;
+		uint64_t id;
+		double d;
+	};
+	std::vector<E> v; // ascending (key, seq); v[head..] is the live queue
+	size_t head = 0;
+	uint32_t next_seq = 0;
 	size_t n = 0;
 	void push(double d, uint64_t id) {
-		uint64_t k = orc_total_key(d);
-		m[k].push_back(id);
-		dist_of[k] = d;
+		E e{orc_total_key(d), next_seq++, id, d};
+		auto it = std::upper_bound(
+		    v.begin() + head, v.end(), e, [](const E &a, const E &b) {
+			    return a.key != b.key ? a.key < b.key : a.seq < b.seq;
+		    });
+		v.insert(it, e);
 		n++;
 	}
 	bool pop_first(double *d, uint64_t *id) {
-		if (m.empty()) return false;
-		auto it = m.begin();
-		*d = dist_of[it->first];
-		*id = it->second.front();
-		it->second.pop_front();
-		if (it->second.empty()) { dist_of.erase(it->first); m.erase(it); }
+		if (n == 0) return false;
+		*d = v[head].d;
+		*id = v[head].id;
+		head++;
 		n--;
 		return true;
 	}
 	bool pop_last(double *d, uint64_t *id) {
-		if (m.empty()) return false;
-		auto it = std::prev(m.end());
-		*d = dist_of[it->first];
-		*id = it->second.back();
-		it->second.pop_back();
-		if (it->second.empty()) { dist_of.erase(it->first); m.erase(it); }
+		if (n == 0) return false;
+		*d = v.back().d;
+		*id = v.back().id;
+		v.pop_back();
 		n--;
 		return true;
 	}
 	bool peek_first(double *d, uint64_t *id) const {
-		if (m.empty()) return false;
-		auto it = m.begin();
-		*d = dist_of.at(it->first);
-		*id = it->second.front();
+		if (n == 0) return false;
+		*d = v[head].d;
+		*id = v[head].id;
 		return true;
 	}
 	double peek_last_dist(double fallback) const {
-		if (m.empty()) return fallback;
-		return dist_of.at(std::prev(m.end())->first);
+		return n == 0 ? fallback : v.back().d;
 	}
 	std::vector<std::pair<double, uint64_t>> to_vec() const {
-		std::vector<std::pair<double, uint64_t>> v;
-		for (auto &e : m)
-			for (auto id : e.second)
-				v.push_back({dist_of.at(e.first), id});
-		return v;
+		std::vector<std::pair<double, uint64_t>> out;
+		out.reserve(n);
+		for (size_t i = head; i < v.size(); i++)
+			out.push_back({v[i].d, v[i].id});
+		return out;
 	}
 };
 
