@@ -160,7 +160,9 @@ def main():
             return one_batch(qi)
         if hnsw_index is not None:
             q = queries[qi % len(queries)]
-            if args.hnsw_perhop:
+            # persistent kernel caps ef at 512 (LDS queue); larger ef takes
+            # the per-hop path
+            if args.hnsw_perhop or args.ef > 512:
                 ids, dists = hnsw_index.knn_search(q, args.k, args.ef)
             else:
                 # persistent kernel (whole best-first loop in ONE launch —
