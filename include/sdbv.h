@@ -161,6 +161,11 @@ int sdbv_hnsw_finalize(sdbv_hnsw *, uint64_t table);
  * expansion. Results ascending (dist total_cmp, id), truncated to k. */
 int sdbv_hnsw_knn(sdbv_hnsw *, const float *q, uint32_t k, uint32_t ef,
                   uint64_t *out_ids, double *out_dists, uint32_t *out_n);
+/* Host-side knn_search (no device): the build path's search algorithm over
+ * the host graph — same exact result contract. For CPU-side tests, quality
+ * audits and boxes without a staged device table. */
+int sdbv_hnsw_knn_host(sdbv_hnsw *, const float *q, uint32_t k, uint32_t ef,
+                       uint64_t *out_ids, double *out_dists, uint32_t *out_n);
 /* Batched ef-search on the persistent kernel (one query per workgroup, the
  * whole layer-0 best-first loop in-kernel — same exact result contract as
  * sdbv_hnsw_knn). Q is b x d; outputs are b x k (+ out_ns per query).
