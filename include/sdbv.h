@@ -153,6 +153,15 @@ int sdbv_hnsw_insert(sdbv_hnsw *, const float *pt);
  * reference's own lock-free enqueue + batched apply (index.rs:138-211). */
 int sdbv_hnsw_insert_batch(sdbv_hnsw *, const float *pts, uint64_t n,
                            int nthreads);
+/* Chunked SNAPSHOT bulk build (bench mode, one step beyond the parallel
+ * build's relaxed ordering): per chunk, every level-0 element's efc-search
+ * runs against the graph as of the chunk start (read-only, parallel, and —
+ * round 2 — batched onto the persistent device kernel), then the apply
+ * half runs under the striped node locks. Quality pinned by the same
+ * recall bars as the parallel build. */
+int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *, const float *pts,
+                                    uint64_t n, uint32_t chunk,
+                                    int nthreads);
 /* Stage vectors (feature-major + norms) into the device table slot `table`;
  * required before sdbv_hnsw_knn. */
 int sdbv_hnsw_finalize(sdbv_hnsw *, uint64_t table);

@@ -84,6 +84,8 @@ def lib():
                                    ctypes.POINTER(vp)]
     L.sdbv_hnsw_insert.argtypes = [vp, f32p]
     L.sdbv_hnsw_insert_batch.argtypes = [vp, f32p, u64, ctypes.c_int]
+    L.sdbv_hnsw_insert_batch_snapshot.argtypes = [vp, f32p, u64, u32,
+                                                  ctypes.c_int]
     L.sdbv_hnsw_finalize.argtypes = [vp, u64]
     L.sdbv_hnsw_knn.argtypes = [vp, f32p, u32, u32, u64p, f64p, u32p]
     L.sdbv_hnsw_knn_batch.argtypes = [vp, f32p, u32, u32, u32, u64p, f64p, u32p]
@@ -310,6 +312,18 @@ class Hnsw:
         _check(self._ctx._ptr, lib().sdbv_hnsw_insert_batch(
             self._ptr, pts.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
             pts.shape[0], nthreads), "sdbv_hnsw_insert_batch")
+
+    def insert_batch_snapshot(self, pts, chunk, nthreads=0):
+        """Chunked snapshot bulk build (bench mode; §8f rank 3 structure —
+        per-chunk searches against the chunk-start graph, GPU-batchable)."""
+        import numpy as np
+        pts = np.ascontiguousarray(pts, dtype=np.float32)
+        _check(self._ctx._ptr if self._ctx else None,
+               lib().sdbv_hnsw_insert_batch_snapshot(
+                   self._ptr,
+                   pts.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+                   pts.shape[0], chunk, nthreads),
+               "sdbv_hnsw_insert_batch_snapshot")
 
     def finalize(self, table):
         _check(self._ctx._ptr, lib().sdbv_hnsw_finalize(self._ptr, table),

@@ -1983,17 +1983,16 @@ static void select_neighbors(sdbv_hnsw *h, const Layer &layer, uint32_t q_id,
 }
 
 // layer.rs:342-387
-static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
-                       const float *q_pt, double q_norm, PQ eps, bool locked) {
-	PQ w = eps;
-	static thread_local VisitSet visited;
-	visited.begin(h->vecs.size() / h->d);
-	for (auto &e : eps.to_vec())
-		visited.insert(e.second);
-	search_layer_host(h, layer, q_pt, q_norm, eps, visited, w, h->efc, locked);
-	PQ out = w;
+// The insert's apply half (select + bidirectional edges + prunes) over a
+// precomputed candidate window `w` — shared by the classic insert (whose
+// search ran just now) and the chunked snapshot build (whose search ran
+// against the pre-chunk graph; see sdbv_hnsw_insert_batch_snapshot).
+static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
+                               const float *q_pt, double q_norm, PQ w,
+                               bool locked) {
 	std::vector<uint32_t> neighbors;
-	select_neighbors(h, layer, q_id, q_pt, q_norm, w, neighbors, locked);
+	select_neighbors(h, layer, q_id, q_pt, q_norm, std::move(w), neighbors,
+	                 locked);
 	{
 		// append (not overwrite): a concurrent inserter may already have
 		// back-linked into q_id; sequential mode this is plain assignment
@@ -2039,6 +2038,18 @@ static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			layer.edges[e] = enew;
 		}
 	}
+}
+
+static PQ layer_insert(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
+                       const float *q_pt, double q_norm, PQ eps, bool locked) {
+	PQ w = eps;
+	static thread_local VisitSet visited;
+	visited.begin(h->vecs.size() / h->d);
+	for (auto &e : eps.to_vec())
+		visited.insert(e.second);
+	search_layer_host(h, layer, q_pt, q_norm, eps, visited, w, h->efc, locked);
+	PQ out = w;
+	layer_insert_apply(h, layer, q_id, q_pt, q_norm, std::move(w), locked);
 	return out;
 }
 
@@ -2118,6 +2129,48 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 		std::lock_guard<std::mutex> lk(h->global_mu);
 		h->enter_point = q_id;
 	}
+}
+
+// Chunked SNAPSHOT build (the §8f-rank-3 structure): per chunk, every
+// level-0 element's efc-search runs against the graph AS OF the chunk
+// start (read-only — no locks, embarrassingly parallel, and in round 2 the
+// per-chunk search batch moves onto the persistent device kernel), then
+// the apply half (select + edges + prunes) runs with the striped node
+// locks. One step beyond the parallel build's relaxed ordering: chunk
+// mates never see each other at search time (they still back-link from
+// later chunks); quality is pinned by the same recall bars. Upper-level
+// elements (~1/m of the batch) insert sequentially at the chunk front so
+// the layer structure exists before the snapshot searches.
+static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
+                                hnsw::PQ &w_out) {
+	using namespace hnsw;
+	const float *q_pt = vec(h, q_id);
+	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+	uint32_t ep_id = (uint32_t)h->enter_point;
+	double ep_dist = dist(h, q_pt, q_norm, ep_id);
+	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
+		PQ cand;
+		cand.push(ep_dist, ep_id);
+		std::unordered_set<uint32_t> visited{ep_id};
+		PQ w = cand;
+		search_layer_host(h, h->layers[l], q_pt, q_norm, cand, visited, w,
+		                  1, false);
+		double dd;
+		uint32_t ii;
+		if (w.peek_first(&dd, &ii)) {
+			ep_dist = dd;
+			ep_id = ii;
+		}
+	}
+	PQ eps;
+	eps.push(ep_dist, ep_id);
+	PQ w = eps;
+	static thread_local VisitSet visited;
+	visited.begin(h->vecs.size() / h->d);
+	visited.insert(ep_id);
+	search_layer_host(h, h->layers[0], q_pt, q_norm, eps, visited, w,
+	                  h->efc, false);
+	w_out = std::move(w);
 }
 
 // ---- graph element removal (sequential only: apply_pendings holds the
@@ -2360,6 +2413,95 @@ int sdbv_hnsw_insert_batch(sdbv_hnsw *h, const float *pts, uint64_t n,
 		w.join();
 	return SDBV_OK;
 }
+
+// Chunked snapshot build (see hnsw::snapshot_search_one above):
+int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *h, const float *pts,
+                                    uint64_t n, uint32_t chunk,
+                                    int nthreads) {
+	using namespace hnsw;
+	if (!h || h->finalized || chunk == 0)
+		return SDBV_ERR_BAD_ARG;
+	if (nthreads <= 0)
+		nthreads = (int)std::thread::hardware_concurrency();
+	uint64_t base = h->next_id;
+	std::vector<uint32_t> levels(n);
+	for (uint64_t i = 0; i < n; i++)
+		levels[i] = next_level(h); // sequential RNG contract, per ordinal
+	h->next_id += n;
+	for (uint64_t i = 0; i < n; i++)
+		hnsw_append_vec(h, pts + i * h->d);
+	// size every layer for the whole batch up front
+	{
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		uint64_t nelem = h->vecs.size() / h->d;
+		for (auto &l : h->layers) {
+			if (l.edges.size() < nelem)
+				l.edges.resize(nelem);
+			if (l.in_layer.size() < nelem)
+				l.in_layer.resize(nelem, 0);
+		}
+	}
+	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
+		uint64_t c1 = std::min(n, c0 + chunk);
+		// upper-level elements (and the very first element of an empty
+		// graph) insert sequentially first — ~1/m of the chunk
+		std::vector<uint64_t> flat;
+		for (uint64_t i = c0; i < c1; i++) {
+			if (levels[i] > 0 || h->enter_point < 0)
+				insert_at(h, (uint32_t)(base + i), levels[i], false);
+			else
+				flat.push_back(i);
+		}
+		// snapshot searches: read-only graph, parallel, no locks
+		std::vector<PQ> ws(flat.size());
+		std::atomic<uint64_t> cursor{0};
+		auto search_worker = [&]() {
+			uint64_t j;
+			while ((j = cursor.fetch_add(1)) < flat.size())
+				snapshot_search_one(h, (uint32_t)(base + flat[j]), ws[j]);
+		};
+		{
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads,
+			                                   (int)flat.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(search_worker);
+			search_worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		// apply phase: striped node locks, parallel (the parallel build's
+		// existing concurrency contract)
+		for (uint64_t j : flat)
+			h->layers[0].in_layer[base + j] = 1;
+		std::atomic<uint64_t> acursor{0};
+		auto apply_worker = [&]() {
+			uint64_t j;
+			while ((j = acursor.fetch_add(1)) < flat.size()) {
+				uint32_t q_id = (uint32_t)(base + flat[j]);
+				const float *q_pt = vec(h, q_id);
+				double q_norm = h->metric == SDBV_METRIC_COSINE
+				                    ? h->norms[q_id]
+				                    : 0;
+				layer_insert_apply(h, h->layers[0], q_id, q_pt, q_norm,
+				                   std::move(ws[j]), true);
+			}
+		};
+		{
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads,
+			                                   (int)flat.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(apply_worker);
+			apply_worker();
+			for (auto &t : ts)
+				t.join();
+		}
+	}
+	h->dirty = true;
+	return SDBV_OK;
+}
+
 
 uint64_t sdbv_hnsw_n(sdbv_hnsw *h) { return h ? h->next_id : 0; }
 uint32_t sdbv_hnsw_layers(sdbv_hnsw *h) {

@@ -106,3 +106,81 @@ def test_host_search_matches_oracle_bitexact():
         assert np.array_equal(gids, oids[order])
         assert np.array_equal(gdists, odists[order])
     h.destroy()
+
+
+def test_snapshot_build_recall_bars():
+    """Chunked snapshot build (§8f rank 3 structure: per-chunk searches
+    against the chunk-start graph — the GPU-batchable form): must meet the
+    same efs=40 recall bar as the parallel build on the reference's golden
+    dataset, across chunk sizes."""
+    import gzip
+    import json
+    import os
+    import numpy as np
+    import surrealdb_amd
+    golden = os.path.join(os.path.dirname(__file__), "golden",
+                          "hnsw-random-9000-20-euclidean.gz")
+    rows = []
+    with gzip.open(golden, "rt") as f:
+        for i, line in enumerate(f):
+            if i >= 2000:
+                break
+            rows.append(json.loads(line))
+    ingest = np.array(rows, dtype=np.float32)
+    queries_all = ingest[:100] + np.float32(0.05)
+
+    def rec_at(chunk):
+        h = surrealdb_amd.hnsw_create_host(20, metric="euclidean", m=8,
+                                           m0=16, efc=100, seed=0x5DB1)
+        h.insert_batch_snapshot(ingest, chunk=chunk, nthreads=4)
+        total = 0.0
+        for q in queries_all:
+            ids, _ = h.knn_search_host(q, 10, 40)
+            bf, _ = oracle.topk_f32("euclidean", ingest, q, 10)
+            total += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
+        offsets, edges = h.l0_csr()
+        deg = np.diff(offsets.astype(np.int64))
+        assert deg.max() <= 16
+        h.destroy()
+        return total / len(queries_all)
+
+    # Quality contract: the recall loss is bounded by the invisible
+    # fraction chunk/n (chunk mates are absent from each other's snapshot
+    # searches). At representative ratios (<= ~2%, the bench regime —
+    # chunk 4096 of 10M rows is 0.04%) the parallel-build bar holds:
+    assert rec_at(16) >= 0.99   # 0.8% invisible
+    assert rec_at(32) >= 0.99   # 1.6% invisible
+    # and the contract boundary is real: at an absurd ratio (26%) recall
+    # visibly degrades — chunk must be sized << n
+    assert rec_at(512) < 0.99
+
+
+def test_snapshot_build_matches_quality_of_parallel_768d():
+    """At bench-like dimensionality the snapshot build's recall must be in
+    line with the classic parallel build (both searched on the host)."""
+    import numpy as np
+    import surrealdb_amd
+    d, n = 768, 3000
+    rows = oracle.gen_f32(0x5DB1, 0, n, d)
+    queries = oracle.gen_f32(0xBEEF, 0, 30, d)
+
+    def build_recall(kind):
+        h = surrealdb_amd.hnsw_create_host(d, metric="cosine", m=16, m0=32,
+                                           efc=150, seed=0x5DB1)
+        if kind == "snapshot":
+            h.insert_batch_snapshot(rows, chunk=64, nthreads=4)
+        else:
+            h.insert_batch(rows, nthreads=4)
+        tot = 0.0
+        for q in queries:
+            ids, _ = h.knn_search_host(q, 10, 64)
+            bf, _, _ = oracle.topk_f32_mt("cosine", rows, q, 10)
+            tot += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
+        h.destroy()
+        return tot / len(queries)
+
+    r_par = build_recall("parallel")
+    r_snap = build_recall("snapshot")
+    # uniform 768-dim data has intrinsically low recall (see DESIGN); the
+    # snapshot relaxation must not degrade it materially
+    assert r_snap >= r_par - 0.08, (r_snap, r_par)
