@@ -2411,6 +2411,25 @@ int sdbv_hnsw_insert_batch(sdbv_hnsw *h, const float *pts, uint64_t n,
 		});
 	for (auto &w : workers)
 		w.join();
+	// Promote the enter point to a max-level element of this batch: with
+	// pre-created layers insert_at's q_level > top_up promotion never
+	// fires, which would leave the upper layers unreachable from a
+	// level-0 enter point (greedy descents would pass through them).
+	{
+		uint32_t maxl = 0;
+		for (auto l : levels)
+			maxl = std::max(maxl, l);
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		uint32_t cur_top = (uint32_t)h->layers.size() - 1;
+		if (h->enter_point >= 0 && maxl >= cur_top &&
+		    !h->layers[cur_top].has((uint32_t)h->enter_point)) {
+			for (uint64_t i = 0; i < n; i++)
+				if (levels[i] == maxl) {
+					h->enter_point = (int64_t)(base + i);
+					break;
+				}
+		}
+	}
 	return SDBV_OK;
 }
 
@@ -2430,9 +2449,16 @@ int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *h, const float *pts,
 	h->next_id += n;
 	for (uint64_t i = 0; i < n; i++)
 		hnsw_append_vec(h, pts + i * h->d);
-	// size every layer for the whole batch up front
+	// pre-create every layer the batch will need and size them up front:
+	// concurrent insert_at must never grow h->layers (vector reallocation
+	// under readers — same discipline as the classic parallel batch path)
 	{
 		std::lock_guard<std::mutex> lk(h->global_mu);
+		uint32_t max_level = 0;
+		for (uint64_t i = 0; i < n; i++)
+			max_level = std::max(max_level, levels[i]);
+		while (h->layers.size() <= max_level)
+			h->layers.push_back(hnsw::Layer{{}, h->m});
 		uint64_t nelem = h->vecs.size() / h->d;
 		for (auto &l : h->layers) {
 			if (l.edges.size() < nelem)
@@ -2443,14 +2469,35 @@ int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *h, const float *pts,
 	}
 	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
 		uint64_t c1 = std::min(n, c0 + chunk);
-		// upper-level elements (and the very first element of an empty
-		// graph) insert sequentially first — ~1/m of the chunk
-		std::vector<uint64_t> flat;
+		// upper-level elements first (~1/m of the chunk) so the layer
+		// structure exists before the snapshot searches; inserted with the
+		// classic parallel path (striped locks) — the first element of an
+		// empty graph goes alone to establish the enter point
+		std::vector<uint64_t> upper, flat;
 		for (uint64_t i = c0; i < c1; i++) {
-			if (levels[i] > 0 || h->enter_point < 0)
+			if (h->enter_point < 0)
 				insert_at(h, (uint32_t)(base + i), levels[i], false);
+			else if (levels[i] > 0)
+				upper.push_back(i);
 			else
 				flat.push_back(i);
+		}
+		if (!upper.empty()) {
+			std::atomic<uint64_t> ucursor{0};
+			auto upper_worker = [&]() {
+				uint64_t j;
+				while ((j = ucursor.fetch_add(1)) < upper.size())
+					insert_at(h, (uint32_t)(base + upper[j]),
+					          levels[upper[j]], true);
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads,
+			                                   (int)upper.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(upper_worker);
+			upper_worker();
+			for (auto &t : ts)
+				t.join();
 		}
 		// snapshot searches: read-only graph, parallel, no locks
 		std::vector<PQ> ws(flat.size());
