@@ -2665,7 +2665,11 @@ int sdbv_hnsw_knn_batch(sdbv_hnsw *h, const float *Q, uint32_t b, uint32_t k,
                         uint32_t ef, uint64_t *out_ids, double *out_dists,
                         uint32_t *out_ns) {
 	using namespace hnsw;
-	if (!h || !h->finalized || b == 0 || k == 0 || k > MAX_K ||
+	// k is bounded by the in-kernel w window (HQ_EF_CAP), not the scan's
+	// MAX_K: the output extraction loops generically over k slots. k <= 64
+	// is the GPU-validated regime; larger k (the snapshot build's efc
+	// window, round 2) shares the same code path.
+	if (!h || !h->finalized || b == 0 || k == 0 || k > HQ_EF_CAP ||
 	    ef > HQ_EF_CAP)
 		return SDBV_ERR_BAD_ARG;
 	if (h->enter_point < 0) {
