@@ -2738,13 +2738,13 @@ int sdbv_hnsw_knn_batch(sdbv_hnsw *h, const float *Q, uint32_t b, uint32_t k,
 			if (p)
 				(void)hipFree(p);
 	};
-	hipMemcpyAsync(Qd, Q, (uint64_t)b * h->d * sizeof(float),
+	(void)hipMemcpyAsync(Qd, Q, (uint64_t)b * h->d * sizeof(float),
 	               hipMemcpyHostToDevice, ctx->stream);
-	hipMemcpyAsync(qnd, qnorms.data(), b * sizeof(double),
+	(void)hipMemcpyAsync(qnd, qnorms.data(), b * sizeof(double),
 	               hipMemcpyHostToDevice, ctx->stream);
-	hipMemcpyAsync(epdd, epd.data(), b * sizeof(double),
+	(void)hipMemcpyAsync(epdd, epd.data(), b * sizeof(double),
 	               hipMemcpyHostToDevice, ctx->stream);
-	hipMemcpyAsync(epsd, eps.data(), b * sizeof(uint32_t),
+	(void)hipMemcpyAsync(epsd, eps.data(), b * sizeof(uint32_t),
 	               hipMemcpyHostToDevice, ctx->stream);
 
 	auto t0 = std::chrono::steady_clock::now();
@@ -2754,13 +2754,13 @@ int sdbv_hnsw_knn_batch(sdbv_hnsw *h, const float *Q, uint32_t b, uint32_t k,
 	                   h->vis_dev, vwords, k, ef, outr, outd, outc, outf);
 	std::vector<uint32_t> h_rows((uint64_t)b * k), h_cnt(b), h_flags(b);
 	std::vector<double> h_d((uint64_t)b * k);
-	hipMemcpyAsync(h_rows.data(), outr, h_rows.size() * sizeof(uint32_t),
+	(void)hipMemcpyAsync(h_rows.data(), outr, h_rows.size() * sizeof(uint32_t),
 	               hipMemcpyDeviceToHost, ctx->stream);
-	hipMemcpyAsync(h_d.data(), outd, h_d.size() * sizeof(double),
+	(void)hipMemcpyAsync(h_d.data(), outd, h_d.size() * sizeof(double),
 	               hipMemcpyDeviceToHost, ctx->stream);
-	hipMemcpyAsync(h_cnt.data(), outc, b * sizeof(uint32_t),
+	(void)hipMemcpyAsync(h_cnt.data(), outc, b * sizeof(uint32_t),
 	               hipMemcpyDeviceToHost, ctx->stream);
-	hipMemcpyAsync(h_flags.data(), outf, b * sizeof(uint32_t),
+	(void)hipMemcpyAsync(h_flags.data(), outf, b * sizeof(uint32_t),
 	               hipMemcpyDeviceToHost, ctx->stream);
 	if (hipStreamSynchronize(ctx->stream) != hipSuccess ||
 	    hipGetLastError() != hipSuccess) {
