@@ -274,6 +274,10 @@ int sdbv_kvload_finish_index(sdbv_kvload *, uint64_t table, sdbv_index **out);
 void sdbv_kvload_abort(sdbv_kvload *);
 int sdbv_index_bind_doc_key(sdbv_index *, uint64_t doc_id,
                             uint64_t record_key);
+/* Export the doc-id <-> record-key map (the hi/hd state the host persists
+ * for cold starts). Returns the total entry count; fills up to cap. */
+uint64_t sdbv_index_doc_keys(sdbv_index *, uint64_t *out_docs,
+                             uint64_t *out_keys, uint64_t cap);
 /* Dump the graph / index back to reference-format KV pairs (round-trip +
  * migration tooling). The callback returns non-zero to abort. */
 typedef int (*sdbv_kv_write_cb)(void *user, const uint8_t *key, uint64_t klen,

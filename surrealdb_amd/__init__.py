@@ -131,6 +131,8 @@ def lib():
     L.sdbv_kvload_finish_index.argtypes = [vp, u64, ctypes.POINTER(vp)]
     L.sdbv_kvload_abort.argtypes = [vp]
     L.sdbv_index_bind_doc_key.argtypes = [vp, u64, u64]
+    L.sdbv_index_doc_keys.restype = u64
+    L.sdbv_index_doc_keys.argtypes = [vp, u64p, u64p, u64]
     global KV_WRITE_CB
     KV_WRITE_CB = ctypes.CFUNCTYPE(ctypes.c_int, vp, u8pp, u64, u8pp, u64)
     L.sdbv_hnsw_dump_kv.argtypes = [vp, u32, u32, ctypes.c_char_p, u32,
@@ -589,6 +591,18 @@ class Index:
 
     def doc_count(self):
         return lib().sdbv_index_doc_count(self._ptr)
+
+    def doc_keys(self):
+        """The doc-id -> record-key map (the hi/hd state a host persists
+        for cold starts; pairs sdbv_index_bind_doc_key on reload)."""
+        import numpy as np
+        n = lib().sdbv_index_doc_keys(self._ptr, None, None, 0)
+        docs = np.empty(max(n, 1), dtype=np.uint64)
+        keys = np.empty(max(n, 1), dtype=np.uint64)
+        lib().sdbv_index_doc_keys(
+            self._ptr, docs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            keys.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)), n)
+        return {int(d): int(k) for d, k in zip(docs[:n], keys[:n])}
 
     def pending_count(self):
         return lib().sdbv_index_pending_count(self._ptr)

@@ -4629,6 +4629,25 @@ int sdbv_kvload_finish_index(sdbv_kvload *L, uint64_t table,
 	return SDBV_OK;
 }
 
+// Export the doc-id <-> record-key-handle map (the host's hi/hd state —
+// what it persists so a cold start can re-bind). Returns the entry count;
+// writes up to `cap` pairs.
+uint64_t sdbv_index_doc_keys(sdbv_index *ix, uint64_t *out_docs,
+                             uint64_t *out_keys, uint64_t cap) {
+	if (!ix)
+		return 0;
+	std::lock_guard<std::mutex> lk(ix->mu);
+	uint64_t i = 0;
+	for (auto &e : ix->doc2key) {
+		if (i < cap) {
+			out_docs[i] = e.first;
+			out_keys[i] = e.second;
+		}
+		i++;
+	}
+	return (uint64_t)ix->doc2key.size();
+}
+
 // Re-binds one record-key handle to a doc id (the host's hi/hd entries)
 // after a cold-start load.
 int sdbv_index_bind_doc_key(sdbv_index *ix, uint64_t doc_id,

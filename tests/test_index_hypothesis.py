@@ -75,3 +75,56 @@ def test_program_equivalence(ops, metric):
         assert np.array_equal(po, oo) and np.array_equal(pe, oe)
     finally:
         p.destroy()
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(ops=ops, reload_at=st.integers(min_value=0, max_value=50))
+def test_cold_start_mid_workload_is_identity(ops, reload_at):
+    """A dump -> load_kv_index -> rebind cycle in the middle of a write
+    workload must be an identity for every later observable (the oracle
+    never reloads; the product does at `reload_at`)."""
+    p = sa.index_create_host(D, metric="euclidean", m=4, m0=8, efc=24, seed=3)
+    o = oracle.Index(D, metric="euclidean", m=4, m0=8, efc=24, seed=3)
+    try:
+        live = {}
+        for step, (op, key, ridx) in enumerate(ops):
+            if step == reload_at:
+                # cold start: outstanding pendings + graph + docs travel;
+                # the key<->doc map is the hi/hd state the host persists
+                # (sdbv_index_doc_keys) and re-binds on load
+                pairs = p.dump_kv()
+                bindings = p.doc_keys()
+                newp = sa.load_kv_index(pairs, 0, D, metric="euclidean",
+                                        m=4, m0=8, efc=24, seed=3,
+                                        doc_keys=bindings)
+                p.destroy()
+                p = newp
+            if op in ("write", "dup_write"):
+                v = ROWS[0] if op == "dup_write" else ROWS[ridx]
+                old = live.get(key)
+                p.enqueue(key, old, v)
+                o.enqueue(key, old, v)
+                live[key] = v
+            elif op == "delete":
+                if key in live:
+                    p.enqueue(key, live[key], None)
+                    o.enqueue(key, live[key], None)
+                    del live[key]
+            elif op == "apply":
+                assert p.apply_pendings() == o.apply_pendings()
+                assert p.doc_count() == o.doc_count()
+            else:
+                q = ROWS[ridx] + np.float32(0.05)
+                pk, pi, pd = p.knn_search(q, 5, 12)
+                ok, oi, od = o.knn_search(q, 5, 12)
+                assert np.array_equal(pk, ok)
+                assert np.array_equal(pi, oi)
+                assert np.array_equal(pd, od)
+        assert p.apply_pendings() == o.apply_pendings()
+        ph, oh = p.hnsw(), o.hnsw()
+        po, pe = ph.l0_csr()
+        oo, oe = oh.l0_csr()
+        assert np.array_equal(po, oo) and np.array_equal(pe, oe)
+    finally:
+        p.destroy()
