@@ -3638,6 +3638,9 @@ static bool dec_vec_f32_cursor(const uint8_t *&p, const uint8_t *end,
 		return false;
 	if (!get_varint(p, end, &len))
 		return false;
+	// bound before reserving: every element needs >= 4 payload bytes
+	if (len > (uint64_t)(end - p) / 4 + 1)
+		return false;
 	out.clear();
 	out.reserve(len);
 	for (uint64_t i = 0; i < len; i++) {
