@@ -4613,10 +4613,19 @@ int sdbv_kvload_finish_index(sdbv_kvload *L, uint64_t table,
 				max_doc = d;
 		}
 	}
-	// HnswDocsState (hd root) stays host-side; allocation resumes past the
-	// highest referenced doc id (recycled holes are NOT reconstructed —
-	// the host re-binds key<->doc pairs via sdbv_index_bind_doc_key).
+	// HnswDocsState (hd root) stays host-side, but the allocator is
+	// reconstructed EXACTLY up to allocation equivalence. Doc ids
+	// allocate densely (docs.rs:78-90), so the reference's persisted
+	// state is always {available = [0, next) \ bound, next}; min-first
+	// draws make any trailing run of `available` foldable into `next`
+	// without changing a single future allocation. Hence: seed
+	// available with every id up to the highest Hv-referenced one (Hv
+	// references include RECYCLED zombies — the dropped-old-removal
+	// quirk — which the reference also holds in `available`), and let
+	// sdbv_index_bind_doc_key carve out the live (hi/hd-bound) ids.
 	ix->next_doc_id = any_doc ? max_doc + 1 : 0;
+	for (uint64_t dd = 0; dd < ix->next_doc_id; dd++)
+		ix->available.insert(dd);
 	// outstanding Hp pendings, in appending order (the reference drains
 	// the Hp range in key order, index.rs:195-205)
 	for (auto &e : L->hp) {
@@ -4660,8 +4669,13 @@ int sdbv_index_bind_doc_key(sdbv_index *ix, uint64_t doc_id,
 	std::lock_guard<std::mutex> lk(ix->mu);
 	ix->key2doc[record_key] = doc_id;
 	ix->doc2key[doc_id] = record_key;
-	if (doc_id >= ix->next_doc_id)
+	// keep the reconstructed allocator consistent: a bound id is live
+	ix->available.erase(doc_id);
+	if (doc_id >= ix->next_doc_id) {
+		for (uint64_t dd = ix->next_doc_id; dd < doc_id; dd++)
+			ix->available.insert(dd);
 		ix->next_doc_id = doc_id + 1;
+	}
 	return SDBV_OK;
 }
 
