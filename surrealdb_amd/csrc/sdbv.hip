@@ -4624,6 +4624,10 @@ int sdbv_kvload_finish_index(sdbv_kvload *L, uint64_t table,
 	// quirk — which the reference also holds in `available`), and let
 	// sdbv_index_bind_doc_key carve out the live (hi/hd-bound) ids.
 	ix->next_doc_id = any_doc ? max_doc + 1 : 0;
+	if (ix->next_doc_id > (1ull << 32)) { // corrupt input guard: doc ids
+		delete ix;                         // are dense, so this is absurd
+		return SDBV_ERR_UNSUPPORTED;
+	}
 	for (uint64_t dd = 0; dd < ix->next_doc_id; dd++)
 		ix->available.insert(dd);
 	// outstanding Hp pendings, in appending order (the reference drains
@@ -4672,6 +4676,8 @@ int sdbv_index_bind_doc_key(sdbv_index *ix, uint64_t doc_id,
 	// keep the reconstructed allocator consistent: a bound id is live
 	ix->available.erase(doc_id);
 	if (doc_id >= ix->next_doc_id) {
+		if (doc_id - ix->next_doc_id > (1ull << 32))
+			return SDBV_ERR_BAD_ARG; // dense ids: absurd gap = corrupt
 		for (uint64_t dd = ix->next_doc_id; dd < doc_id; dd++)
 			ix->available.insert(dd);
 		ix->next_doc_id = doc_id + 1;
