@@ -109,3 +109,36 @@ def test_bruteforce_knn_new_executor_surql():
                                       corpus.astype(np.float64), q
                                       .astype(np.float64), 2)
     assert idsn.tolist() == [1, 2] and distsn.tolist() == [1.0, 2.0]
+
+
+def test_bruteforce_knn_with_filter_surql():
+    """bruteforce_knn_with_filter_new_executor.surql: the predicate runs
+    BEFORE KnnTopK (TableScan [predicate: active = true] -> KnnTopK in the
+    committed plan), so only active rows compete for the top-K:
+    [(2f, pts:3), (9f, pts:1)] — pts:2 at distance 1 is inactive."""
+    pts = np.array([[10, 0, 0, 0], [2, 0, 0, 0], [3, 0, 0, 0],
+                    [100, 0, 0, 0], [50, 0, 0, 0]], dtype=np.float32)
+    active = np.array([True, False, True, True, False])
+    q = np.array([1, 0, 0, 0], dtype=np.float32)
+    # boundary form: the host streams only predicate-matching rows into
+    # the scan (a filtered sub-corpus with its original ids)
+    sub = np.ascontiguousarray(pts[active])
+    sub_ids = np.flatnonzero(active)
+    ids, dists = oracle.topk_f32("euclidean", sub, q, 2)
+    got = [(float(d), int(sub_ids[i])) for i, d in zip(ids, dists)]
+    assert got == [(2.0, 2), (9.0, 0)]  # pts:3, pts:1 (0-based rows)
+
+
+def test_bruteforce_knn_multisource_filter_surql():
+    """bruteforce_knn_multisource_filter_new_executor.surql: Union of two
+    tables -> Filter -> KnnTopK; expected [(2f, pts:3), (3f, pts2:2)] —
+    the nearest rows overall (pts2:1 at 0.5, pts:2 at 1) are inactive."""
+    union = np.array([[10, 0, 0, 0], [2, 0, 0, 0], [3, 0, 0, 0],
+                      [1.5, 0, 0, 0], [4, 0, 0, 0]], dtype=np.float32)
+    active = np.array([True, False, True, False, True])
+    q = np.array([1, 0, 0, 0], dtype=np.float32)
+    sub = np.ascontiguousarray(union[active])
+    sub_ids = np.flatnonzero(active)
+    ids, dists = oracle.topk_f32("euclidean", sub, q, 2)
+    got = [(float(d), int(sub_ids[i])) for i, d in zip(ids, dists)]
+    assert got == [(2.0, 2), (3.0, 4)]  # pts:3, pts2:2
