@@ -106,7 +106,10 @@ void sdbv_gen_f32(uint64_t seed, uint64_t row0, uint64_t nrows, uint32_t d,
  * Results sorted ascending by (distance f64-total_cmp, id).
  * Distances follow Distance::calculate F32 semantics (vector.rs:244-249 /
  * 282-283): f32 accumulation per the restated ndarray contract, f64 finish.
- * k <= 64 in this revision. *out_n = min(k, n). */
+ * k <= 64 runs fully on-device (block-local exact top-K); larger k
+ * (the reference accepts any) takes one all-distances launch + exact host
+ * selection — same ordering contract, costing an n-f64 transfer.
+ * *out_n = min(k, n). */
 int sdbv_knn_bruteforce(sdbv_ctx *, uint64_t table, const float *q, uint32_t d,
                         uint32_t k, uint64_t *out_ids, double *out_dists,
                         uint32_t *out_n);

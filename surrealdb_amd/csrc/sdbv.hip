@@ -51,6 +51,7 @@
 #include <cstring>
 #include <map>
 #include <mutex>
+#include <queue>
 #include <set>
 #include <string>
 #include <unordered_map>
@@ -1337,6 +1338,78 @@ static int ensure_query_scratch(sdbv_ctx *ctx, uint32_t d, uint64_t nblocks,
 	return SDBV_OK;
 }
 
+// k beyond the scan kernel's LDS top-K window (MAX_K): one all-distances
+// launch + exact host selection. ids are strictly increasing, so
+// (dist, row) order equals the contract's (dist, id) order. An edge-case
+// path — the reference accepts any k — costing one n-f64 D2H per query.
+// Caller holds ctx->mu.
+static int knn_large_k(sdbv_ctx *ctx, Table &t, const float *q, uint32_t d,
+                       uint32_t k, uint64_t *out_ids, double *out_dists,
+                       uint32_t *out_n) {
+	int rc = ensure_query_scratch(ctx, d, 1, 1);
+	if (rc != SDBV_OK)
+		return rc;
+	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
+	                              hipMemcpyHostToDevice, ctx->stream));
+	double q_norm = sqrt((double)h_sumsq_f32(q, d));
+	double *dout = nullptr;
+	HIP_CHECK(ctx, hipMalloc(&dout, t.n * sizeof(double)));
+	uint64_t nb = (t.n + THREADS - 1) / THREADS;
+	HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
+	if (t.metric == SDBV_METRIC_COSINE)
+		hipLaunchKernelGGL(k_all_dists<0>, dim3((uint32_t)nb), dim3(THREADS),
+		                   0, ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,
+		                   ctx->q_dev, q_norm, dout);
+	else
+		hipLaunchKernelGGL(k_all_dists<1>, dim3((uint32_t)nb), dim3(THREADS),
+		                   0, ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,
+		                   ctx->q_dev, q_norm, dout);
+	HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
+	std::vector<double> hd(t.n);
+	std::vector<uint64_t> hids(t.n);
+	if (hipMemcpyAsync(hd.data(), dout, t.n * sizeof(double),
+	                   hipMemcpyDeviceToHost, ctx->stream) != hipSuccess ||
+	    hipMemcpyAsync(hids.data(), t.ids_dev, t.n * sizeof(uint64_t),
+	                   hipMemcpyDeviceToHost, ctx->stream) != hipSuccess ||
+	    hipStreamSynchronize(ctx->stream) != hipSuccess ||
+	    hipGetLastError() != hipSuccess) {
+		(void)hipFree(dout);
+		ctx->err = "knn_large_k: device error";
+		return SDBV_ERR_HIP;
+	}
+	(void)hipFree(dout);
+	float ms = 0;
+	(void)hipEventElapsedTime(&ms, ctx->ev0, ctx->ev1);
+	ctx->stats.last_scan_kernel_ms = ms;
+	ctx->stats.last_merge_kernel_ms = 0;
+	ctx->stats.last_rows_scanned = t.n;
+	// bounded max-heap of (total_key(dist), row): keep the k smallest
+	auto tkey = [](double x) {
+		uint64_t bits;
+		std::memcpy(&bits, &x, 8);
+		return (bits >> 63) ? ~bits : (bits | 0x8000000000000000ULL);
+	};
+	std::priority_queue<std::pair<uint64_t, uint64_t>> heap;
+	for (uint64_t r = 0; r < t.n; r++) {
+		std::pair<uint64_t, uint64_t> e{tkey(hd[r]), r};
+		if (heap.size() < k) {
+			heap.push(e);
+		} else if (e < heap.top()) {
+			heap.pop();
+			heap.push(e);
+		}
+	}
+	uint32_t m = (uint32_t)heap.size();
+	*out_n = m;
+	for (uint32_t i = m; i-- > 0;) {
+		auto e = heap.top();
+		heap.pop();
+		out_ids[i] = hids[e.second];
+		out_dists[i] = hd[e.second];
+	}
+	return SDBV_OK;
+}
+
 int sdbv_knn_bruteforce(sdbv_ctx *ctx, uint64_t table, const float *q,
                         uint32_t d, uint32_t k, uint64_t *out_ids,
                         double *out_dists, uint32_t *out_n) {
@@ -1347,8 +1420,10 @@ int sdbv_knn_bruteforce(sdbv_ctx *ctx, uint64_t table, const float *q,
 	Table &t = it->second;
 	if (d != t.d)
 		return SDBV_ERR_BAD_ARG;
-	if (k == 0 || k > MAX_K)
+	if (k == 0 || k > (1u << 22))
 		return SDBV_ERR_BAD_ARG;
+	if (k > MAX_K)
+		return knn_large_k(ctx, t, q, d, k, out_ids, out_dists, out_n);
 
 	// pick a grid: >=2048 blocks to fill 256 CUs, contiguous spans per block
 	uint64_t tiles = (t.n + TILE - 1) / TILE;

@@ -140,3 +140,20 @@ def test_smoke_stats(ctx):
     assert s["last_scan_kernel_ms"] > 0
     assert s["last_rows_scanned"] == 100_000
     ctx.drop_table(10)
+
+
+@pytest.mark.parametrize("metric", ["cosine", "euclidean"])
+def test_knn_large_k_matches_oracle(ctx, metric):
+    """k beyond the scan kernel's top-K window (MAX_K=64): the
+    all-distances + host-selection path must keep the exact ordering
+    contract (bit-exact ids/ranks/distances vs the oracle)."""
+    n, d = 20_000, 128
+    corpus = oracle.gen_f32(0x5DB1, 0, n, d)
+    ctx.stage_corpus(17, corpus, metric=metric)
+    q = oracle.gen_f32(0xBEEF, 0, 1, d)[0]
+    for k in (65, 100, 500):
+        gids, gdists = ctx.knn_bruteforce(17, q, k)
+        oids, odists = oracle.topk_f32(metric, corpus, q, k)
+        assert np.array_equal(gids, oids), f"k={k} ids"
+        assert np.array_equal(gdists, odists), f"k={k} dist bits"
+    ctx.drop_table(17)
