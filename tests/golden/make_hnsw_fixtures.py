@@ -1,7 +1,7 @@
 """Generate committed HNSW oracle fixtures for the GPU parity tests.
 
 The oracle build (oracle/sdbv_oracle.cpp, the restated reference algorithm)
-is deterministic but slow at 768-dim (~19 ms/insert), so the GPU test suite
+is deterministic but slower than the product at 768-dim, so the GPU test suite
 does NOT rebuild it on the GPU box: this script builds each test
 configuration ONCE on a CPU box and commits the expected graph (layer-0 CSR,
 entry point, layer count) and the builder-sorted search results. The GPU

@@ -2,7 +2,7 @@
 the persistent in-kernel search) vs committed ORACLE FIXTURES, plus the
 reference's recall bars on its golden datasets.
 
-The oracle graph builds at 768-dim are slow (~19 ms/insert), so the oracle
+The oracle graph builds at 768-dim are slow, so the oracle
 side of each configuration was built ONCE by tests/golden/make_hnsw_fixtures.py
 (committed script) and its expected graph + builder-sorted search results are
 committed as tests/golden/hnsw_fix_*.npz. Here only the product build (the
