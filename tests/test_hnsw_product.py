@@ -186,3 +186,31 @@ def test_snapshot_build_matches_quality_of_parallel_768d():
     # uniform 768-dim data has intrinsically low recall (see DESIGN); the
     # snapshot relaxation must not degrade it materially
     assert r_snap >= r_par - 0.08, (r_snap, r_par)
+
+
+def test_host_search_matches_committed_768d_fixture():
+    """Product sequential build + host search against the committed
+    4000x768 oracle fixture: graph CSR and every builder-sorted search
+    result bit-exact — the CPU twin of the GPU fixture tests."""
+    import os
+    import numpy as np
+    import surrealdb_amd
+    fx = np.load(os.path.join(os.path.dirname(__file__), "golden",
+                              "hnsw_fix_seq768_cos.npz"))
+    d, n, m = int(fx["d"]), int(fx["n"]), int(fx["m"])
+    rows = oracle.gen_f32(int(fx["data_seed"]), 0, n, d)
+    h = surrealdb_amd.hnsw_create_host(
+        d, metric=str(fx["metric"]), m=m, m0=int(fx["m0"]),
+        efc=int(fx["efc"]), ml=1.0 / math.log(m), seed=int(fx["seed"]))
+    h.insert_batch(rows, nthreads=1)
+    po, pe = h.l0_csr()
+    assert np.array_equal(po, fx["l0_offsets"])
+    assert np.array_equal(pe, fx["l0_edges"])
+    k = int(fx["k"])
+    Q = oracle.gen_f32(int(fx["query_seed"]), 0, fx["ids_ef64"].shape[0], d)
+    for j, q in enumerate(Q):
+        ids, dists = h.knn_search_host(q, k, 64)
+        nn = int(fx["n_ef64"][j])
+        assert np.array_equal(ids, fx["ids_ef64"][j][:nn]), f"q{j}"
+        assert np.array_equal(dists, fx["dists_ef64"][j][:nn]), f"q{j}"
+    h.destroy()
