@@ -28,10 +28,15 @@ ops = st.lists(
 
 @settings(max_examples=150, deadline=None,
           suppress_health_check=[HealthCheck.too_slow])
-@given(ops=ops, metric=st.sampled_from(["euclidean", "cosine"]))
-def test_program_equivalence(ops, metric):
-    p = sa.index_create_host(D, metric=metric, m=4, m0=8, efc=24, seed=3)
-    o = oracle.Index(D, metric=metric, m=4, m0=8, efc=24, seed=3)
+@given(ops=ops, metric=st.sampled_from(["euclidean", "cosine"]),
+       extend=st.booleans(), keep=st.booleans())
+def test_program_equivalence(ops, metric, extend, keep):
+    # extend/keep toggle the Ext/Keep heuristic variants
+    # (heuristic.rs:52-57) — independent restatements on both sides
+    p = sa.index_create_host(D, metric=metric, m=4, m0=8, efc=24, seed=3,
+                             extend=extend, keep=keep)
+    o = oracle.Index(D, metric=metric, m=4, m0=8, efc=24, seed=3,
+                     extend=extend, keep=keep)
     try:
         live = {}
         for op, key, ridx in ops:
