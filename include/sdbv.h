@@ -252,6 +252,12 @@ uint64_t sdbv_index_pending_count(sdbv_index *);
 int sdbv_index_check_props(sdbv_index *, uint64_t expected_count);
 /* Test access to the underlying graph (CSR parity vs the oracle). */
 sdbv_hnsw *sdbv_index_hnsw(sdbv_index *);
+/* Level-RNG state carry-over (extension): the reference's insert-level RNG
+ * is entropy-seeded per process (hnsw/mod.rs:263-266) so any state is
+ * conformant; persisting it keeps this framework's same-seed determinism
+ * across cold starts. */
+uint64_t sdbv_index_level_rng(sdbv_index *);
+void sdbv_index_set_level_rng(sdbv_index *, uint64_t state);
 
 /* ------------------------------------------------------------------------
  * KV codec + cold-start staging pipeline: bulk-load a

@@ -133,6 +133,9 @@ def lib():
     L.sdbv_index_bind_doc_key.argtypes = [vp, u64, u64]
     L.sdbv_index_doc_keys.restype = u64
     L.sdbv_index_doc_keys.argtypes = [vp, u64p, u64p, u64]
+    L.sdbv_index_level_rng.restype = u64
+    L.sdbv_index_level_rng.argtypes = [vp]
+    L.sdbv_index_set_level_rng.argtypes = [vp, u64]
     global KV_WRITE_CB
     KV_WRITE_CB = ctypes.CFUNCTYPE(ctypes.c_int, vp, u8pp, u64, u8pp, u64)
     L.sdbv_hnsw_dump_kv.argtypes = [vp, u32, u32, ctypes.c_char_p, u32,
@@ -591,6 +594,15 @@ class Index:
 
     def doc_count(self):
         return lib().sdbv_index_doc_count(self._ptr)
+
+    def level_rng(self):
+        """Level-RNG state (extension; persist + restore across cold
+        starts for same-seed graph determinism — any state is
+        reference-conformant, the reference reseeds from entropy)."""
+        return lib().sdbv_index_level_rng(self._ptr)
+
+    def set_level_rng(self, state):
+        lib().sdbv_index_set_level_rng(self._ptr, state)
 
     def doc_keys(self):
         """The doc-id -> record-key map (the hi/hd state a host persists

@@ -4222,6 +4222,20 @@ void sdbv_index_destroy(sdbv_index *ix) {
 }
 
 sdbv_hnsw *sdbv_index_hnsw(sdbv_index *ix) { return ix ? ix->h : nullptr; }
+
+// Level-RNG state carry-over (an EXTENSION record): the reference draws
+// insert levels from an entropy-seeded SmallRng (hnsw/mod.rs:263-266), so
+// the sequence restarts arbitrarily on every process start and ANY state
+// is reference-conformant. For this framework's determinism convention
+// (same seed => same graph) a host may persist the state across cold
+// starts and restore it after load.
+uint64_t sdbv_index_level_rng(sdbv_index *ix) {
+	return ix ? ix->h->rng_state : 0;
+}
+void sdbv_index_set_level_rng(sdbv_index *ix, uint64_t state) {
+	if (ix)
+		ix->h->rng_state = state;
+}
 uint64_t sdbv_index_doc_count(sdbv_index *ix) { return ix->doc2key.size(); }
 uint64_t sdbv_index_pending_count(sdbv_index *ix) {
 	return ix->pendings.size();
