@@ -854,9 +854,15 @@ static OrcPQ layer_insert(orc_hnsw *h, OrcLayer &layer, uint32_t q_id,
 		auto &conn = layer.edges[e_id];
 		if (conn.size() > layer.m_max) {
 			const float *e_pt = hvec(h, e_id);
-			OrcPQ e_c; // build_priority_list (layer.rs:390-404)
-			for (uint32_t n_id : conn)
+			OrcPQ e_c; // build_priority_list (layer.rs:390-404):
+			           // get_vector -> None skips removed elements, so a
+			           // prune also cleanses dangling edges
+			for (uint32_t n_id : conn) {
+				if (n_id < h->elem_present.size() &&
+				    !h->elem_present[n_id])
+					continue;
 				e_c.push(hdist(h, e_pt, hvec(h, n_id)), n_id);
+			}
 			std::vector<uint32_t> e_new;
 			select_neighbors(h, layer, e_id, e_pt, std::move(e_c), e_new);
 			layer.edges[e_id] = e_new;

@@ -2095,10 +2095,16 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 				conn = ee;
 		}
 		if (!conn.empty()) {
-			// prune (layer.rs:363-377) — distances computed outside the lock
+			// prune (layer.rs:363-377) — distances computed outside the
+			// lock. build_priority_list (layer.rs:389-404) SKIPS removed
+			// elements (get_vector -> None), so a prune also cleanses any
+			// dangling edges out of e's list.
 			PQ ec;
-			for (uint32_t nid : conn)
+			for (uint32_t nid : conn) {
+				if (nid < h->elem_present.size() && !h->elem_present[nid])
+					continue;
 				ec.push(dist_ee(h, e, nid), nid);
+			}
 			std::vector<uint32_t> enew;
 			select_neighbors(h, layer, e, vec(h, e),
 			                 h->metric == SDBV_METRIC_COSINE ? h->norms[e] : 0,
