@@ -255,3 +255,62 @@ def test_pendings_survive_cold_start():
     assert np.array_equal(i1, i2) and d1[0] == d2_[0] == 0.0
     ix.destroy()
     ix2.destroy()
+
+
+@pytest.mark.parametrize("case", range(8))
+def test_reload_write_cycle_across_params(case):
+    """Randomized dump -> reload (bindings + level-RNG carry) -> apply ->
+    write-more cycles across index parameters; the reloaded index must
+    track a never-reloaded twin exactly. This cycle found two real bugs
+    in-round (the allocator reconstruction and the prune's removed-element
+    gate), so it stays as a committed regression net."""
+    rng = np.random.default_rng(42000 + case)
+    d = int(rng.choice([8, 16]))
+    m = int(rng.choice([3, 4]))
+    m0 = int(rng.choice([m, 2 * m]))
+    efc = int(rng.choice([8, 24, 48]))
+    metric = str(rng.choice(["euclidean", "cosine"]))
+    ext = bool(rng.integers(0, 2))
+    keep = bool(rng.integers(0, 2))
+    seed = int(rng.integers(1, 2**31))
+    n = int(rng.integers(40, 120))
+    rows = oracle.gen_f32(seed ^ 0xABC, 0, 256, d)
+    ix = sa.index_create_host(d, metric=metric, m=m, m0=m0, efc=efc,
+                              extend=ext, keep=keep, seed=seed)
+    live = {}
+    for i in range(n):
+        key = int(rng.integers(0, 64))
+        r = rng.integers(0, 4)
+        if r < 2 or key not in live:
+            v = rows[int(rng.integers(0, 256))]
+            ix.enqueue(key, live.get(key), v)
+            live[key] = v
+        elif r == 2:
+            ix.enqueue(key, live[key], None)
+            del live[key]
+        else:
+            ix.apply_pendings()
+    pairs = ix.dump_kv()
+    ix2 = sa.load_kv_index(pairs, 0, d, metric=metric, m=m, m0=m0, efc=efc,
+                           extend=ext, keep=keep, seed=seed,
+                           doc_keys=ix.doc_keys())
+    ix2.set_level_rng(ix.level_rng())
+    assert ix.pending_count() == ix2.pending_count()
+    assert ix.apply_pendings() == ix2.apply_pendings()
+    a, b = ix.hnsw().l0_csr(), ix2.hnsw().l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    for i in range(15):  # and keep writing on both
+        key = int(rng.integers(0, 64))
+        v = rows[int(rng.integers(0, 256))]
+        ix.enqueue(key, live.get(key), v)
+        ix2.enqueue(key, live.get(key), v)
+        live[key] = v
+    assert ix.apply_pendings() == ix2.apply_pendings()
+    a, b = ix.hnsw().l0_csr(), ix2.hnsw().l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    q = rows[0] + np.float32(0.01)
+    r1, r2 = ix.knn_search(q, 8, 24), ix2.knn_search(q, 8, 24)
+    for x, y in zip(r1, r2):
+        assert np.array_equal(x, y)
+    ix.destroy()
+    ix2.destroy()
