@@ -240,3 +240,41 @@ def test_empty_and_edge_cases():
     assert p.doc_count() == o.doc_count()
     assert_same_search(p, o, v, 5, 10, "after apply")
     p.destroy()
+
+
+@pytest.mark.parametrize("case", range(4))
+def test_graph_remove_param_sweep(case):
+    """Randomized removal orders + reinserts across parameter combos
+    (metric x ext/keep heuristics x m0=m): bit-identical repaired graphs
+    between the two independent restatements at every step."""
+    import math
+    rng = np.random.default_rng(88000 + case)
+    d = int(rng.choice([8, 16, 24]))
+    m = int(rng.choice([3, 4, 8]))
+    m0 = int(rng.choice([m, 2 * m]))
+    efc = int(rng.choice([10, 30, 60]))
+    metric = str(rng.choice(["euclidean", "cosine"]))
+    ext = bool(rng.integers(0, 2))
+    keep = bool(rng.integers(0, 2))
+    seed = int(rng.integers(1, 2**31))
+    n = int(rng.integers(30, 150))
+    rows = oracle.gen_f32(seed ^ 0x77, 0, n, d)
+    h = sa.hnsw_create_host(d, metric=metric, m=m, m0=m0, efc=efc,
+                            extend=ext, keep=keep, seed=seed)
+    o = oracle.Hnsw(d, metric=metric, m=m, m0=m0, efc=efc, extend=ext,
+                    keep=keep, seed=seed, ml=1.0 / math.log(m))
+    h.insert_batch(rows, nthreads=1)
+    for r in rows:
+        o.insert(r)
+    order = rng.permutation(n)
+    nrem = int(rng.integers(1, n))
+    for j, e in enumerate(order[:nrem]):
+        assert h.remove(int(e)) == o.remove(int(e)), (case, j)
+    a, b = h.l0_csr(), o.l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    for r in oracle.gen_f32(seed ^ 0x99, 0, 10, d):
+        h.insert(r)
+        o.insert(r)
+    a, b = h.l0_csr(), o.l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    h.destroy()
