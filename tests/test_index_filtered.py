@@ -134,3 +134,55 @@ def test_expire_signal_fires_on_eviction():
     assert logs[0] == logs[1]
     assert logs[0], "12 pendings + graph into a k=5 builder must evict"
     p.destroy()
+
+
+@pytest.mark.parametrize("case", range(4))
+def test_filtered_param_sweep(case):
+    """Randomized filtered searches (with pendings outstanding and expire
+    logs) across parameter combos — the bounded form of the exploration
+    cycle in tools/explore.py."""
+    rng = np.random.default_rng(77000 + case)
+    d = int(rng.choice([8, 16, 32]))
+    m = int(rng.choice([3, 4, 8]))
+    m0 = int(rng.choice([m, 2 * m]))
+    efc = int(rng.choice([8, 24, 48]))
+    metric = str(rng.choice(["euclidean", "cosine"]))
+    ext = bool(rng.integers(0, 2))
+    keep = bool(rng.integers(0, 2))
+    seed = int(rng.integers(1, 2**31))
+    n = int(rng.integers(20, 150))
+    mod = int(rng.integers(2, 6))
+    rows = oracle.gen_f32(seed ^ 0x123, 0, 256, d)
+    p = sa.index_create_host(d, metric=metric, m=m, m0=m0, efc=efc,
+                             extend=ext, keep=keep, seed=seed)
+    o = oracle.Index(d, metric=metric, m=m, m0=m0, efc=efc, extend=ext,
+                     keep=keep, seed=seed)
+    live = {}
+    for i in range(n):
+        key = int(rng.integers(0, 48))
+        r = rng.integers(0, 4)
+        if r < 2 or key not in live:
+            v = rows[int(rng.integers(0, 256))]
+            p.enqueue(key, live.get(key), v)
+            o.enqueue(key, live.get(key), v)
+            live[key] = v
+        elif r == 2:
+            p.enqueue(key, live[key], None)
+            o.enqueue(key, live[key], None)
+            del live[key]
+        else:
+            assert p.apply_pendings() == o.apply_pendings()
+    pred = lambda kind, i: (int(i) % mod) != 0
+    plog, elog = [], []
+    for j in range(5):
+        q = rows[int(rng.integers(0, 256))] + np.float32(0.01)
+        k = int(rng.integers(1, 10))
+        ef = int(rng.integers(k, 40))
+        a = p.knn_search_filtered(q, k, ef, pred,
+                                  expire=lambda kk, ii: plog.append((kk, ii)))
+        b = o.knn_search_filtered(q, k, ef, pred,
+                                  expire=lambda kk, ii: elog.append((kk, ii)))
+        for x, y in zip(a, b):
+            assert np.array_equal(x, y), (case, j)
+    assert sorted(plog) == sorted(elog), case
+    p.destroy()
