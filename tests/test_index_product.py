@@ -278,3 +278,37 @@ def test_graph_remove_param_sweep(case):
     a, b = h.l0_csr(), o.l0_csr()
     assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
     h.destroy()
+
+
+def test_extreme_small_params_match_oracle():
+    """Corner parameters (m=2, m0=2, efc=2, ef=1): the degenerate windows
+    exercise the prune/eviction edges hardest; both restatements must
+    still agree bit-exactly."""
+    d = 4
+    rows = oracle.gen_f32(0xC0FFEE, 0, 64, d)
+    p = sa.index_create_host(d, metric="euclidean", m=2, m0=2, efc=2,
+                             seed=11)
+    o = oracle.Index(d, metric="euclidean", m=2, m0=2, efc=2, seed=11)
+    rng = np.random.default_rng(5)
+    live = {}
+    for i in range(120):
+        key = int(rng.integers(0, 16))
+        r = rng.integers(0, 5)
+        if r < 2 or key not in live:
+            v = rows[int(rng.integers(0, 64))]
+            p.enqueue(key, live.get(key), v)
+            o.enqueue(key, live.get(key), v)
+            live[key] = v
+        elif r == 2:
+            p.enqueue(key, live[key], None)
+            o.enqueue(key, live[key], None)
+            del live[key]
+        elif r == 3:
+            assert p.apply_pendings() == o.apply_pendings()
+        else:
+            q = rows[int(rng.integers(0, 64))]
+            assert_same_search(p, o, q, int(rng.integers(1, 5)), 1,
+                               f"step {i} ef=1")
+    assert p.apply_pendings() == o.apply_pendings()
+    assert_same_graph(p, o, "extreme params final")
+    p.destroy()
