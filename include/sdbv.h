@@ -17,9 +17,19 @@
  *    HBM-resident staging of a table's vectors.
  *  - sdbv_hnsw_* replace the layer-0 best-first search loop
  *    (surrealdb/core/src/idx/trees/hnsw/layer.rs:184-223) behind
- *    HnswIndex::knn_search (surrealdb/core/src/idx/trees/hnsw/index.rs:270-335).
- *    The host keeps: pendings merge (index.rs:372-420), doc-id expansion
- *    (Ids64, idx/trees/knn.rs:170-326), cond filtering (hnsw/filter.rs).
+ *    HnswIndex::knn_search (surrealdb/core/src/idx/trees/hnsw/index.rs:270-335),
+ *    plus the graph build (sequential = parity; parallel/snapshot = bench
+ *    mode) and element removal with neighbour repair.
+ *  - sdbv_index_* replace the whole HnswIndex operator surface: the Hp
+ *    pendings queue (index.rs:138-257), VecDocs/Ids64 doc expansion
+ *    (docs.rs, knn.rs:163-326), doc-id allocation with recycling, the
+ *    pendings-merged knn_search (index.rs:270-335) and the cond_filter
+ *    flow (hnsw/filter.rs) behind a host truthiness callback. The host
+ *    keeps only what touches the KV transaction: value->vector
+ *    extraction, the RecordIdKey<->handle map, WHERE evaluation, record
+ *    materialisation.
+ *  - sdbv_kvload_* / sdbv_*_dump_kv speak the reference's on-disk formats
+ *    (key/index/{he,hn,hs,hv,hp}.rs) for cold-start bulk loads and dumps.
  *
  * Plain pointers and sizes only; no torch types. All calls are synchronous
  * w.r.t. results and thread-safe (internal per-context mutex). The caller
