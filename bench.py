@@ -57,9 +57,31 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    n_gpus = max(world, args.gpus)
 
     import torch
+
+    if world == 1 and args.gpus > 1:
+        # Not under torchrun: self-launch one rank per GPU, or fail loudly.
+        # NEVER fall through to single-rank work reporting n_gpus > 1
+        # (VERDICT r01 weak #3).
+        ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        if ndev < args.gpus:
+            print(f"bench.py: --gpus {args.gpus} requested but only {ndev} "
+                  "visible GPU(s) and WORLD_SIZE is unset; refusing to run "
+                  "single-rank work as if it were multi-GPU", file=sys.stderr)
+            sys.exit(2)
+        import socket
+        import subprocess
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               f"--nproc-per-node={args.gpus}", "--master-addr", "127.0.0.1",
+               f"--master-port={port}", os.path.abspath(__file__)]
+        cmd += sys.argv[1:]
+        sys.exit(subprocess.call(cmd))
+
+    n_gpus = world
     dist = None
     if world > 1:
         import torch.distributed as tdist

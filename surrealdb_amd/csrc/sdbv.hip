@@ -2423,6 +2423,8 @@ void sdbv_hnsw_destroy(sdbv_hnsw *h) {
 	delete h;
 }
 
+static void hnsw_free_device_state(sdbv_hnsw *h); // defined below finalize
+
 static void hnsw_append_vec(sdbv_hnsw *h, const float *pt) {
 	h->vecs.insert(h->vecs.end(), pt, pt + h->d);
 	h->elem_present.push_back(1);
@@ -2431,8 +2433,17 @@ static void hnsw_append_vec(sdbv_hnsw *h, const float *pt) {
 }
 
 int sdbv_hnsw_insert(sdbv_hnsw *h, const float *pt) {
-	if (!h || h->finalized)
+	if (!h)
 		return SDBV_ERR_BAD_ARG;
+	// Writes on a finalized graph invalidate the device copy (same contract
+	// as hnsw_remove): drop it and let the next search re-finalize. The
+	// round-1 BAD_ARG guard here made idx_vd_insert silently skip the graph
+	// insert for every update applied after a GPU search (driver GPUTEST_r01
+	// failure: updated doc missing from post-apply top-K).
+	if (h->finalized) {
+		hnsw_free_device_state(h);
+		h->dirty = true;
+	}
 	uint32_t q_id = (uint32_t)h->next_id++;
 	hnsw_append_vec(h, pt);
 	hnsw::insert_at(h, q_id, hnsw::next_level(h), /*locked=*/false);
@@ -3387,12 +3398,14 @@ static void idx_docs_remove(sdbv_index *ix, uint64_t doc_id) {
 }
 
 // docs.rs:363-393 VecDocs::insert.
-static void idx_vd_insert(sdbv_index *ix, const float *v, uint64_t doc_id) {
+static int idx_vd_insert(sdbv_index *ix, const float *v, uint64_t doc_id) {
 	std::string key((const char *)v, (size_t)ix->h->d * 4);
 	auto it = ix->vd.find(key);
 	if (it == ix->vd.end()) {
 		uint32_t e_id = (uint32_t)ix->h->next_id;
-		sdbv_hnsw_insert(ix->h, v);
+		int rc = sdbv_hnsw_insert(ix->h, v);
+		if (rc != SDBV_OK)
+			return rc; // never silent: a dropped insert corrupts VecDocs
 		auto r = ix->vd.emplace(std::move(key), sdbv_index::ED{e_id, {}});
 		r.first->second.docs.v = {doc_id};
 		ix->by_elem[e_id] = &r.first->first;
@@ -3401,6 +3414,7 @@ static void idx_vd_insert(sdbv_index *ix, const float *v, uint64_t doc_id) {
 		if (ed.docs.insert_ret_variant(doc_id))
 			it->second = ed; // persisted only on a new variant
 	}
+	return SDBV_OK;
 }
 
 // docs.rs:424-449 VecDocs::remove.
@@ -3593,8 +3607,10 @@ static int idx_graph_search_gpu(sdbv_index *ix, const float *q, uint32_t k,
 // golden bytes). Values use the `revision` crate 0.17.0 wire format
 // (dependency pinned by Cargo.lock, source not vendored): varint revision
 // tag + varint enum variant + varint lengths + fixed little-endian
-// primitives — pinned by the hv.rs golden vectors at small sizes; the
-// varint extrapolation beyond 127 follows the crate's published LEB128.
+// primitives — pinned by the hv.rs golden vectors at small sizes; beyond
+// the single-byte range (values >= 251) the varints use bincode-style
+// markers (0xFB + u16 LE, 0xFC + u32 LE, 0xFD + u64 LE), pinned by the
+// catalog compat fixtures' 900/3600/86400 durations.
 // ===========================================================================
 
 namespace kvc {
@@ -3820,18 +3836,25 @@ static bool dec_state(const uint8_t *p, const uint8_t *end, HnswStateKV *s) {
 	return true;
 }
 
-// revisioned ElementDocs (docs.rs:164-177): e_id u64 LE + Ids64 (variant
-// varint: Empty=0 One=1 Vec2=2..Vec8=8 Bits=9; fixed u64 LE ids; Bits
-// (RoaringTreemap) unsupported -> false).
-static void enc_element_docs(std::vector<uint8_t> &b, uint64_t e_id,
+// revisioned ElementDocs (docs.rs:164-177): e_id varint + Ids64 (variant
+// varint: Empty=0 One=1 Vec2=2..Vec8=8 Bits=9; doc ids as unsigned
+// varints like every unsigned field). The reference's Bits variant is a
+// serialized RoaringTreemap — not implemented here, so encode REFUSES a
+// bits-mode set (callers return SDBV_ERR_UNSUPPORTED rather than emit a
+// record a real surrealdb could not deserialize) and decode rejects
+// variant 9.
+static bool enc_element_docs(std::vector<uint8_t> &b, uint64_t e_id,
                              const vdocs::Ids64 &docs) {
+	size_t n = docs.v.size();
+	if (docs.bits || n > 8)
+		return false; // Bits (roaring) — refuse, never emit malformed
 	put_varint(b, 1);     // revision
 	put_varint(b, e_id);  // ElementId: unsigned -> varint
 	put_varint(b, 1);     // Ids64 revision
-	size_t n = docs.v.size();
-	put_varint(b, n == 0 ? 0 : (n <= 8 ? n : 9));
+	put_varint(b, n);
 	for (uint64_t d : docs.v)
 		put_varint(b, d); // DocId: unsigned -> varint
+	return true;
 }
 static bool dec_element_docs(const uint8_t *p, const uint8_t *end,
                              uint64_t *e_id, vdocs::Ids64 *docs) {
@@ -4289,8 +4312,11 @@ int sdbv_index_apply_pendings(sdbv_index *ix, uint64_t *out_count) {
 		if (!p.news.empty()) {
 			uint64_t doc_id =
 			    (p.kind == 0) ? p.id : idx_docs_resolve(ix, p.id);
-			for (size_t i = 0; i * d < p.news.size(); i++)
-				idx_vd_insert(ix, p.news.data() + i * d, doc_id);
+			for (size_t i = 0; i * d < p.news.size(); i++) {
+				int rc = idx_vd_insert(ix, p.news.data() + i * d, doc_id);
+				if (rc != SDBV_OK)
+					return rc;
+			}
 		}
 		count++;
 	}
@@ -4627,6 +4653,14 @@ static int kvload_build_graph(sdbv_kvload *L, sdbv_hnsw **out) {
 	if (rc)
 		return rc;
 	uint64_t n = L->state.next_element_id;
+	// Untrusted Hs fields bound the allocations below (n*d floats, one
+	// n-sized edge vector per layer): reject absurd values up front like
+	// the 2^32 doc-id guard, instead of letting bad_alloc abort through
+	// the extern "C" boundary.
+	if (n > (1ull << 32) || L->state.n_upper_layers > 64) {
+		sdbv_hnsw_destroy(h);
+		return SDBV_ERR_UNSUPPORTED;
+	}
 	h->next_id = n;
 	h->vecs.assign((size_t)n * L->d, 0.0f);
 	h->elem_present.assign(n, 0);
@@ -4664,6 +4698,14 @@ static int kvload_build_graph(sdbv_kvload *L, sdbv_hnsw **out) {
 				sdbv_hnsw_destroy(h);
 				return SDBV_ERR_BAD_ARG;
 			}
+		// The GPU per-hop scratch (rows_pinned/rows_dev/dists_pinned and
+		// the search fdists vector) is sized m_max+1; a dump whose degree
+		// exceeds the declared layer cap (corrupt, or an honest dump
+		// loaded under smaller m/m0) would overflow it — reject here.
+		if (nd.second.size() > h->layers[layer].m_max) {
+			sdbv_hnsw_destroy(h);
+			return SDBV_ERR_BAD_ARG;
+		}
 		h->layers[layer].edges[node] = nd.second;
 		h->layers[layer].in_layer[node] = 1;
 	}
@@ -4845,7 +4887,8 @@ int sdbv_index_dump_kv(sdbv_index *ix, uint32_t ns, uint32_t db,
 		v.clear();
 		kvc::key_hv(k, ns, db, tb, ix_id, (const float *)e.first.data(),
 		            ix->h->d);
-		kvc::enc_element_docs(v, e.second.e_id, e.second.docs);
+		if (!kvc::enc_element_docs(v, e.second.e_id, e.second.docs))
+			return SDBV_ERR_UNSUPPORTED; // Ids64 in Bits (roaring) mode
 		if (write(user, k.data(), k.size(), v.data(), v.size()))
 			return SDBV_ERR_BAD_ARG;
 	}

@@ -32,6 +32,15 @@ def load_golden(name, limit):
 
 @pytest.fixture(scope="module")
 def golden_index():
+    # The reference's OWN scale: test_recall_euclidean passes ingest
+    # limit 1000 and query limit 300 (hnsw/mod.rs:1146-1152 —
+    # `test_recall("hnsw-random-9000-20-euclidean.gz", 1000,
+    # "hnsw-random-5000-20-euclidean.gz", 300, p, &[(10,0.98),(40,1.0)])`).
+    # The recall bars are defined at THESE limits only: at the full 9000
+    # rows recall@efs10 drops below 0.98 for this algorithm regardless of
+    # implementation (measured 0.959 on the bit-exact restatement) — see
+    # test_recall_full_files_floor below, which pins the full-file
+    # behaviour without inventing a bar the reference never states.
     ingest = load_golden("hnsw-random-9000-20-euclidean.gz", 1000)
     queries = load_golden("hnsw-random-5000-20-euclidean.gz", 300)
     h = oracle.Hnsw(20, metric="euclidean", m=8, m0=16, efc=100,
@@ -71,11 +80,13 @@ def test_recall_efs40_exact(golden_index):
 
 
 def test_heuristic_variants_recall():
-    """Restates test_recall_euclidean_keep_pruned_connections / _full
-    (hnsw/mod.rs:1158-1184) at reduced size."""
-    ingest = load_golden("hnsw-random-9000-20-euclidean.gz", 500)
-    queries = load_golden("hnsw-random-5000-20-euclidean.gz", 100)
-    for extend, keep in [(False, True), (True, True)]:
+    """Restates test_recall_euclidean_keep_pruned_connections (ingest 750 /
+    queries 200) and test_recall_euclidean_full (500 / 100) at the
+    reference's own per-variant limits (hnsw/mod.rs:1158-1184)."""
+    for extend, keep, n_ing, n_q in [(False, True, 750, 200),
+                                     (True, True, 500, 100)]:
+        ingest = load_golden("hnsw-random-9000-20-euclidean.gz", n_ing)
+        queries = load_golden("hnsw-random-5000-20-euclidean.gz", n_q)
         h = oracle.Hnsw(20, metric="euclidean", m=8, m0=16, efc=100,
                         extend=extend, keep=keep, ml=1.0 / math.log(8.0))
         for row in ingest:
@@ -87,6 +98,28 @@ def test_heuristic_variants_recall():
             bids, _ = oracle.topk_f32("euclidean", ingest, q, k)
             total += len(set(ids.tolist()) & set(bids.tolist())) / k
         assert total / len(queries) == 1.0, (extend, keep)
+
+
+def test_recall_full_files_floor():
+    """The FULL golden files (9000 ingest / 5000 queries — a scale the
+    reference's own tests never run): no reference bar exists here, so pin
+    a floor on the measured behaviour of the bit-exact restatement
+    (recall@10 efs=10 was 0.9593 when this was written) to catch build or
+    search regressions at scale."""
+    ingest = load_golden("hnsw-random-9000-20-euclidean.gz", 9000)
+    queries = load_golden("hnsw-random-5000-20-euclidean.gz", 5000)
+    h = oracle.Hnsw(20, metric="euclidean", m=8, m0=16, efc=100,
+                    ml=1.0 / math.log(8.0), seed=0x5DB1)
+    for row in ingest:
+        h.insert(row)
+    assert h.check_props() == 0
+    k, total = 10, 0.0
+    for q in queries:
+        ids, _ = h.search(q, k, 10)
+        bids, _ = oracle.topk_f32("euclidean", ingest, q, k)
+        total += len(set(ids.tolist()) & set(bids.tolist())) / k
+    recall = total / len(queries)
+    assert recall >= 0.95, recall
 
 
 def test_upper_layer_structure(golden_index):

@@ -314,3 +314,61 @@ def test_reload_write_cycle_across_params(case):
         assert np.array_equal(x, y)
     ix.destroy()
     ix2.destroy()
+
+
+def test_dump_refuses_bits_mode_ids64():
+    """An Ids64 past 8 docs collapses to Bits (a serialized RoaringTreemap
+    in the reference, knn.rs:170-326) which this codec does not emit:
+    dump_kv must REFUSE with SDBV_ERR_UNSUPPORTED rather than write a
+    malformed variant-9 record a real surrealdb could not deserialize
+    (round-1 advisor finding, medium)."""
+    d = 8
+    ix = sa.index_create_host(d, metric="euclidean", m=4, m0=8, efc=20,
+                              seed=1)
+    v = oracle.gen_f32(0x1, 0, 1, d)[0]
+    for doc in range(12):  # 12 docs on ONE vector -> Bits mode
+        ix.enqueue(doc, None, v)
+    ix.apply_pendings()
+    with pytest.raises(sa.SdbvError):
+        ix.dump_kv()
+    ix.destroy()
+
+
+def test_loader_rejects_degree_above_declared_m0():
+    """An honest m0=16 dump loaded under declared m0=4: nodes with more
+    edges than the declared layer cap would overflow the m0+1-sized GPU
+    per-hop scratch — the loader must reject them (round-1 advisor
+    finding, high)."""
+    d, n = 8, 200
+    rows = oracle.gen_f32(0x55, 0, n, d)
+    g = sa.hnsw_create_host(d, metric="euclidean", m=8, m0=16, efc=40,
+                            seed=3)
+    g.insert_batch(rows, nthreads=1)
+    pairs = g.dump_kv()
+    # same dump, smaller declared caps -> SDBV_ERR_BAD_ARG
+    with pytest.raises(sa.SdbvError):
+        sa.load_kv_hnsw(pairs, d, metric="euclidean", m=2, m0=4, efc=40,
+                        seed=3)
+    g.destroy()
+
+
+def test_loader_bounds_untrusted_state_fields():
+    """Corrupt Hs fields (absurd next_element_id / n_upper_layers) must be
+    rejected with an error, not trigger multi-terabyte allocations whose
+    bad_alloc aborts through the C ABI (round-1 advisor finding, low)."""
+    d = 8
+    g = sa.hnsw_create_host(d, metric="euclidean", m=4, m0=8, efc=20, seed=2)
+    g.insert(np.zeros(d, dtype=np.float32))
+    pairs = g.dump_kv()
+    hs_key = PREFIX + b"s"
+    others = [(k, v) for (k, v) in pairs if k != hs_key]
+    assert len(others) == len(pairs) - 1
+    # hand-encode an Hs with next_element_id ~ 2^40 (varint 0xfd + u64 LE):
+    # rev=1, version u16 varint, ep Option tag 0, next_element_id,
+    # layer0 LayerState(rev=1, version, chunks=0), n_upper_layers=0
+    bad_state = bytes([1, 1, 0, 0xFD]) + (1 << 40).to_bytes(8, "little") + \
+        bytes([1, 1, 0, 0])
+    with pytest.raises(sa.SdbvError):
+        sa.load_kv_hnsw(others + [(hs_key, bad_state)], d,
+                        metric="euclidean", m=4, m0=8, efc=20, seed=2)
+    g.destroy()
