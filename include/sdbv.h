@@ -175,6 +175,15 @@ int sdbv_hnsw_insert_batch(sdbv_hnsw *, const float *pts, uint64_t n,
 int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *, const float *pts,
                                     uint64_t n, uint32_t chunk,
                                     int nthreads);
+/* GPU-accelerated chunked snapshot build (hnsw/mod.rs:230-394 build hot
+ * loop): same algorithm and same resulting graph as
+ * sdbv_hnsw_insert_batch_snapshot, but each chunk's level-0 efc-searches
+ * run as one persistent-kernel launch against a delta-updated device
+ * adjacency; the select/edges/prune apply half is the identical host code.
+ * Requires a device context; efc <= 512 (the in-kernel w window). */
+int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *, const float *pts,
+                                        uint64_t n, uint32_t chunk,
+                                        int nthreads);
 /* Stage vectors (feature-major + norms) into the device table slot `table`;
  * required before sdbv_hnsw_knn. */
 int sdbv_hnsw_finalize(sdbv_hnsw *, uint64_t table);
@@ -195,6 +204,14 @@ int sdbv_hnsw_knn_host(sdbv_hnsw *, const float *q, uint32_t k, uint32_t ef,
 int sdbv_hnsw_knn_batch(sdbv_hnsw *, const float *Q, uint32_t b, uint32_t k,
                         uint32_t ef, uint64_t *out_ids, double *out_dists,
                         uint32_t *out_ns);
+/* Full per-layer graph export (CSR + membership + enter point + zero-copy
+ * vector view): lets a host-side searcher (the bench's oracle cpu_baseline
+ * leg) run on the exact graph the device searches. */
+uint64_t sdbv_hnsw_layer_edge_count(sdbv_hnsw *, uint32_t layer);
+void sdbv_hnsw_layer_export(sdbv_hnsw *, uint32_t layer, uint32_t *offsets,
+                            uint32_t *edges, uint8_t *in_layer);
+int64_t sdbv_hnsw_enter_point(sdbv_hnsw *);
+const float *sdbv_hnsw_vecs_ptr(sdbv_hnsw *);
 void sdbv_hnsw_destroy(sdbv_hnsw *);
 /* Remove one graph element with neighbour repair (Hnsw::remove,
  * hnsw/mod.rs:398-455 + layer.rs:408-460). Pre-finalize host graphs only;

@@ -93,6 +93,12 @@ def _load():
     lib.orc_hnsw_entry_point.argtypes = [ctypes.c_void_p]
     lib.orc_hnsw_remove.restype = ctypes.c_int
     lib.orc_hnsw_remove.argtypes = [ctypes.c_void_p, u64]
+    lib.orc_hnsw_import.restype = ctypes.c_void_p
+    lib.orc_hnsw_import.argtypes = [u32, u8, ctypes.c_double, u32, u32, u32,
+                                    u64, f32p, ctypes.c_int64, u32]
+    lib.orc_hnsw_import_layer.restype = ctypes.c_int
+    lib.orc_hnsw_import_layer.argtypes = [ctypes.c_void_p, u32, u32p, u32p,
+                                          ctypes.POINTER(u8)]
     # index layer (hnsw/index.rs + docs.rs + knn.rs Ids64)
     u8p = ctypes.POINTER(u8)
     lib.orc_index_new.restype = ctypes.c_void_p
@@ -278,6 +284,34 @@ class Hnsw:
     def remove(self, e_id):
         """Hnsw::remove (hnsw/mod.rs:398-455). True if removed."""
         return bool(lib().orc_hnsw_remove(self._h, e_id))
+
+    @classmethod
+    def import_graph(cls, d, metric, m, m0, efc, vecs, enter_point, layers,
+                     order=0.0):
+        """Build an oracle searcher over a graph built elsewhere (the
+        bench cpu_baseline leg searches the product's own graph). `vecs` is
+        the n x d f32 row store; `layers` a list of (offsets u32[n+1],
+        edges u32[], in_layer u8[n]) from layer 0 up."""
+        vecs = np.ascontiguousarray(vecs, dtype=np.float32)
+        n = vecs.shape[0]
+        self = cls.__new__(cls)
+        self.d = d
+        self._h = lib().orc_hnsw_import(d, METRICS[metric], order, m, m0,
+                                        efc, n, _f32p(vecs), enter_point,
+                                        len(layers))
+        u32p = ctypes.POINTER(ctypes.c_uint32)
+        u8p = ctypes.POINTER(ctypes.c_uint8)
+        for l, (offsets, edges, in_layer) in enumerate(layers):
+            offsets = np.ascontiguousarray(offsets, dtype=np.uint32)
+            edges = np.ascontiguousarray(edges, dtype=np.uint32)
+            if edges.size == 0:
+                edges = np.zeros(1, dtype=np.uint32)
+            in_layer = np.ascontiguousarray(in_layer, dtype=np.uint8)
+            rc = lib().orc_hnsw_import_layer(
+                self._h, l, offsets.ctypes.data_as(u32p),
+                edges.ctypes.data_as(u32p), in_layer.ctypes.data_as(u8p))
+            assert rc == 0, rc
+        return self
 
     def __del__(self):
         try:

@@ -1092,6 +1092,42 @@ void orc_hnsw_l0_export(orc_hnsw *h, uint32_t *offsets, uint32_t *edges) {
 	offsets[h->next_id] = off;
 }
 
+// Import a prebuilt graph (vectors + per-layer adjacency) so the oracle
+// can SEARCH a graph built elsewhere — the bench's cpu_baseline leg runs
+// orc_hnsw_search on the very graph the GPU searches (test infrastructure,
+// like everything in this file). Layers load via orc_hnsw_import_layer.
+orc_hnsw *orc_hnsw_import(uint32_t d, uint8_t metric, double order,
+                          uint32_t m, uint32_t m0, uint32_t efc,
+                          uint64_t n, const float *vecs, int64_t enter_point,
+                          uint32_t nlayers) {
+	auto *h = orc_hnsw_new(d, metric, order, m, m0, efc, 0, 0, 0, 0.0);
+	h->vecs.assign(vecs, vecs + n * d);
+	h->next_id = n;
+	h->elem_present.assign(n, 1);
+	h->top_layer.assign(n, 0);
+	h->enter_point = enter_point;
+	h->layers.clear();
+	for (uint32_t l = 0; l < (nlayers ? nlayers : 1); l++) {
+		OrcLayer L{std::vector<std::vector<uint32_t>>(n), l == 0 ? m0 : m};
+		L.in_layer.assign(n, 0);
+		h->layers.push_back(std::move(L));
+	}
+	return h;
+}
+int orc_hnsw_import_layer(orc_hnsw *h, uint32_t l, const uint32_t *offsets,
+                          const uint32_t *edges, const uint8_t *in_layer) {
+	if (!h || l >= h->layers.size())
+		return -1;
+	OrcLayer &L = h->layers[l];
+	for (uint64_t i = 0; i < h->next_id; i++) {
+		L.in_layer[i] = in_layer[i];
+		L.edges[i].assign(edges + offsets[i], edges + offsets[i + 1]);
+		if (in_layer[i] && (int32_t)l > h->top_layer[i])
+			h->top_layer[i] = (int32_t)l;
+	}
+	return 0;
+}
+
 // Host-side upper-layer descent (search_ep, hnsw/mod.rs:521-548) for the GPU
 // path: returns the layer-0 entry element and its distance.
 void orc_hnsw_search_ep(orc_hnsw *h, const float *q, uint64_t *ep_id_out,

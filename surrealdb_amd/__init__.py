@@ -86,6 +86,8 @@ def lib():
     L.sdbv_hnsw_insert_batch.argtypes = [vp, f32p, u64, ctypes.c_int]
     L.sdbv_hnsw_insert_batch_snapshot.argtypes = [vp, f32p, u64, u32,
                                                   ctypes.c_int]
+    L.sdbv_hnsw_insert_batch_snapshot_gpu.argtypes = [vp, f32p, u64, u32,
+                                                      ctypes.c_int]
     L.sdbv_hnsw_finalize.argtypes = [vp, u64]
     L.sdbv_hnsw_knn.argtypes = [vp, f32p, u32, u32, u64p, f64p, u32p]
     L.sdbv_hnsw_knn_batch.argtypes = [vp, f32p, u32, u32, u32, u64p, f64p, u32p]
@@ -97,6 +99,14 @@ def lib():
     L.sdbv_hnsw_l0_edge_count.restype = u64
     L.sdbv_hnsw_l0_edge_count.argtypes = [vp]
     L.sdbv_hnsw_l0_export.argtypes = [vp, u32p, u32p]
+    L.sdbv_hnsw_layer_edge_count.restype = u64
+    L.sdbv_hnsw_layer_edge_count.argtypes = [vp, u32]
+    L.sdbv_hnsw_layer_export.argtypes = [vp, u32, u32p, u32p,
+                                         ctypes.POINTER(u8)]
+    L.sdbv_hnsw_enter_point.restype = ctypes.c_int64
+    L.sdbv_hnsw_enter_point.argtypes = [vp]
+    L.sdbv_hnsw_vecs_ptr.restype = ctypes.POINTER(ctypes.c_float)
+    L.sdbv_hnsw_vecs_ptr.argtypes = [vp]
     L.sdbv_hnsw_remove.restype = ctypes.c_int
     L.sdbv_hnsw_remove.argtypes = [vp, u64]
     L.sdbv_hnsw_knn_host.argtypes = [vp, f32p, u32, u32, u64p, f64p, u32p]
@@ -330,6 +340,19 @@ class Hnsw:
                    pts.shape[0], chunk, nthreads),
                "sdbv_hnsw_insert_batch_snapshot")
 
+    def insert_batch_snapshot_gpu(self, pts, chunk, nthreads=0):
+        """GPU-accelerated chunked snapshot build: per-chunk level-0
+        efc-searches as one persistent-kernel launch; same resulting graph
+        as insert_batch_snapshot (device context required)."""
+        import numpy as np
+        pts = np.ascontiguousarray(pts, dtype=np.float32)
+        _check(self._ctx._ptr,
+               lib().sdbv_hnsw_insert_batch_snapshot_gpu(
+                   self._ptr,
+                   pts.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+                   pts.shape[0], chunk, nthreads),
+               "sdbv_hnsw_insert_batch_snapshot_gpu")
+
     def finalize(self, table):
         _check(self._ctx._ptr, lib().sdbv_hnsw_finalize(self._ptr, table),
                "sdbv_hnsw_finalize")
@@ -404,6 +427,31 @@ class Hnsw:
         """Dump the graph as reference-format (key, value) KV pairs
         (He/Hn/Hs; key/index/{he,hn,hs}.rs byte layouts)."""
         return _dump_kv(lib().sdbv_hnsw_dump_kv, self._ptr, ns, db, tb, ix)
+
+    def layer_csr(self, l):
+        """(offsets, edges, in_layer) of layer l."""
+        import numpy as np
+        n = self.n()
+        ec = lib().sdbv_hnsw_layer_edge_count(self._ptr, l)
+        offsets = np.empty(n + 1, dtype=np.uint32)
+        edges = np.empty(max(ec, 1), dtype=np.uint32)
+        in_layer = np.empty(n, dtype=np.uint8)
+        u32p = ctypes.POINTER(ctypes.c_uint32)
+        lib().sdbv_hnsw_layer_export(
+            self._ptr, l, offsets.ctypes.data_as(u32p),
+            edges.ctypes.data_as(u32p),
+            in_layer.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
+        return offsets, edges[:ec], in_layer
+
+    def enter_point(self):
+        return lib().sdbv_hnsw_enter_point(self._ptr)
+
+    def vecs_view(self):
+        """Zero-copy numpy view of the host n x d f32 row store."""
+        import numpy as np
+        n = self.n()
+        ptr = lib().sdbv_hnsw_vecs_ptr(self._ptr)
+        return np.ctypeslib.as_array(ptr, shape=(n, self.d))
 
     def remove(self, e_id):
         """Hnsw::remove (hnsw/mod.rs:398-455). True if removed. Host graphs

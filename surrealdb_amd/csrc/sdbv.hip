@@ -841,10 +841,17 @@ __global__ void k_batch_exact(const float *__restrict__ cm,
 #define HQ_EF_CAP 512
 #define HQ_FLAG_OVERFLOW 1u
 
+// PADDED=0: offsets/edges are the scrubbed layer-0 CSR (search path after
+// finalize). PADDED=1: `offsets` is a per-node degree array and `edges` a
+// padded [n][stride] adjacency — the GPU snapshot build's delta-updatable
+// graph (sdbv_hnsw_insert_batch_snapshot_gpu), where per-chunk edge
+// changes scatter into rows instead of re-laying-out a CSR.
+template <int PADDED>
 __global__ __launch_bounds__(64) void k_hnsw_search(
     const float *__restrict__ rm, const double *__restrict__ norms,
     uint32_t d, int metric, const uint32_t *__restrict__ offsets,
-    const uint32_t *__restrict__ edges, const float *__restrict__ Q,
+    const uint32_t *__restrict__ edges, uint32_t stride,
+    const float *__restrict__ Q,
     const double *__restrict__ qnorms, const uint32_t *__restrict__ ep_rows,
     const double *__restrict__ ep_dists, uint32_t *__restrict__ visited,
     uint64_t vwords_per_q, uint32_t k, uint32_t ef,
@@ -965,7 +972,14 @@ __global__ __launch_bounds__(64) void k_hnsw_search(
 			c_key[ci] = ~0ULL; // tombstone
 		__syncthreads();
 
-		uint32_t e0 = offsets[doc], e1 = offsets[doc + 1];
+		uint32_t e0, e1;
+		if (PADDED) {
+			e0 = doc * stride;
+			e1 = e0 + offsets[doc]; // offsets = per-node degree
+		} else {
+			e0 = offsets[doc];
+			e1 = offsets[doc + 1];
+		}
 		for (uint32_t base = e0; base < e1; base += 64) {
 			uint32_t my_e = ~0u;
 			double my_d = 0.0;
@@ -1110,6 +1124,27 @@ __global__ __launch_bounds__(64) void k_hnsw_search(
 		out_cnt[qid] = out_m;
 		out_flags[qid] = flags;
 	}
+}
+
+// Scatter per-chunk adjacency deltas into the padded device graph: upd_ids
+// names the dirty nodes, upd_deg their new degrees, upd_edges their new
+// edge rows (cnt x stride, host-packed). One thread per slot.
+__global__ void k_adj_scatter(const uint32_t *__restrict__ upd_ids,
+                              const uint32_t *__restrict__ upd_deg,
+                              const uint32_t *__restrict__ upd_edges,
+                              uint32_t cnt, uint32_t stride,
+                              uint32_t *__restrict__ adj,
+                              uint32_t *__restrict__ deg) {
+	uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+	if (t >= (uint64_t)cnt * stride)
+		return;
+	uint32_t i = (uint32_t)(t / stride);
+	uint32_t s = (uint32_t)(t % stride);
+	uint32_t node = upd_ids[i];
+	if (s == 0)
+		deg[node] = upd_deg[i];
+	if (s < upd_deg[i])
+		adj[(uint64_t)node * stride + s] = upd_edges[t];
 }
 
 // ---------------------------------------------------------------------------
@@ -1871,6 +1906,20 @@ struct sdbv_hnsw {
 	double *norms_dev = nullptr;    // per-element f64 norm (cosine)
 	uint32_t *vis_dev = nullptr;    // visited bitsets scratch
 	uint64_t vis_cap = 0;
+	uint32_t max_deg = 0; // actual max layer-0 degree at finalize (scratch
+	                      // sizing; a parallel build can transiently exceed
+	                      // m0 — see layer_insert_apply's keep-back)
+	// GPU snapshot-build state (padded layer-0 adjacency, delta-updated)
+	uint32_t *adj_dev = nullptr; // [n][adj_stride]
+	uint32_t *deg_dev = nullptr; // [n]
+	uint32_t adj_stride = 0;
+	uint64_t adj_nodes = 0; // allocated node capacity of adj_dev/deg_dev
+	uint64_t dev_rows = 0;  // rows of rm_dev/norms_dev currently uploaded
+	// dirty tracking for the per-chunk adjacency sync: bytes set under the
+	// striped node locks wherever layer-0 edge lists mutate. Only active
+	// (non-empty) inside the GPU snapshot build, where layers are
+	// pre-created so h->layers[0] never reallocates under workers.
+	std::vector<uint8_t> l0_dirty;
 	std::string err;
 };
 
@@ -2065,6 +2114,9 @@ static void select_neighbors(sdbv_hnsw *h, const Layer &layer, uint32_t q_id,
 static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
                                const float *q_pt, double q_norm, PQ w,
                                bool locked) {
+	// dirty tracking for the GPU snapshot build's device adjacency (only
+	// layer 0 lives on the device; flags written under the node locks)
+	const bool track = !h->l0_dirty.empty() && &layer == &h->layers[0];
 	std::vector<uint32_t> neighbors;
 	select_neighbors(h, layer, q_id, q_pt, q_norm, std::move(w), neighbors,
 	                 locked);
@@ -2077,6 +2129,8 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			if (e != q_id &&
 			    std::find(eq.begin(), eq.end(), e) == eq.end())
 				eq.push_back(e);
+		if (track)
+			h->l0_dirty[q_id] = 1;
 	}
 	for (uint32_t e : neighbors) {
 		if (e == q_id)
@@ -2093,6 +2147,8 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 				ee.push_back(q_id);
 			if (ee.size() > layer.m_max)
 				conn = ee;
+			if (track)
+				h->l0_dirty[e] = 1;
 		}
 		if (!conn.empty()) {
 			// prune (layer.rs:363-377) — distances computed outside the
@@ -2117,6 +2173,8 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 				    std::find(enew.begin(), enew.end(), cur) == enew.end())
 					enew.push_back(cur);
 			layer.edges[e] = enew;
+			if (track)
+				h->l0_dirty[e] = 1;
 		}
 	}
 }
@@ -2222,11 +2280,12 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 // later chunks); quality is pinned by the same recall bars. Upper-level
 // elements (~1/m of the batch) insert sequentially at the chunk front so
 // the layer structure exists before the snapshot searches.
-static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
-                                hnsw::PQ &w_out) {
+// Upper-layer greedy descent only (the host half shared by the CPU and GPU
+// snapshot search paths); returns the layer-0 entry point + distance.
+static void snapshot_descend_one(sdbv_hnsw *h, uint32_t q_id, double q_norm,
+                                 uint32_t *ep_out, double *epd_out) {
 	using namespace hnsw;
 	const float *q_pt = vec(h, q_id);
-	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
 	uint32_t ep_id = (uint32_t)h->enter_point;
 	double ep_dist = dist(h, q_pt, q_norm, ep_id);
 	for (size_t l = h->layers.size() - 1; l >= 1; l--) {
@@ -2243,6 +2302,18 @@ static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
 			ep_id = ii;
 		}
 	}
+	*ep_out = ep_id;
+	*epd_out = ep_dist;
+}
+
+static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
+                                hnsw::PQ &w_out) {
+	using namespace hnsw;
+	const float *q_pt = vec(h, q_id);
+	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+	uint32_t ep_id;
+	double ep_dist;
+	snapshot_descend_one(h, q_id, q_norm, &ep_id, &ep_dist);
 	PQ eps;
 	eps.push(ep_dist, ep_id);
 	PQ w = eps;
@@ -2641,6 +2712,356 @@ int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *h, const float *pts,
 	return SDBV_OK;
 }
 
+// GPU-accelerated chunked snapshot build (the configs[2]
+// 10M-row build): identical algorithm and results to
+// sdbv_hnsw_insert_batch_snapshot — upper-level elements insert on the host
+// (same classic locked path), but each chunk's level-0 efc-searches run as
+// ONE persistent-kernel launch (k_hnsw_search<PADDED=1>, one workgroup per
+// chunk element, k = ef = efc so the full w window comes back) against a
+// delta-updated padded device adjacency. The apply half (select + edges +
+// prunes, layer.rs:342-387) is unchanged host code, so graph parity with
+// the host snapshot build is structural: the kernel returns exactly the
+// host search's w set (exact queue semantics, validated bit-exact in
+// tests/test_gpu_insert.py), and everything downstream is the same code.
+int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
+                                        uint64_t n, uint32_t chunk,
+                                        int nthreads) {
+	using namespace hnsw;
+	if (!h || !h->ctx || chunk == 0 || n == 0 || h->efc == 0 ||
+	    h->efc > HQ_EF_CAP)
+		return SDBV_ERR_BAD_ARG;
+	if (h->finalized) { // writes invalidate the finalized device state
+		hnsw_free_device_state(h);
+		h->dirty = true;
+	}
+	if (nthreads <= 0)
+		nthreads = (int)std::thread::hardware_concurrency();
+	sdbv_ctx *ctx = h->ctx;
+	std::lock_guard<std::mutex> ctxlk(ctx->mu);
+	uint64_t base = h->next_id;
+	std::vector<uint32_t> levels(n);
+	for (uint64_t i = 0; i < n; i++)
+		levels[i] = next_level(h); // sequential RNG contract, per ordinal
+	h->next_id += n;
+	h->vecs.reserve(h->vecs.size() + n * h->d);
+	for (uint64_t i = 0; i < n; i++)
+		hnsw_append_vec(h, pts + i * h->d);
+	// pre-create + pre-size every layer (same discipline as the host
+	// snapshot build: h->layers never grows under workers, which also
+	// keeps the &layer == &layers[0] dirty-tracking check stable)
+	{
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		uint32_t max_level = 0;
+		for (uint64_t i = 0; i < n; i++)
+			max_level = std::max(max_level, levels[i]);
+		while (h->layers.size() <= max_level)
+			h->layers.push_back(hnsw::Layer{{}, h->m});
+		uint64_t ne = h->vecs.size() / h->d;
+		for (auto &l : h->layers) {
+			if (l.edges.size() < ne)
+				l.edges.resize(ne);
+			if (l.in_layer.size() < ne)
+				l.in_layer.resize(ne, 0);
+		}
+	}
+	const uint64_t nelem = h->vecs.size() / h->d;
+	const uint32_t d = h->d;
+	// ---- device state: full row-major vector store (rows are immutable
+	// once appended) + padded layer-0 adjacency, rebuilt fresh per call ----
+	for (void **p : {(void **)&h->rm_dev, (void **)&h->norms_dev,
+	                 (void **)&h->adj_dev, (void **)&h->deg_dev})
+		if (*p) {
+			(void)hipFree(*p);
+			*p = nullptr;
+		}
+	HIP_CHECK(ctx, hipMalloc(&h->rm_dev, nelem * d * sizeof(float)));
+	HIP_CHECK(ctx, hipMemcpy(h->rm_dev, h->vecs.data(),
+	                         nelem * d * sizeof(float),
+	                         hipMemcpyHostToDevice));
+	if (h->metric == SDBV_METRIC_COSINE) {
+		HIP_CHECK(ctx, hipMalloc(&h->norms_dev, nelem * sizeof(double)));
+		HIP_CHECK(ctx, hipMemcpy(h->norms_dev, h->norms.data(),
+		                         nelem * sizeof(double),
+		                         hipMemcpyHostToDevice));
+	}
+	h->adj_stride = ((h->m0 + 8) + 7) & ~7u; // headroom: transient >m0
+	HIP_CHECK(ctx, hipMalloc(&h->adj_dev,
+	                         nelem * h->adj_stride * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMalloc(&h->deg_dev, nelem * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMemset(h->deg_dev, 0, nelem * sizeof(uint32_t)));
+	h->adj_nodes = nelem;
+	h->dev_rows = nelem;
+	h->l0_dirty.assign(nelem, 0);
+	for (uint64_t i = 0; i < base; i++) // pre-existing graph: first sync
+		h->l0_dirty[i] = 1;             // uploads the whole adjacency
+	// visited bitsets for the search launches
+	const uint64_t vwords = (nelem + 31) / 32;
+	{
+		uint64_t need = (uint64_t)chunk * vwords * sizeof(uint32_t);
+		if (h->vis_cap < need) {
+			if (h->vis_dev)
+				(void)hipFree(h->vis_dev);
+			h->vis_dev = nullptr;
+			h->vis_cap = 0;
+			HIP_CHECK(ctx, hipMalloc(&h->vis_dev, need));
+			h->vis_cap = need;
+		}
+	}
+	// per-launch scratch (capacity = chunk, reused across chunks)
+	float *Qd = nullptr;
+	double *qnd = nullptr, *epdd = nullptr, *outd = nullptr;
+	uint32_t *epsd = nullptr, *outr = nullptr, *outc = nullptr,
+	         *outf = nullptr;
+	uint32_t *upd_ids_dev = nullptr, *upd_deg_dev = nullptr,
+	         *upd_edges_dev = nullptr;
+	uint64_t upd_cap = 0; // node capacity of the scatter scratch
+	const uint32_t efc = h->efc;
+	auto cleanup = [&] {
+		for (void *p : {(void *)Qd, (void *)qnd, (void *)epdd, (void *)epsd,
+		                (void *)outr, (void *)outd, (void *)outc,
+		                (void *)outf, (void *)upd_ids_dev,
+		                (void *)upd_deg_dev, (void *)upd_edges_dev})
+			if (p)
+				(void)hipFree(p);
+		h->l0_dirty.clear();
+	};
+#define BGPU_CHECK(call)                                                     \
+	do {                                                                     \
+		hipError_t err_ = (call);                                            \
+		if (err_ != hipSuccess) {                                            \
+			ctx->err = std::string("hip: ") + hipGetErrorString(err_);       \
+			cleanup();                                                       \
+			return SDBV_ERR_HIP;                                             \
+		}                                                                    \
+	} while (0)
+	BGPU_CHECK(hipMalloc(&Qd, (uint64_t)chunk * d * sizeof(float)));
+	BGPU_CHECK(hipMalloc(&qnd, chunk * sizeof(double)));
+	BGPU_CHECK(hipMalloc(&epdd, chunk * sizeof(double)));
+	BGPU_CHECK(hipMalloc(&epsd, chunk * sizeof(uint32_t)));
+	BGPU_CHECK(hipMalloc(&outr, (uint64_t)chunk * efc * sizeof(uint32_t)));
+	BGPU_CHECK(hipMalloc(&outd, (uint64_t)chunk * efc * sizeof(double)));
+	BGPU_CHECK(hipMalloc(&outc, chunk * sizeof(uint32_t)));
+	BGPU_CHECK(hipMalloc(&outf, chunk * sizeof(uint32_t)));
+	// host staging reused across chunks
+	std::vector<float> Qh;
+	std::vector<double> qnh, epdh, outdh;
+	std::vector<uint32_t> epsh, outrh, outch, outfh;
+	std::vector<uint32_t> upd_ids, upd_deg, upd_edges;
+
+	// scan the dirty flags and scatter changed adjacency rows to the device
+	auto sync_adj = [&]() -> int {
+		upd_ids.clear();
+		upd_deg.clear();
+		uint32_t maxdeg = 0;
+		auto &l0 = h->layers[0];
+		for (uint64_t i = 0; i < nelem; i++)
+			if (h->l0_dirty[i]) {
+				uint32_t dg = (uint32_t)l0.edges[i].size();
+				maxdeg = std::max(maxdeg, dg);
+				upd_ids.push_back((uint32_t)i);
+				upd_deg.push_back(dg);
+			}
+		if (maxdeg > h->adj_stride) {
+			// rare: transient degree above the headroom — grow the stride
+			// and re-upload every non-empty row
+			uint32_t ns = (maxdeg + 8 + 7) & ~7u;
+			(void)hipFree(h->adj_dev);
+			h->adj_dev = nullptr;
+			BGPU_CHECK(hipMalloc(&h->adj_dev,
+			                     nelem * (uint64_t)ns * sizeof(uint32_t)));
+			h->adj_stride = ns;
+			upd_ids.clear();
+			upd_deg.clear();
+			for (uint64_t i = 0; i < nelem; i++)
+				if (!l0.edges[i].empty() || h->l0_dirty[i]) {
+					upd_ids.push_back((uint32_t)i);
+					upd_deg.push_back((uint32_t)l0.edges[i].size());
+				}
+		}
+		std::fill(h->l0_dirty.begin(), h->l0_dirty.end(), 0);
+		uint64_t cnt = upd_ids.size();
+		if (cnt == 0)
+			return SDBV_OK;
+		const uint32_t stride = h->adj_stride;
+		upd_edges.assign(cnt * stride, 0);
+		for (uint64_t i = 0; i < cnt; i++) {
+			const auto &e = l0.edges[upd_ids[i]];
+			std::copy(e.begin(), e.end(), upd_edges.begin() + i * stride);
+		}
+		if (upd_cap < (uint64_t)cnt * stride) { // capacity in edge slots:
+			for (void **p : {(void **)&upd_ids_dev, (void **)&upd_deg_dev,
+			                 (void **)&upd_edges_dev})
+				if (*p) {                        // stride can grow mid-build
+					(void)hipFree(*p);
+					*p = nullptr;
+				}
+			uint64_t cap = ((uint64_t)cnt * stride * 3) / 2;
+			BGPU_CHECK(hipMalloc(&upd_ids_dev, cap * sizeof(uint32_t)));
+			BGPU_CHECK(hipMalloc(&upd_deg_dev, cap * sizeof(uint32_t)));
+			BGPU_CHECK(hipMalloc(&upd_edges_dev, cap * sizeof(uint32_t)));
+			upd_cap = cap;
+		}
+		BGPU_CHECK(hipMemcpyAsync(upd_ids_dev, upd_ids.data(),
+		                          cnt * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(upd_deg_dev, upd_deg.data(),
+		                          cnt * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(upd_edges_dev, upd_edges.data(),
+		                          cnt * stride * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		uint64_t threads = cnt * stride;
+		hipLaunchKernelGGL(k_adj_scatter,
+		                   dim3((uint32_t)((threads + 255) / 256)), dim3(256),
+		                   0, ctx->stream, upd_ids_dev, upd_deg_dev,
+		                   upd_edges_dev, (uint32_t)cnt, stride, h->adj_dev,
+		                   h->deg_dev);
+		return SDBV_OK;
+	};
+
+	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
+		uint64_t c1 = std::min(n, c0 + chunk);
+		std::vector<uint64_t> upper, flat;
+		for (uint64_t i = c0; i < c1; i++) {
+			if (h->enter_point < 0)
+				insert_at(h, (uint32_t)(base + i), levels[i], false);
+			else if (levels[i] > 0)
+				upper.push_back(i);
+			else
+				flat.push_back(i);
+		}
+		if (!upper.empty()) {
+			std::atomic<uint64_t> ucursor{0};
+			auto upper_worker = [&]() {
+				uint64_t j;
+				while ((j = ucursor.fetch_add(1)) < upper.size())
+					insert_at(h, (uint32_t)(base + upper[j]),
+					          levels[upper[j]], true);
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads, (int)upper.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(upper_worker);
+			upper_worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		if (flat.empty())
+			continue;
+		// device graph := graph as of NOW (prev chunks + this chunk's
+		// upper elements) — the snapshot the searches run against
+		int rc = sync_adj();
+		if (rc)
+			return rc; // cleanup already ran inside BGPU_CHECK
+		const uint32_t b = (uint32_t)flat.size();
+		// host half: query pack + norms + upper-layer descent (parallel)
+		Qh.resize((uint64_t)b * d);
+		qnh.resize(b);
+		epdh.resize(b);
+		epsh.resize(b);
+		{
+			std::atomic<uint64_t> cursor{0};
+			auto descend_worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < b) {
+					uint32_t q_id = (uint32_t)(base + flat[j]);
+					std::memcpy(Qh.data() + j * d, vec(h, q_id),
+					            d * sizeof(float));
+					qnh[j] = h->metric == SDBV_METRIC_COSINE
+					             ? h->norms[q_id]
+					             : 0;
+					snapshot_descend_one(h, q_id, qnh[j], &epsh[j],
+					                     &epdh[j]);
+				}
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads, (int)b));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(descend_worker);
+			descend_worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		BGPU_CHECK(hipMemcpyAsync(Qd, Qh.data(),
+		                          (uint64_t)b * d * sizeof(float),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(qnd, qnh.data(), b * sizeof(double),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(epdd, epdh.data(), b * sizeof(double),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(epsd, epsh.data(), b * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemsetAsync(h->vis_dev, 0,
+		                          (uint64_t)b * vwords * sizeof(uint32_t),
+		                          ctx->stream));
+		// one launch: every chunk element's full efc-window search
+		hipLaunchKernelGGL(k_hnsw_search<1>, dim3(b), dim3(64), 0,
+		                   ctx->stream, h->rm_dev, h->norms_dev, d,
+		                   (int)h->metric, h->deg_dev, h->adj_dev,
+		                   h->adj_stride, Qd, qnd, epsd, epdd, h->vis_dev,
+		                   vwords, efc, efc, outr, outd, outc, outf);
+		outrh.resize((uint64_t)b * efc);
+		outdh.resize((uint64_t)b * efc);
+		outch.resize(b);
+		outfh.resize(b);
+		BGPU_CHECK(hipMemcpyAsync(outrh.data(), outr,
+		                          outrh.size() * sizeof(uint32_t),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(outdh.data(), outd,
+		                          outdh.size() * sizeof(double),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(outch.data(), outc, b * sizeof(uint32_t),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(outfh.data(), outf, b * sizeof(uint32_t),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipStreamSynchronize(ctx->stream));
+		BGPU_CHECK(hipGetLastError());
+		// rebuild each element's w window; candidate-queue overflow (rare,
+		// flagged) falls back to the exact host snapshot search
+		std::vector<PQ> ws(b);
+		for (uint32_t j = 0; j < b; j++) {
+			if (outfh[j] & HQ_FLAG_OVERFLOW) {
+				snapshot_search_one(h, (uint32_t)(base + flat[j]), ws[j]);
+				continue;
+			}
+			// kernel emits ascending (total_key, seq); pushing in that
+			// order reproduces the host PQ's observable state exactly
+			for (uint32_t i = 0; i < outch[j]; i++)
+				ws[j].push(outdh[(uint64_t)j * efc + i],
+				           outrh[(uint64_t)j * efc + i]);
+		}
+		// apply phase: unchanged host code under the striped locks
+		for (uint64_t j : flat)
+			h->layers[0].in_layer[base + j] = 1;
+		std::atomic<uint64_t> acursor{0};
+		auto apply_worker = [&]() {
+			uint64_t j;
+			while ((j = acursor.fetch_add(1)) < b) {
+				uint32_t q_id = (uint32_t)(base + flat[j]);
+				const float *q_pt = vec(h, q_id);
+				double q_norm = h->metric == SDBV_METRIC_COSINE
+				                    ? h->norms[q_id]
+				                    : 0;
+				layer_insert_apply(h, h->layers[0], q_id, q_pt, q_norm,
+				                   std::move(ws[j]), true);
+			}
+		};
+		{
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads, (int)b));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(apply_worker);
+			apply_worker();
+			for (auto &t : ts)
+				t.join();
+		}
+	}
+#undef BGPU_CHECK
+	cleanup();
+	h->dirty = true;
+	return SDBV_OK;
+}
+
 
 uint64_t sdbv_hnsw_n(sdbv_hnsw *h) { return h ? h->next_id : 0; }
 uint32_t sdbv_hnsw_layers(sdbv_hnsw *h) {
@@ -2662,6 +3083,37 @@ void sdbv_hnsw_l0_export(sdbv_hnsw *h, uint32_t *offsets, uint32_t *edges) {
 	}
 	offsets[h->next_id] = off;
 }
+/* Full per-layer export (CSR + membership) — lets the bench's cpu_baseline
+ * leg import the exact product graph into the oracle searcher. */
+uint64_t sdbv_hnsw_layer_edge_count(sdbv_hnsw *h, uint32_t l) {
+	if (!h || l >= h->layers.size())
+		return 0;
+	uint64_t c = 0;
+	for (auto &e : h->layers[l].edges)
+		c += e.size();
+	return c;
+}
+void sdbv_hnsw_layer_export(sdbv_hnsw *h, uint32_t l, uint32_t *offsets,
+                            uint32_t *edges, uint8_t *in_layer) {
+	uint32_t off = 0;
+	const auto &L = h->layers[l];
+	for (uint64_t i = 0; i < h->next_id; i++) {
+		offsets[i] = off;
+		in_layer[i] = L.has((uint32_t)i) ? 1 : 0;
+		if (i < L.edges.size())
+			for (uint32_t e : L.edges[i])
+				edges[off++] = e;
+	}
+	offsets[h->next_id] = off;
+}
+int64_t sdbv_hnsw_enter_point(sdbv_hnsw *h) {
+	return h ? h->enter_point : -1;
+}
+/* Zero-copy view of the host row-major vector store (n x d f32, element id
+ * = row): the oracle import reads rows from here instead of a second copy. */
+const float *sdbv_hnsw_vecs_ptr(sdbv_hnsw *h) {
+	return h ? h->vecs.data() : nullptr;
+}
 
 // Frees the per-index device state so finalize can be called again after
 // host-graph mutations (apply_pendings re-finalizes a dirty index).
@@ -2669,7 +3121,8 @@ static void hnsw_free_device_state(sdbv_hnsw *h) {
 	for (void **p : {(void **)&h->rows_dev, (void **)&h->dout_dev,
 	                 (void **)&h->q_dev, (void **)&h->rm_dev,
 	                 (void **)&h->offsets_dev, (void **)&h->edges_dev,
-	                 (void **)&h->norms_dev, (void **)&h->vis_dev}) {
+	                 (void **)&h->norms_dev, (void **)&h->vis_dev,
+	                 (void **)&h->adj_dev, (void **)&h->deg_dev}) {
 		if (*p)
 			(void)hipFree(*p);
 		*p = nullptr;
@@ -2683,6 +3136,9 @@ static void hnsw_free_device_state(sdbv_hnsw *h) {
 		h->dists_pinned = nullptr;
 	}
 	h->vis_cap = 0;
+	h->adj_stride = 0;
+	h->adj_nodes = 0;
+	h->dev_rows = 0;
 	h->finalized = false;
 }
 
@@ -2840,9 +3296,9 @@ int sdbv_hnsw_knn_batch(sdbv_hnsw *h, const float *Q, uint32_t b, uint32_t k,
 	               hipMemcpyHostToDevice, ctx->stream);
 
 	auto t0 = std::chrono::steady_clock::now();
-	hipLaunchKernelGGL(k_hnsw_search, dim3(b), dim3(64), 0, ctx->stream,
+	hipLaunchKernelGGL(k_hnsw_search<0>, dim3(b), dim3(64), 0, ctx->stream,
 	                   h->rm_dev, h->norms_dev, h->d, (int)h->metric,
-	                   h->offsets_dev, h->edges_dev, Qd, qnd, epsd, epdd,
+	                   h->offsets_dev, h->edges_dev, 0u, Qd, qnd, epsd, epdd,
 	                   h->vis_dev, vwords, k, ef, outr, outd, outc, outf);
 	std::vector<uint32_t> h_rows((uint64_t)b * k), h_cnt(b), h_flags(b);
 	std::vector<double> h_d((uint64_t)b * k);

@@ -1,0 +1,119 @@
+"""GPU-accelerated snapshot build: the per-chunk
+level-0 efc-searches run as one persistent-kernel launch; everything else
+is the host snapshot build's code. With nthreads=1 both builds are fully
+deterministic, so the resulting graphs must be BIT-IDENTICAL; with
+threads, quality is pinned by the same recall bars as the host builds."""
+import gzip
+import json
+import os
+
+import numpy as np
+import pytest
+
+import oracle
+import surrealdb_amd as sa
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    c = sa.Context()
+    yield c
+    c.close()
+
+
+@pytest.mark.parametrize("metric", ["cosine", "euclidean"])
+def test_gpu_snapshot_build_bitexact_vs_host(ctx, metric):
+    d, n, chunk = 64, 4000, 256
+    rows = oracle.gen_f32(0xC0FFEE, 0, n, d)
+    hh = sa.hnsw_create_host(d, metric=metric, m=8, m0=16, efc=60,
+                             seed=0x5DB1)
+    hh.insert_batch_snapshot(rows, chunk=chunk, nthreads=1)
+    hg = ctx.hnsw_create(d, metric=metric, m=8, m0=16, efc=60, seed=0x5DB1)
+    hg.insert_batch_snapshot_gpu(rows, chunk=chunk, nthreads=1)
+    # identical graphs: layer count, layer-0 CSR, and searches
+    assert hh.num_layers() == hg.num_layers()
+    a, b = hh.l0_csr(), hg.l0_csr()
+    assert np.array_equal(a[0], b[0]), "l0 offsets differ"
+    assert np.array_equal(a[1], b[1]), "l0 edges differ"
+    for q in oracle.gen_f32(0xBEEF, 0, 10, d):
+        hi, hd = hh.knn_search_host(q, 10, 40)
+        gi, gd = hg.knn_search_host(q, 10, 40)
+        assert np.array_equal(hi, gi)
+        assert np.array_equal(hd, gd)
+    hh.destroy()
+    hg.destroy()
+
+
+def test_gpu_snapshot_build_incremental_calls(ctx):
+    """Two consecutive GPU build calls (base > 0 second call: the whole
+    pre-existing adjacency re-syncs) == one host build over both halves
+    in two calls, bit-exact."""
+    d, n = 64, 3000
+    rows = oracle.gen_f32(0xAB, 0, n, d)
+    hh = sa.hnsw_create_host(d, metric="cosine", m=8, m0=16, efc=60,
+                             seed=0x11)
+    hh.insert_batch_snapshot(rows[:1800], chunk=256, nthreads=1)
+    hh.insert_batch_snapshot(rows[1800:], chunk=256, nthreads=1)
+    hg = ctx.hnsw_create(d, metric="cosine", m=8, m0=16, efc=60, seed=0x11)
+    hg.insert_batch_snapshot_gpu(rows[:1800], chunk=256, nthreads=1)
+    hg.insert_batch_snapshot_gpu(rows[1800:], chunk=256, nthreads=1)
+    a, b = hh.l0_csr(), hg.l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    for q in oracle.gen_f32(0xBEEF, 0, 6, d):
+        hi, hd = hh.knn_search_host(q, 10, 40)
+        gi, gd = hg.knn_search_host(q, 10, 40)
+        assert np.array_equal(hi, gi) and np.array_equal(hd, gd)
+    hh.destroy()
+    hg.destroy()
+
+
+def test_gpu_snapshot_build_then_device_search(ctx):
+    """GPU build -> finalize -> device searches (per-hop and persistent
+    kernel) agree bit-exactly with the host search on the same graph."""
+    d, n = 64, 3000
+    rows = oracle.gen_f32(0xD1, 0, n, d)
+    hg = ctx.hnsw_create(d, metric="euclidean", m=8, m0=16, efc=60,
+                         seed=0x7)
+    hg.insert_batch_snapshot_gpu(rows, chunk=256, nthreads=0)
+    hg.finalize(90)
+    queries = oracle.gen_f32(0xBEEF, 0, 8, d)
+    for q in queries:
+        hi, hd = hg.knn_search_host(q, 10, 40)
+        gi, gd = hg.knn_search(q, 10, 40)
+        assert np.array_equal(hi, gi) and np.array_equal(hd, gd)
+    bi, bd, bn = hg.knn_search_batch(queries, 10, 40)
+    for j, q in enumerate(queries):
+        hi, hd = hg.knn_search_host(q, 10, 40)
+        assert np.array_equal(hi, bi[j][:bn[j]].astype(np.uint64))
+        assert np.array_equal(hd, bd[j][:bn[j]])
+    hg.destroy()
+    ctx.drop_table(90)
+
+
+def test_gpu_snapshot_build_recall_golden(ctx):
+    """Threaded GPU build on the reference's golden dataset meets the same
+    efs=40 recall bar as the host snapshot build (chunk << n regime)."""
+    golden = os.path.join(os.path.dirname(__file__), "golden",
+                          "hnsw-random-9000-20-euclidean.gz")
+    rows = []
+    with gzip.open(golden, "rt") as f:
+        for i, line in enumerate(f):
+            if i >= 2000:
+                break
+            rows.append(json.loads(line))
+    ingest = np.array(rows, dtype=np.float32)
+    queries = ingest[:100] + np.float32(0.05)
+    h = ctx.hnsw_create(20, metric="euclidean", m=8, m0=16, efc=100,
+                        seed=0x5DB1)
+    h.insert_batch_snapshot_gpu(ingest, chunk=32, nthreads=4)
+    total = 0.0
+    for q in queries:
+        ids, _ = h.knn_search_host(q, 10, 40)
+        bf, _ = oracle.topk_f32("euclidean", ingest, q, 10)
+        total += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
+    offsets, _ = h.l0_csr()
+    assert np.diff(offsets.astype(np.int64)).max() <= 16
+    assert total / len(queries) >= 0.98
+    h.destroy()

@@ -214,3 +214,28 @@ def test_host_search_matches_committed_768d_fixture():
         assert np.array_equal(ids, fx["ids_ef64"][j][:nn]), f"q{j}"
         assert np.array_equal(dists, fx["dists_ef64"][j][:nn]), f"q{j}"
     h.destroy()
+
+
+def test_oracle_import_searches_product_graph_bitexact():
+    """The oracle's graph-import path (the bench cpu_baseline leg): a
+    product-built graph exported layer by layer into orc_hnsw_import must
+    search bit-exactly like the product host search on the same graph."""
+    import numpy as np
+    import surrealdb_amd
+    from surrealdb_amd.shard import total_key
+    d, n = 48, 2500
+    rows = oracle.gen_f32(0xE1, 0, n, d)
+    h = surrealdb_amd.hnsw_create_host(d, metric="cosine", m=8, m0=16,
+                                       efc=60, seed=0x5DB1)
+    h.insert_batch_snapshot(rows, chunk=256, nthreads=1)
+    layers = [h.layer_csr(l) for l in range(h.num_layers())]
+    o = oracle.Hnsw.import_graph(d, "cosine", 8, 16, 60, h.vecs_view(),
+                                 h.enter_point(), layers)
+    assert o.check_props() == 0
+    for q in oracle.gen_f32(0xBEEF, 0, 12, d):
+        hi, hd = h.knn_search_host(q, 10, 40)
+        oi, od = o.search(q, 10, 40)
+        order = np.lexsort((oi, total_key(od)))
+        assert np.array_equal(hi, oi[order])
+        assert np.array_equal(hd, od[order])
+    h.destroy()
