@@ -58,12 +58,23 @@ enum {
 };
 
 /* Distance metric — mirrors catalog::Distance
- * (surrealdb/core/src/catalog/schema/index.rs:250-284).
- * GPU kernels implement COSINE and EUCLIDEAN (the vector-search metrics);
- * the remaining metrics stay on the reference's own CPU path. */
+ * (surrealdb/core/src/catalog/schema/index.rs:250-284), all 8 variants on
+ * the GPU scan. COSINE/EUCLIDEAN run the LDS-tiled k_scan (the headline
+ * path); the others run an all-distances kernel + exact top-k selection
+ * (each restating the reference's F32 chain, vector.rs:206-451 —
+ * including jaccard's bit-pattern sets :317-340 and the F32 |I|/|U|
+ * asymmetry, and pearson's similarity-as-distance :413-440). MINKOWSKI
+ * takes its order from sdbv_table_set_order (default 0 — degenerate, like
+ * the reference with order 0). */
 enum {
 	SDBV_METRIC_COSINE = 0,
 	SDBV_METRIC_EUCLIDEAN = 1,
+	SDBV_METRIC_MANHATTAN = 2,
+	SDBV_METRIC_CHEBYSHEV = 3,
+	SDBV_METRIC_HAMMING = 4,
+	SDBV_METRIC_JACCARD = 5,
+	SDBV_METRIC_MINKOWSKI = 6,
+	SDBV_METRIC_PEARSON = 7,
 };
 
 typedef struct sdbv_ctx sdbv_ctx; /* owns HIP stream, device pools, staged tables */
@@ -106,6 +117,10 @@ int sdbv_stage_synthetic(sdbv_ctx *, uint64_t table, uint64_t n, uint32_t d,
 
 uint64_t sdbv_table_rows(sdbv_ctx *, uint64_t table);
 int sdbv_drop_table(sdbv_ctx *, uint64_t table);
+
+/* Minkowski order for a staged table (Distance::Minkowski(Number),
+ * index.rs:258): call after staging; ignored by other metrics. */
+int sdbv_table_set_order(sdbv_ctx *, uint64_t table, double order);
 
 /* Host-side generator of the committed synthetic-data contract (bench/test
  * input prep; bit-identical to the device staging generator). */

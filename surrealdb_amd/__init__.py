@@ -14,7 +14,17 @@ _SO = os.path.join(_DIR, "libsdbv.so")
 
 METRIC_COSINE = 0
 METRIC_EUCLIDEAN = 1
-METRICS = {"cosine": METRIC_COSINE, "euclidean": METRIC_EUCLIDEAN}
+# catalog::Distance (index.rs:250-284) — all 8 on the GPU scan
+METRICS = {
+    "cosine": 0,
+    "euclidean": 1,
+    "manhattan": 2,
+    "chebyshev": 3,
+    "hamming": 4,
+    "jaccard": 5,
+    "minkowski": 6,
+    "pearson": 7,
+}
 
 TRUTHY_CB = None  # ctypes callback types, set when lib() loads
 EXPIRE_CB = None
@@ -79,6 +89,7 @@ def lib():
     L.sdbv_all_distances.argtypes = [vp, u64, f32p, u32, f64p]
     L.sdbv_gather_distance.argtypes = [vp, u64, u32p, u32, f32p, u32, f64p]
     L.sdbv_knn_batch.argtypes = [vp, u64, f32p, u32, u32, u32, u64p, f64p]
+    L.sdbv_table_set_order.argtypes = [vp, u64, ctypes.c_double]
     L.sdbv_hnsw_create.argtypes = [vp, u32, u8, u32, u32, u32, ctypes.c_int,
                                    ctypes.c_int, u64, ctypes.c_double,
                                    ctypes.POINTER(vp)]
@@ -193,7 +204,8 @@ class Context:
                "sdbv_get_stats")
         return {f: getattr(s, f) for f, _ in s._fields_}
 
-    def stage_corpus(self, table, rows, ids=None, metric="cosine"):
+    def stage_corpus(self, table, rows, ids=None, metric="cosine",
+                     order=None):
         import numpy as np
         rows = np.ascontiguousarray(rows, dtype=np.float32)
         n, d = rows.shape
@@ -204,6 +216,10 @@ class Context:
         _check(self._ptr, lib().sdbv_stage_corpus(
             self._ptr, table, rows.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
             idp, n, d, METRICS[metric]), "sdbv_stage_corpus")
+        if order is not None:
+            _check(self._ptr,
+                   lib().sdbv_table_set_order(self._ptr, table, order),
+                   "sdbv_table_set_order")
 
     def stage_synthetic(self, table, n, d, metric="cosine", seed=0x5DB1,
                         row_offset=0, id_base=0):

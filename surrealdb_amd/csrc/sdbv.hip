@@ -82,6 +82,7 @@ struct Table {
 	uint64_t n_pad = 0;
 	uint32_t d = 0;
 	uint8_t metric = 0;
+	double order = 0.0;       // minkowski order (sdbv_table_set_order)
 	float *cm = nullptr;      // [d][n_pad] feature-major
 	double *norms = nullptr;  // per-row f64 norm (cosine only)
 	float *aux = nullptr;     // per-row f32 selection key aux for the batch
@@ -510,12 +511,18 @@ __global__ __launch_bounds__(THREADS) void k_merge(
 	}
 }
 
-// Debug/parity helper: all-distance dump (small n).
+// All-distance kernel: one lane per row, feature-major coalesced reads.
+// Used for parity dumps, k > MAX_K, and (as the product path proper) the
+// six non-headline metrics — each restating the reference's F32 chain
+// (vector.rs:206-451) operation-for-operation. `order` = minkowski order;
+// `q_mean` = the query's ndarray-mean (pearson, host-precomputed with the
+// same unrolled-8 chain).
 template <int METRIC>
 __global__ void k_all_dists(const float *__restrict__ cm,
                             const double *__restrict__ norms, uint64_t n,
                             uint64_t n_pad, uint32_t d,
                             const float *__restrict__ qg, double q_norm,
+                            double order, double q_mean,
                             double *__restrict__ out) {
 	uint64_t r = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
 	if (r >= n)
@@ -538,14 +545,171 @@ __global__ void k_all_dists(const float *__restrict__ cm,
 		for (; k < d; k++)
 			sum = __fadd_rn(sum, __fmul_rn(cm[(uint64_t)k * n_pad + r], qg[k]));
 		out[r] = 1.0 - (double)sum / (q_norm * norms[r]);
-	} else {
+	} else if (METRIC == 1) {
 		float acc = 0.f;
 		for (uint32_t k = 0; k < d; k++) {
 			float diff = cm[(uint64_t)k * n_pad + r] - qg[k];
 			acc = __fadd_rn(acc, __fmul_rn(diff, diff));
 		}
 		out[r] = sqrt((double)acc);
+	} else if (METRIC == 2) {
+		// manhattan via l1_dist: sequential f32 |a-b| sum (vector.rs:379)
+		float acc = 0.f;
+		for (uint32_t k = 0; k < d; k++)
+			acc = __fadd_rn(acc, fabsf(qg[k] - cm[(uint64_t)k * n_pad + r]));
+		out[r] = (double)acc;
+	} else if (METRIC == 3) {
+		// chebyshev via linf_dist: f32 fmax chain (vector.rs:220-229)
+		float m = 0.f;
+		for (uint32_t k = 0; k < d; k++)
+			m = fmaxf(m, fabsf(qg[k] - cm[(uint64_t)k * n_pad + r]));
+		out[r] = (double)m;
+	} else if (METRIC == 4) {
+		// hamming: count of element-wise != (vector.rs:292-303)
+		uint64_t acc = 0;
+		for (uint32_t k = 0; k < d; k++)
+			if (qg[k] != cm[(uint64_t)k * n_pad + r])
+				acc++;
+		out[r] = (double)acc;
+	} else if (METRIC == 6) {
+		// minkowski: f64 |diff|^order sequential sum, final ^(1/order)
+		// (vector.rs:389-399)
+		double acc = 0.0;
+		for (uint32_t k = 0; k < d; k++)
+			acc += pow(fabs((double)qg[k] -
+			                (double)cm[(uint64_t)k * n_pad + r]),
+			           order);
+		out[r] = pow(acc, 1.0 / order);
+	} else if (METRIC == 7) {
+		// pearson similarity used directly as the metric value
+		// (vector.rs:413-440): row mean via the unrolled-8 f32 chain
+		// (ndarray .mean() = .sum()/len), then sequential f64 moments
+		float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+		uint32_t k = 0;
+		for (; k + 8 <= d; k += 8) {
+#pragma unroll
+			for (uint32_t j = 0; j < 8; j++)
+				p[j] = __fadd_rn(p[j], cm[(uint64_t)(k + j) * n_pad + r]);
+		}
+		float sb = 0.f;
+		sb = __fadd_rn(sb, __fadd_rn(__fadd_rn(p[0], p[4]),
+		                             __fadd_rn(p[1], p[5])));
+		sb = __fadd_rn(sb, __fadd_rn(__fadd_rn(p[2], p[6]),
+		                             __fadd_rn(p[3], p[7])));
+		for (; k < d; k++)
+			sb = __fadd_rn(sb, cm[(uint64_t)k * n_pad + r]);
+		double my = (double)(sb / (float)d);
+		double sum_xy = 0, sum_x2 = 0, sum_y2 = 0;
+		for (uint32_t i = 0; i < d; i++) {
+			double dx = (double)qg[i] - q_mean;
+			double dy = (double)cm[(uint64_t)i * n_pad + r] - my;
+			sum_xy += dx * dy;
+			sum_x2 += dx * dx;
+			sum_y2 += dy * dy;
+		}
+		double den = sqrt(sum_x2 * sum_y2);
+		out[r] = den == 0.0 ? 0.0 : sum_xy / den;
 	}
+}
+
+// Jaccard (vector.rs:317-340): bit-pattern sets — |I|/|U| with the
+// reference's F32 asymmetry restated as-is. One block per row; two LDS
+// open-addressing sets (query uniques, row uniques). The counted
+// quantities are order-independent restatements of the reference's
+// progressive-insert loop: inter = per row element [bits in unique(q)] or
+// [non-first duplicate within the row]; |U| = |unique(q)| + |unique(row)
+// \ unique(q)|. d <= 1024 (LDS capacity; the guard is in the caller).
+#define JACC_SLOTS 2048
+#define JACC_EMPTY 0xFFFFFFFFu
+__global__ __launch_bounds__(256) void k_jaccard_dists(
+    const float *__restrict__ cm, uint64_t n, uint64_t n_pad, uint32_t d,
+    const float *__restrict__ qg, double *__restrict__ out) {
+	__shared__ uint32_t setA[JACC_SLOTS];
+	__shared__ uint32_t setB[JACC_SLOTS];
+	__shared__ uint32_t suA, interS, bNotA;
+	__shared__ int sentA, sentB; // the 0xFFFFFFFF bit pattern, if present
+	uint64_t r = blockIdx.x;
+	if (r >= n)
+		return;
+	for (uint32_t i = threadIdx.x; i < JACC_SLOTS; i += 256) {
+		setA[i] = JACC_EMPTY;
+		setB[i] = JACC_EMPTY;
+	}
+	if (threadIdx.x == 0) {
+		suA = 0;
+		interS = 0;
+		bNotA = 0;
+		sentA = 0;
+		sentB = 0;
+	}
+	__syncthreads();
+	for (uint32_t i = threadIdx.x; i < d; i += 256) {
+		uint32_t bits = __float_as_uint(qg[i]);
+		if (bits == JACC_EMPTY) {
+			if (atomicOr(&sentA, 1) == 0)
+				atomicAdd(&suA, 1);
+			continue;
+		}
+		uint32_t h = (bits * 2654435761u) & (JACC_SLOTS - 1);
+		for (;;) {
+			uint32_t prev = atomicCAS(&setA[h], JACC_EMPTY, bits);
+			if (prev == JACC_EMPTY) {
+				atomicAdd(&suA, 1);
+				break;
+			}
+			if (prev == bits)
+				break;
+			h = (h + 1) & (JACC_SLOTS - 1);
+		}
+	}
+	__syncthreads();
+	for (uint32_t i = threadIdx.x; i < d; i += 256) {
+		uint32_t bits = __float_as_uint(cm[(uint64_t)i * n_pad + r]);
+		bool inA;
+		if (bits == JACC_EMPTY) {
+			inA = sentA != 0;
+		} else {
+			inA = false;
+			uint32_t h = (bits * 2654435761u) & (JACC_SLOTS - 1);
+			for (;;) {
+				uint32_t v = setA[h];
+				if (v == bits) {
+					inA = true;
+					break;
+				}
+				if (v == JACC_EMPTY)
+					break;
+				h = (h + 1) & (JACC_SLOTS - 1);
+			}
+		}
+		if (inA) {
+			atomicAdd(&interS, 1);
+			continue;
+		}
+		if (bits == JACC_EMPTY) {
+			if (atomicOr(&sentB, 1) == 0)
+				atomicAdd(&bNotA, 1);
+			else
+				atomicAdd(&interS, 1);
+			continue;
+		}
+		uint32_t h = (bits * 2654435761u) & (JACC_SLOTS - 1);
+		for (;;) {
+			uint32_t prev = atomicCAS(&setB[h], JACC_EMPTY, bits);
+			if (prev == JACC_EMPTY) {
+				atomicAdd(&bNotA, 1);
+				break;
+			}
+			if (prev == bits) {
+				atomicAdd(&interS, 1);
+				break;
+			}
+			h = (h + 1) & (JACC_SLOTS - 1);
+		}
+	}
+	__syncthreads();
+	if (threadIdx.x == 0)
+		out[r] = (double)interS / (double)(suA + bNotA);
 }
 
 // Gather + distance for HNSW frontier expansion: one block per frontier row,
@@ -1165,6 +1329,23 @@ static float h_sumsq_f32(const float *a, uint32_t d) {
 	return acc;
 }
 
+// ndarray .mean() = unrolled-8 .sum() / len in f32, then widened (the
+// pearson query mean, vector.rs:418-421 via ndarray; same chain as the
+// oracle's restatement).
+static double h_mean_f32(const float *a, uint32_t d) {
+	float p[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+	uint32_t i = 0;
+	for (; i + 8 <= d; i += 8)
+		for (uint32_t j = 0; j < 8; j++)
+			p[j] += a[i + j];
+	float s = 0;
+	s += ((p[0] + p[4]) + (p[1] + p[5]));
+	s += ((p[2] + p[6]) + (p[3] + p[7]));
+	for (; i < d; i++)
+		s += a[i];
+	return (double)(s / (float)d);
+}
+
 static void free_table(Table &t) {
 	if (t.cm)
 		(void)hipFree(t.cm);
@@ -1241,8 +1422,10 @@ static int stage_common(sdbv_ctx *ctx, uint64_t table, uint64_t n, uint32_t d,
                         uint8_t metric, Table **out) {
 	if (d == 0 || d > MAX_D || (d % 4) != 0)
 		return SDBV_ERR_BAD_ARG; // float4 path needs d%4==0 in this revision
-	if (metric > SDBV_METRIC_EUCLIDEAN)
+	if (metric > SDBV_METRIC_PEARSON)
 		return SDBV_ERR_UNSUPPORTED;
+	if (metric == SDBV_METRIC_JACCARD && d > 1024)
+		return SDBV_ERR_UNSUPPORTED; // LDS set capacity (JACC_SLOTS)
 	auto it = ctx->tables.find(table);
 	if (it != ctx->tables.end()) {
 		free_table(it->second);
@@ -1273,11 +1456,15 @@ static int finish_stage(sdbv_ctx *ctx, Table *t) {
 		hipLaunchKernelGGL(k_norms, dim3((uint32_t)nb), dim3(THREADS), 0,
 		                   ctx->stream, t->cm, t->norms, t->n, t->n_pad, t->d);
 	}
-	HIP_CHECK(ctx, hipMalloc(&t->aux, t->n_pad * sizeof(float)));
-	t->bytes += t->n_pad * sizeof(float);
-	hipLaunchKernelGGL(k_aux, dim3((uint32_t)nb), dim3(THREADS), 0, ctx->stream,
-	                   t->cm, t->norms, t->aux, t->n, t->n_pad, t->d,
-	                   (int)t->metric);
+	if (t->metric <= SDBV_METRIC_EUCLIDEAN) {
+		// batch-path selection aux (cosine 1/norm, euclidean sumsq); the
+		// other metrics take the all-distances route and need none
+		HIP_CHECK(ctx, hipMalloc(&t->aux, t->n_pad * sizeof(float)));
+		t->bytes += t->n_pad * sizeof(float);
+		hipLaunchKernelGGL(k_aux, dim3((uint32_t)nb), dim3(THREADS), 0,
+		                   ctx->stream, t->cm, t->norms, t->aux, t->n,
+		                   t->n_pad, t->d, (int)t->metric);
+	}
 	HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
 	HIP_CHECK(ctx, hipGetLastError());
 	uint64_t total = 0;
@@ -1353,6 +1540,15 @@ int sdbv_drop_table(sdbv_ctx *ctx, uint64_t table) {
 	return SDBV_OK;
 }
 
+int sdbv_table_set_order(sdbv_ctx *ctx, uint64_t table, double order) {
+	std::lock_guard<std::mutex> lk(ctx->mu);
+	auto it = ctx->tables.find(table);
+	if (it == ctx->tables.end())
+		return SDBV_ERR_NO_TABLE;
+	it->second.order = order;
+	return SDBV_OK;
+}
+
 static int ensure_query_scratch(sdbv_ctx *ctx, uint32_t d, uint64_t nblocks,
                                 uint32_t k) {
 	if (ctx->q_cap < d) {
@@ -1373,11 +1569,42 @@ static int ensure_query_scratch(sdbv_ctx *ctx, uint32_t d, uint64_t nblocks,
 	return SDBV_OK;
 }
 
-// k beyond the scan kernel's LDS top-K window (MAX_K): one all-distances
-// launch + exact host selection. ids are strictly increasing, so
-// (dist, row) order equals the contract's (dist, id) order. An edge-case
-// path — the reference accepts any k — costing one n-f64 D2H per query.
-// Caller holds ctx->mu.
+// Launch the all-distances computation for any metric (q already staged in
+// ctx->q_dev). Jaccard runs its block-per-row LDS-set kernel; the rest run
+// the per-lane k_all_dists chain.
+static void launch_all_dists(sdbv_ctx *ctx, Table &t, const float *q,
+                             double *dout) {
+	uint64_t nb = (t.n + THREADS - 1) / THREADS;
+	double q_norm = t.metric == SDBV_METRIC_COSINE
+	                    ? sqrt((double)h_sumsq_f32(q, t.d))
+	                    : 0;
+	double q_mean = t.metric == SDBV_METRIC_PEARSON ? h_mean_f32(q, t.d) : 0;
+#define LAUNCH_AD(M)                                                       \
+	hipLaunchKernelGGL(k_all_dists<M>, dim3((uint32_t)nb), dim3(THREADS), \
+	                   0, ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,  \
+	                   ctx->q_dev, q_norm, t.order, q_mean, dout)
+	switch (t.metric) {
+	case SDBV_METRIC_COSINE: LAUNCH_AD(0); break;
+	case SDBV_METRIC_EUCLIDEAN: LAUNCH_AD(1); break;
+	case SDBV_METRIC_MANHATTAN: LAUNCH_AD(2); break;
+	case SDBV_METRIC_CHEBYSHEV: LAUNCH_AD(3); break;
+	case SDBV_METRIC_HAMMING: LAUNCH_AD(4); break;
+	case SDBV_METRIC_JACCARD:
+		hipLaunchKernelGGL(k_jaccard_dists, dim3((uint32_t)t.n), dim3(256),
+		                   0, ctx->stream, t.cm, t.n, t.n_pad, t.d,
+		                   ctx->q_dev, dout);
+		break;
+	case SDBV_METRIC_MINKOWSKI: LAUNCH_AD(6); break;
+	case SDBV_METRIC_PEARSON: LAUNCH_AD(7); break;
+	}
+#undef LAUNCH_AD
+}
+
+// All-distances launch + exact host selection: the route for k beyond the
+// scan kernel's LDS top-K window (MAX_K), and the product path proper for
+// the six non-headline metrics (SURVEY §8 a2 closure). ids are strictly
+// increasing, so (dist, row) order equals the contract's (dist, id)
+// order. Costs one n-f64 D2H per query. Caller holds ctx->mu.
 static int knn_large_k(sdbv_ctx *ctx, Table &t, const float *q, uint32_t d,
                        uint32_t k, uint64_t *out_ids, double *out_dists,
                        uint32_t *out_n) {
@@ -1386,19 +1613,10 @@ static int knn_large_k(sdbv_ctx *ctx, Table &t, const float *q, uint32_t d,
 		return rc;
 	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
 	                              hipMemcpyHostToDevice, ctx->stream));
-	double q_norm = sqrt((double)h_sumsq_f32(q, d));
 	double *dout = nullptr;
 	HIP_CHECK(ctx, hipMalloc(&dout, t.n * sizeof(double)));
-	uint64_t nb = (t.n + THREADS - 1) / THREADS;
 	HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
-	if (t.metric == SDBV_METRIC_COSINE)
-		hipLaunchKernelGGL(k_all_dists<0>, dim3((uint32_t)nb), dim3(THREADS),
-		                   0, ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,
-		                   ctx->q_dev, q_norm, dout);
-	else
-		hipLaunchKernelGGL(k_all_dists<1>, dim3((uint32_t)nb), dim3(THREADS),
-		                   0, ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,
-		                   ctx->q_dev, q_norm, dout);
+	launch_all_dists(ctx, t, q, dout);
 	HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
 	std::vector<double> hd(t.n);
 	std::vector<uint64_t> hids(t.n);
@@ -1457,7 +1675,7 @@ int sdbv_knn_bruteforce(sdbv_ctx *ctx, uint64_t table, const float *q,
 		return SDBV_ERR_BAD_ARG;
 	if (k == 0 || k > (1u << 22))
 		return SDBV_ERR_BAD_ARG;
-	if (k > MAX_K)
+	if (k > MAX_K || t.metric > SDBV_METRIC_EUCLIDEAN)
 		return knn_large_k(ctx, t, q, d, k, out_ids, out_dists, out_n);
 
 	// pick a grid: >=2048 blocks to fill 256 CUs, contiguous spans per block
@@ -1534,18 +1752,9 @@ int sdbv_all_distances(sdbv_ctx *ctx, uint64_t table, const float *q,
 		return rc;
 	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
 	                              hipMemcpyHostToDevice, ctx->stream));
-	double q_norm = sqrt((double)h_sumsq_f32(q, d));
 	double *dout = nullptr;
 	HIP_CHECK(ctx, hipMalloc(&dout, t.n * sizeof(double)));
-	uint64_t nb = (t.n + THREADS - 1) / THREADS;
-	if (t.metric == SDBV_METRIC_COSINE)
-		hipLaunchKernelGGL(k_all_dists<0>, dim3((uint32_t)nb), dim3(THREADS), 0,
-		                   ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,
-		                   ctx->q_dev, q_norm, dout);
-	else
-		hipLaunchKernelGGL(k_all_dists<1>, dim3((uint32_t)nb), dim3(THREADS), 0,
-		                   ctx->stream, t.cm, t.norms, t.n, t.n_pad, t.d,
-		                   ctx->q_dev, q_norm, dout);
+	launch_all_dists(ctx, t, q, dout);
 	HIP_CHECK(ctx, hipMemcpyAsync(out, dout, t.n * sizeof(double),
 	                              hipMemcpyDeviceToHost, ctx->stream));
 	HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
@@ -1615,6 +1824,24 @@ int sdbv_knn_batch(sdbv_ctx *ctx, uint64_t table, const float *Q, uint32_t b,
 	Table &t = it->second;
 	if (d != t.d || b == 0 || k == 0 || k > MAX_K - BATCH_SLACK)
 		return SDBV_ERR_BAD_ARG;
+	if (t.metric > SDBV_METRIC_EUCLIDEAN) {
+		// non-GEMM metrics: per-query all-distances route (exact; the MFMA
+		// batch path is only for the genuinely dense cosine/euclidean case)
+		for (uint32_t j = 0; j < b; j++) {
+			uint32_t m = 0;
+			int rc = knn_large_k(ctx, t, Q + (uint64_t)j * d, d, k,
+			                     out_ids + (uint64_t)j * k,
+			                     out_dists + (uint64_t)j * k, &m);
+			if (rc)
+				return rc;
+			for (uint32_t i = m; i < k; i++) {
+				out_ids[(uint64_t)j * k + i] = ~0ULL;
+				out_dists[(uint64_t)j * k + i] =
+				    std::numeric_limits<double>::infinity();
+			}
+		}
+		return SDBV_OK;
+	}
 	const int kk = (int)k + BATCH_SLACK;
 
 	if (!ctx->blas) {
@@ -2848,6 +3075,15 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	std::vector<uint32_t> epsh, outrh, outch, outfh;
 	std::vector<uint32_t> upd_ids, upd_deg, upd_edges;
 
+	// per-phase wall time, printed at the end (build-tuning diagnostics)
+	double t_upper = 0, t_sync = 0, t_desc = 0, t_kern = 0, t_rebuild = 0,
+	       t_apply = 0;
+	auto now = [] { return std::chrono::steady_clock::now(); };
+	auto secs = [](std::chrono::steady_clock::time_point a,
+	               std::chrono::steady_clock::time_point b) {
+		return std::chrono::duration<double>(b - a).count();
+	};
+
 	// scan the dirty flags and scatter changed adjacency rows to the device
 	auto sync_adj = [&]() -> int {
 		upd_ids.clear();
@@ -2921,6 +3157,7 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 
 	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
 		uint64_t c1 = std::min(n, c0 + chunk);
+		auto tp0 = now();
 		std::vector<uint64_t> upper, flat;
 		for (uint64_t i = c0; i < c1; i++) {
 			if (h->enter_point < 0)
@@ -2946,6 +3183,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 			for (auto &t : ts)
 				t.join();
 		}
+		auto tp1 = now();
+		t_upper += secs(tp0, tp1);
 		if (flat.empty())
 			continue;
 		// device graph := graph as of NOW (prev chunks + this chunk's
@@ -2953,6 +3192,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		int rc = sync_adj();
 		if (rc)
 			return rc; // cleanup already ran inside BGPU_CHECK
+		auto tp2 = now();
+		t_sync += secs(tp1, tp2);
 		const uint32_t b = (uint32_t)flat.size();
 		// host half: query pack + norms + upper-layer descent (parallel)
 		Qh.resize((uint64_t)b * d);
@@ -2982,6 +3223,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 			for (auto &t : ts)
 				t.join();
 		}
+		auto tp3 = now();
+		t_desc += secs(tp2, tp3);
 		BGPU_CHECK(hipMemcpyAsync(Qd, Qh.data(),
 		                          (uint64_t)b * d * sizeof(float),
 		                          hipMemcpyHostToDevice, ctx->stream));
@@ -3016,6 +3259,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		                          hipMemcpyDeviceToHost, ctx->stream));
 		BGPU_CHECK(hipStreamSynchronize(ctx->stream));
 		BGPU_CHECK(hipGetLastError());
+		auto tp4 = now();
+		t_kern += secs(tp3, tp4);
 		// rebuild each element's w window; candidate-queue overflow (rare,
 		// flagged) falls back to the exact host snapshot search
 		std::vector<PQ> ws(b);
@@ -3030,6 +3275,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 				ws[j].push(outdh[(uint64_t)j * efc + i],
 				           outrh[(uint64_t)j * efc + i]);
 		}
+		auto tp5 = now();
+		t_rebuild += secs(tp4, tp5);
 		// apply phase: unchanged host code under the striped locks
 		for (uint64_t j : flat)
 			h->layers[0].in_layer[base + j] = 1;
@@ -3055,8 +3302,15 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 			for (auto &t : ts)
 				t.join();
 		}
+		t_apply += secs(tp5, now());
 	}
 #undef BGPU_CHECK
+	fprintf(stderr,
+	        "[sdbv build_gpu] n=%llu chunk=%u phases: upper=%.1fs "
+	        "sync=%.1fs descend=%.1fs kernel=%.1fs rebuild=%.1fs "
+	        "apply=%.1fs\n",
+	        (unsigned long long)n, chunk, t_upper, t_sync, t_desc, t_kern,
+	        t_rebuild, t_apply);
 	cleanup();
 	h->dirty = true;
 	return SDBV_OK;

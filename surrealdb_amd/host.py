@@ -31,9 +31,11 @@ class KnnTopK:
 
     def __init__(self, ctx: Context, table: int, query_vector, k: int,
                  distance: str = "cosine"):
-        if distance not in ("cosine", "euclidean"):
+        from . import METRICS
+        if distance not in METRICS:
             raise SdbvError(
-                f"KnnTopK GPU path supports cosine/euclidean; got {distance}")
+                f"KnnTopK: unknown distance {distance!r} "
+                f"(catalog::Distance variants: {sorted(METRICS)})")
         self.ctx = ctx
         self.table = table
         self.query_vector = np.ascontiguousarray(query_vector, dtype=np.float32)

@@ -157,3 +157,111 @@ def test_knn_large_k_matches_oracle(ctx, metric):
         assert np.array_equal(gids, oids), f"k={k} ids"
         assert np.array_equal(gdists, odists), f"k={k} dist bits"
     ctx.drop_table(17)
+
+
+# --- the six non-headline metrics (SURVEY §8 a2 closure): all-distances
+# route; chains restate vector.rs:206-451 op-for-op ---
+
+EXTRA_METRICS = ["manhattan", "chebyshev", "hamming", "pearson"]
+
+
+def quantized(seed, n, d):
+    """Coarsely quantized corpus: forces element collisions so hamming /
+    jaccard see real overlaps (random f32 would make every element
+    distinct and every distance equal)."""
+    return np.round(oracle.gen_f32(seed, 0, n, d) * 0.25).astype(np.float32)
+
+
+@pytest.mark.parametrize("metric", EXTRA_METRICS)
+def test_extra_metric_all_distances_bitexact(ctx, metric):
+    n, d = 4000, 64
+    corpus = quantized(0x5DB1, n, d)
+    q = quantized(0xBEEF, 1, d)[0]
+    stage_host(ctx, 21, corpus, metric)
+    gpu = ctx.all_distances(21, q)
+    cpu = np.array([oracle.dist_f32(metric, q, corpus[i]) for i in range(n)])
+    neq = np.nonzero(gpu != cpu)[0]
+    assert neq.size == 0, (
+        f"{metric}: {neq.size}/{n} differ; first {neq[:5]}: "
+        f"gpu={gpu[neq[:5]]} cpu={cpu[neq[:5]]}")
+    ctx.drop_table(21)
+
+
+def test_jaccard_all_distances_bitexact(ctx):
+    n, d = 4000, 64
+    corpus = quantized(0xA5, n, d)
+    # inject rows with heavy element overlap with q and in-row duplicates
+    q = quantized(0xBEEF, 1, d)[0]
+    corpus[7, :32] = q[:32]
+    corpus[11, :] = q
+    corpus[13, :] = corpus[13, 0]  # all-duplicate row
+    stage_host(ctx, 22, corpus, "jaccard")
+    gpu = ctx.all_distances(22, q)
+    cpu = np.array([oracle.dist_f32("jaccard", q, corpus[i])
+                    for i in range(n)])
+    neq = np.nonzero(gpu != cpu)[0]
+    assert neq.size == 0, (
+        f"jaccard: {neq.size}/{n} differ; first {neq[:5]}: "
+        f"gpu={gpu[neq[:5]]} cpu={cpu[neq[:5]]}")
+    ctx.drop_table(22)
+
+
+def test_minkowski_all_distances(ctx):
+    """Minkowski goes through f64 pow on both sides; device libm pow may
+    differ from glibc in the last ulp, so the distance bar is 1e-12 rel
+    (the north_star's score bar is 1e-5) with ranks checked separately in
+    test_extra_metric_knn."""
+    n, d = 4000, 64
+    corpus = oracle.gen_f32(0x5DB1, 0, n, d)
+    q = oracle.gen_f32(0xBEEF, 1, 1, d)[0]
+    stage_host(ctx, 23, corpus, "minkowski")
+    import surrealdb_amd
+    surrealdb_amd.lib().sdbv_table_set_order(ctx._ptr, 23, 3.0)
+    gpu = ctx.all_distances(23, q)
+    cpu = np.array([oracle.dist_f32("minkowski", q, corpus[i], order=3.0)
+                    for i in range(n)])
+    assert np.allclose(gpu, cpu, rtol=1e-12, atol=0)
+    ctx.drop_table(23)
+
+
+@pytest.mark.parametrize("metric,order", [("manhattan", 0.0),
+                                          ("chebyshev", 0.0),
+                                          ("hamming", 0.0),
+                                          ("pearson", 0.0),
+                                          ("jaccard", 0.0),
+                                          ("minkowski", 3.0)])
+def test_extra_metric_knn(ctx, metric, order):
+    """Top-k through sdbv_knn_bruteforce for every metric: ids/ranks equal
+    the oracle's; distances bit-exact (minkowski: 1e-12 rel, see above).
+    pearson's similarity-as-distance quirk (most-negative correlation
+    first) and jaccard's F32 |I|/|U| asymmetry ride through as-is."""
+    n, d, k = 30_000, 64, 10
+    corpus = quantized(0x5DB1, n, d) if metric in ("hamming", "jaccard") \
+        else oracle.gen_f32(0x5DB1, 0, n, d)
+    q = (quantized(0xBEEF, 1, d) if metric in ("hamming", "jaccard")
+         else oracle.gen_f32(0xBEEF, 1, 1, d))[0]
+    ctx.stage_corpus(24, corpus, metric=metric,
+                     order=order if order else None)
+    gids, gdists = ctx.knn_bruteforce(24, q, k)
+    oids, odists = oracle.topk_f32(metric, corpus, q, k, order=order)
+    assert np.array_equal(gids, oids), f"{metric}: ids/ranks"
+    if metric == "minkowski":
+        assert np.allclose(gdists, odists, rtol=1e-12, atol=0)
+    else:
+        assert np.array_equal(gdists, odists), f"{metric}: dist bits"
+    ctx.drop_table(24)
+
+
+def test_extra_metric_batch_route(ctx):
+    """sdbv_knn_batch for a non-GEMM metric routes per query through the
+    all-distances path — results equal per-query bruteforce."""
+    n, d, k, b = 8000, 64, 10, 7
+    corpus = oracle.gen_f32(0x5DB1, 0, n, d)
+    ctx.stage_corpus(25, corpus, metric="manhattan")
+    Q = oracle.gen_f32(0xBEEF, 0, b, d)
+    bids, bdists = ctx.knn_batch(25, Q, k)
+    for j in range(b):
+        sids, sdists = ctx.knn_bruteforce(25, Q[j], k)
+        assert np.array_equal(bids[j], sids)
+        assert np.array_equal(bdists[j], sdists)
+    ctx.drop_table(25)
