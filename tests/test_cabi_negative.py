@@ -16,8 +16,14 @@ def test_hnsw_create_rejects_bad_args():
 
 
 def test_unsupported_metric_rejected():
+    # the scan accepts every catalog::Distance variant; the HNSW graph
+    # engine still takes cosine/euclidean only (the reference's own HNSW
+    # defaults; other metrics on the graph are future work) — rejected
+    # loudly, never silently downgraded
+    with pytest.raises(sa.SdbvError):
+        sa.hnsw_create_host(8, metric="manhattan")
     with pytest.raises(KeyError):
-        sa.hnsw_create_host(8, metric="manhattan")  # GPU path: cos/euc only
+        sa.hnsw_create_host(8, metric="nonsense")
 
 
 def test_hnsw_remove_on_finalized_graph_rejected():
