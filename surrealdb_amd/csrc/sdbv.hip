@@ -117,6 +117,13 @@ struct sdbv_ctx {
 	uint64_t qnorms_cap = 0;
 	double *bdists = nullptr; // [b][kk] exact distances
 	uint64_t bdists_cap = 0;
+	// fused MFMA batch path scratch
+	uint32_t *btheta = nullptr; // [b] per-query kth-best u32 key
+	uint64_t btheta_cap = 0;
+	void *bcand = nullptr; // [b][FMM_CAND_CAP] BCand survivors
+	uint64_t bcand_cap = 0;
+	uint32_t *bcand_cnt = nullptr; // [b]
+	uint64_t bcand_cnt_cap = 0;
 	double ms_gemm = 0, ms_select = 0, ms_exact = 0; // last batch timings
 };
 
@@ -944,6 +951,233 @@ __global__ __launch_bounds__(THREADS) void k_batch_topk(
 	}
 }
 
+// ---------------------------------------------------------------------------
+// Fused MFMA scan (the hand-written batch kernel): scores = corpus x Q^T on
+// v_mfma_f32_32x32x2_f32 (exact f32 at the 157 TF vector rate), with the
+// top-K filter fused into the epilogue — survivors (score key <= the
+// query's bootstrapped kth-best threshold) append to a per-query candidate
+// buffer; the b x n score matrix NEVER touches HBM (the round-1 rocBLAS +
+// k_batch_topk path re-read ~41 GB/step of it). Thresholds bootstrap on the
+// first rows via the legacy path, so expected appends/query ~ kk*ln(n/n0).
+// Selection semantics are identical to k_batch_topk (exact running top-kk
+// by (monotone f32 key, row)): the filter can only drop entries strictly
+// worse than the current kk-th, and k_cand_fold recomputes the exact
+// running top-kk from the survivors.
+// Block tile 128x128 x K=32 LDS stages; 4 waves, each 2x2 subtiles of
+// 32x32. A = cm (feature-major = column-major corpus, lda n_pad), B = Q
+// row-major [b][d].
+// ---------------------------------------------------------------------------
+#define FMM_BM 128
+#define FMM_BN 128
+#define FMM_BK 32
+#define FMM_LDS_PAD 4
+#define FMM_CAND_CAP 4096u
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+__global__ __launch_bounds__(256) void k_mfma_scan_topk(
+    const float *__restrict__ cm, uint64_t n_pad, uint32_t d,
+    const float *__restrict__ Q, uint32_t b, uint64_t row0, uint64_t row1,
+    const float *__restrict__ aux, int metric,
+    const uint32_t *__restrict__ theta, // per-query kth-best u32 key
+    BCand *__restrict__ cand,           // [b][FMM_CAND_CAP]
+    uint32_t *__restrict__ cand_cnt) {  // [b]
+	__shared__ float As[FMM_BK][FMM_BM + FMM_LDS_PAD];
+	__shared__ float Bs[FMM_BK][FMM_BN + FMM_LDS_PAD];
+	const uint32_t wave = threadIdx.x >> 6;
+	const uint32_t lane = threadIdx.x & 63;
+	const uint32_t wm = wave & 1;  // wave row (2x2 wave grid)
+	const uint32_t wn = wave >> 1; // wave col
+	const uint64_t brow = row0 + (uint64_t)blockIdx.x * FMM_BM;
+	const uint32_t bcol = blockIdx.y * FMM_BN;
+	f32x16 acc[2][2] = {};
+
+	const uint32_t tid = threadIdx.x;
+	for (uint32_t k0 = 0; k0 < d; k0 += FMM_BK) {
+		// As[k][m]: contiguous in m (coalesced; corpus rows are the fast
+		// axis of the feature-major store). 256 threads x 16 elems.
+		for (uint32_t i = tid; i < FMM_BK * (FMM_BM / 4); i += 256) {
+			uint32_t k = i / (FMM_BM / 4);
+			uint32_t m4 = (i % (FMM_BM / 4)) * 4;
+			const float4 v = *(const float4 *)(cm +
+			                                   (uint64_t)(k0 + k) * n_pad +
+			                                   brow + m4);
+			As[k][m4 + 0] = v.x;
+			As[k][m4 + 1] = v.y;
+			As[k][m4 + 2] = v.z;
+			As[k][m4 + 3] = v.w;
+		}
+		// Bs[k][j]: Q row-major -> float4 along k per (j, k4)
+		for (uint32_t i = tid; i < (FMM_BK / 4) * FMM_BN; i += 256) {
+			uint32_t j = i / (FMM_BK / 4);
+			uint32_t k4 = (i % (FMM_BK / 4)) * 4;
+			const float4 v = *(const float4 *)(Q +
+			                                   (uint64_t)(bcol + j) * d +
+			                                   k0 + k4);
+			Bs[k4 + 0][j] = v.x;
+			Bs[k4 + 1][j] = v.y;
+			Bs[k4 + 2][j] = v.z;
+			Bs[k4 + 3][j] = v.w;
+		}
+		__syncthreads();
+#pragma unroll
+		for (uint32_t kk = 0; kk < FMM_BK; kk += 2) {
+			// operand map (32x32x2): lane l holds A[i=l&31][k=l>>5],
+			// B[k=l>>5][j=l&31]
+			const uint32_t ai = lane & 31, ak = lane >> 5;
+			float a0 = As[kk + ak][wm * 64 + ai];
+			float a1 = As[kk + ak][wm * 64 + 32 + ai];
+			float b0 = Bs[kk + ak][wn * 64 + ai];
+			float b1 = Bs[kk + ak][wn * 64 + 32 + ai];
+			acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0,
+			                                                 acc[0][0], 0, 0, 0);
+			acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0,
+			                                                 acc[1][0], 0, 0, 0);
+			acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1,
+			                                                 acc[0][1], 0, 0, 0);
+			acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1,
+			                                                 acc[1][1], 0, 0, 0);
+		}
+		__syncthreads();
+	}
+
+	// epilogue: C/D map for 32x32 shapes — col = lane&31,
+	// row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+	const uint32_t ccol = lane & 31;
+	const uint32_t crow_base = 4 * (lane >> 5);
+#pragma unroll
+	for (uint32_t sn = 0; sn < 2; sn++) {
+		uint32_t j = bcol + wn * 64 + sn * 32 + ccol;
+		if (j >= b)
+			continue;
+		uint32_t th = theta[j];
+#pragma unroll
+		for (uint32_t sm = 0; sm < 2; sm++) {
+#pragma unroll
+			for (uint32_t rg = 0; rg < 16; rg++) {
+				uint32_t rsub =
+				    (rg & 3) + 8 * (rg >> 2) + crow_base;
+				uint64_t row =
+				    brow + wm * 64 + sm * 32 + rsub;
+				if (row >= row1)
+					continue;
+				float s = acc[sm][sn][rg];
+				float key = (metric == 0) ? -(s * aux[row])
+				                          : (aux[row] - 2.0f * s);
+				uint32_t tk = (uint32_t)d_total_key32(key);
+				if (tk <= th) {
+					uint32_t idx = atomicAdd(&cand_cnt[j], 1u);
+					if (idx < FMM_CAND_CAP) {
+						cand[(uint64_t)j * FMM_CAND_CAP + idx] =
+						    BCand{key, (uint32_t)row};
+					}
+				}
+			}
+		}
+	}
+}
+
+// Per-query thresholds from the bootstrapped running top-kk state.
+__global__ void k_theta_init(const BCand *__restrict__ state, int kk,
+                             uint32_t b, uint32_t *__restrict__ theta) {
+	uint32_t j = blockIdx.x * blockDim.x + threadIdx.x;
+	if (j >= b)
+		return;
+	BCand kth = state[(uint64_t)j * kk + (kk - 1)];
+	theta[j] = kth.row == ~0u ? 0xFFFFFFFFu
+	                          : (uint32_t)d_total_key32(kth.key);
+}
+
+// Fold each query's candidate buffer into its running top-kk state —
+// identical comparator and tile loop as k_batch_topk, reading precomputed
+// (key, row) pairs instead of a score matrix.
+__global__ __launch_bounds__(THREADS) void k_cand_fold(
+    const BCand *__restrict__ cand, const uint32_t *__restrict__ cand_cnt,
+    BCand *__restrict__ state, int kk) {
+	__shared__ LCand buf[TILE + MAX_K];
+	__shared__ LCand topk[MAX_K];
+	__shared__ int cnt;
+	__shared__ int topk_n;
+	__shared__ uint64_t kth_key;
+	__shared__ uint32_t kth_row;
+	uint32_t j = blockIdx.x;
+	uint32_t m_in = cand_cnt[j];
+	if (m_in > FMM_CAND_CAP)
+		m_in = FMM_CAND_CAP; // overflow: host reruns via the legacy path
+	const BCand *cj = cand + (uint64_t)j * FMM_CAND_CAP;
+	BCand *st = state + (uint64_t)j * kk;
+	if (threadIdx.x == 0) {
+		cnt = 0;
+		topk_n = 0;
+		kth_key = ~0ULL;
+		kth_row = ~0u;
+	}
+	__syncthreads();
+	if (threadIdx.x < (uint32_t)kk) {
+		BCand c = st[threadIdx.x];
+		if (c.row != ~0u) {
+			topk[threadIdx.x].key = d_total_key32(c.key);
+			topk[threadIdx.x].dist = (double)c.key;
+			topk[threadIdx.x].row = c.row;
+			atomicAdd(&topk_n, 1);
+		}
+	}
+	__syncthreads();
+	if (threadIdx.x == 0 && topk_n >= kk) {
+		kth_key = topk[kk - 1].key;
+		kth_row = topk[kk - 1].row;
+	}
+	__syncthreads();
+	for (uint32_t tile = 0; tile < m_in; tile += TILE) {
+		uint32_t i = tile + threadIdx.x;
+		uint64_t kk_key = kth_key;
+		uint32_t kk_row = kth_row;
+		int full = (topk_n >= kk);
+		if (i < m_in) {
+			BCand c = cj[i];
+			uint64_t tk = d_total_key32(c.key);
+			bool take = !full || tk < kk_key ||
+			            (tk == kk_key && c.row < kk_row);
+			if (take) {
+				int idx = atomicAdd(&cnt, 1);
+				buf[idx].key = tk;
+				buf[idx].dist = (double)c.key;
+				buf[idx].row = c.row;
+			}
+		}
+		__syncthreads();
+		if (cnt > 0) {
+			int m = cnt;
+			for (int i2 = threadIdx.x; i2 < topk_n; i2 += THREADS)
+				buf[m + i2] = topk[i2];
+			int total = m + topk_n;
+			__syncthreads();
+			int new_n;
+			block_select_topk(buf, total, topk, kk, &new_n);
+			if (threadIdx.x == 0) {
+				topk_n = new_n;
+				if (new_n >= kk) {
+					kth_key = topk[kk - 1].key;
+					kth_row = topk[kk - 1].row;
+				}
+				cnt = 0;
+			}
+		}
+		__syncthreads();
+	}
+	if (threadIdx.x < (uint32_t)kk) {
+		BCand c;
+		if ((int)threadIdx.x < topk_n) {
+			c.key = (float)topk[threadIdx.x].dist;
+			c.row = topk[threadIdx.x].row;
+		} else {
+			c.key = __uint_as_float(0x7F800000u);
+			c.row = ~0u;
+		}
+		st[threadIdx.x] = c;
+	}
+}
+
 // Exact recompute of each surviving candidate with the restated per-row
 // chain (bitwise identical to the single-query scan path).
 __global__ void k_batch_exact(const float *__restrict__ cm,
@@ -1397,7 +1631,9 @@ void sdbv_shutdown(sdbv_ctx *ctx) {
 	if (ctx->q_dev)
 		(void)hipFree(ctx->q_dev);
 	for (void *p : {(void *)ctx->S, ctx->bstate, (void *)ctx->Q_dev,
-	                (void *)ctx->qnorms, (void *)ctx->bdists})
+	                (void *)ctx->qnorms, (void *)ctx->bdists,
+	                (void *)ctx->btheta, (void *)ctx->bcand,
+	                (void *)ctx->bcand_cnt})
 		if (p)
 			(void)hipFree(p);
 	if (ctx->blas)
@@ -1885,33 +2121,105 @@ int sdbv_knn_batch(sdbv_ctx *ctx, uint64_t table, const float *Q, uint32_t b,
 		                   total);
 	}
 
+	// Fused MFMA path (default): bootstrap per-query thresholds on the
+	// first rows via the legacy rocBLAS + k_batch_topk pass, then ONE
+	// hand-written MFMA launch (k_mfma_scan_topk) covers the rest — the
+	// score matrix never touches HBM. SDBV_BATCH_LEGACY=1 forces the
+	// round-1 path (A/B comparisons); shape constraints fall back too.
+	const bool use_fused = (d % FMM_BK) == 0 && (b % FMM_BN) == 0 &&
+	                       !std::getenv("SDBV_BATCH_LEGACY");
+	uint64_t fused_row0 = t.n; // rows from here on go to the fused kernel
+	if (use_fused && t.n > 65536)
+		fused_row0 = 65536; // multiple of FMM_BM and TILE
+
 	double ms_gemm = 0, ms_select = 0;
-	for (uint64_t row0 = 0; row0 < t.n; row0 += chunk) {
-		uint32_t nc = (uint32_t)std::min<uint64_t>(chunk, t.n_pad - row0);
-		const float alpha = 1.0f, beta = 0.0f;
-		HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
-		// S[nc x b] = corpus[nc x d] (col-major view of cm, lda=n_pad)
-		//           x Q^T[d x b]    (col-major view of row-major Q, ldb=d)
-		if (rocblas_sgemm(ctx->blas, rocblas_operation_none,
-		                  rocblas_operation_none, (rocblas_int)nc,
-		                  (rocblas_int)b, (rocblas_int)d, &alpha, t.cm + row0,
-		                  (rocblas_int)t.n_pad, ctx->Q_dev, (rocblas_int)d,
-		                  &beta, ctx->S, (rocblas_int)nc) !=
-		    rocblas_status_success) {
-			ctx->err = "rocblas_sgemm failed";
-			return SDBV_ERR_HIP;
+	auto legacy_rows = [&](uint64_t from, uint64_t to) -> int {
+		// `to` == t.n means "to the end": extend the last chunk over the
+		// padded rows exactly like the round-1 loop (k_batch_topk's n
+		// guard skips them); a bounded `to` (the TILE-aligned bootstrap)
+		// is covered exactly.
+		for (uint64_t row0 = from; row0 < to; row0 += chunk) {
+			uint64_t lim = (to == t.n) ? t.n_pad : to;
+			uint32_t nc = (uint32_t)std::min<uint64_t>(chunk, lim - row0);
+			const float alpha = 1.0f, beta = 0.0f;
+			HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
+			// S[nc x b] = corpus[nc x d] (col-major view of cm, lda n_pad)
+			//           x Q^T[d x b]    (col-major view of row-major Q)
+			if (rocblas_sgemm(ctx->blas, rocblas_operation_none,
+			                  rocblas_operation_none, (rocblas_int)nc,
+			                  (rocblas_int)b, (rocblas_int)d, &alpha,
+			                  t.cm + row0, (rocblas_int)t.n_pad, ctx->Q_dev,
+			                  (rocblas_int)d, &beta, ctx->S,
+			                  (rocblas_int)nc) != rocblas_status_success) {
+				ctx->err = "rocblas_sgemm failed";
+				return SDBV_ERR_HIP;
+			}
+			HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
+			hipLaunchKernelGGL(k_batch_topk, dim3(b), dim3(THREADS), 0,
+			                   ctx->stream, ctx->S, t.aux, row0, nc, t.n,
+			                   (int)t.metric, (BCand *)ctx->bstate, kk);
+			HIP_CHECK(ctx, hipEventRecord(ctx->ev2, ctx->stream));
+			HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+			float m0 = 0, m1 = 0;
+			(void)hipEventElapsedTime(&m0, ctx->ev0, ctx->ev1);
+			(void)hipEventElapsedTime(&m1, ctx->ev1, ctx->ev2);
+			ms_gemm += m0;
+			ms_select += m1;
 		}
+		return SDBV_OK;
+	};
+	if ((rc = legacy_rows(0, fused_row0)))
+		return rc;
+	if (fused_row0 < t.n) {
+		if ((rc = ensure_cap(ctx, (void **)&ctx->btheta, &ctx->btheta_cap,
+		                     (uint64_t)b * sizeof(uint32_t))))
+			return rc;
+		if ((rc = ensure_cap(ctx, (void **)&ctx->bcand, &ctx->bcand_cap,
+		                     (uint64_t)b * FMM_CAND_CAP * sizeof(BCand))))
+			return rc;
+		if ((rc = ensure_cap(ctx, (void **)&ctx->bcand_cnt,
+		                     &ctx->bcand_cnt_cap,
+		                     (uint64_t)b * sizeof(uint32_t))))
+			return rc;
+		hipLaunchKernelGGL(k_theta_init, dim3((b + 255) / 256), dim3(256),
+		                   0, ctx->stream, (const BCand *)ctx->bstate, kk, b,
+		                   ctx->btheta);
+		HIP_CHECK(ctx, hipMemsetAsync(ctx->bcand_cnt, 0,
+		                              b * sizeof(uint32_t), ctx->stream));
+		uint32_t gx = (uint32_t)((t.n - fused_row0 + FMM_BM - 1) / FMM_BM);
+		HIP_CHECK(ctx, hipEventRecord(ctx->ev0, ctx->stream));
+		hipLaunchKernelGGL(k_mfma_scan_topk, dim3(gx, b / FMM_BN),
+		                   dim3(256), 0, ctx->stream, t.cm, t.n_pad, d,
+		                   ctx->Q_dev, b, fused_row0, t.n, t.aux,
+		                   (int)t.metric, ctx->btheta, (BCand *)ctx->bcand,
+		                   ctx->bcand_cnt);
 		HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
-		hipLaunchKernelGGL(k_batch_topk, dim3(b), dim3(THREADS), 0,
-		                   ctx->stream, ctx->S, t.aux, row0, nc, t.n,
-		                   (int)t.metric, (BCand *)ctx->bstate, kk);
+		hipLaunchKernelGGL(k_cand_fold, dim3(b), dim3(THREADS), 0,
+		                   ctx->stream, (const BCand *)ctx->bcand,
+		                   ctx->bcand_cnt, (BCand *)ctx->bstate, kk);
 		HIP_CHECK(ctx, hipEventRecord(ctx->ev2, ctx->stream));
+		std::vector<uint32_t> h_cnt(b);
+		HIP_CHECK(ctx, hipMemcpyAsync(h_cnt.data(), ctx->bcand_cnt,
+		                              b * sizeof(uint32_t),
+		                              hipMemcpyDeviceToHost, ctx->stream));
 		HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+		HIP_CHECK(ctx, hipGetLastError());
 		float m0 = 0, m1 = 0;
 		(void)hipEventElapsedTime(&m0, ctx->ev0, ctx->ev1);
 		(void)hipEventElapsedTime(&m1, ctx->ev1, ctx->ev2);
 		ms_gemm += m0;
 		ms_select += m1;
+		bool overflow = false;
+		for (uint32_t j = 0; j < b; j++)
+			if (h_cnt[j] > FMM_CAND_CAP)
+				overflow = true;
+		if (overflow) {
+			// a candidate buffer filled (adversarially ordered corpus):
+			// re-present the fused rows through the legacy pass — state
+			// merging is exact, so dropped appends are recovered
+			if ((rc = legacy_rows(fused_row0, t.n)))
+				return rc;
+		}
 	}
 
 	// exact recompute of survivors with the restated chain

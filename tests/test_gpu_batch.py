@@ -69,3 +69,47 @@ def test_batch_duplicates(ctx):
         assert ids[j][0] == orig and ids[j][1] == 1000 + orig
         assert dists[j][0] == 0.0 and dists[j][1] == 0.0
     ctx.drop_table(14)
+
+
+@pytest.mark.parametrize("metric", ["cosine", "euclidean"])
+def test_fused_mfma_matches_oracle_and_legacy(ctx, metric):
+    """The hand-written fused MFMA kernel (b % 128 == 0, n > 65536 engages
+    it) must produce exactly the oracle's results AND exactly the legacy
+    rocBLAS + k_batch_topk path's results."""
+    import os
+    n, d, b, k = 200_000, 768, 128, 10
+    corpus = oracle.gen_f32(0x5DB1, 0, n, d)
+    Q = oracle.gen_f32(0xBEEF, 0, b, d)
+    ctx.stage_corpus(13, corpus, metric=metric)
+    ids_f, dists_f = ctx.knn_batch(13, Q, k)       # fused (default)
+    os.environ["SDBV_BATCH_LEGACY"] = "1"
+    try:
+        ids_l, dists_l = ctx.knn_batch(13, Q, k)   # legacy
+    finally:
+        del os.environ["SDBV_BATCH_LEGACY"]
+    assert np.array_equal(ids_f, ids_l), f"{metric}: fused != legacy ids"
+    assert np.array_equal(dists_f, dists_l), f"{metric}: fused != legacy"
+    for j in range(0, b, 17):
+        oids, odists = oracle.topk_f32(metric, corpus, Q[j], k)
+        assert np.array_equal(ids_f[j], oids), f"{metric} q{j}: ids"
+        assert np.array_equal(dists_f[j], odists), f"{metric} q{j}: bits"
+    ctx.drop_table(13)
+
+
+def test_fused_mfma_duplicate_ties(ctx):
+    """Exact duplicates of the query inside the fused row range tie at
+    distance 0 and must surface in ascending id order."""
+    n, d, b, k = 140_000, 256, 128, 8
+    corpus = oracle.gen_f32(0x99, 0, n, d)
+    Q = oracle.gen_f32(0xAB, 0, b, d)
+    corpus[70_000] = Q[3]  # inside the fused range (> 65536)
+    corpus[100_001] = Q[3]
+    corpus[12_345] = Q[3]  # inside the bootstrap range
+    ctx.stage_corpus(14, corpus, metric="euclidean")
+    ids, dists = ctx.knn_batch(14, Q, k)
+    assert list(ids[3][:3]) == [12_345, 70_000, 100_001]
+    assert dists[3][0] == 0.0 and dists[3][2] == 0.0
+    oids, odists = oracle.topk_f32("euclidean", corpus, Q[3], k)
+    assert np.array_equal(ids[3], oids)
+    assert np.array_equal(dists[3], odists)
+    ctx.drop_table(14)
