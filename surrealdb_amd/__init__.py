@@ -97,6 +97,8 @@ def lib():
     L.sdbv_hnsw_insert_batch.argtypes = [vp, f32p, u64, ctypes.c_int]
     L.sdbv_hnsw_insert_batch_snapshot.argtypes = [vp, f32p, u64, u32,
                                                   ctypes.c_int]
+    L.sdbv_hnsw_insert_batch_snapshot2.argtypes = [vp, f32p, u64, u32,
+                                                   ctypes.c_int]
     L.sdbv_hnsw_insert_batch_snapshot_gpu.argtypes = [vp, f32p, u64, u32,
                                                       ctypes.c_int]
     L.sdbv_hnsw_finalize.argtypes = [vp, u64]
@@ -355,6 +357,19 @@ class Hnsw:
                    pts.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
                    pts.shape[0], chunk, nthreads),
                "sdbv_hnsw_insert_batch_snapshot")
+
+    def insert_batch_snapshot2(self, pts, chunk, nthreads=0):
+        """Snapshot build, batched-apply schedule (host twin of the GPU
+        build: all selects, then appends in element order, then one prune
+        pass; upper elements' layer-0 half joins the batch)."""
+        import numpy as np
+        pts = np.ascontiguousarray(pts, dtype=np.float32)
+        _check(self._ctx._ptr if self._ctx else None,
+               lib().sdbv_hnsw_insert_batch_snapshot2(
+                   self._ptr,
+                   pts.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+                   pts.shape[0], chunk, nthreads),
+               "sdbv_hnsw_insert_batch_snapshot2")
 
     def insert_batch_snapshot_gpu(self, pts, chunk, nthreads=0):
         """GPU-accelerated chunked snapshot build: per-chunk level-0

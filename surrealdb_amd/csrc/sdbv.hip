@@ -1244,6 +1244,10 @@ __global__ void k_batch_exact(const float *__restrict__ cm,
 // padded [n][stride] adjacency — the GPU snapshot build's delta-updatable
 // graph (sdbv_hnsw_insert_batch_snapshot_gpu), where per-chunk edge
 // changes scatter into rows instead of re-laying-out a CSR.
+// `ep_off` selects the seeding mode: NULL = one entry point per query
+// (ep_rows[qid]); non-NULL = the insert path's multi-ep seeding — query
+// qid seeds from ep_rows/ep_dists[ep_off[qid] .. ep_off[qid+1]), already
+// in (dist total_cmp, seq) order (the previous layer's w, layer.rs:342).
 template <int PADDED>
 __global__ __launch_bounds__(64) void k_hnsw_search(
     const float *__restrict__ rm, const double *__restrict__ norms,
@@ -1251,7 +1255,8 @@ __global__ __launch_bounds__(64) void k_hnsw_search(
     const uint32_t *__restrict__ edges, uint32_t stride,
     const float *__restrict__ Q,
     const double *__restrict__ qnorms, const uint32_t *__restrict__ ep_rows,
-    const double *__restrict__ ep_dists, uint32_t *__restrict__ visited,
+    const double *__restrict__ ep_dists, const uint32_t *__restrict__ ep_off,
+    uint32_t *__restrict__ visited,
     uint64_t vwords_per_q, uint32_t k, uint32_t ef,
     uint32_t *__restrict__ out_rows, double *__restrict__ out_dists,
     uint32_t *__restrict__ out_cnt, uint32_t *__restrict__ out_flags) {
@@ -1276,25 +1281,29 @@ __global__ __launch_bounds__(64) void k_hnsw_search(
 	uint64_t fq_key = ~0ULL; // max key in w (or +max when w not full)
 	uint32_t fq_idx = 0;
 
-	// seed with the entry point (search_single, layer.rs:76-90)
+	// seed with the entry point(s) (search_single layer.rs:76-90; insert's
+	// search_multi seeds the whole previous-layer w, layer.rs:342-358)
 	{
-		uint64_t ek = d_total_key(ep_dists[qid]);
-		uint32_t er = ep_rows[qid];
-		if (lane == 0) {
-			c_key[0] = ek;
-			c_seq[0] = 0;
-			c_row[0] = er;
-			w_key[0] = ek;
-			w_seq[0] = 0;
-			w_row[0] = er;
-			uint32_t word = er >> 5;
-			atomicOr(&vis[word], 1u << (er & 31));
+		uint32_t e0 = ep_off ? ep_off[qid] : qid;
+		uint32_t e1 = ep_off ? ep_off[qid + 1] : qid + 1;
+		uint32_t ne = e1 - e0;
+		for (uint32_t i = lane; i < ne; i += 64) {
+			uint64_t ek = d_total_key(ep_dists[e0 + i]);
+			uint32_t er = ep_rows[e0 + i];
+			c_key[i] = ek;
+			c_seq[i] = i;
+			c_row[i] = er;
+			w_key[i] = ek;
+			w_seq[i] = i;
+			w_row[i] = er;
+			atomicOr(&vis[er >> 5], 1u << (er & 31));
 		}
-		c_cnt = 1;
-		w_cnt = 1;
-		seq = 1;
-		fq_key = ek;
-		fq_idx = 0;
+		c_cnt = ne;
+		w_cnt = ne;
+		seq = ne;
+		// eps arrive sorted ascending (key, seq): the max is the last
+		fq_key = d_total_key(ep_dists[e1 - 1]);
+		fq_idx = ne - 1;
 	}
 	__syncthreads();
 
@@ -1543,6 +1552,161 @@ __global__ void k_adj_scatter(const uint32_t *__restrict__ upd_ids,
 		deg[node] = upd_deg[i];
 	if (s < upd_deg[i])
 		adj[(uint64_t)node * stride + s] = upd_edges[t];
+}
+
+// ---------------------------------------------------------------------------
+// Batched-apply kernels for the GPU snapshot build: the host apply phase is
+// RAM-bound on pair distances (select's is_closer + backlink prunes read
+// ~100 MB of random rows per insert), so the pair matrices and the
+// heuristic select itself run on the device; the host keeps only the
+// (cheap, exact) edge bookkeeping. Distances use the restated chain, so a
+// host twin with the same batched schedule is bit-identical.
+// ---------------------------------------------------------------------------
+
+// One block per list. lists = concatenated [focus, c_0, c_1, ...] row ids
+// (loff[i]..loff[i+1]); mats = per-list len^2 f64 distance matrix at
+// moff[i] (row-major, M[a][b] = dist(list[a], list[b])).
+__global__ __launch_bounds__(256) void k_pair_mats(
+    const float *__restrict__ rm, const double *__restrict__ norms,
+    uint32_t d, int metric, const uint32_t *__restrict__ lists,
+    const uint32_t *__restrict__ loff, const uint64_t *__restrict__ moff,
+    double *__restrict__ mats) {
+	uint32_t li = blockIdx.x;
+	uint32_t o0 = loff[li], o1 = loff[li + 1];
+	uint32_t len = o1 - o0;
+	double *M = mats + moff[li];
+	for (uint32_t p = threadIdx.x; p < len * len; p += 256) {
+		uint32_t a = p / len, bb = p % len;
+		if (bb < a)
+			continue; // symmetric: fill upper, mirror below
+		uint32_t ra = lists[o0 + a], rb = lists[o0 + bb];
+		const float *va = rm + (uint64_t)ra * d;
+		const float *vb = rm + (uint64_t)rb * d;
+		double dist;
+		if (metric == 0) {
+			float pp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+			uint32_t k = 0;
+			for (; k + 8 <= d; k += 8)
+#pragma unroll
+				for (uint32_t t = 0; t < 8; t++)
+					pp[t] = __fadd_rn(pp[t],
+					                  __fmul_rn(va[k + t], vb[k + t]));
+			float sum = 0.f;
+			sum = __fadd_rn(sum, __fadd_rn(pp[0], pp[4]));
+			sum = __fadd_rn(sum, __fadd_rn(pp[1], pp[5]));
+			sum = __fadd_rn(sum, __fadd_rn(pp[2], pp[6]));
+			sum = __fadd_rn(sum, __fadd_rn(pp[3], pp[7]));
+			for (; k < d; k++)
+				sum = __fadd_rn(sum, __fmul_rn(va[k], vb[k]));
+			dist = 1.0 - (double)sum / (norms[ra] * norms[rb]);
+		} else {
+			float acc = 0.f;
+			for (uint32_t k = 0; k < d; k++) {
+				float diff = va[k] - vb[k];
+				acc = __fadd_rn(acc, __fmul_rn(diff, diff));
+			}
+			dist = sqrt((double)acc);
+		}
+		M[(uint64_t)a * len + bb] = dist;
+		if (a != bb)
+			M[(uint64_t)bb * len + a] = dist;
+	}
+}
+
+// Heuristic neighbour select on the device (heuristic.rs:35-116 without
+// extend, + keep :92-112; is_closer :193-216): one 64-lane block per list.
+// Candidates = list[1..len), keys = M[0][i], popped in (total_cmp key,
+// insertion seq) order — reproduced by a stable sort on (key, index).
+// out_sel[i][m_max] gets the selected ROW ids, out_cnt[i] the count.
+#define HSEL_CAP 640 // 1 + HQ_EF_CAP + headroom; m_max <= 64
+__global__ __launch_bounds__(64) void k_heur_select(
+    const uint32_t *__restrict__ lists, const uint32_t *__restrict__ loff,
+    const uint64_t *__restrict__ moff, const double *__restrict__ mats,
+    uint32_t m_max, int keep, uint32_t *__restrict__ out_sel,
+    uint32_t *__restrict__ out_cnt) {
+	__shared__ uint64_t skey[HSEL_CAP];
+	__shared__ uint32_t sidx[HSEL_CAP];
+	__shared__ uint32_t res[64];
+	__shared__ uint32_t pruned[HSEL_CAP];
+	uint32_t li = blockIdx.x;
+	uint32_t o0 = loff[li], o1 = loff[li + 1];
+	uint32_t len = o1 - o0;
+	uint32_t n = len - 1; // candidates
+	const double *M = mats + moff[li];
+	const int lane = threadIdx.x;
+	uint32_t *out = out_sel + (uint64_t)li * m_max;
+	for (uint32_t i = lane; i < n; i += 64) {
+		skey[i] = d_total_key(M[1 + i]);
+		sidx[i] = i;
+	}
+	__syncthreads();
+	// odd-even transposition sort on (key, idx) — stable order == the
+	// host PQ's (total_cmp, push seq)
+	for (uint32_t phase = 0; phase < n; phase++) {
+		uint32_t start = phase & 1;
+		for (uint32_t i = start + 2 * lane; i + 1 < n; i += 128) {
+			uint64_t k0 = skey[i], k1 = skey[i + 1];
+			uint32_t i0 = sidx[i], i1 = sidx[i + 1];
+			if (k1 < k0 || (k1 == k0 && i1 < i0)) {
+				skey[i] = k1;
+				skey[i + 1] = k0;
+				sidx[i] = i1;
+				sidx[i + 1] = i0;
+			}
+		}
+		__syncthreads();
+	}
+	if (n <= m_max) {
+		for (uint32_t i = lane; i < n; i += 64)
+			out[i] = lists[o0 + 1 + sidx[i]];
+		if (lane == 0)
+			out_cnt[li] = n;
+		return;
+	}
+	// serial heuristic loop; is_closer's res scan is lane-parallel
+	__shared__ int res_n, pruned_n, done;
+	if (lane == 0) {
+		res_n = 0;
+		pruned_n = 0;
+		done = 0;
+	}
+	__syncthreads();
+	for (uint32_t i = 0; i < n; i++) {
+		if (done)
+			break;
+		uint32_t ci = sidx[i];          // list index - 1
+		double ed = M[1 + ci];          // dist(focus, cand)
+		// is_closer: fails if ed > dist(cand, res_r) for ANY r
+		int bad = 0;
+		if (lane < res_n) {
+			double dr = M[(uint64_t)(1 + ci) * len + (1 + res[lane])];
+			bad = ed > dr;
+		}
+		uint64_t anybad = __ballot(bad);
+		if (lane == 0) {
+			if (!anybad) {
+				res[res_n++] = ci;
+				if ((uint32_t)res_n == m_max)
+					done = 1;
+			} else if (keep) {
+				pruned[pruned_n++] = ci;
+			}
+		}
+		__syncthreads();
+	}
+	if (keep) {
+		__syncthreads();
+		if (lane == 0) {
+			int nmore = (int)m_max - res_n;
+			for (int i = 0; i < nmore && i < pruned_n; i++)
+				res[res_n++] = pruned[i];
+		}
+		__syncthreads();
+	}
+	for (int i = lane; i < res_n; i += 64)
+		out[i] = lists[o0 + 1 + res[i]];
+	if (lane == 0)
+		out_cnt[li] = (uint32_t)res_n;
 }
 
 // ---------------------------------------------------------------------------
@@ -2805,6 +2969,65 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 	}
 }
 
+// insert_at with the LAYER-0 HALF DEFERRED (the batched-apply builds):
+// descend + insert into layers >= 1 exactly as insert_at, but return the
+// eps that would seed the layer-0 search (the layer-1 w — layer.rs:358)
+// instead of running it; the chunk's batched layer-0 search+apply takes
+// over from there.
+static void insert_at_upper(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
+                            bool locked, PQ *eps_out) {
+	const float *q_pt = vec(h, q_id);
+	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+	uint32_t top_up;
+	{
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		top_up = (uint32_t)h->layers.size() - 1;
+		uint64_t nelem = h->vecs.size() / h->d;
+		for (auto &l : h->layers) {
+			if (l.edges.size() <= q_id)
+				l.edges.resize(nelem);
+			if (l.in_layer.size() < nelem)
+				l.in_layer.resize(nelem, 0);
+		}
+		for (uint32_t l = 0;
+		     l < (uint32_t)h->layers.size() && l <= q_level; l++)
+			h->layers[l].in_layer[q_id] = 1;
+		h->dirty = true;
+		if (h->enter_point < 0) {
+			h->enter_point = q_id;
+			PQ eps;
+			eps.push(0.0, q_id); // self-seed; batched search is a no-op
+			*eps_out = std::move(eps);
+			return;
+		}
+	}
+	uint64_t ep_id = (uint64_t)h->enter_point;
+	double ep_dist = dist(h, q_pt, q_norm, (uint32_t)ep_id);
+	if (q_level < top_up) {
+		for (uint32_t l = top_up; l > q_level; l--) {
+			PQ cand;
+			cand.push(ep_dist, (uint32_t)ep_id);
+			std::unordered_set<uint32_t> visited{(uint32_t)ep_id};
+			PQ w = cand;
+			search_layer_host(h, h->layers[l], q_pt, q_norm, cand, visited,
+			                  w, 1, locked);
+			double dd;
+			uint32_t ii;
+			if (w.peek_first(&dd, &ii)) {
+				ep_dist = dd;
+				ep_id = ii;
+			}
+		}
+	}
+	PQ eps;
+	eps.push(ep_dist, (uint32_t)ep_id);
+	uint32_t ins_to = std::min(q_level, top_up);
+	for (uint32_t l = ins_to; l >= 1; l--)
+		eps = layer_insert(h, h->layers[l], q_id, q_pt, q_norm,
+		                   std::move(eps), locked);
+	*eps_out = std::move(eps);
+}
+
 // Chunked SNAPSHOT build (the §8f-rank-3 structure): per chunk, every
 // level-0 element's efc-search runs against the graph AS OF the chunk
 // start (read-only — no locks, embarrassingly parallel, and in round 2 the
@@ -2858,6 +3081,165 @@ static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
 	search_layer_host(h, h->layers[0], q_pt, q_norm, eps, visited, w,
 	                  h->efc, false);
 	w_out = std::move(w);
+}
+
+// Layer-0 efc-search from a given eps window (the insert path's multi-ep
+// seeding, layer.rs:342-358) against the snapshot graph.
+static void snapshot_search_eps(sdbv_hnsw *h, uint32_t q_id, const PQ &eps_in,
+                                hnsw::PQ &w_out) {
+	using namespace hnsw;
+	const float *q_pt = vec(h, q_id);
+	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+	PQ eps = eps_in;
+	PQ w = eps;
+	static thread_local VisitSet visited;
+	visited.begin(h->vecs.size() / h->d);
+	for (auto &e : eps_in.to_vec())
+		visited.insert(e.second);
+	search_layer_host(h, h->layers[0], q_pt, q_norm, eps, visited, w,
+	                  h->efc, false);
+	w_out = std::move(w);
+}
+
+// ---- batched layer-0 apply (snapshot-build v2) ----
+// The same select/prune algorithm as layer_insert_apply, but in three
+// bulk phases per chunk: (A) every element's neighbour select against the
+// post-search graph, read-only and parallel; (B) all edge appends, in
+// element order, sequential (deterministic and cheap); (C) one prune pass
+// over every node that ended over m_max, parallel (prunes are
+// independent: enew ⊆ conn, no cascading). This is ONE valid
+// serialization of the parallel interleaved apply — same algorithm, a
+// fixed schedule — and the form whose distance work (the RAM-bound part)
+// batches onto the device in the GPU build. Quality is pinned by the same
+// recall bars; the GPU twin must match this host twin bit-exactly.
+struct ApplyItem {
+	uint32_t q_id;
+	PQ w;
+	std::vector<uint32_t> neighbors; // phase-A output
+};
+
+static void batched_apply_l0_links(sdbv_hnsw *h,
+                                   std::vector<ApplyItem> &items,
+                                   int nthreads);
+static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
+                                                  std::vector<ApplyItem> &it);
+static void batched_apply_phaseC_host(sdbv_hnsw *h,
+                                      const std::vector<uint32_t> &overfull,
+                                      int nthreads);
+
+static void batched_apply_l0_host(sdbv_hnsw *h, std::vector<ApplyItem> &items,
+                                  int nthreads) {
+	Layer &l0 = h->layers[0];
+	// phase A: selects (read-only graph)
+	{
+		std::atomic<uint64_t> cursor{0};
+		auto worker = [&]() {
+			uint64_t j;
+			while ((j = cursor.fetch_add(1)) < items.size()) {
+				ApplyItem &it = items[j];
+				const float *q_pt = vec(h, it.q_id);
+				double q_norm = h->metric == SDBV_METRIC_COSINE
+				                    ? h->norms[it.q_id]
+				                    : 0;
+				select_neighbors(h, l0, it.q_id, q_pt, q_norm, it.w,
+				                 it.neighbors, false);
+			}
+		};
+		std::vector<std::thread> ts;
+		int nt = std::max(1, std::min<int>(nthreads, (int)items.size()));
+		for (int t = 1; t < nt; t++)
+			ts.emplace_back(worker);
+		worker();
+		for (auto &t : ts)
+			t.join();
+	}
+	batched_apply_l0_links(h, items, nthreads);
+}
+
+// phase B: appends, element order (graph.rs:52-64 entry().or_insert);
+// returns the deduped list of nodes that ended over m_max (prune targets)
+static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
+                                                  std::vector<ApplyItem> &items) {
+	Layer &l0 = h->layers[0];
+	const bool track = !h->l0_dirty.empty();
+	std::vector<uint32_t> overfull;
+	for (auto &it : items) {
+		auto &eq = l0.edges[it.q_id];
+		for (uint32_t e : it.neighbors)
+			if (e != it.q_id &&
+			    std::find(eq.begin(), eq.end(), e) == eq.end())
+				eq.push_back(e);
+		l0.in_layer[it.q_id] = 1;
+		if (track)
+			h->l0_dirty[it.q_id] = 1;
+	}
+	for (auto &it : items) {
+		for (uint32_t e : it.neighbors) {
+			if (e == it.q_id)
+				continue;
+			auto &ee = l0.edges[e];
+			if (e < l0.in_layer.size())
+				l0.in_layer[e] = 1;
+			if (std::find(ee.begin(), ee.end(), it.q_id) == ee.end()) {
+				ee.push_back(it.q_id);
+				if (ee.size() > l0.m_max)
+					overfull.push_back(e); // deduped below
+				if (track)
+					h->l0_dirty[e] = 1;
+			}
+		}
+	}
+	std::sort(overfull.begin(), overfull.end());
+	overfull.erase(std::unique(overfull.begin(), overfull.end()),
+	               overfull.end());
+	return overfull;
+}
+
+// phase C: prunes (layer.rs:363-377), parallel over distinct nodes
+static void batched_apply_phaseC_host(sdbv_hnsw *h,
+                                      const std::vector<uint32_t> &overfull,
+                                      int nthreads) {
+	Layer &l0 = h->layers[0];
+	const bool track = !h->l0_dirty.empty();
+	std::atomic<uint64_t> cursor{0};
+	auto worker = [&]() {
+		uint64_t i;
+		while ((i = cursor.fetch_add(1)) < overfull.size()) {
+			uint32_t e = overfull[i];
+			const auto conn = l0.edges[e]; // copy (read-only source)
+			PQ ec;
+			for (uint32_t nid : conn) {
+				if (nid < h->elem_present.size() &&
+				    !h->elem_present[nid])
+					continue;
+				ec.push(dist_ee(h, e, nid), nid);
+			}
+			std::vector<uint32_t> enew;
+			select_neighbors(h, l0, e, vec(h, e),
+			                 h->metric == SDBV_METRIC_COSINE
+			                     ? h->norms[e]
+			                     : 0,
+			                 std::move(ec), enew, false);
+			l0.edges[e] = enew;
+			if (track)
+				h->l0_dirty[e] = 1;
+		}
+	};
+	std::vector<std::thread> ts;
+	int nt = std::max(1, std::min<int>(nthreads,
+	                                   (int)std::max<size_t>(overfull.size(),
+	                                                         1)));
+	for (int t = 1; t < nt; t++)
+		ts.emplace_back(worker);
+	worker();
+	for (auto &t : ts)
+		t.join();
+}
+
+static void batched_apply_l0_links(sdbv_hnsw *h, std::vector<ApplyItem> &items,
+                                   int nthreads) {
+	auto overfull = batched_apply_phaseB(h, items);
+	batched_apply_phaseC_host(h, overfull, nthreads);
 }
 
 // ---- graph element removal (sequential only: apply_pendings holds the
@@ -3247,17 +3629,173 @@ int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *h, const float *pts,
 	return SDBV_OK;
 }
 
-// GPU-accelerated chunked snapshot build (the configs[2]
-// 10M-row build): identical algorithm and results to
-// sdbv_hnsw_insert_batch_snapshot — upper-level elements insert on the host
-// (same classic locked path), but each chunk's level-0 efc-searches run as
-// ONE persistent-kernel launch (k_hnsw_search<PADDED=1>, one workgroup per
-// chunk element, k = ef = efc so the full w window comes back) against a
-// delta-updated padded device adjacency. The apply half (select + edges +
-// prunes, layer.rs:342-387) is unchanged host code, so graph parity with
-// the host snapshot build is structural: the kernel returns exactly the
-// host search's w set (exact queue semantics, validated bit-exact in
-// tests/test_gpu_insert.py), and everything downstream is the same code.
+// Enter-point reachability fix shared by the batch builds (the classic
+// parallel build's promotion, generalized): with pre-created layers the
+// insert-time promotion never fires, which can leave the upper layers
+// unreachable from a level-0 enter point; promote any top-layer member.
+static void hnsw_promote_ep(sdbv_hnsw *h) {
+	using namespace hnsw;
+	if (h->enter_point < 0 || h->layers.size() < 2)
+		return;
+	std::lock_guard<std::mutex> lk(h->global_mu);
+	uint32_t top = (uint32_t)h->layers.size() - 1;
+	// the top non-empty layer
+	while (top >= 1) {
+		const auto &in = h->layers[top].in_layer;
+		if (h->layers[top].has((uint32_t)h->enter_point))
+			return;
+		for (uint64_t i = 0; i < in.size(); i++)
+			if (in[i]) {
+				h->enter_point = (int64_t)i;
+				return;
+			}
+		top--; // layer empty (pre-created): look lower
+	}
+}
+
+// Host twin of the GPU batched-apply snapshot build ("snapshot2"): same
+// chunked snapshot searches as sdbv_hnsw_insert_batch_snapshot, but the
+// apply half runs the BATCHED schedule (phase A selects, phase B appends
+// in element order, phase C one prune pass — batched_apply_l0_host), and
+// upper-level elements defer their layer-0 half into the same batch
+// (insert_at_upper). This is the bit-exact CPU reference for
+// sdbv_hnsw_insert_batch_snapshot_gpu (identical schedule, identical
+// restated distance chains); quality is pinned by the same recall bars as
+// the v1 builds.
+int sdbv_hnsw_insert_batch_snapshot2(sdbv_hnsw *h, const float *pts,
+                                     uint64_t n, uint32_t chunk,
+                                     int nthreads) {
+	using namespace hnsw;
+	if (!h || h->finalized || chunk == 0 || n == 0)
+		return SDBV_ERR_BAD_ARG;
+	if (nthreads <= 0)
+		nthreads = (int)std::thread::hardware_concurrency();
+	uint64_t base = h->next_id;
+	std::vector<uint32_t> levels(n);
+	for (uint64_t i = 0; i < n; i++)
+		levels[i] = next_level(h); // sequential RNG contract, per ordinal
+	h->next_id += n;
+	h->vecs.reserve(h->vecs.size() + n * h->d);
+	for (uint64_t i = 0; i < n; i++)
+		hnsw_append_vec(h, pts + i * h->d);
+	{
+		std::lock_guard<std::mutex> lk(h->global_mu);
+		uint32_t max_level = 0;
+		for (uint64_t i = 0; i < n; i++)
+			max_level = std::max(max_level, levels[i]);
+		while (h->layers.size() <= max_level)
+			h->layers.push_back(hnsw::Layer{{}, h->m});
+		uint64_t ne = h->vecs.size() / h->d;
+		for (auto &l : h->layers) {
+			if (l.edges.size() < ne)
+				l.edges.resize(ne);
+			if (l.in_layer.size() < ne)
+				l.in_layer.resize(ne, 0);
+		}
+	}
+	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
+		uint64_t c1 = std::min(n, c0 + chunk);
+		std::vector<ApplyItem> items;
+		std::vector<PQ> eps_of; // parallel to items
+		std::vector<uint32_t> kinds; // 1 = upper (eps ready), 0 = flat
+		for (uint64_t i = c0; i < c1; i++) {
+			if (h->enter_point < 0) {
+				insert_at(h, (uint32_t)(base + i), levels[i], false);
+				continue;
+			}
+			items.push_back(ApplyItem{(uint32_t)(base + i), PQ{}, {}});
+			eps_of.emplace_back();
+			kinds.push_back(levels[i] > 0 ? 1 : 0);
+		}
+		if (items.empty())
+			continue;
+		// phase 1: upper inserts (layers >= 1, striped locks) — BEFORE any
+		// descent reads those layers (mutation/read separation, like v1)
+		{
+			std::atomic<uint64_t> cursor{0};
+			auto worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < items.size()) {
+					if (!kinds[j])
+						continue;
+					uint32_t q_id = items[j].q_id;
+					insert_at_upper(h, q_id, levels[q_id - base], true,
+					                &eps_of[j]);
+				}
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1,
+			                  std::min<int>(nthreads, (int)items.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(worker);
+			worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		// phase 2: flat descents (read-only upper layers, parallel)
+		{
+			std::atomic<uint64_t> cursor{0};
+			auto worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < items.size()) {
+					if (kinds[j])
+						continue;
+					uint32_t q_id = items[j].q_id;
+					double qn = h->metric == SDBV_METRIC_COSINE
+					                ? h->norms[q_id]
+					                : 0;
+					uint32_t ep;
+					double epd;
+					snapshot_descend_one(h, q_id, qn, &ep, &epd);
+					eps_of[j].push(epd, ep);
+				}
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1,
+			                  std::min<int>(nthreads, (int)items.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(worker);
+			worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		// snapshot layer-0 searches (read-only, parallel)
+		{
+			std::atomic<uint64_t> cursor{0};
+			auto worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < items.size())
+					snapshot_search_eps(h, items[j].q_id, eps_of[j],
+					                    items[j].w);
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1,
+			                  std::min<int>(nthreads, (int)items.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(worker);
+			worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		batched_apply_l0_host(h, items, nthreads);
+	}
+	hnsw_promote_ep(h);
+	h->dirty = true;
+	return SDBV_OK;
+}
+
+// GPU-accelerated chunked snapshot build, v2 (the
+// configs[2] 10M-row build). Identical algorithm and bit-identical results
+// to the host twin sdbv_hnsw_insert_batch_snapshot2 (same batched
+// schedule, same restated distance chains), with every RAM-bound phase on
+// the device:
+//  - layer-0 efc-searches: ONE k_hnsw_search<PADDED=1> launch per chunk
+//    (multi-ep seeding carries the insert path's layer-1 w windows);
+//  - phase-A neighbour selects: k_pair_mats (candidate pair-distance
+//    matrices) + k_heur_select (the exact heuristic, one wave per element);
+//  - phase-C prunes: the same two kernels over the over-degree nodes.
+// The host keeps the upper-layer inserts (tiny layers), the deterministic
+// phase-B edge appends, and the delta sync of the padded device adjacency.
 int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
                                         uint64_t n, uint32_t chunk,
                                         int nthreads) {
@@ -3265,6 +3803,11 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	if (!h || !h->ctx || chunk == 0 || n == 0 || h->efc == 0 ||
 	    h->efc > HQ_EF_CAP)
 		return SDBV_ERR_BAD_ARG;
+	if (h->extend || h->m0 > 64 || h->m > 64 ||
+	    h->efc + 1 > HSEL_CAP)
+		// extend-candidates (or oversized select windows) stay on the
+		// host twin — same algorithm, host distances
+		return sdbv_hnsw_insert_batch_snapshot2(h, pts, n, chunk, nthreads);
 	if (h->finalized) { // writes invalidate the finalized device state
 		hnsw_free_device_state(h);
 		h->dirty = true;
@@ -3281,9 +3824,6 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	h->vecs.reserve(h->vecs.size() + n * h->d);
 	for (uint64_t i = 0; i < n; i++)
 		hnsw_append_vec(h, pts + i * h->d);
-	// pre-create + pre-size every layer (same discipline as the host
-	// snapshot build: h->layers never grows under workers, which also
-	// keeps the &layer == &layers[0] dirty-tracking check stable)
 	{
 		std::lock_guard<std::mutex> lk(h->global_mu);
 		uint32_t max_level = 0;
@@ -3301,8 +3841,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	}
 	const uint64_t nelem = h->vecs.size() / h->d;
 	const uint32_t d = h->d;
-	// ---- device state: full row-major vector store (rows are immutable
-	// once appended) + padded layer-0 adjacency, rebuilt fresh per call ----
+	const uint32_t efc = h->efc;
+	// ---- device state: vectors + padded adjacency (delta-synced) ----
 	for (void **p : {(void **)&h->rm_dev, (void **)&h->norms_dev,
 	                 (void **)&h->adj_dev, (void **)&h->deg_dev})
 		if (*p) {
@@ -3319,7 +3859,7 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		                         nelem * sizeof(double),
 		                         hipMemcpyHostToDevice));
 	}
-	h->adj_stride = ((h->m0 + 8) + 7) & ~7u; // headroom: transient >m0
+	h->adj_stride = ((h->m0 + 8) + 7) & ~7u;
 	HIP_CHECK(ctx, hipMalloc(&h->adj_dev,
 	                         nelem * h->adj_stride * sizeof(uint32_t)));
 	HIP_CHECK(ctx, hipMalloc(&h->deg_dev, nelem * sizeof(uint32_t)));
@@ -3327,9 +3867,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	h->adj_nodes = nelem;
 	h->dev_rows = nelem;
 	h->l0_dirty.assign(nelem, 0);
-	for (uint64_t i = 0; i < base; i++) // pre-existing graph: first sync
-		h->l0_dirty[i] = 1;             // uploads the whole adjacency
-	// visited bitsets for the search launches
+	for (uint64_t i = 0; i < base; i++)
+		h->l0_dirty[i] = 1; // first sync uploads the pre-existing graph
 	const uint64_t vwords = (nelem + 31) / 32;
 	{
 		uint64_t need = (uint64_t)chunk * vwords * sizeof(uint32_t);
@@ -3342,20 +3881,27 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 			h->vis_cap = need;
 		}
 	}
-	// per-launch scratch (capacity = chunk, reused across chunks)
+	// ---- per-launch scratch (capacity = chunk, reused) ----
 	float *Qd = nullptr;
 	double *qnd = nullptr, *epdd = nullptr, *outd = nullptr;
-	uint32_t *epsd = nullptr, *outr = nullptr, *outc = nullptr,
-	         *outf = nullptr;
+	uint32_t *epsd = nullptr, *epoffd = nullptr, *qrowsd = nullptr;
+	uint32_t *outr = nullptr, *outc = nullptr, *outf = nullptr;
 	uint32_t *upd_ids_dev = nullptr, *upd_deg_dev = nullptr,
 	         *upd_edges_dev = nullptr;
-	uint64_t upd_cap = 0; // node capacity of the scatter scratch
-	const uint32_t efc = h->efc;
+	uint32_t *listsd = nullptr, *loffd = nullptr, *seld = nullptr,
+	         *selcntd = nullptr;
+	uint64_t *moffd = nullptr;
+	double *matsd = nullptr;
+	uint64_t upd_cap = 0, lists_cap = 0, mats_cap = 0, sel_cap = 0,
+	         loff_cap = 0;
 	auto cleanup = [&] {
 		for (void *p : {(void *)Qd, (void *)qnd, (void *)epdd, (void *)epsd,
-		                (void *)outr, (void *)outd, (void *)outc,
-		                (void *)outf, (void *)upd_ids_dev,
-		                (void *)upd_deg_dev, (void *)upd_edges_dev})
+		                (void *)epoffd, (void *)qrowsd, (void *)outr,
+		                (void *)outd, (void *)outc, (void *)outf,
+		                (void *)upd_ids_dev, (void *)upd_deg_dev,
+		                (void *)upd_edges_dev, (void *)listsd,
+		                (void *)loffd, (void *)seld, (void *)selcntd,
+		                (void *)moffd, (void *)matsd})
 			if (p)
 				(void)hipFree(p);
 		h->l0_dirty.clear();
@@ -3371,28 +3917,32 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	} while (0)
 	BGPU_CHECK(hipMalloc(&Qd, (uint64_t)chunk * d * sizeof(float)));
 	BGPU_CHECK(hipMalloc(&qnd, chunk * sizeof(double)));
-	BGPU_CHECK(hipMalloc(&epdd, chunk * sizeof(double)));
-	BGPU_CHECK(hipMalloc(&epsd, chunk * sizeof(uint32_t)));
+	BGPU_CHECK(hipMalloc(&epdd,
+	                     (uint64_t)chunk * (efc + 1) * sizeof(double)));
+	BGPU_CHECK(hipMalloc(&epsd,
+	                     (uint64_t)chunk * (efc + 1) * sizeof(uint32_t)));
+	BGPU_CHECK(hipMalloc(&epoffd, (chunk + 1) * sizeof(uint32_t)));
+	BGPU_CHECK(hipMalloc(&qrowsd, chunk * sizeof(uint32_t)));
 	BGPU_CHECK(hipMalloc(&outr, (uint64_t)chunk * efc * sizeof(uint32_t)));
 	BGPU_CHECK(hipMalloc(&outd, (uint64_t)chunk * efc * sizeof(double)));
 	BGPU_CHECK(hipMalloc(&outc, chunk * sizeof(uint32_t)));
 	BGPU_CHECK(hipMalloc(&outf, chunk * sizeof(uint32_t)));
-	// host staging reused across chunks
+	// host staging
 	std::vector<float> Qh;
-	std::vector<double> qnh, epdh, outdh;
-	std::vector<uint32_t> epsh, outrh, outch, outfh;
+	std::vector<double> qnh, epdh;
+	std::vector<uint32_t> epsh, epoffh, qrowsh, outch, outfh;
 	std::vector<uint32_t> upd_ids, upd_deg, upd_edges;
+	std::vector<uint32_t> listsh, loffh, selh, selcnth;
+	std::vector<uint64_t> moffh;
 
-	// per-phase wall time, printed at the end (build-tuning diagnostics)
-	double t_upper = 0, t_sync = 0, t_desc = 0, t_kern = 0, t_rebuild = 0,
-	       t_apply = 0;
+	double t_upper = 0, t_sync = 0, t_desc = 0, t_kern = 0, t_selA = 0,
+	       t_link = 0, t_selC = 0;
 	auto now = [] { return std::chrono::steady_clock::now(); };
 	auto secs = [](std::chrono::steady_clock::time_point a,
 	               std::chrono::steady_clock::time_point b) {
 		return std::chrono::duration<double>(b - a).count();
 	};
 
-	// scan the dirty flags and scatter changed adjacency rows to the device
 	auto sync_adj = [&]() -> int {
 		upd_ids.clear();
 		upd_deg.clear();
@@ -3406,8 +3956,6 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 				upd_deg.push_back(dg);
 			}
 		if (maxdeg > h->adj_stride) {
-			// rare: transient degree above the headroom — grow the stride
-			// and re-upload every non-empty row
 			uint32_t ns = (maxdeg + 8 + 7) & ~7u;
 			(void)hipFree(h->adj_dev);
 			h->adj_dev = nullptr;
@@ -3429,13 +3977,13 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		const uint32_t stride = h->adj_stride;
 		upd_edges.assign(cnt * stride, 0);
 		for (uint64_t i = 0; i < cnt; i++) {
-			const auto &e = l0.edges[upd_ids[i]];
+			const auto &e = h->layers[0].edges[upd_ids[i]];
 			std::copy(e.begin(), e.end(), upd_edges.begin() + i * stride);
 		}
-		if (upd_cap < (uint64_t)cnt * stride) { // capacity in edge slots:
+		if (upd_cap < (uint64_t)cnt * stride) {
 			for (void **p : {(void **)&upd_ids_dev, (void **)&upd_deg_dev,
 			                 (void **)&upd_edges_dev})
-				if (*p) {                        // stride can grow mid-build
+				if (*p) {
 					(void)hipFree(*p);
 					*p = nullptr;
 				}
@@ -3463,104 +4011,210 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		return SDBV_OK;
 	};
 
+	// run phase-A/phase-C selects on the device for a group of lists
+	// (host-assembled for phase C; device-assembled upstream for phase A
+	// is folded into the same host path for simplicity — list data is
+	// small). Returns selections via selh/selcnth.
+	auto device_select = [&](const std::vector<uint32_t> &lists,
+	                         const std::vector<uint32_t> &loff,
+	                         uint32_t nlists, uint32_t m_max) -> int {
+		// matrices offsets
+		moffh.resize(nlists + 1);
+		uint64_t mo = 0;
+		for (uint32_t i = 0; i < nlists; i++) {
+			moffh[i] = mo;
+			uint64_t len = loff[i + 1] - loff[i];
+			mo += len * len;
+		}
+		moffh[nlists] = mo;
+		if (lists_cap < lists.size()) {
+			if (listsd)
+				(void)hipFree(listsd);
+			listsd = nullptr;
+			BGPU_CHECK(hipMalloc(&listsd,
+			                     lists.size() * 3 / 2 * sizeof(uint32_t)));
+			lists_cap = lists.size() * 3 / 2;
+		}
+		if (loff_cap < nlists + 1) {
+			for (void **p : {(void **)&loffd, (void **)&moffd,
+			                 (void **)&selcntd})
+				if (*p) {
+					(void)hipFree(*p);
+					*p = nullptr;
+				}
+			uint64_t cap = (nlists + 1) * 3 / 2;
+			BGPU_CHECK(hipMalloc(&loffd, cap * sizeof(uint32_t)));
+			BGPU_CHECK(hipMalloc(&moffd, cap * sizeof(uint64_t)));
+			BGPU_CHECK(hipMalloc(&selcntd, cap * sizeof(uint32_t)));
+			loff_cap = cap;
+		}
+		if (mats_cap < mo) {
+			if (matsd)
+				(void)hipFree(matsd);
+			matsd = nullptr;
+			BGPU_CHECK(hipMalloc(&matsd, mo * 5 / 4 * sizeof(double)));
+			mats_cap = mo * 5 / 4;
+		}
+		if (sel_cap < (uint64_t)nlists * m_max) {
+			if (seld)
+				(void)hipFree(seld);
+			seld = nullptr;
+			BGPU_CHECK(hipMalloc(&seld, (uint64_t)nlists * m_max * 3 / 2 *
+			                                sizeof(uint32_t)));
+			sel_cap = (uint64_t)nlists * m_max * 3 / 2;
+		}
+		BGPU_CHECK(hipMemcpyAsync(listsd, lists.data(),
+		                          lists.size() * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(loffd, loff.data(),
+		                          (nlists + 1) * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(moffd, moffh.data(),
+		                          (nlists + 1) * sizeof(uint64_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		hipLaunchKernelGGL(k_pair_mats, dim3(nlists), dim3(256), 0,
+		                   ctx->stream, h->rm_dev, h->norms_dev, d,
+		                   (int)h->metric, listsd, loffd, moffd, matsd);
+		hipLaunchKernelGGL(k_heur_select, dim3(nlists), dim3(64), 0,
+		                   ctx->stream, listsd, loffd, moffd, matsd, m_max,
+		                   h->keep ? 1 : 0, seld, selcntd);
+		selh.resize((uint64_t)nlists * m_max);
+		selcnth.resize(nlists);
+		BGPU_CHECK(hipMemcpyAsync(selh.data(), seld,
+		                          (uint64_t)nlists * m_max *
+		                              sizeof(uint32_t),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(selcnth.data(), selcntd,
+		                          nlists * sizeof(uint32_t),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipStreamSynchronize(ctx->stream));
+		BGPU_CHECK(hipGetLastError());
+		return SDBV_OK;
+	};
+
 	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
 		uint64_t c1 = std::min(n, c0 + chunk);
 		auto tp0 = now();
-		std::vector<uint64_t> upper, flat;
+		std::vector<ApplyItem> items;
+		std::vector<PQ> eps_of;
+		std::vector<uint32_t> kinds;
 		for (uint64_t i = c0; i < c1; i++) {
-			if (h->enter_point < 0)
+			if (h->enter_point < 0) {
 				insert_at(h, (uint32_t)(base + i), levels[i], false);
-			else if (levels[i] > 0)
-				upper.push_back(i);
-			else
-				flat.push_back(i);
+				continue;
+			}
+			items.push_back(ApplyItem{(uint32_t)(base + i), PQ{}, {}});
+			eps_of.emplace_back();
+			kinds.push_back(levels[i] > 0 ? 1 : 0);
 		}
-		if (!upper.empty()) {
-			std::atomic<uint64_t> ucursor{0};
-			auto upper_worker = [&]() {
-				uint64_t j;
-				while ((j = ucursor.fetch_add(1)) < upper.size())
-					insert_at(h, (uint32_t)(base + upper[j]),
-					          levels[upper[j]], true);
-			};
-			std::vector<std::thread> ts;
-			int nt = std::max(1, std::min<int>(nthreads, (int)upper.size()));
-			for (int t = 1; t < nt; t++)
-				ts.emplace_back(upper_worker);
-			upper_worker();
-			for (auto &t : ts)
-				t.join();
-		}
-		auto tp1 = now();
-		t_upper += secs(tp0, tp1);
-		if (flat.empty())
+		if (items.empty())
 			continue;
-		// device graph := graph as of NOW (prev chunks + this chunk's
-		// upper elements) — the snapshot the searches run against
-		int rc = sync_adj();
-		if (rc)
-			return rc; // cleanup already ran inside BGPU_CHECK
-		auto tp2 = now();
-		t_sync += secs(tp1, tp2);
-		const uint32_t b = (uint32_t)flat.size();
-		// host half: query pack + norms + upper-layer descent (parallel)
-		Qh.resize((uint64_t)b * d);
-		qnh.resize(b);
-		epdh.resize(b);
-		epsh.resize(b);
+		const uint32_t b = (uint32_t)items.size();
+		// phase 1: upper inserts (layers >= 1)
 		{
 			std::atomic<uint64_t> cursor{0};
-			auto descend_worker = [&]() {
+			auto worker = [&]() {
 				uint64_t j;
 				while ((j = cursor.fetch_add(1)) < b) {
-					uint32_t q_id = (uint32_t)(base + flat[j]);
-					std::memcpy(Qh.data() + j * d, vec(h, q_id),
-					            d * sizeof(float));
-					qnh[j] = h->metric == SDBV_METRIC_COSINE
-					             ? h->norms[q_id]
-					             : 0;
-					snapshot_descend_one(h, q_id, qnh[j], &epsh[j],
-					                     &epdh[j]);
+					if (!kinds[j])
+						continue;
+					uint32_t q_id = items[j].q_id;
+					insert_at_upper(h, q_id, levels[q_id - base], true,
+					                &eps_of[j]);
 				}
 			};
 			std::vector<std::thread> ts;
 			int nt = std::max(1, std::min<int>(nthreads, (int)b));
 			for (int t = 1; t < nt; t++)
-				ts.emplace_back(descend_worker);
-			descend_worker();
+				ts.emplace_back(worker);
+			worker();
 			for (auto &t : ts)
 				t.join();
 		}
+		auto tp1 = now();
+		t_upper += secs(tp0, tp1);
+		// phase 2: flat descents
+		{
+			std::atomic<uint64_t> cursor{0};
+			auto worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < b) {
+					if (kinds[j])
+						continue;
+					uint32_t q_id = items[j].q_id;
+					double qn = h->metric == SDBV_METRIC_COSINE
+					                ? h->norms[q_id]
+					                : 0;
+					uint32_t ep;
+					double epd;
+					snapshot_descend_one(h, q_id, qn, &ep, &epd);
+					eps_of[j].push(epd, ep);
+				}
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1, std::min<int>(nthreads, (int)b));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(worker);
+			worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		auto tp2 = now();
+		t_desc += secs(tp1, tp2);
+		// device graph := graph as of now (incl. this chunk's upper edges)
+		int rc = sync_adj();
+		if (rc)
+			return rc;
 		auto tp3 = now();
-		t_desc += secs(tp2, tp3);
+		t_sync += secs(tp2, tp3);
+		// layer-0 searches: one multi-ep persistent-kernel launch
+		Qh.resize((uint64_t)b * d);
+		qnh.resize(b);
+		qrowsh.resize(b);
+		epoffh.resize(b + 1);
+		epsh.clear();
+		epdh.clear();
+		for (uint32_t j = 0; j < b; j++) {
+			uint32_t q_id = items[j].q_id;
+			qrowsh[j] = q_id;
+			std::memcpy(Qh.data() + (uint64_t)j * d, vec(h, q_id),
+			            d * sizeof(float));
+			qnh[j] = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
+			epoffh[j] = (uint32_t)epsh.size();
+			for (auto &e : eps_of[j].to_vec()) {
+				epsh.push_back(e.second);
+				epdh.push_back(e.first);
+			}
+		}
+		epoffh[b] = (uint32_t)epsh.size();
 		BGPU_CHECK(hipMemcpyAsync(Qd, Qh.data(),
 		                          (uint64_t)b * d * sizeof(float),
 		                          hipMemcpyHostToDevice, ctx->stream));
 		BGPU_CHECK(hipMemcpyAsync(qnd, qnh.data(), b * sizeof(double),
 		                          hipMemcpyHostToDevice, ctx->stream));
-		BGPU_CHECK(hipMemcpyAsync(epdd, epdh.data(), b * sizeof(double),
+		BGPU_CHECK(hipMemcpyAsync(epsd, epsh.data(),
+		                          epsh.size() * sizeof(uint32_t),
 		                          hipMemcpyHostToDevice, ctx->stream));
-		BGPU_CHECK(hipMemcpyAsync(epsd, epsh.data(), b * sizeof(uint32_t),
+		BGPU_CHECK(hipMemcpyAsync(epdd, epdh.data(),
+		                          epdh.size() * sizeof(double),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(epoffd, epoffh.data(),
+		                          (b + 1) * sizeof(uint32_t),
+		                          hipMemcpyHostToDevice, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(qrowsd, qrowsh.data(),
+		                          b * sizeof(uint32_t),
 		                          hipMemcpyHostToDevice, ctx->stream));
 		BGPU_CHECK(hipMemsetAsync(h->vis_dev, 0,
 		                          (uint64_t)b * vwords * sizeof(uint32_t),
 		                          ctx->stream));
-		// one launch: every chunk element's full efc-window search
 		hipLaunchKernelGGL(k_hnsw_search<1>, dim3(b), dim3(64), 0,
 		                   ctx->stream, h->rm_dev, h->norms_dev, d,
 		                   (int)h->metric, h->deg_dev, h->adj_dev,
-		                   h->adj_stride, Qd, qnd, epsd, epdd, h->vis_dev,
-		                   vwords, efc, efc, outr, outd, outc, outf);
-		outrh.resize((uint64_t)b * efc);
-		outdh.resize((uint64_t)b * efc);
+		                   h->adj_stride, Qd, qnd, epsd, epdd, epoffd,
+		                   h->vis_dev, vwords, efc, efc, outr, outd, outc,
+		                   outf);
 		outch.resize(b);
 		outfh.resize(b);
-		BGPU_CHECK(hipMemcpyAsync(outrh.data(), outr,
-		                          outrh.size() * sizeof(uint32_t),
-		                          hipMemcpyDeviceToHost, ctx->stream));
-		BGPU_CHECK(hipMemcpyAsync(outdh.data(), outd,
-		                          outdh.size() * sizeof(double),
-		                          hipMemcpyDeviceToHost, ctx->stream));
 		BGPU_CHECK(hipMemcpyAsync(outch.data(), outc, b * sizeof(uint32_t),
 		                          hipMemcpyDeviceToHost, ctx->stream));
 		BGPU_CHECK(hipMemcpyAsync(outfh.data(), outf, b * sizeof(uint32_t),
@@ -3569,56 +4223,102 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		BGPU_CHECK(hipGetLastError());
 		auto tp4 = now();
 		t_kern += secs(tp3, tp4);
-		// rebuild each element's w window; candidate-queue overflow (rare,
-		// flagged) falls back to the exact host snapshot search
-		std::vector<PQ> ws(b);
-		for (uint32_t j = 0; j < b; j++) {
-			if (outfh[j] & HQ_FLAG_OVERFLOW) {
-				snapshot_search_one(h, (uint32_t)(base + flat[j]), ws[j]);
-				continue;
+		// phase A on the device: lists = [q, w rows] per element. List
+		// data is assembled host-side from outr (copied in bulk).
+		{
+			std::vector<uint32_t> outrh((uint64_t)b * efc);
+			BGPU_CHECK(hipMemcpyAsync(outrh.data(), outr,
+			                          outrh.size() * sizeof(uint32_t),
+			                          hipMemcpyDeviceToHost, ctx->stream));
+			BGPU_CHECK(hipStreamSynchronize(ctx->stream));
+			listsh.clear();
+			loffh.assign(b + 1, 0);
+			for (uint32_t j = 0; j < b; j++) {
+				loffh[j] = (uint32_t)listsh.size();
+				listsh.push_back(items[j].q_id);
+				if (!(outfh[j] & HQ_FLAG_OVERFLOW))
+					for (uint32_t i = 0; i < outch[j]; i++)
+						listsh.push_back(outrh[(uint64_t)j * efc + i]);
 			}
-			// kernel emits ascending (total_key, seq); pushing in that
-			// order reproduces the host PQ's observable state exactly
-			for (uint32_t i = 0; i < outch[j]; i++)
-				ws[j].push(outdh[(uint64_t)j * efc + i],
-				           outrh[(uint64_t)j * efc + i]);
+			loffh[b] = (uint32_t)listsh.size();
+			rc = device_select(listsh, loffh, b, h->m0);
+			if (rc)
+				return rc;
+			for (uint32_t j = 0; j < b; j++) {
+				if (outfh[j] & HQ_FLAG_OVERFLOW) {
+					// exact host fallback (rare): snapshot search + select
+					PQ w;
+					snapshot_search_eps(h, items[j].q_id, eps_of[j], w);
+					const float *q_pt = vec(h, items[j].q_id);
+					double qn = h->metric == SDBV_METRIC_COSINE
+					                ? h->norms[items[j].q_id]
+					                : 0;
+					select_neighbors(h, h->layers[0], items[j].q_id,
+					                 q_pt, qn, std::move(w),
+					                 items[j].neighbors, false);
+					continue;
+				}
+				items[j].neighbors.assign(
+				    selh.begin() + (uint64_t)j * h->m0,
+				    selh.begin() + (uint64_t)j * h->m0 + selcnth[j]);
+			}
 		}
 		auto tp5 = now();
-		t_rebuild += secs(tp4, tp5);
-		// apply phase: unchanged host code under the striped locks
-		for (uint64_t j : flat)
-			h->layers[0].in_layer[base + j] = 1;
-		std::atomic<uint64_t> acursor{0};
-		auto apply_worker = [&]() {
-			uint64_t j;
-			while ((j = acursor.fetch_add(1)) < b) {
-				uint32_t q_id = (uint32_t)(base + flat[j]);
-				const float *q_pt = vec(h, q_id);
-				double q_norm = h->metric == SDBV_METRIC_COSINE
-				                    ? h->norms[q_id]
-				                    : 0;
-				layer_insert_apply(h, h->layers[0], q_id, q_pt, q_norm,
-				                   std::move(ws[j]), true);
+		t_selA += secs(tp4, tp5);
+		// phase B (host, deterministic) + phase C on the device
+		auto overfull = batched_apply_phaseB(h, items);
+		auto tp6 = now();
+		t_link += secs(tp5, tp6);
+		if (!overfull.empty()) {
+			listsh.clear();
+			loffh.clear();
+			std::vector<uint32_t> host_prunes, dev_nodes;
+			for (uint32_t e : overfull) {
+				const auto &conn = h->layers[0].edges[e];
+				if (1 + conn.size() > HSEL_CAP) {
+					host_prunes.push_back(e); // oversized: host prune
+					continue;
+				}
+				dev_nodes.push_back(e);
+				loffh.push_back((uint32_t)listsh.size());
+				listsh.push_back(e);
+				for (uint32_t nid : conn) {
+					if (nid < h->elem_present.size() &&
+					    !h->elem_present[nid])
+						continue;
+					listsh.push_back(nid);
+				}
 			}
-		};
-		{
-			std::vector<std::thread> ts;
-			int nt = std::max(1, std::min<int>(nthreads, (int)b));
-			for (int t = 1; t < nt; t++)
-				ts.emplace_back(apply_worker);
-			apply_worker();
-			for (auto &t : ts)
-				t.join();
+			loffh.push_back((uint32_t)listsh.size());
+			if (!dev_nodes.empty()) {
+				rc = device_select(listsh, loffh,
+				                   (uint32_t)dev_nodes.size(), h->m0);
+				if (rc)
+					return rc;
+				const bool track = !h->l0_dirty.empty();
+				for (uint32_t li = 0; li < dev_nodes.size(); li++) {
+					uint32_t e = dev_nodes[li];
+					auto &ee = h->layers[0].edges[e];
+					ee.assign(selh.begin() + (uint64_t)li * h->m0,
+					          selh.begin() + (uint64_t)li * h->m0 +
+					              selcnth[li]);
+					if (track)
+						h->l0_dirty[e] = 1;
+				}
+			}
+			if (!host_prunes.empty())
+				batched_apply_phaseC_host(h, host_prunes, nthreads);
 		}
-		t_apply += secs(tp5, now());
+		t_selC += secs(tp6, now());
 	}
 #undef BGPU_CHECK
+	hnsw_promote_ep(h);
 	fprintf(stderr,
-	        "[sdbv build_gpu] n=%llu chunk=%u phases: upper=%.1fs "
-	        "sync=%.1fs descend=%.1fs kernel=%.1fs rebuild=%.1fs "
-	        "apply=%.1fs\n",
-	        (unsigned long long)n, chunk, t_upper, t_sync, t_desc, t_kern,
-	        t_rebuild, t_apply);
+	        "[sdbv build_gpu2] n=%llu chunk=%u phases: upper=%.1fs "
+	        "descend=%.1fs sync=%.1fs kernel=%.1fs selA=%.1fs link=%.1fs "
+	        "selC=%.1fs\n",
+	        (unsigned long long)n, chunk, t_upper, t_desc, t_sync, t_kern,
+	        t_selA, t_link, t_selC);
 	cleanup();
 	h->dirty = true;
 	return SDBV_OK;
@@ -3861,6 +4561,7 @@ int sdbv_hnsw_knn_batch(sdbv_hnsw *h, const float *Q, uint32_t b, uint32_t k,
 	hipLaunchKernelGGL(k_hnsw_search<0>, dim3(b), dim3(64), 0, ctx->stream,
 	                   h->rm_dev, h->norms_dev, h->d, (int)h->metric,
 	                   h->offsets_dev, h->edges_dev, 0u, Qd, qnd, epsd, epdd,
+	                   (const uint32_t *)nullptr,
 	                   h->vis_dev, vwords, k, ef, outr, outd, outc, outf);
 	std::vector<uint32_t> h_rows((uint64_t)b * k), h_cnt(b), h_flags(b);
 	std::vector<double> h_d((uint64_t)b * k);

@@ -1,8 +1,10 @@
 """GPU-accelerated snapshot build: the per-chunk
 level-0 efc-searches run as one persistent-kernel launch; everything else
-is the host snapshot build's code. With nthreads=1 both builds are fully
-deterministic, so the resulting graphs must be BIT-IDENTICAL; with
-threads, quality is pinned by the same recall bars as the host builds."""
+is the host snapshot build's code. The GPU build runs the BATCHED apply
+schedule, whose bit-exact host reference is insert_batch_snapshot2. With
+nthreads=1 both are fully deterministic, so the resulting graphs must be
+BIT-IDENTICAL; with threads, quality is pinned by the same recall bars as
+the host builds."""
 import gzip
 import json
 import os
@@ -29,7 +31,7 @@ def test_gpu_snapshot_build_bitexact_vs_host(ctx, metric):
     rows = oracle.gen_f32(0xC0FFEE, 0, n, d)
     hh = sa.hnsw_create_host(d, metric=metric, m=8, m0=16, efc=60,
                              seed=0x5DB1)
-    hh.insert_batch_snapshot(rows, chunk=chunk, nthreads=1)
+    hh.insert_batch_snapshot2(rows, chunk=chunk, nthreads=1)
     hg = ctx.hnsw_create(d, metric=metric, m=8, m0=16, efc=60, seed=0x5DB1)
     hg.insert_batch_snapshot_gpu(rows, chunk=chunk, nthreads=1)
     # identical graphs: layer count, layer-0 CSR, and searches
@@ -54,8 +56,8 @@ def test_gpu_snapshot_build_incremental_calls(ctx):
     rows = oracle.gen_f32(0xAB, 0, n, d)
     hh = sa.hnsw_create_host(d, metric="cosine", m=8, m0=16, efc=60,
                              seed=0x11)
-    hh.insert_batch_snapshot(rows[:1800], chunk=256, nthreads=1)
-    hh.insert_batch_snapshot(rows[1800:], chunk=256, nthreads=1)
+    hh.insert_batch_snapshot2(rows[:1800], chunk=256, nthreads=1)
+    hh.insert_batch_snapshot2(rows[1800:], chunk=256, nthreads=1)
     hg = ctx.hnsw_create(d, metric="cosine", m=8, m0=16, efc=60, seed=0x11)
     hg.insert_batch_snapshot_gpu(rows[:1800], chunk=256, nthreads=1)
     hg.insert_batch_snapshot_gpu(rows[1800:], chunk=256, nthreads=1)

@@ -239,3 +239,69 @@ def test_oracle_import_searches_product_graph_bitexact():
         assert np.array_equal(hi, oi[order])
         assert np.array_equal(hd, od[order])
     h.destroy()
+
+
+def test_snapshot2_build_recall_bars():
+    """The batched-apply schedule (snapshot2, the GPU build's host twin)
+    meets the same recall bars as the interleaved snapshot build on the
+    reference's golden dataset."""
+    import gzip
+    import json
+    import os
+    import numpy as np
+    import surrealdb_amd
+    golden = os.path.join(os.path.dirname(__file__), "golden",
+                          "hnsw-random-9000-20-euclidean.gz")
+    rows = []
+    with gzip.open(golden, "rt") as f:
+        for i, line in enumerate(f):
+            if i >= 2000:
+                break
+            rows.append(json.loads(line))
+    ingest = np.array(rows, dtype=np.float32)
+    queries = ingest[:100] + np.float32(0.05)
+    for chunk in (16, 32):
+        h = surrealdb_amd.hnsw_create_host(20, metric="euclidean", m=8,
+                                           m0=16, efc=100, seed=0x5DB1)
+        h.insert_batch_snapshot2(ingest, chunk=chunk, nthreads=4)
+        total = 0.0
+        for q in queries:
+            ids, _ = h.knn_search_host(q, 10, 40)
+            bf, _ = oracle.topk_f32("euclidean", ingest, q, 10)
+            total += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
+        offsets, edges = h.l0_csr()
+        deg = np.diff(offsets.astype(np.int64))
+        assert deg.max() <= 16, chunk
+        assert total / len(queries) >= 0.98, (chunk, total / len(queries))
+        h.destroy()
+
+
+def test_snapshot2_keep_flag_and_768d_quality():
+    """keep_pruned_connections through the batched schedule, plus
+    768-dim quality parity with the classic parallel build."""
+    import numpy as np
+    import surrealdb_amd
+    d, n = 768, 2000
+    rows = oracle.gen_f32(0x5DB1, 0, n, d)
+    queries = oracle.gen_f32(0xBEEF, 0, 20, d)
+
+    def build_recall(kind, keep=False):
+        h = surrealdb_amd.hnsw_create_host(d, metric="cosine", m=16, m0=32,
+                                           efc=150, keep=keep, seed=0x5DB1)
+        if kind == "snapshot2":
+            h.insert_batch_snapshot2(rows, chunk=64, nthreads=4)
+        else:
+            h.insert_batch(rows, nthreads=4)
+        tot = 0.0
+        for q in queries:
+            ids, _ = h.knn_search_host(q, 10, 64)
+            bf, _, _ = oracle.topk_f32_mt("cosine", rows, q, 10)
+            tot += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
+        h.destroy()
+        return tot / len(queries)
+
+    r_par = build_recall("parallel")
+    r_s2 = build_recall("snapshot2")
+    r_s2k = build_recall("snapshot2", keep=True)
+    assert r_s2 >= r_par - 0.08, (r_s2, r_par)
+    assert r_s2k >= r_par - 0.12, (r_s2k, r_par)
