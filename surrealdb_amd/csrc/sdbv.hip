@@ -102,6 +102,7 @@ struct sdbv_ctx {
 	void *block_out = nullptr; // [max_blocks][MAX_K] Cand
 	uint64_t block_out_cap = 0;
 	void *final_out = nullptr; // [MAX_K] Cand + ids
+	void *merge_tmp = nullptr; // [32][MAX_K] Cand (tree-merge level 1)
 	float *q_dev = nullptr;
 	uint32_t q_cap = 0;
 	hipEvent_t ev0, ev1, ev2;
@@ -459,19 +460,34 @@ __global__ __launch_bounds__(THREADS) void k_scan(
 // Final merge: one block selects global top-k from nblocks*k candidates.
 // block_out is mutable scratch: selected entries are poisoned to +inf
 // (same-workgroup global visibility ordered by __syncthreads()).
+// Exact k-selection over a candidate slab. Each BLOCK merges its slice
+// [per_block*g, per_block*(g+1)) into out[g*k..]: grid 1 with per_block =
+// total is the classic single-block merge; a two-level tree (G slices,
+// then one block over G*k) cuts the 1954-block scan's merge latency
+// (top-k of per-slice top-ks == the global top-k, exactly).
 __global__ __launch_bounds__(THREADS) void k_merge(
     Cand *__restrict__ block_out, uint64_t total, uint32_t k,
-    Cand *__restrict__ out) {
+    uint64_t per_block, Cand *__restrict__ out) {
 	__shared__ uint64_t red_key[THREADS / 64];
 	__shared__ uint64_t red_id[THREADS / 64];
 	__shared__ long red_idx[THREADS / 64];
 	const int lane = threadIdx.x & 63;
 	const int wave = threadIdx.x >> 6;
+	const uint64_t s0 = (uint64_t)blockIdx.x * per_block;
+	const uint64_t s1 = s0 + per_block < total ? s0 + per_block : total;
+	out += (uint64_t)blockIdx.x * k;
+	if (s0 >= total) {
+		for (uint32_t i = threadIdx.x; i < k; i += THREADS) {
+			out[i].dist = __longlong_as_double(0x7FF0000000000000LL);
+			out[i].id = ~0ULL;
+		}
+		return;
+	}
 	for (int slot = 0; slot < (int)k; slot++) {
 		uint64_t bk = ~0ULL;
 		uint64_t bid = ~0ULL;
 		long bi = -1;
-		for (uint64_t i = threadIdx.x; i < total; i += THREADS) {
+		for (uint64_t i = s0 + threadIdx.x; i < s1; i += THREADS) {
 			uint64_t ck = d_total_key(block_out[i].dist);
 			uint64_t cid = block_out[i].id;
 			if (ck < bk || (ck == bk && cid < bid)) {
@@ -509,10 +525,16 @@ __global__ __launch_bounds__(THREADS) void k_merge(
 					best = red_idx[w];
 				}
 			}
-			out[slot] = block_out[best];
-			block_out[best].dist =
-			    __longlong_as_double(0x7FF0000000000000LL);
-			block_out[best].id = ~0ULL;
+			if (best < 0) { // slice shorter than k: pad
+				out[slot].dist =
+				    __longlong_as_double(0x7FF0000000000000LL);
+				out[slot].id = ~0ULL;
+			} else {
+				out[slot] = block_out[best];
+				block_out[best].dist =
+				    __longlong_as_double(0x7FF0000000000000LL);
+				block_out[best].id = ~0ULL;
+			}
 		}
 		__syncthreads();
 	}
@@ -1792,6 +1814,8 @@ void sdbv_shutdown(sdbv_ctx *ctx) {
 		(void)hipFree(ctx->block_out);
 	if (ctx->final_out)
 		(void)hipFree(ctx->final_out);
+	if (ctx->merge_tmp)
+		(void)hipFree(ctx->merge_tmp);
 	if (ctx->q_dev)
 		(void)hipFree(ctx->q_dev);
 	for (void *p : {(void *)ctx->S, ctx->bstate, (void *)ctx->Q_dev,
@@ -1966,6 +1990,9 @@ static int ensure_query_scratch(sdbv_ctx *ctx, uint32_t d, uint64_t nblocks,
 	}
 	if (!ctx->final_out)
 		HIP_CHECK(ctx, hipMalloc(&ctx->final_out, MAX_K * sizeof(Cand)));
+	if (!ctx->merge_tmp)
+		HIP_CHECK(ctx,
+		          hipMalloc(&ctx->merge_tmp, 32 * MAX_K * sizeof(Cand)));
 	return SDBV_OK;
 }
 
@@ -2107,9 +2134,25 @@ int sdbv_knn_bruteforce(sdbv_ctx *ctx, uint64_t table, const float *q,
 		                   ctx->q_dev, q_norm, k, rows_per_block,
 		                   (Cand *)ctx->block_out, t.ids_dev);
 	HIP_CHECK(ctx, hipEventRecord(ctx->ev1, ctx->stream));
-	hipLaunchKernelGGL(k_merge, dim3(1), dim3(THREADS), 0, ctx->stream,
-	                   (Cand *)ctx->block_out, nblocks * k, k,
-	                   (Cand *)ctx->final_out);
+	{
+		uint64_t total = nblocks * k;
+		const uint64_t G = 32; // level-1 fan-in
+		if (total > 4096 && nblocks > G) {
+			// two-level tree merge: G slices in parallel, then one block
+			// over G*k (scratch lives past final_out's k entries)
+			uint64_t per = ((nblocks + G - 1) / G) * k;
+			hipLaunchKernelGGL(k_merge, dim3((uint32_t)G), dim3(THREADS),
+			                   0, ctx->stream, (Cand *)ctx->block_out,
+			                   total, k, per, (Cand *)ctx->merge_tmp);
+			hipLaunchKernelGGL(k_merge, dim3(1), dim3(THREADS), 0,
+			                   ctx->stream, (Cand *)ctx->merge_tmp, G * k,
+			                   k, G * k, (Cand *)ctx->final_out);
+		} else {
+			hipLaunchKernelGGL(k_merge, dim3(1), dim3(THREADS), 0,
+			                   ctx->stream, (Cand *)ctx->block_out, total,
+			                   k, total, (Cand *)ctx->final_out);
+		}
+	}
 	HIP_CHECK(ctx, hipEventRecord(ctx->ev2, ctx->stream));
 	Cand host_out[MAX_K];
 	HIP_CHECK(ctx, hipMemcpyAsync(host_out, ctx->final_out, k * sizeof(Cand),
