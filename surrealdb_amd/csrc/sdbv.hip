@@ -104,6 +104,7 @@ struct sdbv_ctx {
 	void *final_out = nullptr; // [MAX_K] Cand + ids
 	void *merge_tmp = nullptr; // [32][MAX_K] Cand (tree-merge level 1)
 	float *q_dev = nullptr;
+	float *q_pin = nullptr; // pinned H2D staging for the per-query upload
 	uint32_t q_cap = 0;
 	hipEvent_t ev0, ev1, ev2;
 	// batched path
@@ -1015,33 +1016,49 @@ __global__ __launch_bounds__(256) void k_mfma_scan_topk(
 	f32x16 acc[2][2] = {};
 
 	const uint32_t tid = threadIdx.x;
+	// software pipeline: prefetch K-stage t+1 into registers while MFMA
+	// consumes stage t from LDS (the naive load->sync->compute cycle
+	// measured 73 TF; the guide's untuned same-shape reference is 122)
+	// each thread stages 4 float4 of A (k = a_kq + 8r, fixed m4) and 4 of
+	// B (j = b_jq + 32r, fixed k4): A has FMM_BK*FMM_BM/4 = 1024 float4
+	// per stage over 256 threads, B likewise
+	const uint32_t a_m4 = (tid % (FMM_BM / 4)) * 4; // 0..124
+	const uint32_t b_k4 = (tid % (FMM_BK / 4)) * 4; // 0..28
+	const uint32_t a_kq = tid / (FMM_BM / 4); // 0..7 (k quarter index)
+	const uint32_t b_jq = tid / (FMM_BK / 4); // 0..31 (j quarter index)
+	float4 pa[4], pb[4];
+	auto prefetch = [&](uint32_t k0) {
+#pragma unroll
+		for (uint32_t r = 0; r < 4; r++) {
+			uint32_t k = a_kq + 8 * r;
+			pa[r] = *(const float4 *)(cm + (uint64_t)(k0 + k) * n_pad +
+			                          brow + a_m4);
+			uint32_t j = b_jq + 32 * r;
+			pb[r] = *(const float4 *)(Q + (uint64_t)(bcol + j) * d + k0 +
+			                          b_k4);
+		}
+	};
+	auto stage_lds = [&]() {
+#pragma unroll
+		for (uint32_t r = 0; r < 4; r++) {
+			uint32_t k = a_kq + 8 * r;
+			As[k][a_m4 + 0] = pa[r].x;
+			As[k][a_m4 + 1] = pa[r].y;
+			As[k][a_m4 + 2] = pa[r].z;
+			As[k][a_m4 + 3] = pa[r].w;
+			uint32_t j = b_jq + 32 * r;
+			Bs[b_k4 + 0][j] = pb[r].x;
+			Bs[b_k4 + 1][j] = pb[r].y;
+			Bs[b_k4 + 2][j] = pb[r].z;
+			Bs[b_k4 + 3][j] = pb[r].w;
+		}
+	};
+	prefetch(0);
+	stage_lds();
+	__syncthreads();
 	for (uint32_t k0 = 0; k0 < d; k0 += FMM_BK) {
-		// As[k][m]: contiguous in m (coalesced; corpus rows are the fast
-		// axis of the feature-major store). 256 threads x 16 elems.
-		for (uint32_t i = tid; i < FMM_BK * (FMM_BM / 4); i += 256) {
-			uint32_t k = i / (FMM_BM / 4);
-			uint32_t m4 = (i % (FMM_BM / 4)) * 4;
-			const float4 v = *(const float4 *)(cm +
-			                                   (uint64_t)(k0 + k) * n_pad +
-			                                   brow + m4);
-			As[k][m4 + 0] = v.x;
-			As[k][m4 + 1] = v.y;
-			As[k][m4 + 2] = v.z;
-			As[k][m4 + 3] = v.w;
-		}
-		// Bs[k][j]: Q row-major -> float4 along k per (j, k4)
-		for (uint32_t i = tid; i < (FMM_BK / 4) * FMM_BN; i += 256) {
-			uint32_t j = i / (FMM_BK / 4);
-			uint32_t k4 = (i % (FMM_BK / 4)) * 4;
-			const float4 v = *(const float4 *)(Q +
-			                                   (uint64_t)(bcol + j) * d +
-			                                   k0 + k4);
-			Bs[k4 + 0][j] = v.x;
-			Bs[k4 + 1][j] = v.y;
-			Bs[k4 + 2][j] = v.z;
-			Bs[k4 + 3][j] = v.w;
-		}
-		__syncthreads();
+		if (k0 + FMM_BK < d)
+			prefetch(k0 + FMM_BK); // in flight during the MFMA block
 #pragma unroll
 		for (uint32_t kk = 0; kk < FMM_BK; kk += 2) {
 			// operand map (32x32x2): lane l holds A[i=l&31][k=l>>5],
@@ -1061,6 +1078,10 @@ __global__ __launch_bounds__(256) void k_mfma_scan_topk(
 			                                                 acc[1][1], 0, 0, 0);
 		}
 		__syncthreads();
+		if (k0 + FMM_BK < d) {
+			stage_lds();
+			__syncthreads();
+		}
 	}
 
 	// epilogue: C/D map for 32x32 shapes — col = lane&31,
@@ -1979,6 +2000,9 @@ static int ensure_query_scratch(sdbv_ctx *ctx, uint32_t d, uint64_t nblocks,
 		if (ctx->q_dev)
 			(void)hipFree(ctx->q_dev);
 		HIP_CHECK(ctx, hipMalloc(&ctx->q_dev, d * sizeof(float)));
+		if (ctx->q_pin)
+			(void)hipHostFree(ctx->q_pin);
+		HIP_CHECK(ctx, hipHostMalloc(&ctx->q_pin, d * sizeof(float)));
 		ctx->q_cap = d;
 	}
 	uint64_t need = nblocks * k * sizeof(Cand);
@@ -2038,7 +2062,8 @@ static int knn_large_k(sdbv_ctx *ctx, Table &t, const float *q, uint32_t d,
 	int rc = ensure_query_scratch(ctx, d, 1, 1);
 	if (rc != SDBV_OK)
 		return rc;
-	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
+	std::memcpy(ctx->q_pin, q, d * sizeof(float)); // pinned staging
+	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, ctx->q_pin, d * sizeof(float),
 	                              hipMemcpyHostToDevice, ctx->stream));
 	double *dout = nullptr;
 	HIP_CHECK(ctx, hipMalloc(&dout, t.n * sizeof(double)));
@@ -2117,7 +2142,8 @@ int sdbv_knn_bruteforce(sdbv_ctx *ctx, uint64_t table, const float *q,
 	int rc = ensure_query_scratch(ctx, d, nblocks, k);
 	if (rc != SDBV_OK)
 		return rc;
-	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
+	std::memcpy(ctx->q_pin, q, d * sizeof(float)); // pinned staging
+	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, ctx->q_pin, d * sizeof(float),
 	                              hipMemcpyHostToDevice, ctx->stream));
 	double q_norm = sqrt((double)h_sumsq_f32(q, d));
 
@@ -2193,7 +2219,8 @@ int sdbv_all_distances(sdbv_ctx *ctx, uint64_t table, const float *q,
 	int rc = ensure_query_scratch(ctx, d, 1, 1);
 	if (rc != SDBV_OK)
 		return rc;
-	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
+	std::memcpy(ctx->q_pin, q, d * sizeof(float)); // pinned staging
+	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, ctx->q_pin, d * sizeof(float),
 	                              hipMemcpyHostToDevice, ctx->stream));
 	double *dout = nullptr;
 	HIP_CHECK(ctx, hipMalloc(&dout, t.n * sizeof(double)));
@@ -2219,7 +2246,8 @@ int sdbv_gather_distance(sdbv_ctx *ctx, uint64_t table, const uint32_t *rows,
 	int rc = ensure_query_scratch(ctx, d, 1, 1);
 	if (rc != SDBV_OK)
 		return rc;
-	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, q, d * sizeof(float),
+	std::memcpy(ctx->q_pin, q, d * sizeof(float)); // pinned staging
+	HIP_CHECK(ctx, hipMemcpyAsync(ctx->q_dev, ctx->q_pin, d * sizeof(float),
 	                              hipMemcpyHostToDevice, ctx->stream));
 	double q_norm = sqrt((double)h_sumsq_f32(q, d));
 	uint32_t *rows_dev = nullptr;
