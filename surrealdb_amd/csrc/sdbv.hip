@@ -992,7 +992,7 @@ __global__ __launch_bounds__(THREADS) void k_batch_topk(
 // ---------------------------------------------------------------------------
 #define FMM_BM 128
 #define FMM_BN 128
-#define FMM_BK 16 /* double-buffered: 2 stages of 16 keep 4 WGs/CU in LDS */
+#define FMM_BK 32
 #define FMM_LDS_PAD 4
 #define FMM_CAND_CAP 4096u
 
@@ -1005,8 +1005,8 @@ __global__ __launch_bounds__(256) void k_mfma_scan_topk(
     const uint32_t *__restrict__ theta, // per-query kth-best u32 key
     BCand *__restrict__ cand,           // [b][FMM_CAND_CAP]
     uint32_t *__restrict__ cand_cnt) {  // [b]
-	__shared__ float As[2][FMM_BK][FMM_BM + FMM_LDS_PAD];
-	__shared__ float Bs[2][FMM_BK][FMM_BN + FMM_LDS_PAD];
+	__shared__ float As[FMM_BK][FMM_BM + FMM_LDS_PAD];
+	__shared__ float Bs[FMM_BK][FMM_BN + FMM_LDS_PAD];
 	const uint32_t wave = threadIdx.x >> 6;
 	const uint32_t lane = threadIdx.x & 63;
 	const uint32_t wm = wave & 1;  // wave row (2x2 wave grid)
@@ -1016,58 +1016,58 @@ __global__ __launch_bounds__(256) void k_mfma_scan_topk(
 	f32x16 acc[2][2] = {};
 
 	const uint32_t tid = threadIdx.x;
-	// software pipeline + LDS double buffer: prefetch K-stage t+1 into
-	// registers during stage t's MFMA block, write it to the OTHER LDS
-	// buffer (no barrier needed before the writes), one barrier per stage.
-	// The naive load->sync->compute cycle measured 73 TF; register
-	// prefetch alone 104; this removes the staging stall entirely.
-	// Each thread stages 2 float4 of A (k = a_kq + 8r, fixed m4) and 2 of
-	// B (j = b_jq + 64r, fixed k4): BK*BM/4 = 512 float4 per stage.
+	// software pipeline: prefetch K-stage t+1 into registers while MFMA
+	// consumes stage t from LDS (the naive load->sync->compute cycle
+	// measured 73 TF; the guide's untuned same-shape reference is 122)
+	// each thread stages 4 float4 of A (k = a_kq + 8r, fixed m4) and 4 of
+	// B (j = b_jq + 32r, fixed k4): A has FMM_BK*FMM_BM/4 = 1024 float4
+	// per stage over 256 threads, B likewise
 	const uint32_t a_m4 = (tid % (FMM_BM / 4)) * 4; // 0..124
-	const uint32_t b_k4 = (tid % (FMM_BK / 4)) * 4; // 0..12
-	const uint32_t a_kq = tid / (FMM_BM / 4); // 0..7
-	const uint32_t b_jq = tid / (FMM_BK / 4); // 0..63
-	float4 pa[2], pb[2];
+	const uint32_t b_k4 = (tid % (FMM_BK / 4)) * 4; // 0..28
+	const uint32_t a_kq = tid / (FMM_BM / 4); // 0..7 (k quarter index)
+	const uint32_t b_jq = tid / (FMM_BK / 4); // 0..31 (j quarter index)
+	float4 pa[4], pb[4];
 	auto prefetch = [&](uint32_t k0) {
 #pragma unroll
-		for (uint32_t r = 0; r < 2; r++) {
+		for (uint32_t r = 0; r < 4; r++) {
 			uint32_t k = a_kq + 8 * r;
 			pa[r] = *(const float4 *)(cm + (uint64_t)(k0 + k) * n_pad +
 			                          brow + a_m4);
-			uint32_t j = b_jq + 64 * r;
+			uint32_t j = b_jq + 32 * r;
 			pb[r] = *(const float4 *)(Q + (uint64_t)(bcol + j) * d + k0 +
 			                          b_k4);
 		}
 	};
-	auto stage_lds = [&](uint32_t buf) {
+	auto stage_lds = [&]() {
 #pragma unroll
-		for (uint32_t r = 0; r < 2; r++) {
+		for (uint32_t r = 0; r < 4; r++) {
 			uint32_t k = a_kq + 8 * r;
-			*(float4 *)&As[buf][k][a_m4] = pa[r];
-			uint32_t j = b_jq + 64 * r;
-			Bs[buf][b_k4 + 0][j] = pb[r].x;
-			Bs[buf][b_k4 + 1][j] = pb[r].y;
-			Bs[buf][b_k4 + 2][j] = pb[r].z;
-			Bs[buf][b_k4 + 3][j] = pb[r].w;
+			As[k][a_m4 + 0] = pa[r].x;
+			As[k][a_m4 + 1] = pa[r].y;
+			As[k][a_m4 + 2] = pa[r].z;
+			As[k][a_m4 + 3] = pa[r].w;
+			uint32_t j = b_jq + 32 * r;
+			Bs[b_k4 + 0][j] = pb[r].x;
+			Bs[b_k4 + 1][j] = pb[r].y;
+			Bs[b_k4 + 2][j] = pb[r].z;
+			Bs[b_k4 + 3][j] = pb[r].w;
 		}
 	};
 	prefetch(0);
-	stage_lds(0);
+	stage_lds();
 	__syncthreads();
-	const uint32_t nstages = d / FMM_BK;
-	for (uint32_t s = 0; s < nstages; s++) {
-		const uint32_t buf = s & 1;
-		if (s + 1 < nstages)
-			prefetch((s + 1) * FMM_BK); // in flight during the MFMA block
+	for (uint32_t k0 = 0; k0 < d; k0 += FMM_BK) {
+		if (k0 + FMM_BK < d)
+			prefetch(k0 + FMM_BK); // in flight during the MFMA block
 #pragma unroll
 		for (uint32_t kk = 0; kk < FMM_BK; kk += 2) {
 			// operand map (32x32x2): lane l holds A[i=l&31][k=l>>5],
 			// B[k=l>>5][j=l&31]
 			const uint32_t ai = lane & 31, ak = lane >> 5;
-			float a0 = As[buf][kk + ak][wm * 64 + ai];
-			float a1 = As[buf][kk + ak][wm * 64 + 32 + ai];
-			float b0 = Bs[buf][kk + ak][wn * 64 + ai];
-			float b1 = Bs[buf][kk + ak][wn * 64 + 32 + ai];
+			float a0 = As[kk + ak][wm * 64 + ai];
+			float a1 = As[kk + ak][wm * 64 + 32 + ai];
+			float b0 = Bs[kk + ak][wn * 64 + ai];
+			float b1 = Bs[kk + ak][wn * 64 + 32 + ai];
 			acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0,
 			                                                 acc[0][0], 0, 0, 0);
 			acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0,
@@ -1077,9 +1077,11 @@ __global__ __launch_bounds__(256) void k_mfma_scan_topk(
 			acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1,
 			                                                 acc[1][1], 0, 0, 0);
 		}
-		if (s + 1 < nstages)
-			stage_lds(buf ^ 1); // other buffer: no barrier needed first
 		__syncthreads();
+		if (k0 + FMM_BK < d) {
+			stage_lds();
+			__syncthreads();
+		}
 	}
 
 	// epilogue: C/D map for 32x32 shapes — col = lane&31,
