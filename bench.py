@@ -51,7 +51,7 @@ def main():
     ap.add_argument("--host-build", action="store_true",
                     help="HNSW mode: host chunked snapshot build instead of "
                          "the GPU-accelerated one")
-    ap.add_argument("--chunk", type=int, default=4096,
+    ap.add_argument("--chunk", type=int, default=8192,
                     help="HNSW snapshot build chunk size")
     ap.add_argument("--seed", type=lambda x: int(x, 0), default=0x5DB1)
     ap.add_argument("--cpu-sample-rows", type=int, default=10_000_000,

@@ -2686,10 +2686,23 @@ struct sdbv_hnsw {
 	uint64_t adj_nodes = 0; // allocated node capacity of adj_dev/deg_dev
 	uint64_t dev_rows = 0;  // rows of rm_dev/norms_dev currently uploaded
 	// dirty tracking for the per-chunk adjacency sync: bytes set under the
-	// striped node locks wherever layer-0 edge lists mutate. Only active
+	// striped node locks wherever layer-0 edge lists mutate, plus an
+	// append-only id list so the sync never scans all nelem flags (that
+	// scan was O(nelem x chunks) — quadratic at 10M rows). Only active
 	// (non-empty) inside the GPU snapshot build, where layers are
 	// pre-created so h->layers[0] never reallocates under workers.
 	std::vector<uint8_t> l0_dirty;
+	std::vector<uint32_t> l0_dirty_list;
+	std::atomic<uint32_t> l0_dirty_n{0};
+	// flag+append; callers serialize per node (node locks / sequential
+	// loops), so the flag test-and-set cannot race for one node
+	inline void mark_l0_dirty(uint32_t e) {
+		if (l0_dirty.empty() || l0_dirty[e])
+			return;
+		l0_dirty[e] = 1;
+		l0_dirty_list[l0_dirty_n.fetch_add(
+		    1, std::memory_order_relaxed)] = e;
+	}
 	std::string err;
 };
 
@@ -2900,7 +2913,7 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			    std::find(eq.begin(), eq.end(), e) == eq.end())
 				eq.push_back(e);
 		if (track)
-			h->l0_dirty[q_id] = 1;
+			h->mark_l0_dirty(q_id);
 	}
 	for (uint32_t e : neighbors) {
 		if (e == q_id)
@@ -2918,7 +2931,7 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			if (ee.size() > layer.m_max)
 				conn = ee;
 			if (track)
-				h->l0_dirty[e] = 1;
+				h->mark_l0_dirty(e);
 		}
 		if (!conn.empty()) {
 			// prune (layer.rs:363-377) — distances computed outside the
@@ -2944,7 +2957,7 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 					enew.push_back(cur);
 			layer.edges[e] = enew;
 			if (track)
-				h->l0_dirty[e] = 1;
+				h->mark_l0_dirty(e);
 		}
 	}
 }
@@ -3242,7 +3255,7 @@ static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
 				eq.push_back(e);
 		l0.in_layer[it.q_id] = 1;
 		if (track)
-			h->l0_dirty[it.q_id] = 1;
+			h->mark_l0_dirty(it.q_id);
 	}
 	for (auto &it : items) {
 		for (uint32_t e : it.neighbors) {
@@ -3256,7 +3269,7 @@ static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
 				if (ee.size() > l0.m_max)
 					overfull.push_back(e); // deduped below
 				if (track)
-					h->l0_dirty[e] = 1;
+					h->mark_l0_dirty(e);
 			}
 		}
 	}
@@ -3293,7 +3306,7 @@ static void batched_apply_phaseC_host(sdbv_hnsw *h,
 			                 std::move(ec), enew, false);
 			l0.edges[e] = enew;
 			if (track)
-				h->l0_dirty[e] = 1;
+				h->mark_l0_dirty(e);
 		}
 	};
 	std::vector<std::thread> ts;
@@ -3938,8 +3951,10 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	h->adj_nodes = nelem;
 	h->dev_rows = nelem;
 	h->l0_dirty.assign(nelem, 0);
+	h->l0_dirty_list.assign(nelem, 0);
+	h->l0_dirty_n.store(0, std::memory_order_relaxed);
 	for (uint64_t i = 0; i < base; i++)
-		h->l0_dirty[i] = 1; // first sync uploads the pre-existing graph
+		h->mark_l0_dirty((uint32_t)i); // first sync: pre-existing graph
 	const uint64_t vwords = (nelem + 31) / 32;
 	{
 		uint64_t need = (uint64_t)chunk * vwords * sizeof(uint32_t);
@@ -3976,6 +3991,8 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 			if (p)
 				(void)hipFree(p);
 		h->l0_dirty.clear();
+		h->l0_dirty_list.clear();
+		h->l0_dirty_list.shrink_to_fit();
 	};
 #define BGPU_CHECK(call)                                                     \
 	do {                                                                     \
@@ -4019,13 +4036,16 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		upd_deg.clear();
 		uint32_t maxdeg = 0;
 		auto &l0 = h->layers[0];
-		for (uint64_t i = 0; i < nelem; i++)
-			if (h->l0_dirty[i]) {
-				uint32_t dg = (uint32_t)l0.edges[i].size();
-				maxdeg = std::max(maxdeg, dg);
-				upd_ids.push_back((uint32_t)i);
-				upd_deg.push_back(dg);
-			}
+		// dirty LIST, not a full flag scan (the scan was O(nelem) per
+		// chunk -> quadratic across the build at 10M rows)
+		uint32_t nd = h->l0_dirty_n.load(std::memory_order_relaxed);
+		for (uint32_t li = 0; li < nd; li++) {
+			uint32_t i = h->l0_dirty_list[li];
+			uint32_t dg = (uint32_t)l0.edges[i].size();
+			maxdeg = std::max(maxdeg, dg);
+			upd_ids.push_back(i);
+			upd_deg.push_back(dg);
+		}
 		if (maxdeg > h->adj_stride) {
 			uint32_t ns = (maxdeg + 8 + 7) & ~7u;
 			(void)hipFree(h->adj_dev);
@@ -4041,7 +4061,9 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 					upd_deg.push_back((uint32_t)l0.edges[i].size());
 				}
 		}
-		std::fill(h->l0_dirty.begin(), h->l0_dirty.end(), 0);
+		for (uint32_t li = 0; li < nd; li++)
+			h->l0_dirty[h->l0_dirty_list[li]] = 0;
+		h->l0_dirty_n.store(0, std::memory_order_relaxed);
 		uint64_t cnt = upd_ids.size();
 		if (cnt == 0)
 			return SDBV_OK;
@@ -4374,7 +4396,7 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 					          selh.begin() + (uint64_t)li * h->m0 +
 					              selcnth[li]);
 					if (track)
-						h->l0_dirty[e] = 1;
+						h->mark_l0_dirty(e);
 				}
 			}
 			if (!host_prunes.empty())
