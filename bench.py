@@ -110,8 +110,12 @@ def main():
     if args.hnsw:
         # HNSW does not shard (sequential graph traversal):
         # replicas only — each rank builds/holds a replica of its shard.
+        def mark(what):
+            print(f"[bench hnsw] {what} at +{time.perf_counter()-t0:.0f}s",
+                  file=sys.stderr, flush=True)
         from surrealdb_amd.synth import gen_f32 as _gen
         pts = _gen(args.seed, row_offset, rows, args.dim)
+        mark("generated")
         hnsw_index = ctx.hnsw_create(args.dim, metric=args.metric, m=16,
                                      m0=32, efc=150, seed=args.seed)
         # chunked snapshot build (chunk/n <= 0.4% at bench scales — quality
@@ -123,8 +127,10 @@ def main():
         else:
             hnsw_index.insert_batch_snapshot_gpu(pts, chunk=args.chunk,
                                                  nthreads=os.cpu_count())
+        mark("built")
         del pts
         hnsw_index.finalize(1)
+        mark("finalized")
     else:
         ctx.stage_synthetic(1, rows, args.dim, metric=args.metric,
                             seed=args.seed, row_offset=row_offset,
@@ -280,6 +286,10 @@ def main():
     qps = args.steps * queries_per_step / t_total
     p50_ms = float(np.percentile(np.array(step_times) * 1e3, 50))
     p95_ms = float(np.percentile(np.array(step_times) * 1e3, 95))
+    # early echo so a timeout during the (post-region) baseline leg still
+    # leaves the core measurement on record
+    print(f"[bench] value={qps:.3f} q/s p50={p50_ms:.3f}ms "
+          f"p95={p95_ms:.3f}ms", file=sys.stderr, flush=True)
 
     scan_ms_avg = scan_ms_acc / args.steps
     traffic = os.environ.get("SDBV_TRAFFIC_BYTES_PER_LAUNCH")

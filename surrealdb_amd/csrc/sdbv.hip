@@ -4403,6 +4403,12 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 				batched_apply_phaseC_host(h, host_prunes, nthreads);
 		}
 		t_selC += secs(tp6, now());
+		if (((c0 / chunk) & 63) == 63)
+			fprintf(stderr,
+			        "[sdbv build_gpu2] %llu/%llu upper=%.1f desc=%.1f "
+			        "sync=%.1f kern=%.1f selA=%.1f link=%.1f selC=%.1f\n",
+			        (unsigned long long)c1, (unsigned long long)n, t_upper,
+			        t_desc, t_sync, t_kern, t_selA, t_link, t_selC);
 	}
 #undef BGPU_CHECK
 	hnsw_promote_ep(h);
@@ -4506,13 +4512,21 @@ int sdbv_hnsw_finalize(sdbv_hnsw *h, uint64_t table) {
 	if (rc)
 		return rc;
 	sdbv_ctx *ctx = h->ctx;
-	HIP_CHECK(ctx, hipMalloc(&h->rows_dev, (h->m0 + 1) * sizeof(uint32_t)));
-	HIP_CHECK(ctx, hipMalloc(&h->dout_dev, (h->m0 + 1) * sizeof(double)));
+	// per-hop scratch sized from the ACTUAL max layer-0 degree: a
+	// parallel/batched build can leave a node transiently above m0 (the
+	// keep-back fold in layer_insert_apply), and a frontier copy larger
+	// than the scratch would overflow it (round-1 advisor class of bug)
+	h->max_deg = h->m0 + 1;
+	for (uint64_t i = 0; i < h->next_id; i++)
+		h->max_deg = std::max<uint32_t>(
+		    h->max_deg, (uint32_t)h->layers[0].edges[i].size() + 1);
+	HIP_CHECK(ctx, hipMalloc(&h->rows_dev, h->max_deg * sizeof(uint32_t)));
+	HIP_CHECK(ctx, hipMalloc(&h->dout_dev, h->max_deg * sizeof(double)));
 	HIP_CHECK(ctx, hipMalloc(&h->q_dev, h->d * sizeof(float)));
 	HIP_CHECK(ctx, hipHostMalloc(&h->rows_pinned,
-	                             (h->m0 + 1) * sizeof(uint32_t)));
+	                             h->max_deg * sizeof(uint32_t)));
 	HIP_CHECK(ctx, hipHostMalloc(&h->dists_pinned,
-	                             (h->m0 + 1) * sizeof(double)));
+	                             h->max_deg * sizeof(double)));
 	// persistent-kernel graph state: row-major vectors + layer-0 CSR
 	uint64_t n = h->next_id;
 	HIP_CHECK(ctx, hipMalloc(&h->rm_dev, n * h->d * sizeof(float)));
@@ -4785,7 +4799,7 @@ int sdbv_hnsw_knn(sdbv_hnsw *h, const float *q, uint32_t k, uint32_t ef,
 	visited[ep_id] = true;
 	double fq = w.peek_last_dist(DBL_MAX);
 	std::vector<uint32_t> frontier;
-	std::vector<double> fdists(h->m0 + 1);
+	std::vector<double> fdists(h->max_deg); // actual max degree
 	double cd;
 	uint32_t doc;
 	double gpu_ms = 0;
@@ -5343,7 +5357,7 @@ static int idx_graph_search_gpu(sdbv_index *ix, const float *q, uint32_t k,
 	visited[ep_id] = true;
 	double fq = w.peek_last_dist(DBL_MAX);
 	std::vector<uint32_t> frontier;
-	std::vector<double> fdists(h->m0 + 1);
+	std::vector<double> fdists(h->max_deg); // actual max degree
 	double cd;
 	uint32_t doc;
 	while (candidates.pop_first(&cd, &doc)) {
@@ -5984,7 +5998,7 @@ static int idx_search_l0_filter_gpu(sdbv_index *ix, const float *q,
 	const Layer &l0 = h->layers[0];
 	double f_dist = w.peek_last_dist(DBL_MAX);
 	std::vector<uint32_t> frontier;
-	std::vector<double> fdists(h->m0 + 1);
+	std::vector<double> fdists(h->max_deg); // actual max degree
 	double cd;
 	uint32_t doc;
 	while (candidates.pop_first(&cd, &doc)) {
