@@ -45,6 +45,7 @@
 #include <cstdint>
 #include <deque>
 #include <limits>
+#include <memory>
 #include <thread>
 #include <unordered_set>
 #include <cstdio>
@@ -2685,23 +2686,28 @@ struct sdbv_hnsw {
 	uint32_t adj_stride = 0;
 	uint64_t adj_nodes = 0; // allocated node capacity of adj_dev/deg_dev
 	uint64_t dev_rows = 0;  // rows of rm_dev/norms_dev currently uploaded
-	// dirty tracking for the per-chunk adjacency sync: bytes set under the
-	// striped node locks wherever layer-0 edge lists mutate, plus an
-	// append-only id list so the sync never scans all nelem flags (that
-	// scan was O(nelem x chunks) — quadratic at 10M rows). Only active
-	// (non-empty) inside the GPU snapshot build, where layers are
-	// pre-created so h->layers[0] never reallocates under workers.
-	std::vector<uint8_t> l0_dirty;
-	std::vector<uint32_t> l0_dirty_list;
-	std::atomic<uint32_t> l0_dirty_n{0};
+	// Per-layer dirty tracking for the per-chunk adjacency syncs of the
+	// GPU build (v3 keeps a padded device adjacency for EVERY layer):
+	// flag dedup + append-only id list, so a sync never scans all nelem
+	// flags (that scan was O(nelem x chunks) — quadratic at 10M rows).
+	// Only active (allocated) inside the GPU snapshot build, where layers
+	// are pre-created so h->layers never reallocates under workers.
+	struct DirtyTrack {
+		std::vector<uint8_t> flag;
+		std::vector<uint32_t> list;
+		std::atomic<uint32_t> n{0};
+	};
+	std::vector<std::unique_ptr<DirtyTrack>> dtrack; // indexed by layer
 	// flag+append; callers serialize per node (node locks / sequential
 	// loops), so the flag test-and-set cannot race for one node
-	inline void mark_l0_dirty(uint32_t e) {
-		if (l0_dirty.empty() || l0_dirty[e])
+	inline void mark_dirty(uint32_t layer, uint32_t e) {
+		if (layer >= dtrack.size() || !dtrack[layer])
 			return;
-		l0_dirty[e] = 1;
-		l0_dirty_list[l0_dirty_n.fetch_add(
-		    1, std::memory_order_relaxed)] = e;
+		DirtyTrack &D = *dtrack[layer];
+		if (D.flag[e])
+			return;
+		D.flag[e] = 1;
+		D.list[D.n.fetch_add(1, std::memory_order_relaxed)] = e;
 	}
 	std::string err;
 };
@@ -2897,9 +2903,10 @@ static void select_neighbors(sdbv_hnsw *h, const Layer &layer, uint32_t q_id,
 static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
                                const float *q_pt, double q_norm, PQ w,
                                bool locked) {
-	// dirty tracking for the GPU snapshot build's device adjacency (only
-	// layer 0 lives on the device; flags written under the node locks)
-	const bool track = !h->l0_dirty.empty() && &layer == &h->layers[0];
+	// device-adjacency dirty marking for the GPU build's classic-path
+	// inserts (no-op unless a build allocated dtrack for this layer);
+	// layers never reallocates during builds, so the index is stable
+	const uint32_t li = (uint32_t)(&layer - h->layers.data());
 	std::vector<uint32_t> neighbors;
 	select_neighbors(h, layer, q_id, q_pt, q_norm, std::move(w), neighbors,
 	                 locked);
@@ -2912,8 +2919,7 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			if (e != q_id &&
 			    std::find(eq.begin(), eq.end(), e) == eq.end())
 				eq.push_back(e);
-		if (track)
-			h->mark_l0_dirty(q_id);
+		h->mark_dirty(li, q_id);
 	}
 	for (uint32_t e : neighbors) {
 		if (e == q_id)
@@ -2926,12 +2932,12 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 			if (e < layer.in_layer.size())
 				layer.in_layer[e] = 1;
 			auto &ee = layer.edges[e];
-			if (std::find(ee.begin(), ee.end(), q_id) == ee.end())
+			if (std::find(ee.begin(), ee.end(), q_id) == ee.end()) {
 				ee.push_back(q_id);
+				h->mark_dirty(li, e);
+			}
 			if (ee.size() > layer.m_max)
 				conn = ee;
-			if (track)
-				h->mark_l0_dirty(e);
 		}
 		if (!conn.empty()) {
 			// prune (layer.rs:363-377) — distances computed outside the
@@ -2956,8 +2962,7 @@ static void layer_insert_apply(sdbv_hnsw *h, Layer &layer, uint32_t q_id,
 				    std::find(enew.begin(), enew.end(), cur) == enew.end())
 					enew.push_back(cur);
 			layer.edges[e] = enew;
-			if (track)
-				h->mark_l0_dirty(e);
+			h->mark_dirty(li, e);
 		}
 	}
 }
@@ -3053,80 +3058,15 @@ static void insert_at(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
 	}
 }
 
-// insert_at with the LAYER-0 HALF DEFERRED (the batched-apply builds):
-// descend + insert into layers >= 1 exactly as insert_at, but return the
-// eps that would seed the layer-0 search (the layer-1 w — layer.rs:358)
-// instead of running it; the chunk's batched layer-0 search+apply takes
-// over from there.
-static void insert_at_upper(sdbv_hnsw *h, uint32_t q_id, uint32_t q_level,
-                            bool locked, PQ *eps_out) {
-	const float *q_pt = vec(h, q_id);
-	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
-	uint32_t top_up;
-	{
-		std::lock_guard<std::mutex> lk(h->global_mu);
-		top_up = (uint32_t)h->layers.size() - 1;
-		uint64_t nelem = h->vecs.size() / h->d;
-		for (auto &l : h->layers) {
-			if (l.edges.size() <= q_id)
-				l.edges.resize(nelem);
-			if (l.in_layer.size() < nelem)
-				l.in_layer.resize(nelem, 0);
-		}
-		for (uint32_t l = 0;
-		     l < (uint32_t)h->layers.size() && l <= q_level; l++)
-			h->layers[l].in_layer[q_id] = 1;
-		h->dirty = true;
-		if (h->enter_point < 0) {
-			h->enter_point = q_id;
-			PQ eps;
-			eps.push(0.0, q_id); // self-seed; batched search is a no-op
-			*eps_out = std::move(eps);
-			return;
-		}
-	}
-	uint64_t ep_id = (uint64_t)h->enter_point;
-	double ep_dist = dist(h, q_pt, q_norm, (uint32_t)ep_id);
-	if (q_level < top_up) {
-		for (uint32_t l = top_up; l > q_level; l--) {
-			PQ cand;
-			cand.push(ep_dist, (uint32_t)ep_id);
-			std::unordered_set<uint32_t> visited{(uint32_t)ep_id};
-			PQ w = cand;
-			search_layer_host(h, h->layers[l], q_pt, q_norm, cand, visited,
-			                  w, 1, locked);
-			double dd;
-			uint32_t ii;
-			if (w.peek_first(&dd, &ii)) {
-				ep_dist = dd;
-				ep_id = ii;
-			}
-		}
-	}
-	PQ eps;
-	eps.push(ep_dist, (uint32_t)ep_id);
-	uint32_t ins_to = std::min(q_level, top_up);
-	for (uint32_t l = ins_to; l >= 1; l--)
-		eps = layer_insert(h, h->layers[l], q_id, q_pt, q_norm,
-		                   std::move(eps), locked);
-	*eps_out = std::move(eps);
-}
+// Chunked SNAPSHOT searches (the §8f-rank-3 structure): per chunk, every
+// element's efc-search runs against the graph AS OF the chunk start
+// (read-only — no locks, embarrassingly parallel; the GPU build batches
+// the same searches onto the persistent kernel).
 
-// Chunked SNAPSHOT build (the §8f-rank-3 structure): per chunk, every
-// level-0 element's efc-search runs against the graph AS OF the chunk
-// start (read-only — no locks, embarrassingly parallel, and in round 2 the
-// per-chunk search batch moves onto the persistent device kernel), then
-// the apply half (select + edges + prunes) runs with the striped node
-// locks. One step beyond the parallel build's relaxed ordering: chunk
-// mates never see each other at search time (they still back-link from
-// later chunks); quality is pinned by the same recall bars. Upper-level
-// elements (~1/m of the batch) insert sequentially at the chunk front so
-// the layer structure exists before the snapshot searches.
-// Upper-layer greedy descent only (the host half shared by the CPU and GPU
-// snapshot search paths); returns the layer-0 entry point + distance.
+// Upper-layer greedy descent only (host half shared by the CPU and GPU
+// snapshot paths); returns the layer-0 entry point + distance.
 static void snapshot_descend_one(sdbv_hnsw *h, uint32_t q_id, double q_norm,
                                  uint32_t *ep_out, double *epd_out) {
-	using namespace hnsw;
 	const float *q_pt = vec(h, q_id);
 	uint32_t ep_id = (uint32_t)h->enter_point;
 	double ep_dist = dist(h, q_pt, q_norm, ep_id);
@@ -3148,9 +3088,7 @@ static void snapshot_descend_one(sdbv_hnsw *h, uint32_t q_id, double q_norm,
 	*epd_out = ep_dist;
 }
 
-static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
-                                hnsw::PQ &w_out) {
-	using namespace hnsw;
+static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id, PQ &w_out) {
 	const float *q_pt = vec(h, q_id);
 	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
 	uint32_t ep_id;
@@ -3167,11 +3105,11 @@ static void snapshot_search_one(sdbv_hnsw *h, uint32_t q_id,
 	w_out = std::move(w);
 }
 
-// Layer-0 efc-search from a given eps window (the insert path's multi-ep
-// seeding, layer.rs:342-358) against the snapshot graph.
-static void snapshot_search_eps(sdbv_hnsw *h, uint32_t q_id, const PQ &eps_in,
-                                hnsw::PQ &w_out) {
-	using namespace hnsw;
+// ef-search at ANY layer from a given eps window (the insert path's
+// multi-ep seeding, layer.rs:342-358) against the snapshot graph.
+static void snapshot_search_layer_eps(sdbv_hnsw *h, uint32_t layer_idx,
+                                      uint32_t q_id, const PQ &eps_in,
+                                      uint32_t ef, PQ &w_out) {
 	const float *q_pt = vec(h, q_id);
 	double q_norm = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
 	PQ eps = eps_in;
@@ -3180,18 +3118,48 @@ static void snapshot_search_eps(sdbv_hnsw *h, uint32_t q_id, const PQ &eps_in,
 	visited.begin(h->vecs.size() / h->d);
 	for (auto &e : eps_in.to_vec())
 		visited.insert(e.second);
-	search_layer_host(h, h->layers[0], q_pt, q_norm, eps, visited, w,
-	                  h->efc, false);
+	search_layer_host(h, h->layers[layer_idx], q_pt, q_norm, eps, visited,
+	                  w, ef, false);
 	w_out = std::move(w);
 }
 
-// ---- batched layer-0 apply (snapshot-build v2) ----
+static void snapshot_search_eps(sdbv_hnsw *h, uint32_t q_id,
+                                const PQ &eps_in, PQ &w_out) {
+	snapshot_search_layer_eps(h, 0, q_id, eps_in, h->efc, w_out);
+}
+
+// Greedy descent through layers top..2 only (ef=1; the layer-1 hop is the
+// batched builds' device work).
+static void descend_to_layer2(sdbv_hnsw *h, uint32_t q_id, double q_norm,
+                              uint32_t *ep_out, double *epd_out) {
+	const float *q_pt = vec(h, q_id);
+	uint32_t ep_id = (uint32_t)h->enter_point;
+	double ep_dist = dist(h, q_pt, q_norm, ep_id);
+	for (size_t l = h->layers.size() - 1; l >= 2; l--) {
+		PQ cand;
+		cand.push(ep_dist, ep_id);
+		std::unordered_set<uint32_t> visited{ep_id};
+		PQ w = cand;
+		search_layer_host(h, h->layers[l], q_pt, q_norm, cand, visited, w,
+		                  1, false);
+		double dd;
+		uint32_t ii;
+		if (w.peek_first(&dd, &ii)) {
+			ep_dist = dd;
+			ep_id = ii;
+		}
+	}
+	*ep_out = ep_id;
+	*epd_out = ep_dist;
+}
+
+// ---- batched per-layer apply (snapshot-build v3) ----
 // The same select/prune algorithm as layer_insert_apply, but in three
-// bulk phases per chunk: (A) every element's neighbour select against the
-// post-search graph, read-only and parallel; (B) all edge appends, in
-// element order, sequential (deterministic and cheap); (C) one prune pass
-// over every node that ended over m_max, parallel (prunes are
-// independent: enew ⊆ conn, no cascading). This is ONE valid
+// bulk phases per chunk AND PER LAYER: (A) every element's neighbour
+// select against the post-search graph, read-only and parallel; (B) all
+// edge appends, in element order, sequential (deterministic and cheap);
+// (C) one prune pass over every node that ended over m_max, parallel
+// (prunes are independent: enew ⊆ conn, no cascading). This is ONE valid
 // serialization of the parallel interleaved apply — same algorithm, a
 // fixed schedule — and the form whose distance work (the RAM-bound part)
 // batches onto the device in the GPU build. Quality is pinned by the same
@@ -3202,30 +3170,32 @@ struct ApplyItem {
 	std::vector<uint32_t> neighbors; // phase-A output
 };
 
-static void batched_apply_l0_links(sdbv_hnsw *h,
-                                   std::vector<ApplyItem> &items,
-                                   int nthreads);
 static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
-                                                  std::vector<ApplyItem> &it);
-static void batched_apply_phaseC_host(sdbv_hnsw *h,
+                                                  uint32_t layer_idx,
+                                                  std::vector<ApplyItem *> &it);
+static void batched_apply_phaseC_host(sdbv_hnsw *h, uint32_t layer_idx,
                                       const std::vector<uint32_t> &overfull,
                                       int nthreads);
 
-static void batched_apply_l0_host(sdbv_hnsw *h, std::vector<ApplyItem> &items,
-                                  int nthreads) {
-	Layer &l0 = h->layers[0];
+// host phase A + B + C (the twin's apply; the GPU build replaces A and C
+// with k_pair_mats/k_heur_select)
+static void batched_apply_host(sdbv_hnsw *h, uint32_t layer_idx,
+                               std::vector<ApplyItem *> &items,
+                               int nthreads) {
+	Layer &L = h->layers[layer_idx];
 	// phase A: selects (read-only graph)
 	{
 		std::atomic<uint64_t> cursor{0};
 		auto worker = [&]() {
 			uint64_t j;
 			while ((j = cursor.fetch_add(1)) < items.size()) {
-				ApplyItem &it = items[j];
+				ApplyItem &it = *items[j];
 				const float *q_pt = vec(h, it.q_id);
 				double q_norm = h->metric == SDBV_METRIC_COSINE
 				                    ? h->norms[it.q_id]
 				                    : 0;
-				select_neighbors(h, l0, it.q_id, q_pt, q_norm, it.w,
+				it.neighbors.clear();
+				select_neighbors(h, L, it.q_id, q_pt, q_norm, it.w,
 				                 it.neighbors, false);
 			}
 		};
@@ -3237,39 +3207,42 @@ static void batched_apply_l0_host(sdbv_hnsw *h, std::vector<ApplyItem> &items,
 		for (auto &t : ts)
 			t.join();
 	}
-	batched_apply_l0_links(h, items, nthreads);
+	auto overfull = batched_apply_phaseB(h, layer_idx, items);
+	batched_apply_phaseC_host(h, layer_idx, overfull, nthreads);
 }
 
-// phase B: appends, element order (graph.rs:52-64 entry().or_insert);
+// phase B: appends, element order (graph.rs:52-64 entry().or_insert —
+// including the implicit creation of a missing back-edge target, e.g. an
+// eps node from the layer above that is not a member of THIS layer);
 // returns the deduped list of nodes that ended over m_max (prune targets)
 static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
-                                                  std::vector<ApplyItem> &items) {
-	Layer &l0 = h->layers[0];
-	const bool track = !h->l0_dirty.empty();
+                                                  uint32_t layer_idx,
+                                                  std::vector<ApplyItem *> &items) {
+	Layer &L = h->layers[layer_idx];
 	std::vector<uint32_t> overfull;
-	for (auto &it : items) {
-		auto &eq = l0.edges[it.q_id];
+	for (auto *itp : items) {
+		auto &it = *itp;
+		auto &eq = L.edges[it.q_id];
 		for (uint32_t e : it.neighbors)
 			if (e != it.q_id &&
 			    std::find(eq.begin(), eq.end(), e) == eq.end())
 				eq.push_back(e);
-		l0.in_layer[it.q_id] = 1;
-		if (track)
-			h->mark_l0_dirty(it.q_id);
+		L.in_layer[it.q_id] = 1;
+		h->mark_dirty(layer_idx, it.q_id);
 	}
-	for (auto &it : items) {
+	for (auto *itp : items) {
+		auto &it = *itp;
 		for (uint32_t e : it.neighbors) {
 			if (e == it.q_id)
 				continue;
-			auto &ee = l0.edges[e];
-			if (e < l0.in_layer.size())
-				l0.in_layer[e] = 1;
+			auto &ee = L.edges[e];
+			if (e < L.in_layer.size())
+				L.in_layer[e] = 1;
 			if (std::find(ee.begin(), ee.end(), it.q_id) == ee.end()) {
 				ee.push_back(it.q_id);
-				if (ee.size() > l0.m_max)
+				if (ee.size() > L.m_max)
 					overfull.push_back(e); // deduped below
-				if (track)
-					h->mark_l0_dirty(e);
+				h->mark_dirty(layer_idx, e);
 			}
 		}
 	}
@@ -3280,17 +3253,16 @@ static std::vector<uint32_t> batched_apply_phaseB(sdbv_hnsw *h,
 }
 
 // phase C: prunes (layer.rs:363-377), parallel over distinct nodes
-static void batched_apply_phaseC_host(sdbv_hnsw *h,
+static void batched_apply_phaseC_host(sdbv_hnsw *h, uint32_t layer_idx,
                                       const std::vector<uint32_t> &overfull,
                                       int nthreads) {
-	Layer &l0 = h->layers[0];
-	const bool track = !h->l0_dirty.empty();
+	Layer &L = h->layers[layer_idx];
 	std::atomic<uint64_t> cursor{0};
 	auto worker = [&]() {
 		uint64_t i;
 		while ((i = cursor.fetch_add(1)) < overfull.size()) {
 			uint32_t e = overfull[i];
-			const auto conn = l0.edges[e]; // copy (read-only source)
+			const auto conn = L.edges[e]; // copy (read-only source)
 			PQ ec;
 			for (uint32_t nid : conn) {
 				if (nid < h->elem_present.size() &&
@@ -3299,14 +3271,13 @@ static void batched_apply_phaseC_host(sdbv_hnsw *h,
 				ec.push(dist_ee(h, e, nid), nid);
 			}
 			std::vector<uint32_t> enew;
-			select_neighbors(h, l0, e, vec(h, e),
+			select_neighbors(h, L, e, vec(h, e),
 			                 h->metric == SDBV_METRIC_COSINE
 			                     ? h->norms[e]
 			                     : 0,
 			                 std::move(ec), enew, false);
-			l0.edges[e] = enew;
-			if (track)
-				h->mark_l0_dirty(e);
+			L.edges[e] = enew;
+			h->mark_dirty(layer_idx, e);
 		}
 	};
 	std::vector<std::thread> ts;
@@ -3318,12 +3289,6 @@ static void batched_apply_phaseC_host(sdbv_hnsw *h,
 	worker();
 	for (auto &t : ts)
 		t.join();
-}
-
-static void batched_apply_l0_links(sdbv_hnsw *h, std::vector<ApplyItem> &items,
-                                   int nthreads) {
-	auto overfull = batched_apply_phaseB(h, items);
-	batched_apply_phaseC_host(h, overfull, nthreads);
 }
 
 // ---- graph element removal (sequential only: apply_pendings holds the
@@ -3737,15 +3702,17 @@ static void hnsw_promote_ep(sdbv_hnsw *h) {
 	}
 }
 
-// Host twin of the GPU batched-apply snapshot build ("snapshot2"): same
-// chunked snapshot searches as sdbv_hnsw_insert_batch_snapshot, but the
-// apply half runs the BATCHED schedule (phase A selects, phase B appends
-// in element order, phase C one prune pass — batched_apply_l0_host), and
-// upper-level elements defer their layer-0 half into the same batch
-// (insert_at_upper). This is the bit-exact CPU reference for
-// sdbv_hnsw_insert_batch_snapshot_gpu (identical schedule, identical
-// restated distance chains); quality is pinned by the same recall bars as
-// the v1 builds.
+// Host twin of the GPU batched-apply snapshot build ("snapshot2", v3):
+// the full reference insert algorithm with a fixed batched schedule at
+// EVERY layer. Per chunk, top-down through the layers: each element
+// either greedy-descends (ef=1, layers above its level — search_ep,
+// hnsw/mod.rs:521-548) or runs the efc-search whose w both seeds the
+// next layer (layer.rs:358's cascade) and feeds the batched apply
+// (select/append/prune) at that layer; layer 0 takes every element.
+// Searches see the graph as of the chunk start (snapshot semantics at
+// every layer — the same chunk/n quality relaxation the v1 snapshot
+// build pinned with recall bars). This is the bit-exact CPU reference
+// for sdbv_hnsw_insert_batch_snapshot_gpu.
 int sdbv_hnsw_insert_batch_snapshot2(sdbv_hnsw *h, const float *pts,
                                      uint64_t n, uint32_t chunk,
                                      int nthreads) {
@@ -3777,60 +3744,69 @@ int sdbv_hnsw_insert_batch_snapshot2(sdbv_hnsw *h, const float *pts,
 				l.in_layer.resize(ne, 0);
 		}
 	}
+	const uint32_t top = (uint32_t)h->layers.size() - 1;
+	(void)top;
+	// warm-up (the classic parallel build's bootstrap): while the graph is
+	// near-empty every batched element can only select the enter point,
+	// whose single prune then orphans most of them — the first elements
+	// go in classically so batching starts on a connected graph
+	const uint64_t warm_until =
+	    h->enter_point < 0 ? base + 64 : 0;
 	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
 		uint64_t c1 = std::min(n, c0 + chunk);
 		std::vector<ApplyItem> items;
-		std::vector<PQ> eps_of; // parallel to items
-		std::vector<uint32_t> kinds; // 1 = upper (eps ready), 0 = flat
+		std::vector<PQ> eps_of;
+		std::vector<uint32_t> lvl_of;
+		std::vector<std::pair<uint32_t, uint32_t>> upper2; // (q_id, level)
 		for (uint64_t i = c0; i < c1; i++) {
-			if (h->enter_point < 0) {
+			if (h->enter_point < 0 || base + i < warm_until) {
 				insert_at(h, (uint32_t)(base + i), levels[i], false);
+				continue;
+			}
+			if (levels[i] >= 2) {
+				// levels >= 2 (0.4% of elements): FULL classic inserts —
+				// progressive at every layer. Half-inserted elements must
+				// never be search-visible: a descent terminating on a
+				// node without lower-layer edges gives a degenerate ep
+				// (measured: out-degree-1 orphans, recall cliff)
+				upper2.push_back({(uint32_t)(base + i), levels[i]});
 				continue;
 			}
 			items.push_back(ApplyItem{(uint32_t)(base + i), PQ{}, {}});
 			eps_of.emplace_back();
-			kinds.push_back(levels[i] > 0 ? 1 : 0);
+			lvl_of.push_back(levels[i]);
 		}
-		if (items.empty())
-			continue;
-		// phase 1: upper inserts (layers >= 1, striped locks) — BEFORE any
-		// descent reads those layers (mutation/read separation, like v1)
-		{
+		if (!upper2.empty()) {
 			std::atomic<uint64_t> cursor{0};
 			auto worker = [&]() {
 				uint64_t j;
-				while ((j = cursor.fetch_add(1)) < items.size()) {
-					if (!kinds[j])
-						continue;
-					uint32_t q_id = items[j].q_id;
-					insert_at_upper(h, q_id, levels[q_id - base], true,
-					                &eps_of[j]);
-				}
+				while ((j = cursor.fetch_add(1)) < upper2.size())
+					insert_at(h, upper2[j].first, upper2[j].second, true);
 			};
 			std::vector<std::thread> ts;
 			int nt = std::max(1,
-			                  std::min<int>(nthreads, (int)items.size()));
+			                  std::min<int>(nthreads, (int)upper2.size()));
 			for (int t = 1; t < nt; t++)
 				ts.emplace_back(worker);
 			worker();
 			for (auto &t : ts)
 				t.join();
 		}
-		// phase 2: flat descents (read-only upper layers, parallel)
+		if (items.empty())
+			continue;
+		// levels <= 1: greedy descents through layers top..2 (read-only)
 		{
 			std::atomic<uint64_t> cursor{0};
 			auto worker = [&]() {
 				uint64_t j;
 				while ((j = cursor.fetch_add(1)) < items.size()) {
-					if (kinds[j])
-						continue;
 					uint32_t q_id = items[j].q_id;
 					double qn = h->metric == SDBV_METRIC_COSINE
 					                ? h->norms[q_id]
 					                : 0;
 					uint32_t ep;
 					double epd;
-					snapshot_descend_one(h, q_id, qn, &ep, &epd);
+					descend_to_layer2(h, q_id, qn, &ep, &epd);
 					eps_of[j].push(epd, ep);
 				}
 			};
@@ -3843,14 +3819,59 @@ int sdbv_hnsw_insert_batch_snapshot2(sdbv_hnsw *h, const float *pts,
 			for (auto &t : ts)
 				t.join();
 		}
-		// snapshot layer-0 searches (read-only, parallel)
+		// layer 1 (batched): search phase (snapshot, parallel), then the
+		// apply for every element with level >= 1
+		if (h->layers.size() >= 2) {
+			{
+				std::atomic<uint64_t> cursor{0};
+				auto worker = [&]() {
+					uint64_t j;
+					while ((j = cursor.fetch_add(1)) < items.size()) {
+						uint32_t q_id = items[j].q_id;
+						if (lvl_of[j] >= 1) {
+							snapshot_search_layer_eps(
+							    h, 1, q_id, eps_of[j], h->efc,
+							    items[j].w);
+							eps_of[j] = items[j].w; // cascade
+						} else {
+							PQ w;
+							snapshot_search_layer_eps(
+							    h, 1, q_id, eps_of[j], 1, w);
+							double dd;
+							uint32_t ii;
+							if (w.peek_first(&dd, &ii)) {
+								PQ ne2;
+								ne2.push(dd, ii);
+								eps_of[j] = std::move(ne2);
+							}
+						}
+					}
+				};
+				std::vector<std::thread> ts;
+				int nt = std::max(
+				    1, std::min<int>(nthreads, (int)items.size()));
+				for (int t = 1; t < nt; t++)
+					ts.emplace_back(worker);
+				worker();
+				for (auto &t : ts)
+					t.join();
+			}
+			std::vector<ApplyItem *> ins;
+			for (uint64_t j = 0; j < items.size(); j++)
+				if (lvl_of[j] >= 1)
+					ins.push_back(&items[j]);
+			if (!ins.empty())
+				batched_apply_host(h, 1, ins, nthreads);
+		}
+		// layer 0: every element
 		{
 			std::atomic<uint64_t> cursor{0};
 			auto worker = [&]() {
 				uint64_t j;
 				while ((j = cursor.fetch_add(1)) < items.size())
-					snapshot_search_eps(h, items[j].q_id, eps_of[j],
-					                    items[j].w);
+					snapshot_search_layer_eps(h, 0, items[j].q_id,
+					                          eps_of[j], h->efc,
+					                          items[j].w);
 			};
 			std::vector<std::thread> ts;
 			int nt = std::max(1,
@@ -3861,25 +3882,30 @@ int sdbv_hnsw_insert_batch_snapshot2(sdbv_hnsw *h, const float *pts,
 			for (auto &t : ts)
 				t.join();
 		}
-		batched_apply_l0_host(h, items, nthreads);
+		std::vector<ApplyItem *> all;
+		for (auto &it : items)
+			all.push_back(&it);
+		batched_apply_host(h, 0, all, nthreads);
 	}
 	hnsw_promote_ep(h);
 	h->dirty = true;
 	return SDBV_OK;
 }
 
-// GPU-accelerated chunked snapshot build, v2 (the
-// configs[2] 10M-row build). Identical algorithm and bit-identical results
-// to the host twin sdbv_hnsw_insert_batch_snapshot2 (same batched
-// schedule, same restated distance chains), with every RAM-bound phase on
-// the device:
-//  - layer-0 efc-searches: ONE k_hnsw_search<PADDED=1> launch per chunk
-//    (multi-ep seeding carries the insert path's layer-1 w windows);
-//  - phase-A neighbour selects: k_pair_mats (candidate pair-distance
-//    matrices) + k_heur_select (the exact heuristic, one wave per element);
-//  - phase-C prunes: the same two kernels over the over-degree nodes.
-// The host keeps the upper-layer inserts (tiny layers), the deterministic
-// phase-B edge appends, and the delta sync of the padded device adjacency.
+// GPU-accelerated chunked snapshot build, v3 (the
+// configs[2] 10M-row build). Bit-identical to the host twin
+// sdbv_hnsw_insert_batch_snapshot2 (same batched schedule at EVERY layer,
+// same restated distance chains), with all RAM-bound work on the device:
+//  - per layer, top-down: one ef=1 multi-query launch carries the greedy
+//    descents and one efc launch the insert searches (multi-ep seeding =
+//    the layer-above w cascade), against per-layer padded device
+//    adjacencies kept in sync by k_adj_scatter deltas;
+//  - neighbour selects and prunes: k_pair_mats + k_heur_select (the
+//    exact heuristic) at every layer.
+// The host keeps the deterministic phase-B edge appends and the queue
+// bookkeeping. The round-2 10M trace showed the HOST upper-layer inserts
+// of v2 were the remaining wall (~110 s per 1M elements); v3 moves them
+// here.
 int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
                                         uint64_t n, uint32_t chunk,
                                         int nthreads) {
@@ -3887,8 +3913,7 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	if (!h || !h->ctx || chunk == 0 || n == 0 || h->efc == 0 ||
 	    h->efc > HQ_EF_CAP)
 		return SDBV_ERR_BAD_ARG;
-	if (h->extend || h->m0 > 64 || h->m > 64 ||
-	    h->efc + 1 > HSEL_CAP)
+	if (h->extend || h->m0 > 64 || h->m > 64 || h->efc + 1 > HSEL_CAP)
 		// extend-candidates (or oversized select windows) stay on the
 		// host twin — same algorithm, host distances
 		return sdbv_hnsw_insert_batch_snapshot2(h, pts, n, chunk, nthreads);
@@ -3926,7 +3951,9 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	const uint64_t nelem = h->vecs.size() / h->d;
 	const uint32_t d = h->d;
 	const uint32_t efc = h->efc;
-	// ---- device state: vectors + padded adjacency (delta-synced) ----
+	const uint32_t nlayers = (uint32_t)h->layers.size();
+	const uint32_t top = nlayers - 1;
+	// ---- device state: vectors + per-layer padded adjacencies ----
 	for (void **p : {(void **)&h->rm_dev, (void **)&h->norms_dev,
 	                 (void **)&h->adj_dev, (void **)&h->deg_dev})
 		if (*p) {
@@ -3943,18 +3970,39 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		                         nelem * sizeof(double),
 		                         hipMemcpyHostToDevice));
 	}
-	h->adj_stride = ((h->m0 + 8) + 7) & ~7u;
-	HIP_CHECK(ctx, hipMalloc(&h->adj_dev,
-	                         nelem * h->adj_stride * sizeof(uint32_t)));
-	HIP_CHECK(ctx, hipMalloc(&h->deg_dev, nelem * sizeof(uint32_t)));
-	HIP_CHECK(ctx, hipMemset(h->deg_dev, 0, nelem * sizeof(uint32_t)));
+	// layer 0 lives in h->adj_dev/deg_dev; layers >= 1 in build-local
+	// arrays (all freed at the end; finalize re-exports the CSR)
+	// device adjacency for layers 0 and 1 only: levels >= 2 stay on the
+	// classic host path (tiny skeleton layers, progressive for quality)
+	const uint32_t ndev_layers = std::min<uint32_t>(nlayers, 2);
+	std::vector<uint32_t *> adjL(ndev_layers, nullptr),
+	    degL(ndev_layers, nullptr);
+	std::vector<uint32_t> strideL(ndev_layers);
+	std::vector<uint8_t> has_members(ndev_layers, 0);
+	h->dtrack.clear();
+	h->dtrack.resize(ndev_layers);
+	for (uint32_t l = 0; l < ndev_layers; l++) {
+		strideL[l] = (((l == 0 ? h->m0 : h->m) + 8) + 7) & ~7u;
+		HIP_CHECK(ctx, hipMalloc(&adjL[l],
+		                         nelem * (uint64_t)strideL[l] *
+		                             sizeof(uint32_t)));
+		HIP_CHECK(ctx, hipMalloc(&degL[l], nelem * sizeof(uint32_t)));
+		HIP_CHECK(ctx, hipMemset(degL[l], 0, nelem * sizeof(uint32_t)));
+		h->dtrack[l] = std::make_unique<sdbv_hnsw::DirtyTrack>();
+		h->dtrack[l]->flag.assign(nelem, 0);
+		h->dtrack[l]->list.assign(nelem, 0);
+		for (uint64_t i = 0; i < base; i++) { // pre-existing graph
+			if (h->layers[l].has((uint32_t)i)) {
+				h->mark_dirty(l, (uint32_t)i);
+				has_members[l] = 1;
+			}
+		}
+	}
+	h->adj_dev = adjL[0];
+	h->deg_dev = degL[0];
+	h->adj_stride = strideL[0];
 	h->adj_nodes = nelem;
 	h->dev_rows = nelem;
-	h->l0_dirty.assign(nelem, 0);
-	h->l0_dirty_list.assign(nelem, 0);
-	h->l0_dirty_n.store(0, std::memory_order_relaxed);
-	for (uint64_t i = 0; i < base; i++)
-		h->mark_l0_dirty((uint32_t)i); // first sync: pre-existing graph
 	const uint64_t vwords = (nelem + 31) / 32;
 	{
 		uint64_t need = (uint64_t)chunk * vwords * sizeof(uint32_t);
@@ -3970,7 +4018,7 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	// ---- per-launch scratch (capacity = chunk, reused) ----
 	float *Qd = nullptr;
 	double *qnd = nullptr, *epdd = nullptr, *outd = nullptr;
-	uint32_t *epsd = nullptr, *epoffd = nullptr, *qrowsd = nullptr;
+	uint32_t *epsd = nullptr, *epoffd = nullptr;
 	uint32_t *outr = nullptr, *outc = nullptr, *outf = nullptr;
 	uint32_t *upd_ids_dev = nullptr, *upd_deg_dev = nullptr,
 	         *upd_edges_dev = nullptr;
@@ -3981,18 +4029,23 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	uint64_t upd_cap = 0, lists_cap = 0, mats_cap = 0, sel_cap = 0,
 	         loff_cap = 0;
 	auto cleanup = [&] {
+		adjL[0] = nullptr; // owned by h->adj_dev (hnsw_free_device_state)
+		degL[0] = nullptr;
+		for (uint32_t l = 1; l < ndev_layers; l++) {
+			if (adjL[l])
+				(void)hipFree(adjL[l]);
+			if (degL[l])
+				(void)hipFree(degL[l]);
+		}
 		for (void *p : {(void *)Qd, (void *)qnd, (void *)epdd, (void *)epsd,
-		                (void *)epoffd, (void *)qrowsd, (void *)outr,
-		                (void *)outd, (void *)outc, (void *)outf,
-		                (void *)upd_ids_dev, (void *)upd_deg_dev,
-		                (void *)upd_edges_dev, (void *)listsd,
-		                (void *)loffd, (void *)seld, (void *)selcntd,
-		                (void *)moffd, (void *)matsd})
+		                (void *)epoffd, (void *)outr, (void *)outd,
+		                (void *)outc, (void *)outf, (void *)upd_ids_dev,
+		                (void *)upd_deg_dev, (void *)upd_edges_dev,
+		                (void *)listsd, (void *)loffd, (void *)seld,
+		                (void *)selcntd, (void *)moffd, (void *)matsd})
 			if (p)
 				(void)hipFree(p);
-		h->l0_dirty.clear();
-		h->l0_dirty_list.clear();
-		h->l0_dirty_list.shrink_to_fit();
+		h->dtrack.clear();
 	};
 #define BGPU_CHECK(call)                                                     \
 	do {                                                                     \
@@ -4010,7 +4063,6 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	BGPU_CHECK(hipMalloc(&epsd,
 	                     (uint64_t)chunk * (efc + 1) * sizeof(uint32_t)));
 	BGPU_CHECK(hipMalloc(&epoffd, (chunk + 1) * sizeof(uint32_t)));
-	BGPU_CHECK(hipMalloc(&qrowsd, chunk * sizeof(uint32_t)));
 	BGPU_CHECK(hipMalloc(&outr, (uint64_t)chunk * efc * sizeof(uint32_t)));
 	BGPU_CHECK(hipMalloc(&outd, (uint64_t)chunk * efc * sizeof(double)));
 	BGPU_CHECK(hipMalloc(&outc, chunk * sizeof(uint32_t)));
@@ -4018,59 +4070,63 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	// host staging
 	std::vector<float> Qh;
 	std::vector<double> qnh, epdh;
-	std::vector<uint32_t> epsh, epoffh, qrowsh, outch, outfh;
+	std::vector<uint32_t> epsh, epoffh, outch, outfh, outrh;
+	std::vector<double> outdh;
 	std::vector<uint32_t> upd_ids, upd_deg, upd_edges;
 	std::vector<uint32_t> listsh, loffh, selh, selcnth;
 	std::vector<uint64_t> moffh;
 
-	double t_upper = 0, t_sync = 0, t_desc = 0, t_kern = 0, t_selA = 0,
-	       t_link = 0, t_selC = 0;
+	double t_epinit = 0, t_sync = 0, t_kern = 0, t_selA = 0, t_link = 0,
+	       t_selC = 0, t_upperk = 0;
 	auto now = [] { return std::chrono::steady_clock::now(); };
 	auto secs = [](std::chrono::steady_clock::time_point a,
 	               std::chrono::steady_clock::time_point b) {
 		return std::chrono::duration<double>(b - a).count();
 	};
 
-	auto sync_adj = [&]() -> int {
+	auto sync_adj = [&](uint32_t l) -> int {
+		auto &D = *h->dtrack[l];
 		upd_ids.clear();
 		upd_deg.clear();
 		uint32_t maxdeg = 0;
-		auto &l0 = h->layers[0];
-		// dirty LIST, not a full flag scan (the scan was O(nelem) per
-		// chunk -> quadratic across the build at 10M rows)
-		uint32_t nd = h->l0_dirty_n.load(std::memory_order_relaxed);
+		auto &L = h->layers[l];
+		uint32_t nd = D.n.load(std::memory_order_relaxed);
 		for (uint32_t li = 0; li < nd; li++) {
-			uint32_t i = h->l0_dirty_list[li];
-			uint32_t dg = (uint32_t)l0.edges[i].size();
+			uint32_t i = D.list[li];
+			uint32_t dg = (uint32_t)L.edges[i].size();
 			maxdeg = std::max(maxdeg, dg);
 			upd_ids.push_back(i);
 			upd_deg.push_back(dg);
 		}
-		if (maxdeg > h->adj_stride) {
+		if (maxdeg > strideL[l]) {
 			uint32_t ns = (maxdeg + 8 + 7) & ~7u;
-			(void)hipFree(h->adj_dev);
-			h->adj_dev = nullptr;
-			BGPU_CHECK(hipMalloc(&h->adj_dev,
+			(void)hipFree(adjL[l]);
+			adjL[l] = nullptr;
+			BGPU_CHECK(hipMalloc(&adjL[l],
 			                     nelem * (uint64_t)ns * sizeof(uint32_t)));
-			h->adj_stride = ns;
+			strideL[l] = ns;
+			if (l == 0) {
+				h->adj_dev = adjL[0];
+				h->adj_stride = ns;
+			}
 			upd_ids.clear();
 			upd_deg.clear();
 			for (uint64_t i = 0; i < nelem; i++)
-				if (!l0.edges[i].empty() || h->l0_dirty[i]) {
+				if (!L.edges[i].empty() || D.flag[i]) {
 					upd_ids.push_back((uint32_t)i);
-					upd_deg.push_back((uint32_t)l0.edges[i].size());
+					upd_deg.push_back((uint32_t)L.edges[i].size());
 				}
 		}
 		for (uint32_t li = 0; li < nd; li++)
-			h->l0_dirty[h->l0_dirty_list[li]] = 0;
-		h->l0_dirty_n.store(0, std::memory_order_relaxed);
+			D.flag[D.list[li]] = 0;
+		D.n.store(0, std::memory_order_relaxed);
 		uint64_t cnt = upd_ids.size();
 		if (cnt == 0)
 			return SDBV_OK;
-		const uint32_t stride = h->adj_stride;
+		const uint32_t stride = strideL[l];
 		upd_edges.assign(cnt * stride, 0);
 		for (uint64_t i = 0; i < cnt; i++) {
-			const auto &e = h->layers[0].edges[upd_ids[i]];
+			const auto &e = L.edges[upd_ids[i]];
 			std::copy(e.begin(), e.end(), upd_edges.begin() + i * stride);
 		}
 		if (upd_cap < (uint64_t)cnt * stride) {
@@ -4099,19 +4155,14 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		hipLaunchKernelGGL(k_adj_scatter,
 		                   dim3((uint32_t)((threads + 255) / 256)), dim3(256),
 		                   0, ctx->stream, upd_ids_dev, upd_deg_dev,
-		                   upd_edges_dev, (uint32_t)cnt, stride, h->adj_dev,
-		                   h->deg_dev);
+		                   upd_edges_dev, (uint32_t)cnt, stride, adjL[l],
+		                   degL[l]);
 		return SDBV_OK;
 	};
 
-	// run phase-A/phase-C selects on the device for a group of lists
-	// (host-assembled for phase C; device-assembled upstream for phase A
-	// is folded into the same host path for simplicity — list data is
-	// small). Returns selections via selh/selcnth.
 	auto device_select = [&](const std::vector<uint32_t> &lists,
 	                         const std::vector<uint32_t> &loff,
 	                         uint32_t nlists, uint32_t m_max) -> int {
-		// matrices offsets
 		moffh.resize(nlists + 1);
 		uint64_t mo = 0;
 		for (uint32_t i = 0; i < nlists; i++) {
@@ -4185,96 +4236,27 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		return SDBV_OK;
 	};
 
-	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
-		uint64_t c1 = std::min(n, c0 + chunk);
-		auto tp0 = now();
-		std::vector<ApplyItem> items;
-		std::vector<PQ> eps_of;
-		std::vector<uint32_t> kinds;
-		for (uint64_t i = c0; i < c1; i++) {
-			if (h->enter_point < 0) {
-				insert_at(h, (uint32_t)(base + i), levels[i], false);
-				continue;
-			}
-			items.push_back(ApplyItem{(uint32_t)(base + i), PQ{}, {}});
-			eps_of.emplace_back();
-			kinds.push_back(levels[i] > 0 ? 1 : 0);
-		}
-		if (items.empty())
-			continue;
-		const uint32_t b = (uint32_t)items.size();
-		// phase 1: upper inserts (layers >= 1)
-		{
-			std::atomic<uint64_t> cursor{0};
-			auto worker = [&]() {
-				uint64_t j;
-				while ((j = cursor.fetch_add(1)) < b) {
-					if (!kinds[j])
-						continue;
-					uint32_t q_id = items[j].q_id;
-					insert_at_upper(h, q_id, levels[q_id - base], true,
-					                &eps_of[j]);
-				}
-			};
-			std::vector<std::thread> ts;
-			int nt = std::max(1, std::min<int>(nthreads, (int)b));
-			for (int t = 1; t < nt; t++)
-				ts.emplace_back(worker);
-			worker();
-			for (auto &t : ts)
-				t.join();
-		}
-		auto tp1 = now();
-		t_upper += secs(tp0, tp1);
-		// phase 2: flat descents
-		{
-			std::atomic<uint64_t> cursor{0};
-			auto worker = [&]() {
-				uint64_t j;
-				while ((j = cursor.fetch_add(1)) < b) {
-					if (kinds[j])
-						continue;
-					uint32_t q_id = items[j].q_id;
-					double qn = h->metric == SDBV_METRIC_COSINE
-					                ? h->norms[q_id]
-					                : 0;
-					uint32_t ep;
-					double epd;
-					snapshot_descend_one(h, q_id, qn, &ep, &epd);
-					eps_of[j].push(epd, ep);
-				}
-			};
-			std::vector<std::thread> ts;
-			int nt = std::max(1, std::min<int>(nthreads, (int)b));
-			for (int t = 1; t < nt; t++)
-				ts.emplace_back(worker);
-			worker();
-			for (auto &t : ts)
-				t.join();
-		}
-		auto tp2 = now();
-		t_desc += secs(tp1, tp2);
-		// device graph := graph as of now (incl. this chunk's upper edges)
-		int rc = sync_adj();
-		if (rc)
-			return rc;
-		auto tp3 = now();
-		t_sync += secs(tp2, tp3);
-		// layer-0 searches: one multi-ep persistent-kernel launch
+	// launch the persistent-kernel search for a subset of chunk items at
+	// layer l: queries/norms/eps packed from item indices; results land in
+	// outr/outd/outc/outf at the SUBSET ordinal positions.
+	std::vector<ApplyItem> items;
+	std::vector<PQ> eps_of;
+	std::vector<uint32_t> lvl_of;
+	auto launch_search = [&](uint32_t l, const std::vector<uint32_t> &sub,
+	                         uint32_t k, uint32_t ef) -> int {
+		uint32_t b = (uint32_t)sub.size();
 		Qh.resize((uint64_t)b * d);
 		qnh.resize(b);
-		qrowsh.resize(b);
 		epoffh.resize(b + 1);
 		epsh.clear();
 		epdh.clear();
 		for (uint32_t j = 0; j < b; j++) {
-			uint32_t q_id = items[j].q_id;
-			qrowsh[j] = q_id;
+			uint32_t q_id = items[sub[j]].q_id;
 			std::memcpy(Qh.data() + (uint64_t)j * d, vec(h, q_id),
 			            d * sizeof(float));
 			qnh[j] = h->metric == SDBV_METRIC_COSINE ? h->norms[q_id] : 0;
 			epoffh[j] = (uint32_t)epsh.size();
-			for (auto &e : eps_of[j].to_vec()) {
+			for (auto &e : eps_of[sub[j]].to_vec()) {
 				epsh.push_back(e.second);
 				epdh.push_back(e.first);
 			}
@@ -4294,82 +4276,88 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 		BGPU_CHECK(hipMemcpyAsync(epoffd, epoffh.data(),
 		                          (b + 1) * sizeof(uint32_t),
 		                          hipMemcpyHostToDevice, ctx->stream));
-		BGPU_CHECK(hipMemcpyAsync(qrowsd, qrowsh.data(),
-		                          b * sizeof(uint32_t),
-		                          hipMemcpyHostToDevice, ctx->stream));
 		BGPU_CHECK(hipMemsetAsync(h->vis_dev, 0,
 		                          (uint64_t)b * vwords * sizeof(uint32_t),
 		                          ctx->stream));
 		hipLaunchKernelGGL(k_hnsw_search<1>, dim3(b), dim3(64), 0,
 		                   ctx->stream, h->rm_dev, h->norms_dev, d,
-		                   (int)h->metric, h->deg_dev, h->adj_dev,
-		                   h->adj_stride, Qd, qnd, epsd, epdd, epoffd,
-		                   h->vis_dev, vwords, efc, efc, outr, outd, outc,
-		                   outf);
+		                   (int)h->metric, degL[l], adjL[l], strideL[l], Qd,
+		                   qnd, epsd, epdd, epoffd, h->vis_dev, vwords, k,
+		                   ef, outr, outd, outc, outf);
 		outch.resize(b);
 		outfh.resize(b);
+		outrh.resize((uint64_t)b * k);
+		outdh.resize((uint64_t)b * k);
 		BGPU_CHECK(hipMemcpyAsync(outch.data(), outc, b * sizeof(uint32_t),
 		                          hipMemcpyDeviceToHost, ctx->stream));
 		BGPU_CHECK(hipMemcpyAsync(outfh.data(), outf, b * sizeof(uint32_t),
 		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(outrh.data(), outr,
+		                          (uint64_t)b * k * sizeof(uint32_t),
+		                          hipMemcpyDeviceToHost, ctx->stream));
+		BGPU_CHECK(hipMemcpyAsync(outdh.data(), outd,
+		                          (uint64_t)b * k * sizeof(double),
+		                          hipMemcpyDeviceToHost, ctx->stream));
 		BGPU_CHECK(hipStreamSynchronize(ctx->stream));
 		BGPU_CHECK(hipGetLastError());
-		auto tp4 = now();
-		t_kern += secs(tp3, tp4);
-		// phase A on the device: lists = [q, w rows] per element. List
-		// data is assembled host-side from outr (copied in bulk).
-		{
-			std::vector<uint32_t> outrh((uint64_t)b * efc);
-			BGPU_CHECK(hipMemcpyAsync(outrh.data(), outr,
-			                          outrh.size() * sizeof(uint32_t),
-			                          hipMemcpyDeviceToHost, ctx->stream));
-			BGPU_CHECK(hipStreamSynchronize(ctx->stream));
-			listsh.clear();
-			loffh.assign(b + 1, 0);
-			for (uint32_t j = 0; j < b; j++) {
-				loffh[j] = (uint32_t)listsh.size();
-				listsh.push_back(items[j].q_id);
-				if (!(outfh[j] & HQ_FLAG_OVERFLOW))
-					for (uint32_t i = 0; i < outch[j]; i++)
-						listsh.push_back(outrh[(uint64_t)j * efc + i]);
-			}
-			loffh[b] = (uint32_t)listsh.size();
-			rc = device_select(listsh, loffh, b, h->m0);
-			if (rc)
-				return rc;
-			for (uint32_t j = 0; j < b; j++) {
-				if (outfh[j] & HQ_FLAG_OVERFLOW) {
-					// exact host fallback (rare): snapshot search + select
-					PQ w;
-					snapshot_search_eps(h, items[j].q_id, eps_of[j], w);
-					const float *q_pt = vec(h, items[j].q_id);
-					double qn = h->metric == SDBV_METRIC_COSINE
-					                ? h->norms[items[j].q_id]
-					                : 0;
-					select_neighbors(h, h->layers[0], items[j].q_id,
-					                 q_pt, qn, std::move(w),
-					                 items[j].neighbors, false);
-					continue;
-				}
-				items[j].neighbors.assign(
-				    selh.begin() + (uint64_t)j * h->m0,
-				    selh.begin() + (uint64_t)j * h->m0 + selcnth[j]);
-			}
+		return SDBV_OK;
+	};
+
+	// device select + phase B/C for the insert set at layer l. The w
+	// windows were just searched into outrh/outdh (subset order).
+	auto apply_layer = [&](uint32_t l, const std::vector<uint32_t> &ins)
+	    -> int {
+		uint32_t b = (uint32_t)ins.size();
+		uint32_t m_max = h->layers[l].m_max;
+		listsh.clear();
+		loffh.assign(b + 1, 0);
+		for (uint32_t j = 0; j < b; j++) {
+			loffh[j] = (uint32_t)listsh.size();
+			listsh.push_back(items[ins[j]].q_id);
+			if (!(outfh[j] & HQ_FLAG_OVERFLOW))
+				for (uint32_t i = 0; i < outch[j]; i++)
+					listsh.push_back(outrh[(uint64_t)j * efc + i]);
 		}
-		auto tp5 = now();
-		t_selA += secs(tp4, tp5);
-		// phase B (host, deterministic) + phase C on the device
-		auto overfull = batched_apply_phaseB(h, items);
-		auto tp6 = now();
-		t_link += secs(tp5, tp6);
+		loffh[b] = (uint32_t)listsh.size();
+		int rc = device_select(listsh, loffh, b, m_max);
+		if (rc)
+			return rc;
+		for (uint32_t j = 0; j < b; j++) {
+			ApplyItem &it = items[ins[j]];
+			if (outfh[j] & HQ_FLAG_OVERFLOW) {
+				// exact host fallback (rare)
+				PQ w;
+				snapshot_search_layer_eps(h, l, it.q_id, eps_of[ins[j]],
+				                          efc, w);
+				eps_of[ins[j]] = w;
+				const float *q_pt = vec(h, it.q_id);
+				double qn = h->metric == SDBV_METRIC_COSINE
+				                ? h->norms[it.q_id]
+				                : 0;
+				it.neighbors.clear();
+				select_neighbors(h, h->layers[l], it.q_id, q_pt, qn,
+				                 std::move(w), it.neighbors, false);
+				continue;
+			}
+			it.neighbors.assign(
+			    selh.begin() + (uint64_t)j * m_max,
+			    selh.begin() + (uint64_t)j * m_max + selcnth[j]);
+		}
+		std::vector<ApplyItem *> ptrs;
+		ptrs.reserve(b);
+		for (uint32_t j = 0; j < b; j++)
+			ptrs.push_back(&items[ins[j]]);
+		auto tb0 = now();
+		auto overfull = batched_apply_phaseB(h, l, ptrs);
+		t_link += secs(tb0, now());
 		if (!overfull.empty()) {
 			listsh.clear();
 			loffh.clear();
 			std::vector<uint32_t> host_prunes, dev_nodes;
 			for (uint32_t e : overfull) {
-				const auto &conn = h->layers[0].edges[e];
+				const auto &conn = h->layers[l].edges[e];
 				if (1 + conn.size() > HSEL_CAP) {
-					host_prunes.push_back(e); // oversized: host prune
+					host_prunes.push_back(e);
 					continue;
 				}
 				dev_nodes.push_back(e);
@@ -4385,39 +4373,224 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 			loffh.push_back((uint32_t)listsh.size());
 			if (!dev_nodes.empty()) {
 				rc = device_select(listsh, loffh,
-				                   (uint32_t)dev_nodes.size(), h->m0);
+				                   (uint32_t)dev_nodes.size(), m_max);
 				if (rc)
 					return rc;
-				const bool track = !h->l0_dirty.empty();
 				for (uint32_t li = 0; li < dev_nodes.size(); li++) {
 					uint32_t e = dev_nodes[li];
-					auto &ee = h->layers[0].edges[e];
-					ee.assign(selh.begin() + (uint64_t)li * h->m0,
-					          selh.begin() + (uint64_t)li * h->m0 +
+					auto &ee = h->layers[l].edges[e];
+					ee.assign(selh.begin() + (uint64_t)li * m_max,
+					          selh.begin() + (uint64_t)li * m_max +
 					              selcnth[li]);
-					if (track)
-						h->mark_l0_dirty(e);
+					h->mark_dirty(l, e);
 				}
 			}
 			if (!host_prunes.empty())
-				batched_apply_phaseC_host(h, host_prunes, nthreads);
+				batched_apply_phaseC_host(h, l, host_prunes, nthreads);
 		}
-		t_selC += secs(tp6, now());
+		return SDBV_OK;
+	};
+
+	std::vector<std::pair<uint32_t, uint32_t>> upper2; // (q_id, level)
+	// warm-up: see the host twin — the first elements of an empty graph
+	// insert classically so batching starts on a connected graph
+	const uint64_t warm_until = h->enter_point < 0 ? base + 64 : 0;
+	for (uint64_t c0 = 0; c0 < n; c0 += chunk) {
+		uint64_t c1 = std::min(n, c0 + chunk);
+		auto tp0 = now();
+		items.clear();
+		eps_of.clear();
+		lvl_of.clear();
+		upper2.clear();
+		for (uint64_t i = c0; i < c1; i++) {
+			if (h->enter_point < 0 || base + i < warm_until) {
+				insert_at(h, (uint32_t)(base + i), levels[i], false);
+				for (uint32_t l = 0;
+				     l < ndev_layers && l <= levels[i]; l++) {
+					h->mark_dirty(l, (uint32_t)(base + i));
+					has_members[l] = 1;
+				}
+				continue;
+			}
+			if (levels[i] >= 2) {
+				// levels >= 2 (0.4% of elements): FULL classic host
+				// inserts — progressive at every layer (half-inserted
+				// elements must never be search-visible; their classic
+				// applies mark the device adjacency via
+				// layer_insert_apply's dirty hooks)
+				upper2.push_back({(uint32_t)(base + i), levels[i]});
+				continue;
+			}
+			items.push_back(ApplyItem{(uint32_t)(base + i), PQ{}, {}});
+			eps_of.emplace_back();
+			lvl_of.push_back(levels[i]);
+		}
+		if (!upper2.empty()) {
+			std::atomic<uint64_t> cursor{0};
+			auto worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < upper2.size())
+					insert_at(h, upper2[j].first, upper2[j].second, true);
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1,
+			                  std::min<int>(nthreads, (int)upper2.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(worker);
+			worker();
+			for (auto &t : ts)
+				t.join();
+			for (uint32_t l = 0; l < ndev_layers; l++)
+				has_members[l] = 1; // classic inserts joined every layer
+		}
+		if (items.empty())
+			continue;
+		// levels <= 1: host greedy descents through layers top..2
+		{
+			std::atomic<uint64_t> cursor{0};
+			auto worker = [&]() {
+				uint64_t j;
+				while ((j = cursor.fetch_add(1)) < items.size()) {
+					uint32_t q_id = items[j].q_id;
+					double qn = h->metric == SDBV_METRIC_COSINE
+					                ? h->norms[q_id]
+					                : 0;
+					uint32_t ep;
+					double epd;
+					descend_to_layer2(h, q_id, qn, &ep, &epd);
+					eps_of[j].push(epd, ep);
+				}
+			};
+			std::vector<std::thread> ts;
+			int nt = std::max(1,
+			                  std::min<int>(nthreads, (int)items.size()));
+			for (int t = 1; t < nt; t++)
+				ts.emplace_back(worker);
+			worker();
+			for (auto &t : ts)
+				t.join();
+		}
+		auto tp1 = now();
+		t_epinit += secs(tp0, tp1);
+		// layer 1 (device): ef=1 descents for level-0 elements, efc
+		// searches + batched apply for level >= 1
+		if (nlayers >= 2) {
+			const uint32_t l = 1;
+			std::vector<uint32_t> desc, ins;
+			for (uint32_t j = 0; j < (uint32_t)items.size(); j++)
+				(lvl_of[j] >= l ? ins : desc).push_back(j);
+			auto tu0 = now();
+			int rc = sync_adj(l);
+			if (rc)
+				return rc;
+			t_sync += secs(tu0, now());
+			if (has_members[l] && !desc.empty()) {
+				auto tk = now();
+				rc = launch_search(l, desc, 1, 1);
+				if (rc)
+					return rc;
+				t_upperk += secs(tk, now());
+				for (uint32_t j = 0; j < desc.size(); j++) {
+					if (outch[j] == 0)
+						continue; // keep previous eps
+					PQ ne2;
+					ne2.push(outdh[(uint64_t)j * 1], outrh[j]);
+					eps_of[desc[j]] = std::move(ne2);
+				}
+			}
+			if (!ins.empty()) {
+				if (has_members[l]) {
+					auto tk = now();
+					rc = launch_search(l, ins, efc, efc);
+					if (rc)
+						return rc;
+					t_upperk += secs(tk, now());
+					// cascade eps = w
+					for (uint32_t j = 0; j < ins.size(); j++) {
+						if (outfh[j] & HQ_FLAG_OVERFLOW)
+							continue; // handled in apply_layer
+						PQ ne2;
+						for (uint32_t i = 0; i < outch[j]; i++)
+							ne2.push(outdh[(uint64_t)j * efc + i],
+							         outrh[(uint64_t)j * efc + i]);
+						eps_of[ins[j]] = std::move(ne2);
+					}
+				} else {
+					// empty layer: search degenerates to w = eps
+					outch.assign(ins.size(), 0);
+					outfh.assign(ins.size(), 0);
+					outrh.resize(ins.size() * (uint64_t)efc);
+					outdh.resize(ins.size() * (uint64_t)efc);
+					for (uint32_t j = 0; j < ins.size(); j++) {
+						auto v = eps_of[ins[j]].to_vec();
+						outch[j] = (uint32_t)std::min<size_t>(v.size(),
+						                                      efc);
+						for (uint32_t i = 0; i < outch[j]; i++) {
+							outrh[(uint64_t)j * efc + i] = v[i].second;
+							outdh[(uint64_t)j * efc + i] = v[i].first;
+						}
+					}
+				}
+				auto ta = now();
+				rc = apply_layer(l, ins);
+				if (rc)
+					return rc;
+				t_selA += secs(ta, now());
+				has_members[l] = 1;
+			}
+		}
+		// layer 0: every item
+		{
+			std::vector<uint32_t> all(items.size());
+			for (uint32_t j = 0; j < (uint32_t)items.size(); j++)
+				all[j] = j;
+			auto tu0 = now();
+			int rc = sync_adj(0);
+			if (rc)
+				return rc;
+			t_sync += secs(tu0, now());
+			auto tk = now();
+			if (has_members[0]) {
+				rc = launch_search(0, all, efc, efc);
+				if (rc)
+					return rc;
+			} else {
+				outch.assign(all.size(), 0);
+				outfh.assign(all.size(), 0);
+				outrh.resize(all.size() * (uint64_t)efc);
+				outdh.resize(all.size() * (uint64_t)efc);
+				for (uint32_t j = 0; j < all.size(); j++) {
+					auto v = eps_of[j].to_vec();
+					outch[j] = (uint32_t)std::min<size_t>(v.size(), efc);
+					for (uint32_t i = 0; i < outch[j]; i++) {
+						outrh[(uint64_t)j * efc + i] = v[i].second;
+						outdh[(uint64_t)j * efc + i] = v[i].first;
+					}
+				}
+			}
+			t_kern += secs(tk, now());
+			auto ta = now();
+			rc = apply_layer(0, all);
+			if (rc)
+				return rc;
+			t_selA += secs(ta, now());
+			has_members[0] = 1;
+		}
 		if (((c0 / chunk) & 63) == 63)
 			fprintf(stderr,
-			        "[sdbv build_gpu2] %llu/%llu upper=%.1f desc=%.1f "
-			        "sync=%.1f kern=%.1f selA=%.1f link=%.1f selC=%.1f\n",
-			        (unsigned long long)c1, (unsigned long long)n, t_upper,
-			        t_desc, t_sync, t_kern, t_selA, t_link, t_selC);
+			        "[sdbv build_gpu3] %llu/%llu epinit=%.1f sync=%.1f "
+			        "upperk=%.1f kern0=%.1f apply=%.1f link=%.1f\n",
+			        (unsigned long long)c1, (unsigned long long)n,
+			        t_epinit, t_sync, t_upperk, t_kern, t_selA, t_link);
 	}
 #undef BGPU_CHECK
 	hnsw_promote_ep(h);
 	fprintf(stderr,
-	        "[sdbv build_gpu2] n=%llu chunk=%u phases: upper=%.1fs "
-	        "descend=%.1fs sync=%.1fs kernel=%.1fs selA=%.1fs link=%.1fs "
-	        "selC=%.1fs\n",
-	        (unsigned long long)n, chunk, t_upper, t_desc, t_sync, t_kern,
-	        t_selA, t_link, t_selC);
+	        "[sdbv build_gpu3] n=%llu chunk=%u phases: epinit=%.1fs "
+	        "sync=%.1fs upper-kernels=%.1fs kernel0=%.1fs "
+	        "apply(selA+selC)=%.1fs link=%.1fs\n",
+	        (unsigned long long)n, chunk, t_epinit, t_sync, t_upperk,
+	        t_kern, t_selA, t_link);
 	cleanup();
 	h->dirty = true;
 	return SDBV_OK;
