@@ -305,3 +305,37 @@ def test_snapshot2_keep_flag_and_768d_quality():
     r_s2k = build_recall("snapshot2", keep=True)
     assert r_s2 >= r_par - 0.08, (r_s2, r_par)
     assert r_s2k >= r_par - 0.12, (r_s2k, r_par)
+
+
+def test_snapshot2_deterministic_and_incremental():
+    """snapshot2 at nthreads=1 is fully deterministic (same seed, same
+    data -> bit-identical graphs), and two incremental calls equal...
+    themselves deterministically (the GPU twin's incremental test mirrors
+    this on hardware)."""
+    import numpy as np
+    import surrealdb_amd
+    d, n = 32, 1500
+    rows = oracle.gen_f32(0xFEED, 0, n, d)
+
+    def build(split):
+        h = surrealdb_amd.hnsw_create_host(d, metric="cosine", m=8, m0=16,
+                                           efc=60, seed=0x77)
+        if split:
+            h.insert_batch_snapshot2(rows[:900], chunk=128, nthreads=1)
+            h.insert_batch_snapshot2(rows[900:], chunk=128, nthreads=1)
+        else:
+            h.insert_batch_snapshot2(rows, chunk=128, nthreads=1)
+        return h
+
+    a, b = build(False), build(False)
+    ca, cb = a.l0_csr(), b.l0_csr()
+    assert np.array_equal(ca[0], cb[0]) and np.array_equal(ca[1], cb[1])
+    c, e = build(True), build(True)
+    cc, ce = c.l0_csr(), e.l0_csr()
+    assert np.array_equal(cc[0], ce[0]) and np.array_equal(cc[1], ce[1])
+    for q in oracle.gen_f32(0xB, 0, 5, d):
+        i1, d1 = a.knn_search_host(q, 10, 40)
+        i2, d2 = b.knn_search_host(q, 10, 40)
+        assert np.array_equal(i1, i2) and np.array_equal(d1, d2)
+    for h in (a, b, c, e):
+        h.destroy()
