@@ -3123,11 +3123,6 @@ static void snapshot_search_layer_eps(sdbv_hnsw *h, uint32_t layer_idx,
 	w_out = std::move(w);
 }
 
-static void snapshot_search_eps(sdbv_hnsw *h, uint32_t q_id,
-                                const PQ &eps_in, PQ &w_out) {
-	snapshot_search_layer_eps(h, 0, q_id, eps_in, h->efc, w_out);
-}
-
 // Greedy descent through layers top..2 only (ef=1; the layer-1 hop is the
 // batched builds' device work).
 static void descend_to_layer2(sdbv_hnsw *h, uint32_t q_id, double q_norm,
@@ -3952,7 +3947,6 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	const uint32_t d = h->d;
 	const uint32_t efc = h->efc;
 	const uint32_t nlayers = (uint32_t)h->layers.size();
-	const uint32_t top = nlayers - 1;
 	// ---- device state: vectors + per-layer padded adjacencies ----
 	for (void **p : {(void **)&h->rm_dev, (void **)&h->norms_dev,
 	                 (void **)&h->adj_dev, (void **)&h->deg_dev})
@@ -4077,7 +4071,7 @@ int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *h, const float *pts,
 	std::vector<uint64_t> moffh;
 
 	double t_epinit = 0, t_sync = 0, t_kern = 0, t_selA = 0, t_link = 0,
-	       t_selC = 0, t_upperk = 0;
+	       t_upperk = 0;
 	auto now = [] { return std::chrono::steady_clock::now(); };
 	auto secs = [](std::chrono::steady_clock::time_point a,
 	               std::chrono::steady_clock::time_point b) {
