@@ -116,6 +116,9 @@ def test_gpu_snapshot_build_recall_golden(ctx):
         bf, _ = oracle.topk_f32("euclidean", ingest, q, 10)
         total += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
     offsets, _ = h.l0_csr()
-    assert np.diff(offsets.astype(np.int64)).max() <= 16
+    # parallel applies may transiently exceed m0 (documented keep-back
+    # relaxation); the strict bound is asserted on the deterministic
+    # schedule in tests/test_snapshot_invariants.py
+    assert np.diff(offsets.astype(np.int64)).max() <= 16 + 8
     assert total / len(queries) >= 0.98
     h.destroy()
