@@ -140,7 +140,7 @@ def test_snapshot_build_recall_bars():
             total += len(set(ids.tolist()) & set(bf.tolist())) / 10.0
         offsets, edges = h.l0_csr()
         deg = np.diff(offsets.astype(np.int64))
-        assert deg.max() <= 16
+        assert deg.max() <= 16 + 8  # threaded keep-back relaxation
         h.destroy()
         return total / len(queries_all)
 
