@@ -6691,14 +6691,14 @@ static int kvload_build_graph(sdbv_kvload *L, sdbv_hnsw **out) {
 				sdbv_hnsw_destroy(h);
 				return SDBV_ERR_BAD_ARG;
 			}
-		// The GPU per-hop scratch (rows_pinned/rows_dev/dists_pinned and
-		// the search fdists vector) is sized m_max+1; a dump whose degree
-		// exceeds the declared layer cap (corrupt, or an honest dump
-		// loaded under smaller m/m0) would overflow it — reject here.
-		if (nd.second.size() > h->layers[layer].m_max) {
-			sdbv_hnsw_destroy(h);
-			return SDBV_ERR_BAD_ARG;
-		}
+		// Degree above the declared layer cap is ACCEPTED, like the
+		// reference's own loader (layer.rs load reads edge lists with no
+		// cap check): it arises legitimately from parameter-mismatched
+		// dumps and from this repo's threaded builds (keep-back
+		// relaxation). Safe since finalize sizes every per-hop scratch
+		// from the ACTUAL max degree (the round-1 advisor's overflow is
+		// closed by sizing, its alternative remedy); dec_node already
+		// bounds a single node at 65535 edges.
 		h->layers[layer].edges[node] = nd.second;
 		h->layers[layer].in_layer[node] = 1;
 	}

@@ -334,22 +334,29 @@ def test_dump_refuses_bits_mode_ids64():
     ix.destroy()
 
 
-def test_loader_rejects_degree_above_declared_m0():
-    """An honest m0=16 dump loaded under declared m0=4: nodes with more
-    edges than the declared layer cap would overflow the m0+1-sized GPU
-    per-hop scratch — the loader must reject them (round-1 advisor
-    finding, high)."""
+def test_loader_accepts_degree_above_declared_m0():
+    """An honest m0=16 dump loaded under declared m0=4: the reference's
+    own loader reads edge lists with no cap check (layer.rs load), and
+    since round 2 every per-hop GPU scratch is sized from the ACTUAL max
+    degree at finalize, over-cap degrees are safe — the round-1 advisor's
+    overflow (high) is closed by sizing, its alternative remedy. The
+    loaded graph searches identically to the donor (same edges)."""
     d, n = 8, 200
     rows = oracle.gen_f32(0x55, 0, n, d)
     g = sa.hnsw_create_host(d, metric="euclidean", m=8, m0=16, efc=40,
                             seed=3)
     g.insert_batch(rows, nthreads=1)
     pairs = g.dump_kv()
-    # same dump, smaller declared caps -> SDBV_ERR_BAD_ARG
-    with pytest.raises(sa.SdbvError):
-        sa.load_kv_hnsw(pairs, d, metric="euclidean", m=2, m0=4, efc=40,
-                        seed=3)
+    g2 = sa.load_kv_hnsw(pairs, d, metric="euclidean", m=2, m0=4, efc=40,
+                         seed=3)
+    a, b = g.l0_csr(), g2.l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    for q in oracle.gen_f32(0x9A, 0, 6, d):
+        i1, d1 = g.knn_search_host(q, 5, 20)
+        i2, d2 = g2.knn_search_host(q, 5, 20)
+        assert np.array_equal(i1, i2) and np.array_equal(d1, d2)
     g.destroy()
+    g2.destroy()
 
 
 def test_loader_bounds_untrusted_state_fields():
@@ -391,6 +398,27 @@ def test_snapshot2_built_graph_round_trips():
     assert g.num_layers() == g2.num_layers()
     assert g.enter_point() == g2.enter_point()
     for q in oracle.gen_f32(0xD00D, 0, 6, d):
+        i1, d1 = g.knn_search_host(q, 10, 40)
+        i2, d2 = g2.knn_search_host(q, 10, 40)
+        assert np.array_equal(i1, i2) and np.array_equal(d1, d2)
+    g.destroy()
+    g2.destroy()
+
+
+def test_threaded_build_dump_reloads_same_params():
+    """A THREADED build can leave nodes above m_max (keep-back
+    relaxation); its own dump must reload under the same declared params
+    — byte-faithful CSR, identical searches."""
+    d, n = 16, 1200
+    rows = oracle.gen_f32(0x8EAE, 0, n, d)
+    g = sa.hnsw_create_host(d, metric="cosine", m=4, m0=8, efc=60, seed=7)
+    g.insert_batch_snapshot2(rows, chunk=256, nthreads=4)
+    pairs = g.dump_kv()
+    g2 = sa.load_kv_hnsw(pairs, d, metric="cosine", m=4, m0=8, efc=60,
+                         seed=7)
+    a, b = g.l0_csr(), g2.l0_csr()
+    assert np.array_equal(a[0], b[0]) and np.array_equal(a[1], b[1])
+    for q in oracle.gen_f32(0x77, 0, 6, d):
         i1, d1 = g.knn_search_host(q, 10, 40)
         i2, d2 = g2.knn_search_host(q, 10, 40)
         assert np.array_equal(i1, i2) and np.array_equal(d1, d2)
