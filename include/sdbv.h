@@ -190,22 +190,26 @@ int sdbv_hnsw_insert_batch(sdbv_hnsw *, const float *pts, uint64_t n,
 int sdbv_hnsw_insert_batch_snapshot(sdbv_hnsw *, const float *pts,
                                     uint64_t n, uint32_t chunk,
                                     int nthreads);
-/* Snapshot build with the BATCHED apply schedule (one fixed serialization
- * of the parallel apply: all selects, then all appends in element order,
- * then one prune pass; upper-level elements defer their layer-0 half into
- * the batch). Same algorithm, quality pinned by the same recall bars; the
- * bit-exact host reference for the _gpu build below. */
+/* Snapshot build with the BATCHED apply schedule at layers 1 and 0 (one
+ * fixed serialization of the parallel apply per layer: all selects, then
+ * all appends in element order, then one prune pass). Levels >= 2 — the
+ * tiny skeleton layers — insert classically (progressive), and a
+ * 64-element classic warm-up bootstraps an empty graph (both measured
+ * requirements). Same algorithm, quality pinned by the same
+ * recall bars; the bit-exact host reference for the _gpu build below. */
 int sdbv_hnsw_insert_batch_snapshot2(sdbv_hnsw *, const float *pts,
                                      uint64_t n, uint32_t chunk,
                                      int nthreads);
 /* GPU-accelerated chunked snapshot build (hnsw/mod.rs:230-394 build hot
  * loop): bit-identical to
- * sdbv_hnsw_insert_batch_snapshot2 — each chunk's level-0 efc-searches run
- * as one multi-ep persistent-kernel launch against a delta-updated device
- * adjacency, and the select/prune pair-distance work (the RAM-bound part
- * of the apply) runs on device with the exact heuristic
- * (k_pair_mats/k_heur_select). Requires a device context; efc <= 512;
- * extend-candidates falls back to the host twin. */
+ * sdbv_hnsw_insert_batch_snapshot2 — each chunk's layer-1 and layer-0
+ * efc-searches run as multi-ep persistent-kernel launches against
+ * delta-updated padded device adjacencies (ef=1 launches carry the
+ * level-0 elements' layer-1 descent hop), and the select/prune
+ * pair-distance work (the RAM-bound part of the apply) runs on device
+ * with the exact heuristic (k_pair_mats/k_heur_select). Requires a
+ * device context; efc <= 512; extend-candidates falls back to the host
+ * twin. */
 int sdbv_hnsw_insert_batch_snapshot_gpu(sdbv_hnsw *, const float *pts,
                                         uint64_t n, uint32_t chunk,
                                         int nthreads);
